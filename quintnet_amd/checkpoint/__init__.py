@@ -1,0 +1,76 @@
+"""Checkpointing: per-rank shards, offline merge, staged distributed load.
+
+Shard layout parity with the reference (GPT2_Trainer.py:453-507):
+``{out_dir}/{name}_pp{p}_tp{t}.pt`` with ``model_state_dict``,
+``optimizer_state_dict`` and ``parallelism_info`` metadata; the merge
+CLI (merge.py) recombines TP (cat) and PP (key remap) shards into HF
+GPT-2 format.
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Any, Dict, Optional
+
+import torch
+import torch.distributed as dist
+
+from .merge import merge_checkpoints
+from .distributed_loading import load_gpt2_distributed
+
+__all__ = [
+    "save_sharded_checkpoint",
+    "load_sharded_checkpoint",
+    "merge_checkpoints",
+    "load_gpt2_distributed",
+]
+
+
+def save_sharded_checkpoint(
+    model,
+    out_dir: str,
+    name: str = "final_model",
+    pg_manager=None,
+    optimizer=None,
+    config: Optional[Dict[str, Any]] = None,
+) -> Optional[str]:
+    """Write this rank's shard ``{name}_pp{p}_tp{t}.pt`` (dp_rank 0 only)."""
+    pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
+    tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
+    dp_rank = pg_manager.dp_rank if pg_manager is not None else 0
+    pp_size = pg_manager.pp_size if pg_manager is not None else 1
+    tp_size = pg_manager.tp_size if pg_manager is not None else 1
+
+    os.makedirs(out_dir, exist_ok=True)
+    path = None
+    if dp_rank == 0:
+        path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}.pt")
+        payload = {
+            "model_state_dict": {k: v.cpu() for k, v in model.state_dict().items()},
+            "parallelism_info": {
+                "pp_rank": pp_rank,
+                "pp_size": pp_size,
+                "tp_rank": tp_rank,
+                "tp_size": tp_size,
+                "dp_rank": dp_rank,
+            },
+            "config": config or {},
+        }
+        if optimizer is not None and hasattr(optimizer, "state_dict"):
+            try:
+                payload["optimizer_state_dict"] = optimizer.state_dict()
+            except Exception:  # noqa: BLE001 — optimizer state is best-effort
+                pass
+        torch.save(payload, path)
+    if dist.is_initialized():
+        dist.barrier()
+    return path
+
+
+def load_sharded_checkpoint(model, out_dir: str, name: str, pg_manager=None, strict=True):
+    pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
+    tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
+    path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}.pt")
+    ckpt = torch.load(path, map_location="cpu", weights_only=False)
+    model.load_state_dict(ckpt["model_state_dict"], strict=strict)
+    return ckpt

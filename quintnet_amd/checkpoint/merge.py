@@ -1,0 +1,133 @@
+"""Offline merge of ``{name}_pp{p}_tp{t}.pt`` shards into HF GPT-2 format.
+
+Parity with reference merge_checkpoints.py:33-188.  Differences owned
+end-to-end by this framework:
+
+* c_attn/c_fc (column-parallel) shards are merged interleave-aware —
+  each shard's rows are ``[q_loc; k_loc; v_loc]`` (see
+  models/gpt2/attention.py), so the merge splits each shard into its
+  Q/K/V thirds and concatenates per component (a naive cat would
+  scramble head order);
+* c_proj (row-parallel) weights cat along dim 1; its bias (and every
+  replicated tensor) is taken from tp rank 0;
+* PP merge remaps ``blocks.X`` → ``h.{X+offset}``; stage 0 contributes
+  wte/wpe, the last stage ln_f + lm_head;
+* HF conversion adds the ``transformer.`` prefix and re-transposes the
+  Conv1D weights (HF stores [in, out]).
+"""
+
+from __future__ import annotations
+
+import glob
+import os
+import re
+from typing import Dict, Optional
+
+import torch
+
+__all__ = ["load_shards", "merge_tp_shards", "merge_pp_stages", "convert_to_hf_format", "merge_checkpoints"]
+
+_SHARD_RE = re.compile(r"_pp(\d+)_tp(\d+)\.pt$")
+
+
+def load_shards(input_dir: str, prefix: Optional[str] = None) -> Dict[int, Dict[int, dict]]:
+    out: Dict[int, Dict[int, dict]] = {}
+    for path in sorted(glob.glob(os.path.join(input_dir, "*.pt"))):
+        base = os.path.basename(path)
+        if prefix and not base.startswith(prefix):
+            continue
+        m = _SHARD_RE.search(base)
+        if not m:
+            continue
+        pp, tp = int(m.group(1)), int(m.group(2))
+        out.setdefault(pp, {})[tp] = torch.load(path, map_location="cpu", weights_only=False)
+    if not out:
+        raise FileNotFoundError(f"no *_pp*_tp*.pt shards under {input_dir}")
+    return out
+
+
+def _merge_qkv_rows(tensors):
+    """cat column-parallel fused-QKV shards: split each into thirds, cat per part."""
+    parts = [t.chunk(3, dim=0) for t in tensors]
+    return torch.cat(
+        [torch.cat([p[i] for p in parts], dim=0) for i in range(3)], dim=0
+    )
+
+
+def merge_tp_shards(tp_shards: Dict[int, Dict[str, torch.Tensor]]) -> Dict[str, torch.Tensor]:
+    tp_size = len(tp_shards)
+    keys = tp_shards[0].keys()
+    merged: Dict[str, torch.Tensor] = {}
+    for key in keys:
+        ts = [tp_shards[r][key] for r in range(tp_size)]
+        if tp_size == 1:
+            merged[key] = ts[0]
+        elif "attn.c_attn.weight" in key:
+            merged[key] = _merge_qkv_rows(ts)
+        elif "attn.c_attn.bias" in key:
+            merged[key] = _merge_qkv_rows([t.unsqueeze(1) for t in ts]).squeeze(1)
+        elif "c_fc.weight" in key or "c_fc.bias" in key:
+            merged[key] = torch.cat(ts, dim=0)
+        elif "c_proj.weight" in key:
+            merged[key] = torch.cat(ts, dim=1)
+        else:
+            # replicated (LN, embeddings, c_proj bias, lm_head): rank 0 copy
+            merged[key] = ts[0]
+    return merged
+
+
+def merge_pp_stages(
+    pp_stages: Dict[int, Dict[str, torch.Tensor]], n_layer: Optional[int] = None
+) -> Dict[str, torch.Tensor]:
+    pp_size = len(pp_stages)
+    merged: Dict[str, torch.Tensor] = {}
+    offset = 0
+    for pp_rank in sorted(pp_stages):
+        state = pp_stages[pp_rank]
+        max_block = -1
+        for key, value in state.items():
+            new_key = key
+            m = re.match(r"blocks\.(\d+)\.(.*)", key)
+            if m:
+                idx = int(m.group(1))
+                max_block = max(max_block, idx)
+                new_key = f"h.{idx + offset}.{m.group(2)}"
+            elif key.startswith("embedding.wte."):
+                new_key = "wte." + key.split(".")[-1]
+            elif key.startswith("embedding.wpe."):
+                new_key = "wpe." + key.split(".")[-1]
+            elif key.startswith("ln_f."):
+                new_key = key
+            elif key.startswith("lm_head"):
+                new_key = "lm_head.weight"
+            merged[new_key] = value
+        offset += max_block + 1
+    return merged
+
+
+def convert_to_hf_format(merged: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    hf: Dict[str, torch.Tensor] = {}
+    for key, value in merged.items():
+        new_key = key if key.startswith("lm_head") else "transformer." + key
+        if (
+            "c_attn.weight" in key
+            or "c_proj.weight" in key
+            or "c_fc.weight" in key
+        ):
+            value = value.t().contiguous()  # Linear [out,in] -> HF Conv1D [in,out]
+        hf[new_key] = value
+    if "lm_head.weight" not in hf and "transformer.wte.weight" in hf:
+        hf["lm_head.weight"] = hf["transformer.wte.weight"]
+    return hf
+
+
+def merge_checkpoints(input_dir: str, output_path: str, prefix: str = "final_model") -> str:
+    shards = load_shards(input_dir, prefix)
+    pp_stages = {pp: merge_tp_shards({t: s["model_state_dict"] for t, s in tps.items()})
+                 for pp, tps in shards.items()}
+    merged = merge_pp_stages(pp_stages)
+    hf_state = convert_to_hf_format(merged)
+    config = next(iter(next(iter(shards.values())).values())).get("config", {})
+    os.makedirs(os.path.dirname(os.path.abspath(output_path)), exist_ok=True)
+    torch.save({"model_state_dict": hf_state, "config": config}, output_path)
+    return output_path

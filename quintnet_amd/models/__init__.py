@@ -1,0 +1,35 @@
+from .vit import (
+    Model,
+    VisionTransformer,
+    PatchEmbedding,
+    ViTEmbedding,
+    Attention,
+    MLP,
+    TransformerBlock,
+    ClassificationHead,
+)
+from .gpt2 import (
+    GPT2Config,
+    GPT2Embedding,
+    GPT2Attention,
+    GPT2MLP,
+    GPT2Block,
+    GPT2Stage,
+)
+
+__all__ = [
+    "Model",
+    "VisionTransformer",
+    "PatchEmbedding",
+    "ViTEmbedding",
+    "Attention",
+    "MLP",
+    "TransformerBlock",
+    "ClassificationHead",
+    "GPT2Config",
+    "GPT2Embedding",
+    "GPT2Attention",
+    "GPT2MLP",
+    "GPT2Block",
+    "GPT2Stage",
+]

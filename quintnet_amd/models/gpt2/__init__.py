@@ -1,0 +1,15 @@
+from .config import GPT2Config
+from .embeddings import GPT2Embedding
+from .attention import GPT2Attention
+from .mlp import GPT2MLP
+from .block import GPT2Block
+from .stage import GPT2Stage
+
+__all__ = [
+    "GPT2Config",
+    "GPT2Embedding",
+    "GPT2Attention",
+    "GPT2MLP",
+    "GPT2Block",
+    "GPT2Stage",
+]

@@ -1,0 +1,65 @@
+"""GPT-2 causal self-attention with Megatron TP sharding.
+
+Reference parity: utils/GPT2/gpt2_attention.py:24-181.  c_attn is a
+column-parallel fused QKV projection (gather_output=False) and c_proj a
+row-parallel projection whose all-reduce is the block's only TP
+collective on the attention path.  Heads are sharded: each TP rank runs
+n_head/tp heads through the fused attention op.
+
+Local QKV layout note: the rank-local c_attn weight is
+``[q_local; k_local; v_local]`` (each ``n_embd/tp`` rows) — i.e. Q/K/V
+are sharded per-head *separately*, not a naive contiguous slice of the
+[3*n_embd] output (which would put whole Q on rank 0 at tp=2).  The
+staged checkpoint loader and the merge CLI both honor this layout.
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from ...ops import attention as fused_attention
+from ...parallel.tensor_parallel import ColumnParallelLinear, RowParallelLinear
+from .config import GPT2Config
+
+__all__ = ["GPT2Attention"]
+
+
+class GPT2Attention(nn.Module):
+    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
+        super().__init__()
+        self.config = config
+        self.c_attn = ColumnParallelLinear(
+            config.n_embd,
+            3 * config.n_embd,
+            tp_group=tp_group,
+            gather_output=False,
+            device=device,
+            dtype=dtype,
+        )
+        self.c_proj = RowParallelLinear(
+            config.n_embd,
+            config.n_embd,
+            tp_group=tp_group,
+            input_is_parallel=True,
+            device=device,
+            dtype=dtype,
+        )
+        self.tp_size = self.c_attn.tp_size
+        assert config.n_head % self.tp_size == 0, "n_head must divide by tp"
+        self.n_head_local = config.n_head // self.tp_size
+        self.hidden_local = config.n_embd // self.tp_size
+        self.head_dim = config.head_dim
+        self.resid_dropout = nn.Dropout(config.dropout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        B, T, _ = x.shape
+        qkv = self.c_attn(x)  # [B, T, 3*n_embd/tp] = [q_loc | k_loc | v_loc]
+        q, k, v = qkv.split(self.hidden_local, dim=-1)
+
+        def split_heads(t):
+            return t.view(B, T, self.n_head_local, self.head_dim).transpose(1, 2)
+
+        out = fused_attention(split_heads(q), split_heads(k), split_heads(v), causal=True)
+        out = out.transpose(1, 2).reshape(B, T, self.hidden_local)
+        return self.resid_dropout(self.c_proj(out))

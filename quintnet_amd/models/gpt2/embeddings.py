@@ -1,0 +1,32 @@
+"""GPT-2 token + position embedding (reference utils/GPT2/gpt2_embeddings.py).
+
+wte/wpe are replicated across TP ranks (the TP axis shards the block
+GEMMs; the embedding gather is HBM-bandwidth-bound, not FLOP-bound).
+"""
+
+from __future__ import annotations
+
+import torch
+import torch.nn as nn
+
+from .config import GPT2Config
+
+__all__ = ["GPT2Embedding"]
+
+
+class GPT2Embedding(nn.Module):
+    def __init__(self, config: GPT2Config, device=None, dtype=None):
+        super().__init__()
+        kw = {"device": device, "dtype": dtype}
+        self.config = config
+        self.wte = nn.Embedding(config.vocab_size, config.n_embd, **kw)
+        self.wpe = nn.Embedding(config.n_positions, config.n_embd, **kw)
+        self.drop = nn.Dropout(config.dropout)
+        nn.init.normal_(self.wte.weight, std=config.initializer_range)
+        nn.init.normal_(self.wpe.weight, std=config.initializer_range)
+
+    def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
+        T = input_ids.shape[1]
+        pos = torch.arange(T, device=input_ids.device).unsqueeze(0)
+        x = self.wte(input_ids) + self.wpe(pos)
+        return self.drop(x)

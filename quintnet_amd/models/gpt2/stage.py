@@ -1,0 +1,135 @@
+"""GPT-2 pipeline stage container (reference utils/GPT2/gpt2_stage.py).
+
+First stage owns the embedding (wte+wpe); the last stage owns ln_f and
+lm_head — a tied COPY of wte when pp>1 (its gradient is averaged with
+the embedding's over the first+last-stage subgroup each step —
+``sync_tied_weights_grad``; reference gpt2_stage.py:112-141, fixed to a
+dedicated 2-rank group per SURVEY.md §8.3 so pp>2 cannot deadlock).
+With pp==1 lm_head IS wte (true tying, no sync needed).
+
+State-dict naming contract (kept for the merge CLI / checkpoint layout
+parity): ``embedding.wte.weight``, ``embedding.wpe.weight``,
+``blocks.N.{ln_1,attn.c_attn,attn.c_proj,ln_2,mlp.c_fc,mlp.c_proj}.*``,
+``ln_f.*``, ``lm_head.weight``.
+"""
+
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from ...ops import FusedLayerNorm, linear as fused_linear
+from ...parallel.pipeline.wrapper import distribute_layers
+from .block import GPT2Block
+from .config import GPT2Config
+from .embeddings import GPT2Embedding
+
+__all__ = ["GPT2Stage"]
+
+
+class GPT2Stage(nn.Module):
+    def __init__(
+        self,
+        config: GPT2Config,
+        pp_rank: int = 0,
+        pp_size: int = 1,
+        tp_group=None,
+        tied_group=None,
+        device=None,
+        dtype=None,
+    ):
+        super().__init__()
+        self.config = config
+        self.pp_rank = pp_rank
+        self.pp_size = pp_size
+        self.tied_group = tied_group
+        self.is_first_stage = pp_rank == 0
+        self.is_last_stage = pp_rank == pp_size - 1
+        kw = {"device": device, "dtype": dtype}
+
+        self.layer_distribution = distribute_layers(config.n_layer, pp_size)
+        self.my_layers: List[int] = self.layer_distribution[pp_rank]
+
+        if self.is_first_stage:
+            self.embedding = GPT2Embedding(config, **kw)
+        self.blocks = nn.ModuleList(
+            GPT2Block(config, tp_group=tp_group, **kw) for _ in self.my_layers
+        )
+        if self.is_last_stage:
+            self.ln_f = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, **kw)
+            if self.is_first_stage:
+                # pp==1: true weight tying
+                self.lm_head = None
+            else:
+                self.lm_head = nn.Parameter(
+                    torch.empty(config.vocab_size, config.n_embd, **kw)
+                )
+                nn.init.normal_(self.lm_head, std=config.initializer_range)
+
+    # ------------------------------------------------------------------
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if self.is_first_stage:
+            x = self.embedding(x)
+        for blk in self.blocks:
+            x = blk(x)
+        if self.is_last_stage:
+            x = self.ln_f(x)
+            w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
+            x = fused_linear(x, w, None, None)  # logits = h @ wte^T (tied)
+        return x
+
+    # ------------------------------------------------------------------
+    def sync_tied_weights_grad(self) -> None:
+        """Average the tied wte/lm_head gradient between first & last stage."""
+        if self.pp_size == 1 or not (self.is_first_stage or self.is_last_stage):
+            return
+        if not dist.is_initialized():
+            return
+        grad = None
+        if self.is_first_stage and self.embedding.wte.weight.grad is not None:
+            grad = self.embedding.wte.weight.grad
+        elif self.is_last_stage and self.lm_head is not None and self.lm_head.grad is not None:
+            grad = self.lm_head.grad
+        if grad is None:
+            return
+        group = self.tied_group
+        dist.all_reduce(grad, op=dist.ReduceOp.SUM, group=group)
+        grad.div_(2.0)
+
+    # ------------------------------------------------------------------
+    @classmethod
+    def from_sharded_state_dict(
+        cls,
+        config: GPT2Config,
+        state_dict: Dict[str, torch.Tensor],
+        pp_rank: int,
+        pp_size: int,
+        tp_group=None,
+        tied_group=None,
+        device=None,
+        dtype=None,
+    ) -> "GPT2Stage":
+        """Build a stage and load pre-sharded (rank-local) tensors.
+
+        ``state_dict`` uses the stage-local naming contract above with
+        already-TP-sliced c_attn/c_fc/c_proj tensors (see
+        checkpoint/distributed_loading.py).
+        """
+        stage = cls(
+            config,
+            pp_rank=pp_rank,
+            pp_size=pp_size,
+            tp_group=tp_group,
+            tied_group=tied_group,
+            device=device,
+            dtype=dtype,
+        )
+        tgt = stage.state_dict()
+        missing = [k for k in state_dict if k not in tgt]
+        if missing:
+            raise KeyError(f"staged load: unexpected keys {missing[:5]}...")
+        stage.load_state_dict(state_dict, strict=False)
+        return stage

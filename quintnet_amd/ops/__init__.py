@@ -1,0 +1,35 @@
+"""Hand-written CDNA4 (gfx950) op library with CPU test fallbacks.
+
+Every op dispatches to the in-tree HIP extension (quintnet_amd._C) on
+GPU — failing loudly if it is missing — and to a plain PyTorch fp32
+reference on CPU, which is also what the numerics tests compare the
+kernels against.
+"""
+
+from ._backend import ext, has_ext, use_native, force_eager
+from .linear import linear, LinearFunction
+from .layernorm import layer_norm, FusedLayerNorm, LayerNormFunction
+from .attention import attention, AttentionFunction, causal_softmax, softmax_bwd
+from .cross_entropy import cross_entropy, CrossEntropyFunction
+from .adamw import adamw_step_flat, clip_grad_norm_local, l2_norm
+
+__all__ = [
+    "ext",
+    "has_ext",
+    "use_native",
+    "force_eager",
+    "linear",
+    "LinearFunction",
+    "layer_norm",
+    "FusedLayerNorm",
+    "LayerNormFunction",
+    "attention",
+    "AttentionFunction",
+    "causal_softmax",
+    "softmax_bwd",
+    "cross_entropy",
+    "CrossEntropyFunction",
+    "adamw_step_flat",
+    "clip_grad_norm_local",
+    "l2_norm",
+]

@@ -1,0 +1,70 @@
+"""Fused cross-entropy (log-softmax + NLL) with ignore_index.
+
+One hand-written CDNA4 kernel pass per direction over the [N, V] logits
+(V = 50257 for GPT-2): fwd computes per-row max/logsumexp/nll without
+materializing log-probs; bwd writes (softmax - onehot)/n_valid in one
+pass.  Replaces reference nn.CrossEntropyLoss (trainer.py:90,
+GPT2_Trainer.py:109).
+"""
+
+from __future__ import annotations
+
+import torch
+
+from . import _backend
+
+__all__ = ["cross_entropy", "CrossEntropyFunction"]
+
+
+class CrossEntropyFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits, target, ignore_index):
+        # logits: [N, V] (any float dtype), target: [N] int64
+        if _backend.use_native(logits) and _backend.has_ext():
+            loss_sum, n_valid, lse = _backend.ext().cross_entropy_fwd(
+                logits.contiguous(), target.contiguous(), ignore_index
+            )
+        else:
+            lf = logits.float()
+            lse = torch.logsumexp(lf, dim=-1)
+            valid = target != ignore_index
+            n_valid = valid.sum()
+            tgt = target.clamp_min(0)
+            nll = lse - lf.gather(1, tgt.unsqueeze(1)).squeeze(1)
+            nll = torch.where(valid, nll, torch.zeros_like(nll))
+            loss_sum = nll.sum()
+        ctx.save_for_backward(logits, target, lse, n_valid)
+        ctx.ignore_index = ignore_index
+        n = n_valid.clamp_min(1).to(loss_sum.dtype)
+        return loss_sum / n
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        logits, target, lse, n_valid = ctx.saved_tensors
+        if _backend.use_native(logits) and _backend.has_ext():
+            dl = _backend.ext().cross_entropy_bwd(
+                logits.contiguous(), target.contiguous(), lse, int(n_valid.item()),
+                ctx.ignore_index,
+            )
+            dl = dl * grad_output
+        else:
+            lf = logits.float()
+            p = torch.exp(lf - lse.unsqueeze(1))
+            valid = (target != ctx.ignore_index).unsqueeze(1)
+            tgt = target.clamp_min(0)
+            p.scatter_add_(
+                1, tgt.unsqueeze(1), torch.full_like(tgt.unsqueeze(1), -1.0, dtype=p.dtype)
+            )
+            n = n_valid.clamp_min(1).float()
+            dl = torch.where(valid, p / n, torch.zeros_like(p)) * grad_output
+            dl = dl.to(logits.dtype)
+        return dl, None, None
+
+
+def cross_entropy(
+    logits: torch.Tensor, target: torch.Tensor, ignore_index: int = -100
+) -> torch.Tensor:
+    """Mean cross-entropy over target != ignore_index rows."""
+    return CrossEntropyFunction.apply(
+        logits.reshape(-1, logits.shape[-1]), target.reshape(-1), ignore_index
+    )

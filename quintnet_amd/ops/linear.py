@@ -1,0 +1,105 @@
+"""Linear (GEMM) op with fused bias/activation epilogue.
+
+Forward runs the hand-written CDNA4 MFMA GEMM (csrc/gemm.hip — an NT
+GEMM: C[M,N] = A[M,K]·B[N,K]^T matching nn.Linear's [out,in] weight
+layout) with the bias add and optional GELU/ReLU fused into the
+epilogue.  Backward dgrad/dwgrad are plain (unfused) GEMMs and go
+through rocBLAS/hipBLASLt via torch.matmul — per the MI355X design
+split: hand-written kernels for fused hot ops, the vendor GEMM library
+for plain GEMMs.
+
+Replaces the implicit cuBLAS GEMMs at reference
+parallelism/tensor_parallel/layers.py:119,211 and utils/GPT2/*.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from . import _backend
+
+__all__ = ["linear", "LinearFunction"]
+
+_ACT_NONE = 0
+_ACT_GELU = 1
+_ACT_RELU = 2
+_ACT_MAP = {None: _ACT_NONE, "none": _ACT_NONE, "gelu": _ACT_GELU, "relu": _ACT_RELU}
+
+
+def _gelu_tanh(x: torch.Tensor) -> torch.Tensor:
+    return torch.nn.functional.gelu(x, approximate="tanh")
+
+
+def _native_ok(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    # The MFMA kernel covers bf16 with K a multiple of 8 (staging width);
+    # anything else takes the library GEMM path.
+    return (
+        _backend.has_ext()
+        and x.is_cuda
+        and x.dtype == torch.bfloat16
+        and weight.dtype == torch.bfloat16
+        and weight.shape[1] % 8 == 0
+    )
+
+
+class LinearFunction(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, activation):
+        act = _ACT_MAP[activation]
+        x2d = x.reshape(-1, x.shape[-1])
+        ctx.x_shape = x.shape
+        ctx.act = act
+        pre_act = None
+        if _backend.use_native(x) and _native_ok(x, weight):
+            out = _backend.ext().gemm_nt(
+                x2d.contiguous(), weight.contiguous(),
+                bias if bias is not None else None, act,
+            )
+            if act != _ACT_NONE:
+                # epilogue stores pre-activation too when an activation is fused
+                out, pre_act = out
+        else:
+            out = torch.nn.functional.linear(x2d, weight, bias)
+            if act == _ACT_GELU:
+                pre_act = out
+                out = _gelu_tanh(out)
+            elif act == _ACT_RELU:
+                pre_act = out
+                out = torch.relu(out)
+        ctx.save_for_backward(x2d, weight, pre_act if pre_act is not None else torch.empty(0))
+        ctx.has_bias = bias is not None
+        return out.reshape(*x.shape[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, grad_out):
+        x2d, weight, pre_act = ctx.saved_tensors
+        g = grad_out.reshape(-1, grad_out.shape[-1])
+        if ctx.act == _ACT_GELU:
+            pa = pre_act.float()
+            # d/dx gelu_tanh(x)
+            c = 0.7978845608028654  # sqrt(2/pi)
+            t = torch.tanh(c * (pa + 0.044715 * pa.pow(3)))
+            dg = 0.5 * (1.0 + t) + 0.5 * pa * (1.0 - t * t) * c * (1.0 + 3 * 0.044715 * pa.pow(2))
+            g = (g.float() * dg).to(g.dtype)
+        elif ctx.act == _ACT_RELU:
+            g = g * (pre_act > 0).to(g.dtype)
+        grad_x = grad_w = grad_b = None
+        if ctx.needs_input_grad[0]:
+            grad_x = (g @ weight).reshape(ctx.x_shape)
+        if ctx.needs_input_grad[1]:
+            grad_w = g.t() @ x2d
+        if ctx.has_bias and ctx.needs_input_grad[2]:
+            grad_b = g.sum(dim=0)
+        return grad_x, grad_w, grad_b, None
+
+
+def linear(
+    x: torch.Tensor,
+    weight: torch.Tensor,
+    bias: Optional[torch.Tensor] = None,
+    activation: Optional[str] = None,
+) -> torch.Tensor:
+    """y = activation(x @ weight.T + bias), fused on gfx950."""
+    return LinearFunction.apply(x, weight, bias, activation)

@@ -1,0 +1,44 @@
+from .data_parallel import DataParallel, BucketConfig, DistributedConfig, GradientBucket
+from .backends import DistributedBackend, TorchDistributedBackend, LocalBackend
+from .tensor_parallel import (
+    ColumnParallelLinear,
+    RowParallelLinear,
+    VocabParallelEmbedding,
+    apply_tensor_parallel,
+    ensure_divisibility,
+)
+from .pipeline import (
+    PipelineParallelWrapper,
+    PipelineSchedule,
+    AllFwdAllBwdSchedule,
+    OneFOneBSchedule,
+    PipelineTrainer,
+    PipelineDataLoader,
+    distribute_layers,
+)
+
+# Reference-compatible alias (QuintNet exports TensorParallel = apply_tensor_parallel)
+TensorParallel = apply_tensor_parallel
+
+__all__ = [
+    "DataParallel",
+    "BucketConfig",
+    "DistributedConfig",
+    "GradientBucket",
+    "DistributedBackend",
+    "TorchDistributedBackend",
+    "LocalBackend",
+    "ColumnParallelLinear",
+    "RowParallelLinear",
+    "VocabParallelEmbedding",
+    "apply_tensor_parallel",
+    "TensorParallel",
+    "ensure_divisibility",
+    "PipelineParallelWrapper",
+    "PipelineSchedule",
+    "AllFwdAllBwdSchedule",
+    "OneFOneBSchedule",
+    "PipelineTrainer",
+    "PipelineDataLoader",
+    "distribute_layers",
+]

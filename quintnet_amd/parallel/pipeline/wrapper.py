@@ -1,0 +1,100 @@
+"""Pipeline stage wrapper: splits a model into per-rank stages.
+
+Parity with reference parallelism/pipeline_parallel/wrapper.py:40-250.
+Contract: the model exposes ``.embedding``, ``.blocks`` (ModuleList) and
+a head (``.classification_head`` / ``.head`` / ``.lm_head``); blocks are
+distributed evenly with the remainder going to EARLY stages, the first
+stage gets the embedding, the last the head.  A pre-built
+``stage_module`` (e.g. GPT2Stage from a staged checkpoint load) can be
+passed instead.
+"""
+
+from __future__ import annotations
+
+from typing import List, Optional
+
+import torch
+import torch.nn as nn
+
+__all__ = ["PipelineParallelWrapper", "distribute_layers"]
+
+
+def distribute_layers(depth: int, pp_size: int) -> List[List[int]]:
+    """Even block split; remainder blocks go to early stages."""
+    base = depth // pp_size
+    rem = depth % pp_size
+    out: List[List[int]] = []
+    start = 0
+    for r in range(pp_size):
+        n = base + (1 if r < rem else 0)
+        out.append(list(range(start, start + n)))
+        start += n
+    return out
+
+
+def _find_head(model: nn.Module) -> Optional[nn.Module]:
+    for name in ("classification_head", "head", "lm_head"):
+        if hasattr(model, name):
+            return getattr(model, name)
+    return None
+
+
+class PipelineParallelWrapper(nn.Module):
+    def __init__(
+        self,
+        model: Optional[nn.Module] = None,
+        pp_rank: int = 0,
+        pp_group=None,
+        pp_size: int = 1,
+        device: Optional[torch.device] = None,
+        stage_module: Optional[nn.Module] = None,
+    ):
+        super().__init__()
+        self.pp_rank = pp_rank
+        self.pp_size = pp_size
+        self.pp_group = pp_group
+        self.device = device
+        self.is_first_stage = pp_rank == 0
+        self.is_last_stage = pp_rank == pp_size - 1
+
+        if stage_module is not None:
+            self.local_module = stage_module
+            self.layer_distribution = None
+        else:
+            assert model is not None, "need model or stage_module"
+            if not hasattr(model, "blocks"):
+                raise ValueError(
+                    "PipelineParallelWrapper needs a model with a .blocks ModuleList"
+                )
+            depth = len(model.blocks)
+            self.layer_distribution = distribute_layers(depth, pp_size)
+            my_blocks = self.layer_distribution[pp_rank]
+            mods: List[nn.Module] = []
+            if self.is_first_stage and hasattr(model, "embedding"):
+                mods.append(model.embedding)
+            mods.extend(model.blocks[i] for i in my_blocks)
+            if self.is_last_stage:
+                head = _find_head(model)
+                if head is not None:
+                    mods.append(head)
+            self.local_module = nn.Sequential(*mods)
+        if device is not None:
+            self.local_module.to(device)
+
+    # ------------------------------------------------------------------
+    def forward(self, x):
+        return self.local_module(x)
+
+    def backward(self, input_tensor, output_tensor, output_tensor_grad):
+        """Manual micro-batch backward (reference wrapper.py:214-250).
+
+        On the last stage ``output_tensor`` is the (scaled) loss and
+        ``output_tensor_grad`` is None; elsewhere it is the stage output
+        with the grad received from the next stage.
+        """
+        if input_tensor is not None and not input_tensor.requires_grad:
+            raise RuntimeError("pipeline input tensor must require grad")
+        if input_tensor is not None:
+            input_tensor.retain_grad()
+        torch.autograd.backward(output_tensor, grad_tensors=output_tensor_grad)
+        return input_tensor.grad if input_tensor is not None else None

@@ -1,0 +1,231 @@
+"""Generic Trainer: epoch loop + pipeline/plain dispatch + metrics.
+
+Parity with reference trainer.py:66-363 (Trainer for the ViT
+classification path).  The train loader's batch size is the MICRO-batch
+size; one optimizer step consumes ``grad_acc_steps`` micro-batches
+(matching the reference's PipelineDataLoader contract).
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Any, Dict, Optional
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from .ops import clip_grad_norm_local
+from .parallel import DataParallel, PipelineDataLoader, PipelineTrainer
+
+__all__ = ["Trainer"]
+
+
+def _unwrap(model):
+    m = model
+    while hasattr(m, "module"):
+        m = m.module
+    return m
+
+
+class Trainer:
+    def __init__(self, model, train_loader, val_loader, config: Dict[str, Any], pg_manager):
+        self.model = model
+        self.train_loader = train_loader
+        self.val_loader = val_loader
+        self.config = config
+        self.pg = pg_manager
+        self.device = pg_manager.device if pg_manager is not None else (
+            torch.device("cuda") if torch.cuda.is_available() else torch.device("cpu")
+        )
+        self.lr = float(config.get("learning_rate", 1e-3))
+        self.num_epochs = int(config.get("num_epochs", 1))
+        self.grad_acc_steps = int(config.get("grad_acc_steps", 1))
+        self.max_grad_norm = config.get("max_grad_norm", 1.0)
+        self.task_type = config.get("task_type", "classification")
+        self.optimizer = torch.optim.Adam(self.model.parameters(), lr=self.lr)
+        self.criterion = nn.CrossEntropyLoss()
+        self.pp_size = pg_manager.pp_size if pg_manager is not None else 1
+        self.is_pipeline = self.pp_size > 1
+        self.pipeline_trainer: Optional[PipelineTrainer] = None
+        self.tensor_shapes = None
+        if self.is_pipeline:
+            self._setup_pipeline()
+
+    # ------------------------------------------------------------------
+    def _setup_pipeline(self) -> None:
+        micro_b = self._infer_micro_batch()
+        inner = _unwrap(self.model)
+        seq, hidden = self._infer_seq_hidden(inner)
+        self.tensor_shapes = (micro_b, seq, hidden)
+        self.pipeline_trainer = PipelineTrainer(
+            model=self.model,
+            optimizer=self.optimizer,
+            criterion=self.criterion,
+            pp_rank=self.pg.pp_rank,
+            pp_size=self.pg.pp_size,
+            pp_group=self.pg.get_group("pp"),
+            pp_group_ranks=self.pg.get_group_ranks("pp"),
+            schedule=self.config.get("schedule", "1f1b"),
+            task_type=self.task_type,
+            max_grad_norm=self.max_grad_norm,
+        )
+
+    def _infer_micro_batch(self) -> int:
+        try:
+            return int(self.train_loader.batch_size)
+        except (AttributeError, TypeError):
+            batch = next(iter(self.train_loader))
+            if isinstance(batch, dict):
+                key = "images" if "images" in batch else next(iter(batch))
+                return len(batch[key])
+            return len(batch[0])
+
+    def _infer_seq_hidden(self, inner) -> tuple:
+        # local_module path (PP wrapper) or the bare model
+        if hasattr(inner, "local_module"):
+            inner = inner  # PipelineParallelWrapper
+        if hasattr(inner, "seq_len") and hasattr(inner, "hidden_dim"):
+            return inner.seq_len, inner.hidden_dim
+        seq = self.config.get("seq_len")
+        hidden = self.config.get("hidden_dim")
+        if seq is None or hidden is None:
+            # try embedded model attributes
+            m = getattr(inner, "local_module", inner)
+            for mod in ([m] + list(m.children()) if isinstance(m, nn.Module) else []):
+                if hasattr(mod, "seq_len") and hasattr(mod, "hidden_dim"):
+                    return mod.seq_len, mod.hidden_dim
+            raise ValueError("cannot infer pipeline tensor shapes; set config seq_len/hidden_dim")
+        return int(seq), int(hidden)
+
+    @property
+    def _dtype(self) -> torch.dtype:
+        return next(self.model.parameters()).dtype
+
+    # ------------------------------------------------------------------
+    def fit(self) -> Dict[str, float]:
+        history = {}
+        for epoch in range(self.num_epochs):
+            t0 = time.time()
+            train_metrics = self._train_epoch(epoch)
+            val_metrics = self._validate_epoch() if self.val_loader is not None else {}
+            dt = time.time() - t0
+            metrics = self._broadcast_metrics(train_metrics, val_metrics)
+            if self._is_rank0():
+                msg = f"[epoch {epoch+1}/{self.num_epochs}] {dt:.1f}s"
+                for k, v in metrics.items():
+                    msg += f" {k}={v:.4f}"
+                print(msg, flush=True)
+            history = metrics
+        self._save_checkpoint()
+        return history
+
+    def _is_rank0(self) -> bool:
+        return not dist.is_initialized() or dist.get_rank() == 0
+
+    def _broadcast_metrics(self, train: Dict[str, float], val: Dict[str, float]) -> Dict[str, float]:
+        out = {f"train_{k}": v for k, v in train.items()}
+        out.update({f"val_{k}": v for k, v in val.items()})
+        if not dist.is_initialized() or not out:
+            return out
+        # last-stage-only metrics -> everyone (MAX propagates the values;
+        # reference K17 pattern)
+        keys = sorted(out)
+        vals = torch.tensor([out[k] for k in keys], dtype=torch.float64)
+        if self.device.type == "cuda":
+            vals = vals.to(self.device)
+        dist.all_reduce(vals, op=dist.ReduceOp.MAX)
+        return dict(zip(keys, vals.cpu().tolist()))
+
+    # ------------------------------------------------------------------
+    def _train_epoch(self, epoch: int) -> Dict[str, float]:
+        self.model.train()
+        if self.is_pipeline:
+            return self._train_epoch_pipeline()
+        return self._train_epoch_plain()
+
+    def _train_epoch_pipeline(self) -> Dict[str, float]:
+        loader = PipelineDataLoader(self.train_loader, self.grad_acc_steps, self.task_type)
+        num_steps = max(len(self.train_loader) // self.grad_acc_steps, 1)
+        agg: Dict[str, float] = {}
+        for _ in range(num_steps):
+            m = self.pipeline_trainer.train_step(
+                loader, self.tensor_shapes, self.device, self._dtype
+            )
+            for k, v in m.items():
+                agg[k] = agg.get(k, 0.0) + float(v)
+        return {k: v / num_steps for k, v in agg.items()}
+
+    def _train_epoch_plain(self) -> Dict[str, float]:
+        total_loss, correct, total, steps = 0.0, 0, 0, 0
+        for batch in self.train_loader:
+            batch = PipelineDataLoader._normalize(batch)
+            x = batch["images"].to(self.device, non_blocking=True)
+            y = batch["labels"].to(self.device, non_blocking=True)
+            out = self.model(x)
+            loss = self.criterion(out, y)
+            loss.backward()
+            if isinstance(self.model, DataParallel):
+                self.model.finalize_gradients()
+            if self.max_grad_norm:
+                clip_grad_norm_local(
+                    [p for p in self.model.parameters() if p.requires_grad],
+                    self.max_grad_norm,
+                )
+            self.optimizer.step()
+            if hasattr(self.model, "zero_grad") and isinstance(self.model, DataParallel):
+                self.model.zero_grad()
+            else:
+                self.optimizer.zero_grad()
+            total_loss += float(loss.detach())
+            correct += int((out.argmax(-1) == y).sum())
+            total += y.numel()
+            steps += 1
+        return {
+            "loss": total_loss / max(steps, 1),
+            "accuracy": 100.0 * correct / max(total, 1),
+        }
+
+    # ------------------------------------------------------------------
+    @torch.no_grad()
+    def _validate_epoch(self) -> Dict[str, float]:
+        self.model.eval()
+        if self.is_pipeline:
+            return self.pipeline_trainer.evaluate(
+                self.val_loader, self.tensor_shapes, self.device, self._dtype
+            )
+        total_loss, correct, total, steps = 0.0, 0, 0, 0
+        for batch in self.val_loader:
+            batch = PipelineDataLoader._normalize(batch)
+            x = batch["images"].to(self.device, non_blocking=True)
+            y = batch["labels"].to(self.device, non_blocking=True)
+            out = self.model(x)
+            total_loss += float(self.criterion(out, y))
+            correct += int((out.argmax(-1) == y).sum())
+            total += y.numel()
+            steps += 1
+        metrics = {
+            "loss": total_loss / max(steps, 1),
+            "accuracy": 100.0 * correct / max(total, 1),
+        }
+        if dist.is_initialized() and self.pg is not None and self.pg.dp_size > 1:
+            t = torch.tensor([metrics["loss"], metrics["accuracy"]], dtype=torch.float64)
+            if self.device.type == "cuda":
+                t = t.to(self.device)
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.pg.get_group("dp"))
+            t /= self.pg.dp_size
+            metrics = {"loss": float(t[0]), "accuracy": float(t[1])}
+        return metrics
+
+    # ------------------------------------------------------------------
+    def _save_checkpoint(self) -> None:
+        path = self.config.get("checkpoint_dir")
+        if not path:
+            return
+        os.makedirs(path, exist_ok=True)
+        if self._is_rank0():
+            torch.save(
+                {"model_state_dict": _unwrap(self.model).state_dict(), "config": self.config},
+                os.path.join(path, "checkpoint.pt"),
+            )

@@ -1,0 +1,129 @@
+"""PP tests: stage split + full 1F1B/AFAB loss equivalence vs a
+single-process run (a schedule-level oracle the reference lacked)."""
+
+import torch
+
+from conftest import run_distributed
+
+
+def test_distribute_layers():
+    from quintnet_amd.parallel import distribute_layers
+
+    assert distribute_layers(8, 2) == [[0, 1, 2, 3], [4, 5, 6, 7]]
+    assert distribute_layers(7, 2) == [[0, 1, 2, 3], [4, 5, 6]]
+    d = distribute_layers(12, 4)
+    assert sum(len(x) for x in d) == 12 and all(len(x) == 3 for x in d)
+
+
+def test_wrapper_stage_contents():
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel import PipelineParallelWrapper
+
+    m = Model(hidden_dim=32, n_heads=2, depth=4)
+    s0 = PipelineParallelWrapper(m, pp_rank=0, pp_size=2)
+    m2 = Model(hidden_dim=32, n_heads=2, depth=4)
+    s1 = PipelineParallelWrapper(m2, pp_rank=1, pp_size=2)
+    # first stage: embedding + 2 blocks; last: 2 blocks + head
+    assert len(s0.local_module) == 3  # embedding + 2 blocks
+    assert len(s1.local_module) == 3  # 2 blocks + head
+    x = torch.randn(2, 1, 28, 28)
+    h = s0(x)
+    assert h.shape == (2, 50, 32)
+    out = s1(h)
+    assert out.shape == (2, 10)
+    assert torch.isfinite(out).all()
+
+
+def _reference_losses(model, loader_batches, num_steps, grad_acc, lr):
+    """Single-process training oracle: same micro-batch stream, Adam, no clip."""
+    opt = torch.optim.Adam(model.parameters(), lr=lr)
+    crit = torch.nn.CrossEntropyLoss()
+    losses = []
+    it = iter(loader_batches)
+    for _ in range(num_steps):
+        opt.zero_grad()
+        tot = 0.0
+        for _ in range(grad_acc):
+            b = next(it)
+            out = model(b["images"])
+            loss = crit(out, b["labels"])
+            (loss / grad_acc).backward()
+            tot += float(loss.detach())
+        opt.step()
+        losses.append(tot / grad_acc)
+    return losses
+
+
+def _run_schedule(rank, world, schedule_name):
+    import copy
+
+    import torch.distributed as dist
+
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel import (
+        PipelineDataLoader,
+        PipelineParallelWrapper,
+        PipelineTrainer,
+    )
+    from quintnet_amd.utils.data import SyntheticMNIST
+
+    torch.manual_seed(42)
+    model = Model(hidden_dim=32, n_heads=2, depth=4)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    ref_model = copy.deepcopy(model)
+
+    grad_acc, micro_b, num_steps, lr = 4, 2, 3, 1e-3
+    ds = SyntheticMNIST(n=64, seed=9)
+    dl = torch.utils.data.DataLoader(ds, batch_size=micro_b, shuffle=False)
+
+    stage = PipelineParallelWrapper(model, pp_rank=rank, pp_size=world, pp_group=None)
+    opt = torch.optim.Adam(stage.parameters(), lr=lr)
+    pt = PipelineTrainer(
+        model=stage,
+        optimizer=opt,
+        criterion=torch.nn.CrossEntropyLoss(),
+        pp_rank=rank,
+        pp_size=world,
+        pp_group=None,
+        pp_group_ranks=list(range(world)),
+        schedule=schedule_name,
+        task_type="classification",
+        max_grad_norm=None,
+    )
+    loader = PipelineDataLoader(dl, grad_acc_steps=grad_acc, task_type="classification")
+    shapes = (micro_b, 50, 32)
+    losses = []
+    for _ in range(num_steps):
+        m = pt.train_step(loader, shapes, torch.device("cpu"), torch.float32)
+        if rank == world - 1:
+            losses.append(m["loss"])
+
+    if rank == world - 1:
+        batches = [
+            {"images": b["images"], "labels": b["labels"]}
+            for b in PipelineDataLoader(dl, grad_acc, "classification").dataloader
+        ]
+        # rebuild the same infinite stream
+        stream = []
+        while len(stream) < num_steps * grad_acc:
+            stream.extend(batches)
+        ref_losses = _reference_losses(ref_model, stream, num_steps, grad_acc, lr)
+        for a, b in zip(losses, ref_losses):
+            assert abs(a - b) < 1e-4, (losses, ref_losses)
+
+
+def _run_1f1b(rank, world):
+    _run_schedule(rank, world, "1f1b")
+
+
+def _run_afab(rank, world):
+    _run_schedule(rank, world, "afab")
+
+
+def test_1f1b_matches_single_process():
+    run_distributed(_run_1f1b, 2)
+
+
+def test_afab_matches_single_process():
+    run_distributed(_run_afab, 2)
