@@ -57,6 +57,11 @@ class PipelineParallelWrapper(nn.Module):
         self.is_first_stage = pp_rank == 0
         self.is_last_stage = pp_rank == pp_size - 1
 
+        # shape hints for the schedules (inter-stage activation is [B, seq, hidden])
+        src = model if model is not None else stage_module
+        self.seq_len = getattr(src, "seq_len", None)
+        self.hidden_dim = getattr(src, "hidden_dim", None)
+
         if stage_module is not None:
             self.local_module = stage_module
             self.layer_distribution = None

@@ -83,20 +83,12 @@ class Trainer:
             return len(batch[0])
 
     def _infer_seq_hidden(self, inner) -> tuple:
-        # local_module path (PP wrapper) or the bare model
-        if hasattr(inner, "local_module"):
-            inner = inner  # PipelineParallelWrapper
-        if hasattr(inner, "seq_len") and hasattr(inner, "hidden_dim"):
-            return inner.seq_len, inner.hidden_dim
-        seq = self.config.get("seq_len")
-        hidden = self.config.get("hidden_dim")
+        seq = getattr(inner, "seq_len", None) or self.config.get("seq_len")
+        hidden = getattr(inner, "hidden_dim", None) or self.config.get("hidden_dim")
         if seq is None or hidden is None:
-            # try embedded model attributes
-            m = getattr(inner, "local_module", inner)
-            for mod in ([m] + list(m.children()) if isinstance(m, nn.Module) else []):
-                if hasattr(mod, "seq_len") and hasattr(mod, "hidden_dim"):
-                    return mod.seq_len, mod.hidden_dim
-            raise ValueError("cannot infer pipeline tensor shapes; set config seq_len/hidden_dim")
+            raise ValueError(
+                "cannot infer pipeline tensor shapes; set config seq_len/hidden_dim"
+            )
         return int(seq), int(hidden)
 
     @property
@@ -125,12 +117,14 @@ class Trainer:
         return not dist.is_initialized() or dist.get_rank() == 0
 
     def _broadcast_metrics(self, train: Dict[str, float], val: Dict[str, float]) -> Dict[str, float]:
-        out = {f"train_{k}": v for k, v in train.items()}
-        out.update({f"val_{k}": v for k, v in val.items()})
-        if not dist.is_initialized() or not out:
+        # world-uniform key schema (non-last PP stages have no metrics;
+        # MAX all-reduce propagates the last stage's values — K17 pattern)
+        base = ("loss", "ppl") if self.task_type == "clm" else ("loss", "accuracy")
+        out = {f"train_{k}": float(train.get(k, 0.0)) for k in base}
+        if self.val_loader is not None:
+            out.update({f"val_{k}": float(val.get(k, 0.0)) for k in base})
+        if not dist.is_initialized():
             return out
-        # last-stage-only metrics -> everyone (MAX propagates the values;
-        # reference K17 pattern)
         keys = sorted(out)
         vals = torch.tensor([out[k] for k in keys], dtype=torch.float64)
         if self.device.type == "cuda":

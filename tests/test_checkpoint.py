@@ -1,0 +1,96 @@
+"""Checkpoint save → merge → reload round-trip (the reference had no
+such test) + staged distributed loading equivalence."""
+
+import os
+
+import torch
+
+from conftest import run_distributed
+
+
+def _save_and_merge(rank, world, tmpdir):
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.checkpoint import merge_checkpoints, save_sharded_checkpoint
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    pg = init_process_groups("cpu", [2, 2], ["tp", "pp"])
+    cfg = GPT2Config(
+        n_embd=32, n_layer=4, n_head=2, vocab_size=64, n_positions=32, dropout=0.0
+    )
+    torch.manual_seed(100 + rank)  # distinct shards per rank
+    stage = GPT2Stage(
+        cfg,
+        pp_rank=pg.pp_rank,
+        pp_size=pg.pp_size,
+        tp_group=pg.get_group("tp"),
+        tied_group=pg.get_tied_embedding_group(),
+    )
+    save_sharded_checkpoint(stage, tmpdir, name="final_model", pg_manager=pg)
+    dist.barrier()
+    if rank == 0:
+        out = os.path.join(tmpdir, "merged.pt")
+        merge_checkpoints(tmpdir, out, prefix="final_model")
+        merged = torch.load(out, map_location="cpu", weights_only=False)["model_state_dict"]
+        # HF-format keys present with full shapes
+        assert merged["transformer.wte.weight"].shape == (64, 32)
+        assert merged["transformer.h.0.attn.c_attn.weight"].shape == (32, 96)  # Conv1D [in, 3H]
+        assert merged["transformer.h.3.mlp.c_proj.weight"].shape == (128, 32)
+        assert merged["transformer.ln_f.weight"].shape == (32,)
+        assert merged["lm_head.weight"].shape == (64, 32)
+        assert "transformer.h.2.ln_1.weight" in merged  # PP offset remap worked
+
+
+def test_save_merge_roundtrip(tmp_path):
+    run_distributed(_save_and_merge, 4, str(tmp_path))
+
+
+def test_staged_load_matches_full_model(tmp_path):
+    """single-rank staged load of an HF-style checkpoint == direct load."""
+    from quintnet_amd.checkpoint.distributed_loading import load_gpt2_distributed
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import cross_entropy
+
+    cfg = GPT2Config(
+        n_embd=32, n_layer=2, n_head=2, vocab_size=64, n_positions=32, dropout=0.0
+    )
+    torch.manual_seed(0)
+    # build an HF-style state dict (Conv1D layout: [in, out])
+    hf = {}
+    hf["wte.weight"] = torch.randn(64, 32)
+    hf["wpe.weight"] = torch.randn(32, 32)
+    for i in range(2):
+        hf[f"h.{i}.ln_1.weight"] = torch.randn(32)
+        hf[f"h.{i}.ln_1.bias"] = torch.randn(32)
+        hf[f"h.{i}.ln_2.weight"] = torch.randn(32)
+        hf[f"h.{i}.ln_2.bias"] = torch.randn(32)
+        hf[f"h.{i}.attn.c_attn.weight"] = torch.randn(32, 96)
+        hf[f"h.{i}.attn.c_attn.bias"] = torch.randn(96)
+        hf[f"h.{i}.attn.c_proj.weight"] = torch.randn(32, 32)
+        hf[f"h.{i}.attn.c_proj.bias"] = torch.randn(32)
+        hf[f"h.{i}.mlp.c_fc.weight"] = torch.randn(32, 128)
+        hf[f"h.{i}.mlp.c_fc.bias"] = torch.randn(128)
+        hf[f"h.{i}.mlp.c_proj.weight"] = torch.randn(128, 32)
+        hf[f"h.{i}.mlp.c_proj.bias"] = torch.randn(32)
+    hf["ln_f.weight"] = torch.randn(32)
+    hf["ln_f.bias"] = torch.randn(32)
+    path = str(tmp_path / "ckpt.pt")
+    torch.save(hf, path)
+
+    state = load_gpt2_distributed(path, cfg, pp_rank=0, pp_size=1, tp_rank=0, tp_size=1)
+    stage = GPT2Stage.from_sharded_state_dict(cfg, state, pp_rank=0, pp_size=1)
+    ids = torch.randint(0, 64, (2, 8))
+    logits = stage(ids)
+    assert logits.shape == (2, 8, 64)
+    loss = cross_entropy(logits[:, :-1], ids[:, 1:])
+    assert torch.isfinite(loss)
+
+    # manual reference forward of layer 0 ln_1 + qkv to validate slicing
+    x = stage.embedding(ids)
+    import torch.nn.functional as F
+
+    ln = F.layer_norm(x, (32,), hf["h.0.ln_1.weight"], hf["h.0.ln_1.bias"])
+    qkv_ref = ln @ hf["h.0.attn.c_attn.weight"] + hf["h.0.attn.c_attn.bias"]
+    qkv_ours = stage.blocks[0].attn.c_attn(stage.blocks[0].ln_1(x))
+    assert torch.allclose(qkv_ours, qkv_ref, atol=1e-4), (qkv_ours - qkv_ref).abs().max()
