@@ -1,0 +1,290 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: GPT-2 (124M) 3D-parallel training step on MI355X.
+
+Driver contract: ``python bench.py --gpus N --steps K --warmup W`` (run
+under torch.distributed.run for N>1, one rank per GPU over RCCL).  Does
+W untimed warmup steps, times EXACTLY K optimizer steps bracketed by
+barrier + torch.cuda.synchronize on both sides, takes the MAX time over
+ranks, and rank 0 prints ONE JSON line.
+
+Mesh by GPU count (BASELINE.json configs): 1 → [1,1,1]; 2 → dp2;
+4 → dp4; 8 → the named 3D mesh [2,2,2] (dp×tp×pp, 1F1B).  Per-replica
+work is fixed (weak scaling): micro_batch 8 × grad_acc 4 = 32 sequences
+of 1024 tokens per optimizer step per DP replica, synthetic data,
+random-init weights, bf16 compute, ZeRO-1 AdamW.
+
+``--model vit`` benchmarks the ViT-MNIST config instead
+(micro 8 × grad_acc 8 = 64 images/replica/step).
+"""
+
+from __future__ import annotations
+
+import argparse
+import datetime
+import json
+import os
+import time
+
+import torch
+import torch.distributed as dist
+
+
+def parse_args():
+    p = argparse.ArgumentParser()
+    p.add_argument("--gpus", type=int, default=1)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--model", choices=["gpt2", "vit"], default="gpt2")
+    p.add_argument("--micro-batch", type=int, default=None)
+    p.add_argument("--grad-acc", type=int, default=None)
+    p.add_argument("--seq-len", type=int, default=1024)
+    p.add_argument("--tiny", action="store_true", help="tiny model for CPU smoke runs")
+    return p.parse_args()
+
+
+def pick_mesh(n: int):
+    if n == 8:
+        return [2, 2, 2]
+    return [n, 1, 1]
+
+
+def build_gpt2(args, pg, device, dtype):
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        PipelineParallelWrapper,
+    )
+
+    if args.tiny:
+        cfg = GPT2Config(n_embd=64, n_layer=2, n_head=2, vocab_size=512,
+                         n_positions=args.seq_len, dropout=0.0)
+    else:
+        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len))
+    tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
+    stage = GPT2Stage(
+        cfg,
+        pp_rank=pg.pp_rank,
+        pp_size=pg.pp_size,
+        tp_group=tp_group,
+        tied_group=pg.get_tied_embedding_group(),
+        device=device,
+        dtype=dtype,
+    )
+    stage.seq_len = args.seq_len
+    stage.hidden_dim = cfg.n_embd
+    model = stage
+    if pg.pp_size > 1:
+        model = PipelineParallelWrapper(stage_module=stage, pp_rank=pg.pp_rank,
+                                        pp_group=pg.get_group("pp"), pp_size=pg.pp_size,
+                                        device=device)
+        model.seq_len, model.hidden_dim = args.seq_len, cfg.n_embd
+    if pg.dp_size > 1:
+        model = DataParallel(model, DistributedConfig(
+            pg.dp_rank, pg.dp_size, pg.get_group("dp")))
+    return model, cfg
+
+
+def build_vit(args, pg, device, dtype):
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        PipelineParallelWrapper,
+        apply_tensor_parallel,
+    )
+
+    model = Model(hidden_dim=64, n_heads=4, depth=8)
+    model = model.to(device=device, dtype=dtype)
+    if pg.tp_size > 1:
+        apply_tensor_parallel(model, tp_size=pg.tp_size, tp_rank=pg.tp_rank,
+                              tp_group=pg.get_group("tp"), device=device)
+    if pg.pp_size > 1:
+        model = PipelineParallelWrapper(model=model, pp_rank=pg.pp_rank,
+                                        pp_group=pg.get_group("pp"), pp_size=pg.pp_size,
+                                        device=device)
+    if pg.dp_size > 1:
+        model = DataParallel(model, DistributedConfig(
+            pg.dp_rank, pg.dp_size, pg.get_group("dp")))
+    return model, None
+
+
+class _ListLoader:
+    """PipelineDataLoader-compatible infinite loader over device-resident batches."""
+
+    def __init__(self, batches, grad_acc_steps):
+        self.batches = batches
+        self.grad_acc_steps = grad_acc_steps
+        self.i = 0
+
+    def __len__(self):
+        return 1 << 30
+
+    def __iter__(self):
+        return self
+
+    def __next__(self):
+        b = self.batches[self.i % len(self.batches)]
+        self.i += 1
+        return b
+
+
+def main():
+    args = parse_args()
+    n = args.gpus
+    use_cuda = torch.cuda.is_available()
+    dtype = torch.bfloat16 if use_cuda else torch.float32
+    if args.tiny and not use_cuda:
+        dtype = torch.float32
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    if world != n and world > 1:
+        n = world
+    backend = "nccl" if use_cuda else "gloo"
+    if n > 1 and not dist.is_initialized():
+        dist.init_process_group(backend=backend,
+                                timeout=datetime.timedelta(seconds=600))
+    if use_cuda:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.optim import ZeroRedundancyAdamW
+    from quintnet_amd.parallel import PipelineTrainer
+
+    mesh = pick_mesh(n)
+    pg = init_process_groups("cuda" if use_cuda else "cpu", mesh, ["dp", "tp", "pp"])
+    device = pg.device
+    rank = pg.rank
+
+    if args.model == "gpt2":
+        micro_b = args.micro_batch or 8
+        grad_acc = args.grad_acc or 4
+        model, cfg = build_gpt2(args, pg, device, dtype)
+        vocab = cfg.vocab_size
+        g = torch.Generator(device="cpu").manual_seed(1234 + pg.dp_rank)
+        batches = []
+        for _ in range(4):
+            ids = torch.randint(0, vocab, (micro_b, args.seq_len), generator=g).to(device)
+            batches.append({"input_ids": ids, "labels": ids.clone()})
+        task = "clm"
+        seq, hidden = args.seq_len, cfg.n_embd
+        model_name = "gpt2-124M" if not args.tiny else "gpt2-tiny"
+    else:
+        micro_b = args.micro_batch or 8
+        grad_acc = args.grad_acc or 8
+        model, _ = build_vit(args, pg, device, dtype)
+        g = torch.Generator(device="cpu").manual_seed(1234 + pg.dp_rank)
+        batches = []
+        for _ in range(4):
+            x = torch.randn(micro_b, 1, 28, 28, generator=g).to(device=device, dtype=dtype)
+            y = torch.randint(0, 10, (micro_b,), generator=g).to(device)
+            batches.append({"images": x, "labels": y})
+        task = "classification"
+        seq, hidden = 50, 64
+        model_name = "vit-mnist"
+
+    loader = _ListLoader(batches, grad_acc)
+    optimizer = ZeroRedundancyAdamW(
+        model.parameters(), lr=1e-4, weight_decay=0.01,
+        dp_group=pg.get_group("dp") if pg.dp_size > 1 else None,
+    )
+
+    from quintnet_amd.ops import cross_entropy
+    from quintnet_amd.parallel import DataParallel
+
+    criterion = torch.nn.CrossEntropyLoss(ignore_index=-100)
+
+    if pg.pp_size > 1:
+        ptrainer = PipelineTrainer(
+            model=model, optimizer=optimizer, criterion=criterion,
+            pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+            pp_group=pg.get_group("pp"), pp_group_ranks=pg.get_group_ranks("pp"),
+            schedule="1f1b", task_type=task, max_grad_norm=1.0,
+        )
+        shapes = (micro_b, seq, hidden)
+
+        def step():
+            ptrainer.train_step(loader, shapes, device, dtype)
+
+    else:
+        is_ddp = isinstance(model, DataParallel)
+
+        def step():
+            if is_ddp:
+                model.require_backward_grad_sync = False
+            for i in range(grad_acc):
+                b = next(loader)
+                if is_ddp and i == grad_acc - 1:
+                    model.require_backward_grad_sync = True
+                if task == "clm":
+                    logits = model(b["input_ids"])
+                    loss = cross_entropy(logits[:, :-1, :], b["labels"][:, 1:], -100)
+                else:
+                    loss = criterion(model(b["images"]), b["labels"])
+                (loss / grad_acc).backward()
+            if is_ddp:
+                model.finalize_gradients()
+            optimizer.step()
+            if is_ddp:
+                model.zero_grad()
+            else:
+                optimizer.zero_grad()
+
+    def sync():
+        if dist.is_initialized():
+            dist.barrier()
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    for _ in range(args.warmup):
+        step()
+    sync()
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    t = torch.tensor([elapsed], dtype=torch.float64)
+    if dist.is_initialized():
+        if use_cuda:
+            t = t.to(device)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    elapsed = float(t[0])
+
+    samples_per_step = micro_b * grad_acc * pg.dp_size  # whole-job sequences/step
+    value = samples_per_step * args.steps / elapsed
+    ms_per_step = 1000.0 * elapsed / args.steps
+    parallelism = f"dp{pg.dp_size}_tp{pg.tp_size}_pp{pg.pp_size}"
+    result = {
+        "metric": "samples/sec",
+        "value": round(value, 3),
+        "unit": "samples/s",
+        "n_gpus": n,
+        "steps": args.steps,
+        "warmup": args.warmup,
+        "ms_per_step": round(ms_per_step, 3),
+        "higher_is_better": True,
+        "scaling": "weak",
+        "vs_baseline": None,
+        "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+        "data": "synthetic",
+        "config": {
+            "model": model_name,
+            "global_batch": samples_per_step,
+            "seq_len": seq if task == "clm" else None,
+            "micro_batch": micro_b,
+            "grad_acc": grad_acc,
+            "parallelism": parallelism,
+            "optimizer": "zero1-adamw",
+            "tokens_per_sec": round(value * seq, 1) if task == "clm" else None,
+        },
+    }
+    if rank == 0:
+        print(json.dumps(result), flush=True)
+    if dist.is_initialized():
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

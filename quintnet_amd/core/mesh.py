@@ -68,8 +68,12 @@ class MeshGenerator:
     def _setup_group_and_device(self) -> None:
         if not dist.is_initialized():
             backend = _default_backend(self.device_type)
-            init_kwargs = dict(backend=backend, timeout=self.timeout)
-            dist.init_process_group(**init_kwargs)
+            # single-process fallback (tests, bench --gpus 1 without torchrun)
+            os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+            os.environ.setdefault("MASTER_PORT", "29571")
+            os.environ.setdefault("RANK", "0")
+            os.environ.setdefault("WORLD_SIZE", "1")
+            dist.init_process_group(backend=backend, timeout=self.timeout)
         self.rank = dist.get_rank()
         self.world_size = dist.get_world_size()
         if self.mesh.numel() != self.world_size:
