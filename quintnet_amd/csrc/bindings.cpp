@@ -1,0 +1,264 @@
+// Torch bindings for the quintnet_amd CDNA4 kernel library (gfx950).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+#include <hip/hip_runtime.h>
+
+#include <vector>
+
+// launcher decls (defined in the .hip TUs)
+template <typename T>
+void layernorm_fwd_launch(const T*, const T*, const T*, T*, float*, float*, int, int, float, hipStream_t);
+template <typename T>
+void layernorm_bwd_launch(const T*, const T*, const T*, const float*, const float*, T*, float*, float*, int, int, hipStream_t);
+template <typename T>
+void softmax_fwd_launch(const T*, T*, long long, int, int, float, int, hipStream_t);
+template <typename T>
+void softmax_bwd_launch(const T*, const T*, T*, long long, int, float, hipStream_t);
+template <typename T>
+void ce_fwd_launch(const T*, const long long*, float*, float*, long long, int, long long, hipStream_t);
+template <typename T>
+void ce_bwd_launch(const T*, const long long*, const float*, T*, long long, int, long long, float, hipStream_t);
+template <typename TP, typename TG>
+void adamw_launch(TP*, float*, const TG*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
+template <typename T>
+void sumsq_launch(const T*, long long, float*, hipStream_t);
+void gemm_nt_launch(const unsigned short*, const unsigned short*, const unsigned short*,
+                    unsigned short*, unsigned short*, int, int, int, int, hipStream_t);
+
+namespace {
+
+hipStream_t cur_stream() {
+  return (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+}
+
+const unsigned short* bf16p(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+unsigned short* bf16p_mut(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+
+#define CHECK_GPU(t) TORCH_CHECK(t.is_cuda(), #t " must be on GPU")
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> gemm_nt(torch::Tensor x, torch::Tensor w,
+                                   c10::optional<torch::Tensor> bias, int64_t act) {
+  CHECK_GPU(x); CHECK_GPU(w);
+  TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16,
+              "gemm_nt: bf16 only");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "gemm_nt: shape mismatch");
+  TORCH_CHECK(x.size(1) % 8 == 0, "gemm_nt: K must be a multiple of 8");
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  int64_t M = xc.size(0), K = xc.size(1), N = wc.size(0);
+  auto out = torch::empty({M, N}, xc.options());
+  torch::Tensor pre;
+  unsigned short* prep = nullptr;
+  if (act != 0) {
+    pre = torch::empty({M, N}, xc.options());
+    prep = bf16p_mut(pre);
+  }
+  const unsigned short* bp = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value() && bias->defined()) {
+    bc = bias->contiguous();
+    TORCH_CHECK(bc.dtype() == torch::kBFloat16, "bias must be bf16");
+    bp = bf16p(bc);
+  }
+  gemm_nt_launch(bf16p(xc), bf16p(wc), bp, bf16p_mut(out), prep,
+                 (int)M, (int)N, (int)K, (int)act, cur_stream());
+  if (act != 0) return {out, pre};
+  return {out};
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
+                                         torch::Tensor b, double eps) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  int64_t H = xc.size(-1);
+  int64_t rows = xc.numel() / H;
+  auto y = torch::empty_like(xc);
+  auto mean = torch::empty({rows}, xc.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({rows}, xc.options().dtype(torch::kFloat32));
+  auto wc = w.contiguous(); auto bc = b.contiguous();
+  if (xc.dtype() == torch::kBFloat16) {
+    layernorm_fwd_launch<unsigned short>(
+        bf16p(xc), bf16p(wc), bf16p(bc), bf16p_mut(y),
+        mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)rows, (int)H,
+        (float)eps, cur_stream());
+  } else {
+    layernorm_fwd_launch<float>(
+        xc.data_ptr<float>(), wc.data_ptr<float>(), bc.data_ptr<float>(),
+        y.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+        (int)rows, (int)H, (float)eps, cur_stream());
+  }
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
+                                         torch::Tensor w, torch::Tensor mean,
+                                         torch::Tensor rstd) {
+  CHECK_GPU(x);
+  auto dyc = dy.contiguous(); auto xc = x.contiguous(); auto wc = w.contiguous();
+  int64_t H = xc.size(-1);
+  int64_t rows = xc.numel() / H;
+  auto dx = torch::empty_like(xc);
+  auto dwf = torch::zeros({H}, xc.options().dtype(torch::kFloat32));
+  auto dbf = torch::zeros({H}, xc.options().dtype(torch::kFloat32));
+  if (xc.dtype() == torch::kBFloat16) {
+    layernorm_bwd_launch<unsigned short>(
+        bf16p(dyc), bf16p(xc), bf16p(wc), mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), bf16p_mut(dx), dwf.data_ptr<float>(),
+        dbf.data_ptr<float>(), (int)rows, (int)H, cur_stream());
+  } else {
+    layernorm_bwd_launch<float>(
+        dyc.data_ptr<float>(), xc.data_ptr<float>(), wc.data_ptr<float>(),
+        mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
+        dwf.data_ptr<float>(), dbf.data_ptr<float>(), (int)rows, (int)H,
+        cur_stream());
+  }
+  return {dx, dwf.to(w.dtype()), dbf.to(w.dtype())};
+}
+
+// ---------------------------------------------------------------------------
+torch::Tensor softmax_fwd(torch::Tensor scores, double scale, bool causal) {
+  CHECK_GPU(scores);
+  auto sc = scores.contiguous();
+  int64_t S = sc.size(-1);
+  int64_t Tq = sc.size(-2);
+  long long rows = sc.numel() / S;
+  auto out = torch::empty_like(sc);
+  if (sc.dtype() == torch::kBFloat16) {
+    softmax_fwd_launch<unsigned short>(bf16p(sc), bf16p_mut(out), rows, (int)Tq,
+                                       (int)S, (float)scale, causal ? 1 : 0,
+                                       cur_stream());
+  } else {
+    softmax_fwd_launch<float>(sc.data_ptr<float>(), out.data_ptr<float>(), rows,
+                              (int)Tq, (int)S, (float)scale, causal ? 1 : 0,
+                              cur_stream());
+  }
+  return out;
+}
+
+torch::Tensor softmax_bwd(torch::Tensor p, torch::Tensor dp, double scale) {
+  CHECK_GPU(p);
+  auto pc = p.contiguous(); auto dpc = dp.contiguous();
+  int64_t S = pc.size(-1);
+  long long rows = pc.numel() / S;
+  auto ds = torch::empty_like(pc);
+  if (pc.dtype() == torch::kBFloat16) {
+    softmax_bwd_launch<unsigned short>(bf16p(pc), bf16p(dpc), bf16p_mut(ds), rows,
+                                       (int)S, (float)scale, cur_stream());
+  } else {
+    softmax_bwd_launch<float>(pc.data_ptr<float>(), dpc.data_ptr<float>(),
+                              ds.data_ptr<float>(), rows, (int)S, (float)scale,
+                              cur_stream());
+  }
+  return ds;
+}
+
+// ---------------------------------------------------------------------------
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
+                                             torch::Tensor target,
+                                             int64_t ignore_index) {
+  CHECK_GPU(logits);
+  auto lc = logits.contiguous();
+  auto tc = target.contiguous();
+  TORCH_CHECK(tc.dtype() == torch::kInt64, "target must be int64");
+  int64_t V = lc.size(-1);
+  long long rows = lc.numel() / V;
+  auto lse = torch::empty({(int64_t)rows}, lc.options().dtype(torch::kFloat32));
+  auto loss_sum = torch::zeros({}, lc.options().dtype(torch::kFloat32));
+  if (lc.dtype() == torch::kBFloat16) {
+    ce_fwd_launch<unsigned short>(bf16p(lc), (const long long*)tc.data_ptr<int64_t>(),
+                                  lse.data_ptr<float>(), loss_sum.data_ptr<float>(),
+                                  rows, (int)V, ignore_index, cur_stream());
+  } else {
+    ce_fwd_launch<float>(lc.data_ptr<float>(), (const long long*)tc.data_ptr<int64_t>(),
+                         lse.data_ptr<float>(), loss_sum.data_ptr<float>(), rows,
+                         (int)V, ignore_index, cur_stream());
+  }
+  auto n_valid = (tc != ignore_index).sum();
+  return {loss_sum, n_valid, lse};
+}
+
+torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
+                                torch::Tensor lse, int64_t n_valid,
+                                int64_t ignore_index) {
+  CHECK_GPU(logits);
+  auto lc = logits.contiguous();
+  auto tc = target.contiguous();
+  int64_t V = lc.size(-1);
+  long long rows = lc.numel() / V;
+  auto dl = torch::empty_like(lc);
+  float inv_n = 1.0f / (float)std::max<int64_t>(n_valid, 1);
+  if (lc.dtype() == torch::kBFloat16) {
+    ce_bwd_launch<unsigned short>(bf16p(lc), (const long long*)tc.data_ptr<int64_t>(),
+                                  lse.data_ptr<float>(), bf16p_mut(dl), rows,
+                                  (int)V, ignore_index, inv_n, cur_stream());
+  } else {
+    ce_bwd_launch<float>(lc.data_ptr<float>(), (const long long*)tc.data_ptr<int64_t>(),
+                         lse.data_ptr<float>(), dl.data_ptr<float>(), rows,
+                         (int)V, ignore_index, inv_n, cur_stream());
+  }
+  return dl;
+}
+
+// ---------------------------------------------------------------------------
+void adamw_step(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
+                torch::Tensor m, torch::Tensor v, int64_t step, double lr,
+                double beta1, double beta2, double eps, double wd) {
+  CHECK_GPU(param);
+  long long n = param.numel();
+  TORCH_CHECK(master.numel() == n && grad.numel() == n && m.numel() == n && v.numel() == n,
+              "adamw_step: size mismatch");
+  auto st = cur_stream();
+  float* mp = master.data_ptr<float>();
+  float* m1 = m.data_ptr<float>();
+  float* m2 = v.data_ptr<float>();
+  bool pbf = param.dtype() == torch::kBFloat16;
+  bool gbf = grad.dtype() == torch::kBFloat16;
+  if (pbf && gbf)
+    adamw_launch<unsigned short, unsigned short>(bf16p_mut(param), mp, bf16p(grad), m1, m2, n,
+        (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
+  else if (pbf && !gbf)
+    adamw_launch<unsigned short, float>(bf16p_mut(param), mp, grad.data_ptr<float>(), m1, m2, n,
+        (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
+  else if (!pbf && gbf)
+    adamw_launch<float, unsigned short>(param.data_ptr<float>(), mp, bf16p(grad), m1, m2, n,
+        (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
+  else
+    adamw_launch<float, float>(param.data_ptr<float>(), mp, grad.data_ptr<float>(), m1, m2, n,
+        (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
+}
+
+torch::Tensor multi_tensor_sumsq(std::vector<torch::Tensor> tensors) {
+  TORCH_CHECK(!tensors.empty(), "empty tensor list");
+  auto out = torch::zeros({}, tensors[0].options().dtype(torch::kFloat32));
+  auto st = cur_stream();
+  for (auto& t : tensors) {
+    auto tc = t.contiguous();
+    if (tc.dtype() == torch::kBFloat16)
+      sumsq_launch<unsigned short>(bf16p(tc), tc.numel(), out.data_ptr<float>(), st);
+    else
+      sumsq_launch<float>(tc.data_ptr<float>(), tc.numel(), out.data_ptr<float>(), st);
+  }
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("gemm_nt", &gemm_nt, "bf16 MFMA NT GEMM with fused bias/activation");
+  m.def("layernorm_fwd", &layernorm_fwd);
+  m.def("layernorm_bwd", &layernorm_bwd);
+  m.def("softmax_fwd", &softmax_fwd);
+  m.def("softmax_bwd", &softmax_bwd);
+  m.def("cross_entropy_fwd", &cross_entropy_fwd);
+  m.def("cross_entropy_bwd", &cross_entropy_bwd);
+  m.def("adamw_step", &adamw_step);
+  m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
+  m.attr("gfx") = "gfx950";
+}

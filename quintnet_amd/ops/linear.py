@@ -53,13 +53,14 @@ class LinearFunction(torch.autograd.Function):
         ctx.act = act
         pre_act = None
         if _backend.use_native(x) and _native_ok(x, weight):
-            out = _backend.ext().gemm_nt(
+            res = _backend.ext().gemm_nt(
                 x2d.contiguous(), weight.contiguous(),
                 bias if bias is not None else None, act,
             )
+            out = res[0]
             if act != _ACT_NONE:
-                # epilogue stores pre-activation too when an activation is fused
-                out, pre_act = out
+                # epilogue stores the pre-activation too when an act is fused
+                pre_act = res[1]
         else:
             out = torch.nn.functional.linear(x2d, weight, bias)
             if act == _ACT_GELU:

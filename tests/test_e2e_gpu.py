@@ -1,0 +1,71 @@
+"""GPU end-to-end: training steps on the native kernel path reduce loss."""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if not torch.cuda.is_available():
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+DEV = torch.device("cuda", 0)
+
+
+def test_gpt2_tiny_training_loss_decreases():
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import cross_entropy
+    from quintnet_amd.optim import ZeroRedundancyAdamW
+
+    torch.manual_seed(0)
+    cfg = GPT2Config(n_embd=128, n_layer=2, n_head=2, vocab_size=512,
+                     n_positions=128, dropout=0.0)
+    stage = GPT2Stage(cfg, device=DEV, dtype=torch.bfloat16)
+    opt = ZeroRedundancyAdamW(stage.parameters(), lr=3e-3)
+    ids = torch.randint(0, 512, (4, 64), device=DEV)
+    losses = []
+    for _ in range(20):
+        logits = stage(ids)
+        loss = cross_entropy(logits[:, :-1], ids[:, 1:])
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses[:3] + losses[-3:]
+
+
+def test_vit_training_loss_decreases():
+    from quintnet_amd.models import Model
+
+    torch.manual_seed(0)
+    m = Model(hidden_dim=64, n_heads=4, depth=2).to(DEV, torch.bfloat16)
+    opt = torch.optim.Adam(m.parameters(), lr=1e-3)
+    x = torch.randn(16, 1, 28, 28, device=DEV, dtype=torch.bfloat16)
+    y = torch.randint(0, 10, (16,), device=DEV)
+    crit = torch.nn.CrossEntropyLoss()
+    losses = []
+    for _ in range(30):
+        loss = crit(m(x).float(), y)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        losses.append(float(loss))
+    assert losses[-1] < losses[0] * 0.7, losses[:3] + losses[-3:]
+
+
+def test_gpt2_base_one_step_bf16():
+    """One full GPT-2 124M step in bf16 — the bench configuration."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import cross_entropy
+    from quintnet_amd.optim import ZeroRedundancyAdamW
+
+    cfg = GPT2Config(dropout=0.0)
+    stage = GPT2Stage(cfg, device=DEV, dtype=torch.bfloat16)
+    opt = ZeroRedundancyAdamW(stage.parameters(), lr=1e-4)
+    ids = torch.randint(0, cfg.vocab_size, (2, 1024), device=DEV)
+    logits = stage(ids)
+    loss = cross_entropy(logits[:, :-1], ids[:, 1:])
+    loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
+    assert 8.0 < float(loss) < 13.0  # ~ln(50257) at random init

@@ -1,0 +1,181 @@
+"""GPU numerics: each hand-written gfx950 kernel vs a plain PyTorch fp32
+reference of the same op (the SURVEY.md §4 oracle pattern)."""
+
+import math
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    DEV = torch.device("cuda", 0)
+else:
+    pytest.skip("needs MI355X", allow_module_level=True)
+
+
+def _ext():
+    from quintnet_amd.ops import ext, has_ext
+
+    assert has_ext(), "HIP extension must be present on a GPU box"
+    return ext()
+
+
+def test_ext_loaded():
+    assert _ext().gfx == "gfx950"
+
+
+@pytest.mark.parametrize("m,n,k", [(128, 128, 64), (256, 384, 128), (400, 50257 % 512 + 384, 72), (130, 100, 64), (8192 // 16, 768, 384)])
+def test_gemm_nt_vs_fp32(m, n, k):
+    torch.manual_seed(0)
+    a = torch.randn(m, k, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(n, k, device=DEV, dtype=torch.bfloat16)
+    bias = torch.randn(n, device=DEV, dtype=torch.bfloat16)
+    out = _ext().gemm_nt(a, b, bias, 0)[0]
+    ref = (a.float() @ b.float().t() + bias.float())
+    err = (out.float() - ref).abs().max() / max(float(ref.abs().max()), 1.0)
+    assert err < 2e-2, err  # bf16 inputs, fp32 accum
+    assert torch.isfinite(out.float()).all()
+
+
+def test_gemm_nt_gelu_epilogue():
+    torch.manual_seed(1)
+    a = torch.randn(192, 128, device=DEV, dtype=torch.bfloat16)
+    b = torch.randn(96, 128, device=DEV, dtype=torch.bfloat16)
+    bias = torch.randn(96, device=DEV, dtype=torch.bfloat16)
+    out, pre = _ext().gemm_nt(a, b, bias, 1)
+    ref_pre = a.float() @ b.float().t() + bias.float()
+    ref = torch.nn.functional.gelu(ref_pre, approximate="tanh")
+    assert (pre.float() - ref_pre).abs().max() / float(ref_pre.abs().max()) < 2e-2
+    scale = max(float(ref.abs().max()), 1.0)
+    assert (out.float() - ref).abs().max() / scale < 2e-2
+
+
+@pytest.mark.parametrize("rows,h", [(400, 64), (1024, 768), (77, 3072), (65, 100)])
+@pytest.mark.parametrize("dt", [torch.bfloat16, torch.float32])
+def test_layernorm_fwd_bwd(rows, h, dt):
+    torch.manual_seed(2)
+    x = torch.randn(rows, h, device=DEV, dtype=dt)
+    w = torch.randn(h, device=DEV, dtype=dt)
+    b = torch.randn(h, device=DEV, dtype=dt)
+    y, mean, rstd = _ext().layernorm_fwd(x, w, b, 1e-5)
+    ref = torch.nn.functional.layer_norm(x.float(), (h,), w.float(), b.float(), 1e-5)
+    tol = 2e-2 if dt == torch.bfloat16 else 1e-5
+    assert (y.float() - ref).abs().max() < tol * max(float(ref.abs().max()), 1.0)
+
+    dy = torch.randn(rows, h, device=DEV, dtype=dt)
+    dx, dw, db = _ext().layernorm_bwd(dy, x, w, mean, rstd)
+    xf = x.float().requires_grad_(True)
+    wf = w.float().requires_grad_(True)
+    bf = b.float().requires_grad_(True)
+    rr = torch.nn.functional.layer_norm(xf, (h,), wf, bf, 1e-5)
+    gx, gw, gb = torch.autograd.grad(rr, (xf, wf, bf), dy.float())
+    s = max(float(gx.abs().max()), 1.0)
+    dx_tol = 3e-2 * s if dt == torch.bfloat16 else 1e-4 * s
+    assert (dx.float() - gx).abs().max() < dx_tol
+    assert (dw.float() - gw).abs().max() / max(float(gw.abs().max()), 1.0) < 3e-2
+    assert (db.float() - gb).abs().max() / max(float(gb.abs().max()), 1.0) < 3e-2
+
+
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("dt", [torch.bfloat16, torch.float32])
+def test_softmax_fwd_bwd(causal, dt):
+    torch.manual_seed(3)
+    B, H, T, S = 2, 3, 64, 64
+    s = torch.randn(B, H, T, S, device=DEV, dtype=dt)
+    scale = 1.0 / math.sqrt(16)
+    p = _ext().softmax_fwd(s, scale, causal)
+    sf = s.float() * scale
+    if causal:
+        mask = torch.ones(T, S, dtype=torch.bool, device=DEV).tril()
+        sf = sf.masked_fill(~mask, float("-inf"))
+    ref = torch.softmax(sf, dim=-1)
+    tol = 1e-2 if dt == torch.bfloat16 else 1e-5
+    assert (p.float() - ref).abs().max() < tol
+    if causal:
+        assert float(p.float()[..., 0, 1:].abs().max()) == 0.0
+
+    dp = torch.randn_like(s)
+    ds = _ext().softmax_bwd(p, dp, scale)
+    row = (ref * dp.float()).sum(-1, keepdim=True)
+    ref_ds = scale * ref * (dp.float() - row)
+    assert (ds.float() - ref_ds).abs().max() < (3e-2 if dt == torch.bfloat16 else 1e-5)
+
+
+@pytest.mark.parametrize("dt", [torch.bfloat16, torch.float32])
+def test_cross_entropy_fwd_bwd(dt):
+    torch.manual_seed(4)
+    N, V = 512, 50257
+    logits = torch.randn(N, V, device=DEV, dtype=dt)
+    tgt = torch.randint(0, V, (N,), device=DEV)
+    tgt[::5] = -100
+    loss_sum, n_valid, lse = _ext().cross_entropy_fwd(logits, tgt, -100)
+    loss = loss_sum / n_valid.clamp_min(1).float()
+    ref = torch.nn.functional.cross_entropy(logits.float(), tgt, ignore_index=-100)
+    tol = 2e-2 if dt == torch.bfloat16 else 1e-4
+    assert abs(float(loss) - float(ref)) < tol, (float(loss), float(ref))
+
+    dl = _ext().cross_entropy_bwd(logits, tgt, lse, int(n_valid), -100)
+    lf = logits.float().requires_grad_(True)
+    rr = torch.nn.functional.cross_entropy(lf, tgt, ignore_index=-100)
+    (g,) = torch.autograd.grad(rr, lf)
+    assert (dl.float() - g).abs().max() < 1e-3
+
+
+def test_adamw_step_vs_torch():
+    torch.manual_seed(5)
+    n = 10007
+    master = torch.randn(n, device=DEV, dtype=torch.float32)
+    param = master.to(torch.bfloat16)
+    grad = torch.randn(n, device=DEV, dtype=torch.bfloat16)
+    m = torch.zeros(n, device=DEV)
+    v = torch.zeros(n, device=DEV)
+
+    ref = master.clone().requires_grad_(True)
+    opt = torch.optim.AdamW([ref], lr=1e-2, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.01)
+    for step in range(1, 4):
+        _ext().adamw_step(param, master, grad, m, v, step, 1e-2, 0.9, 0.999, 1e-8, 0.01)
+        ref.grad = grad.float()
+        opt.step()
+    assert (master - ref.detach()).abs().max() < 1e-4
+    assert (param.float() - master).abs().max() < 0.02  # bf16 rounding
+
+
+def test_multi_tensor_sumsq():
+    xs = [torch.randn(1000, device=DEV), torch.randn(37, device=DEV, dtype=torch.bfloat16)]
+    out = _ext().multi_tensor_sumsq(xs)
+    ref = sum(float(x.float().pow(2).sum()) for x in xs)
+    assert abs(float(out) - ref) / ref < 1e-2
+
+
+def test_fused_attention_op_vs_sdpa():
+    from quintnet_amd.ops import attention
+
+    torch.manual_seed(6)
+    q = torch.randn(2, 4, 128, 64, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(2, 4, 128, 64, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(2, 4, 128, 64, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    out = attention(q, k, v, causal=True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), k.float(), v.float(), is_causal=True
+    )
+    assert (out.float() - ref).abs().max() < 3e-2
+    out.sum().backward()
+    assert torch.isfinite(q.grad.float()).all()
+
+
+def test_model_linear_uses_native_gemm():
+    """The TP-linear path must run the hand-written GEMM on GPU."""
+    from quintnet_amd.ops import linear
+
+    x = torch.randn(256, 128, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(256, 128, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    b = torch.randn(256, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y = linear(x, w, b, activation="gelu")
+    ref = torch.nn.functional.gelu(
+        torch.nn.functional.linear(x.float(), w.float(), b.float()), approximate="tanh"
+    )
+    assert (y.float() - ref).abs().max() / max(float(ref.abs().max()), 1.0) < 3e-2
+    y.sum().backward()
+    for t in (x, w, b):
+        assert torch.isfinite(t.grad.float()).all()
