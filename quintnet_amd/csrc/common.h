@@ -69,6 +69,7 @@ __device__ __forceinline__ float wave_reduce_max(float v) {
 }
 
 // block reduction via LDS (block = nwaves x 64)
+// Result is returned to EVERY thread of the block (broadcast through LDS).
 template <int MAX_WAVES>
 __device__ __forceinline__ float block_reduce_sum(float v, float* lds_scratch) {
   const int lane = threadIdx.x & (QN_WAVE - 1);
@@ -76,11 +77,16 @@ __device__ __forceinline__ float block_reduce_sum(float v, float* lds_scratch) {
   v = wave_reduce_sum(v);
   if (lane == 0) lds_scratch[wave] = v;
   __syncthreads();
-  const int nwaves = (blockDim.x + QN_WAVE - 1) / QN_WAVE;
-  v = (threadIdx.x < nwaves) ? lds_scratch[threadIdx.x] : 0.f;
-#pragma unroll
-  for (int off = MAX_WAVES / 2; off > 0; off >>= 1) v += __shfl_xor(v, off, QN_WAVE);
-  return __shfl(v, 0, QN_WAVE);
+  if (threadIdx.x == 0) {
+    const int nwaves = (blockDim.x + QN_WAVE - 1) / QN_WAVE;
+    float acc = 0.f;
+    for (int i = 0; i < nwaves; ++i) acc += lds_scratch[i];
+    lds_scratch[0] = acc;
+  }
+  __syncthreads();
+  v = lds_scratch[0];
+  __syncthreads();  // protect scratch for back-to-back reductions
+  return v;
 }
 
 template <int MAX_WAVES>
@@ -90,11 +96,16 @@ __device__ __forceinline__ float block_reduce_max(float v, float* lds_scratch) {
   v = wave_reduce_max(v);
   if (lane == 0) lds_scratch[wave] = v;
   __syncthreads();
-  const int nwaves = (blockDim.x + QN_WAVE - 1) / QN_WAVE;
-  v = (threadIdx.x < nwaves) ? lds_scratch[threadIdx.x] : -INFINITY;
-#pragma unroll
-  for (int off = MAX_WAVES / 2; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, QN_WAVE));
-  return __shfl(v, 0, QN_WAVE);
+  if (threadIdx.x == 0) {
+    const int nwaves = (blockDim.x + QN_WAVE - 1) / QN_WAVE;
+    float acc = -INFINITY;
+    for (int i = 0; i < nwaves; ++i) acc = fmaxf(acc, lds_scratch[i]);
+    lds_scratch[0] = acc;
+  }
+  __syncthreads();
+  v = lds_scratch[0];
+  __syncthreads();
+  return v;
 }
 
 // tanh-approx GELU matching torch.nn.functional.gelu(approximate="tanh")
