@@ -184,13 +184,15 @@ def main():
         model_name = "vit-mnist"
 
     loader = _ListLoader(batches, grad_acc)
-    optimizer = ZeroRedundancyAdamW(
-        model.parameters(), lr=1e-4, weight_decay=0.01,
-        dp_group=pg.get_group("dp") if pg.dp_size > 1 else None,
-    )
-
     from quintnet_amd.ops import cross_entropy
     from quintnet_amd.parallel import DataParallel
+
+    zero_kw = dict(lr=1e-4, weight_decay=0.01,
+                   dp_group=pg.get_group("dp") if pg.dp_size > 1 else None)
+    if isinstance(model, DataParallel):
+        optimizer = ZeroRedundancyAdamW.from_ddp(model, **zero_kw)
+    else:
+        optimizer = ZeroRedundancyAdamW(model.parameters(), **zero_kw)
 
     criterion = torch.nn.CrossEntropyLoss(ignore_index=-100)
 

@@ -17,7 +17,9 @@ void softmax_bwd_launch(const T*, const T*, T*, long long, int, float, hipStream
 template <typename T>
 void ce_fwd_launch(const T*, const long long*, float*, float*, long long, int, long long, hipStream_t);
 template <typename T>
-void ce_bwd_launch(const T*, const long long*, const float*, T*, long long, int, long long, float, hipStream_t);
+void ce_bwd_launch(const T*, const long long*, const float*, const float*, T*, long long, int, long long, float, hipStream_t);
+template <typename T>
+void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
 template <typename TP, typename TG>
 void adamw_launch(TP*, float*, const TG*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
@@ -186,7 +188,8 @@ std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
 
 torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
                                 torch::Tensor lse, int64_t n_valid,
-                                int64_t ignore_index) {
+                                int64_t ignore_index,
+                                c10::optional<torch::Tensor> grad_scale) {
   CHECK_GPU(logits);
   auto lc = logits.contiguous();
   auto tc = target.contiguous();
@@ -194,16 +197,38 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   long long rows = lc.numel() / V;
   auto dl = torch::empty_like(lc);
   float inv_n = 1.0f / (float)std::max<int64_t>(n_valid, 1);
+  const float* gp = nullptr;
+  torch::Tensor gs;
+  if (grad_scale.has_value() && grad_scale->defined()) {
+    gs = grad_scale->to(torch::kFloat32).contiguous();
+    TORCH_CHECK(gs.numel() == 1, "grad_scale must be a scalar tensor");
+    gp = gs.data_ptr<float>();
+  }
   if (lc.dtype() == torch::kBFloat16) {
     ce_bwd_launch<unsigned short>(bf16p(lc), (const long long*)tc.data_ptr<int64_t>(),
-                                  lse.data_ptr<float>(), bf16p_mut(dl), rows,
+                                  lse.data_ptr<float>(), gp, bf16p_mut(dl), rows,
                                   (int)V, ignore_index, inv_n, cur_stream());
   } else {
     ce_bwd_launch<float>(lc.data_ptr<float>(), (const long long*)tc.data_ptr<int64_t>(),
-                         lse.data_ptr<float>(), dl.data_ptr<float>(), rows,
+                         lse.data_ptr<float>(), gp, dl.data_ptr<float>(), rows,
                          (int)V, ignore_index, inv_n, cur_stream());
   }
   return dl;
+}
+
+torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor pre, int64_t act) {
+  CHECK_GPU(dy);
+  auto dyc = dy.contiguous();
+  auto prec = pre.contiguous();
+  TORCH_CHECK(dyc.numel() == prec.numel(), "act_bwd size mismatch");
+  auto dx = torch::empty_like(dyc);
+  if (dyc.dtype() == torch::kBFloat16)
+    act_bwd_launch<unsigned short>(bf16p(dyc), bf16p(prec), bf16p_mut(dx),
+                                   dyc.numel(), (int)act, cur_stream());
+  else
+    act_bwd_launch<float>(dyc.data_ptr<float>(), prec.data_ptr<float>(),
+                          dx.data_ptr<float>(), dyc.numel(), (int)act, cur_stream());
+  return dx;
 }
 
 // ---------------------------------------------------------------------------
@@ -258,6 +283,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("softmax_bwd", &softmax_bwd);
   m.def("cross_entropy_fwd", &cross_entropy_fwd);
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
+  m.def("act_bwd", &act_bwd, "fused activation backward");
   m.def("adamw_step", &adamw_step);
   m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
   m.attr("gfx") = "gfx950";

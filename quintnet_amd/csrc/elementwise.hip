@@ -1,0 +1,39 @@
+// Fused elementwise kernels: activation backward (GELU/ReLU) in one pass.
+// Replaces the ~10-kernel eager float chain the Python GELU backward cost
+// (rocprof r01: tanh/pow/mul/add ≈8% of the GPT-2 step).
+#include "common.h"
+
+__device__ __forceinline__ float dgelu_tanh(float x) {
+  const float c = 0.7978845608028654f;
+  float x2 = x * x;
+  float t = tanhf(c * (x + 0.044715f * x * x2));
+  return 0.5f * (1.f + t) + 0.5f * x * (1.f - t * t) * c * (1.f + 3.f * 0.044715f * x2);
+}
+
+// ACT: 1 = gelu, 2 = relu
+template <typename T, int ACT>
+__global__ void act_bwd_kernel(const T* __restrict__ dy, const T* __restrict__ pre,
+                               T* __restrict__ dx, long long n) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n; i += stride) {
+    float g = ld_as_f32(dy + i);
+    float p = ld_as_f32(pre + i);
+    float v;
+    if constexpr (ACT == 1) v = g * dgelu_tanh(p);
+    else v = (p > 0.f) ? g : 0.f;
+    st_from_f32(dx + i, v);
+  }
+}
+
+template <typename T>
+void act_bwd_launch(const T* dy, const T* pre, T* dx, long long n, int act,
+                    hipStream_t stream) {
+  long long blocks = min((n + 255) / 256, (long long)2048);
+  if (act == 1)
+    hipLaunchKernelGGL((act_bwd_kernel<T, 1>), dim3((unsigned)blocks), dim3(256), 0, stream, dy, pre, dx, n);
+  else
+    hipLaunchKernelGGL((act_bwd_kernel<T, 2>), dim3((unsigned)blocks), dim3(256), 0, stream, dy, pre, dx, n);
+}
+
+template void act_bwd_launch<float>(const float*, const float*, float*, long long, int, hipStream_t);
+template void act_bwd_launch<unsigned short>(const unsigned short*, const unsigned short*, unsigned short*, long long, int, hipStream_t);

@@ -2,11 +2,15 @@
 // Replaces the implicit eager LayerNorm of the reference.
 // Rows are B*T; H ∈ {64, 768, 3072, ...}. bf16 loads vectorized as s16x8
 // (guide G13: scalar bf16 ≈2× slower), one pass of sum/sumsq per row.
+//
+// Backward is two kernels: dx (wave-per-row, no atomics) and a separate
+// dw/db column reduction (block-per-column-tile × row-chunks, one
+// atomicAdd per column per chunk) — the v1 per-element atomic version
+// was 20% of the whole GPT-2 step (rocprof, profiles/r01).
 #include "common.h"
 
 // ---------------------------------------------------------------------------
 // forward: y = (x - mean) * rstd * w + b ; saves mean, rstd (f32 per row)
-// grid: (rows / WPB) blocks of WPB waves; wave r handles row r.
 // ---------------------------------------------------------------------------
 template <typename T, int WPB>
 __global__ void layernorm_fwd_kernel(
@@ -21,22 +25,15 @@ __global__ void layernorm_fwd_kernel(
   T* yr = y + row * H;
 
   float sum = 0.f, sq = 0.f;
-  // strided per-lane pass (vectorize by 8 when H % (64*8) allows)
-  if constexpr (sizeof(T) == 2) {
-    if ((H & 511) == 0) {  // H multiple of 512: 8-wide vector loads
-      for (int i = lane * 8; i < H; i += QN_WAVE * 8) {
-        s16x8 v = *reinterpret_cast<const s16x8*>(xr + i);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          float f = bf16_to_f32((unsigned short)v[j]);
-          sum += f; sq += f * f;
-        }
-      }
-    } else {
-      for (int i = lane; i < H; i += QN_WAVE) {
-        float f = ld_as_f32(reinterpret_cast<const unsigned short*>(xr) + i);
-        sum += f; sq += f * f;
-      }
+  // cache up to 16 elems/lane in registers (H <= 1024) to avoid re-reads
+  float cache[16];
+  const bool cached = H <= QN_WAVE * 16;
+  if (cached) {
+    int n = 0;
+    for (int i = lane; i < H; i += QN_WAVE, ++n) {
+      float f = ld_as_f32(xr + i);
+      cache[n] = f;
+      sum += f; sq += f * f;
     }
   } else {
     for (int i = lane; i < H; i += QN_WAVE) {
@@ -51,26 +48,33 @@ __global__ void layernorm_fwd_kernel(
   const float rstd = rsqrtf(var + eps);
   if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
 
-  for (int i = lane; i < H; i += QN_WAVE) {
-    float f = ld_as_f32(reinterpret_cast<const T*>(xr) + i);
-    float wi = ld_as_f32(w + i);
-    float bi = ld_as_f32(b + i);
-    st_from_f32(yr + i, (f - mean) * rstd * wi + bi);
+  if (cached) {
+    int n = 0;
+    for (int i = lane; i < H; i += QN_WAVE, ++n) {
+      float wi = ld_as_f32(w + i);
+      float bi = ld_as_f32(b + i);
+      st_from_f32(yr + i, (cache[n] - mean) * rstd * wi + bi);
+    }
+  } else {
+    for (int i = lane; i < H; i += QN_WAVE) {
+      float f = ld_as_f32(xr + i);
+      float wi = ld_as_f32(w + i);
+      float bi = ld_as_f32(b + i);
+      st_from_f32(yr + i, (f - mean) * rstd * wi + bi);
+    }
   }
 }
 
 // ---------------------------------------------------------------------------
-// backward:
+// backward dx (wave per row):
 //   xhat = (x - mean) * rstd;  wdy = w * dy
 //   dx = (wdy - mean(wdy) - xhat * mean(wdy * xhat)) * rstd
-//   dw = sum_rows(dy * xhat);  db = sum_rows(dy)        (fp32 atomics)
 // ---------------------------------------------------------------------------
 template <typename T, int WPB>
-__global__ void layernorm_bwd_kernel(
+__global__ void layernorm_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
-    T* __restrict__ dx, float* __restrict__ dw, float* __restrict__ db,
-    int rows, int H) {
+    T* __restrict__ dx, int rows, int H) {
   const int lane = threadIdx.x & (QN_WAVE - 1);
   const int wave = threadIdx.x / QN_WAVE;
   const long long row = (long long)blockIdx.x * WPB + wave;
@@ -80,32 +84,66 @@ __global__ void layernorm_bwd_kernel(
   T* dxr = dx + row * H;
   const float mean = mean_in[row], rstd = rstd_in[row];
 
+  float cx[16], cwdy[16];
+  const bool cached = H <= QN_WAVE * 16;
   float c1 = 0.f, c2 = 0.f;
-  for (int i = lane; i < H; i += QN_WAVE) {
-    float xf = ld_as_f32(xr + i);
-    float dyf = ld_as_f32(dyr + i);
-    float wi = ld_as_f32(w + i);
-    float xhat = (xf - mean) * rstd;
-    float wdy = wi * dyf;
-    c1 += wdy;
-    c2 += wdy * xhat;
+  if (cached) {
+    int n = 0;
+    for (int i = lane; i < H; i += QN_WAVE, ++n) {
+      float xhat = (ld_as_f32(xr + i) - mean) * rstd;
+      float wdy = ld_as_f32(w + i) * ld_as_f32(dyr + i);
+      cx[n] = xhat; cwdy[n] = wdy;
+      c1 += wdy; c2 += wdy * xhat;
+    }
+  } else {
+    for (int i = lane; i < H; i += QN_WAVE) {
+      float xhat = (ld_as_f32(xr + i) - mean) * rstd;
+      float wdy = ld_as_f32(w + i) * ld_as_f32(dyr + i);
+      c1 += wdy; c2 += wdy * xhat;
+    }
   }
   c1 = wave_reduce_sum(c1) / H;
   c2 = wave_reduce_sum(c2) / H;
 
-  for (int i = lane; i < H; i += QN_WAVE) {
-    float xf = ld_as_f32(xr + i);
-    float dyf = ld_as_f32(dyr + i);
-    float wi = ld_as_f32(w + i);
-    float xhat = (xf - mean) * rstd;
-    float wdy = wi * dyf;
-    st_from_f32(dxr + i, (wdy - c1 - xhat * c2) * rstd);
-    atomicAdd(dw + i, dyf * xhat);
-    atomicAdd(db + i, dyf);
+  if (cached) {
+    int n = 0;
+    for (int i = lane; i < H; i += QN_WAVE, ++n)
+      st_from_f32(dxr + i, (cwdy[n] - c1 - cx[n] * c2) * rstd);
+  } else {
+    for (int i = lane; i < H; i += QN_WAVE) {
+      float xhat = (ld_as_f32(xr + i) - mean) * rstd;
+      float wdy = ld_as_f32(w + i) * ld_as_f32(dyr + i);
+      st_from_f32(dxr + i, (wdy - c1 - xhat * c2) * rstd);
+    }
   }
 }
 
-// ---- launchers (called from bindings.cpp) ----------------------------------
+// ---------------------------------------------------------------------------
+// backward dw/db: column reduction.  grid = (ceil(H/256), ROW_CHUNKS);
+// each thread owns one column within its row chunk, one atomicAdd per
+// column per chunk.
+// ---------------------------------------------------------------------------
+template <typename T>
+__global__ void layernorm_bwd_dwdb_kernel(
+    const T* __restrict__ dy, const T* __restrict__ x,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    float* __restrict__ dw, float* __restrict__ db, int rows, int H) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= H) return;
+  const int r0 = blockIdx.y;
+  const int nchunks = gridDim.y;
+  float aw = 0.f, ab = 0.f;
+  for (long long r = r0; r < rows; r += nchunks) {
+    float dyv = ld_as_f32(dy + r * H + col);
+    float xhat = (ld_as_f32(x + r * H + col) - mean_in[r]) * rstd_in[r];
+    aw += dyv * xhat;
+    ab += dyv;
+  }
+  atomicAdd(dw + col, aw);
+  atomicAdd(db + col, ab);
+}
+
+// ---- launchers -------------------------------------------------------------
 template <typename T>
 void layernorm_fwd_launch(const T* x, const T* w, const T* b, T* y,
                           float* mean, float* rstd, int rows, int H, float eps,
@@ -122,8 +160,12 @@ void layernorm_bwd_launch(const T* dy, const T* x, const T* w, const float* mean
                           int rows, int H, hipStream_t stream) {
   constexpr int WPB = 4;
   dim3 grid((rows + WPB - 1) / WPB);
-  hipLaunchKernelGGL((layernorm_bwd_kernel<T, WPB>), grid, dim3(WPB * QN_WAVE), 0,
-                     stream, dy, x, w, mean, rstd, dx, dw, db, rows, H);
+  hipLaunchKernelGGL((layernorm_bwd_dx_kernel<T, WPB>), grid, dim3(WPB * QN_WAVE),
+                     0, stream, dy, x, w, mean, rstd, dx, rows, H);
+  int colblocks = (H + 255) / 256;
+  int chunks = min(max(rows / 64, 1), 256);
+  hipLaunchKernelGGL((layernorm_bwd_dwdb_kernel<T>), dim3(colblocks, chunks),
+                     dim3(256), 0, stream, dy, x, mean, rstd, dw, db, rows, H);
 }
 
 // explicit instantiations

@@ -37,13 +37,12 @@ class GPT2Trainer(Trainer):
             else None
         )
         if use_zero:
-            self.optimizer = ZeroRedundancyAdamW(
-                self.model.parameters(),
-                lr=self.lr,
-                weight_decay=wd,
-                dp_group=dp_group,
-                max_grad_norm=None,  # clipping handled by the schedule
-            )
+            zkw = dict(lr=self.lr, weight_decay=wd, dp_group=dp_group,
+                       max_grad_norm=None)  # clipping handled by the schedule
+            if isinstance(self.model, DataParallel):
+                self.optimizer = ZeroRedundancyAdamW.from_ddp(self.model, **zkw)
+            else:
+                self.optimizer = ZeroRedundancyAdamW(self.model.parameters(), **zkw)
         else:
             self.optimizer = torch.optim.AdamW(
                 self.model.parameters(), lr=self.lr, weight_decay=wd

@@ -77,15 +77,19 @@ class LinearFunction(torch.autograd.Function):
     def backward(ctx, grad_out):
         x2d, weight, pre_act = ctx.saved_tensors
         g = grad_out.reshape(-1, grad_out.shape[-1])
-        if ctx.act == _ACT_GELU:
-            pa = pre_act.float()
-            # d/dx gelu_tanh(x)
-            c = 0.7978845608028654  # sqrt(2/pi)
-            t = torch.tanh(c * (pa + 0.044715 * pa.pow(3)))
-            dg = 0.5 * (1.0 + t) + 0.5 * pa * (1.0 - t * t) * c * (1.0 + 3 * 0.044715 * pa.pow(2))
-            g = (g.float() * dg).to(g.dtype)
-        elif ctx.act == _ACT_RELU:
-            g = g * (pre_act > 0).to(g.dtype)
+        if ctx.act != _ACT_NONE:
+            if _backend.use_native(g) and _backend.has_ext():
+                g = _backend.ext().act_bwd(g, pre_act, ctx.act)  # fused single pass
+            elif ctx.act == _ACT_GELU:
+                pa = pre_act.float()
+                c = 0.7978845608028654  # sqrt(2/pi): d/dx gelu_tanh(x)
+                t = torch.tanh(c * (pa + 0.044715 * pa.pow(3)))
+                dg = 0.5 * (1.0 + t) + 0.5 * pa * (1.0 - t * t) * c * (
+                    1.0 + 3 * 0.044715 * pa.pow(2)
+                )
+                g = (g.float() * dg).to(g.dtype)
+            elif ctx.act == _ACT_RELU:
+                g = g * (pre_act > 0).to(g.dtype)
         grad_x = grad_w = grad_b = None
         if ctx.needs_input_grad[0]:
             grad_x = (g @ weight).reshape(ctx.x_shape)

@@ -7,13 +7,17 @@ BASELINE.json's GPT-2 config requires it.  MI355X-native design:
   (model dtype, padded to dp_size), so the post-step parameter
   all-gather is a single RCCL ``all_gather_into_tensor`` on the flat
   buffer — no per-param traffic;
+* gradients live in ONE matching flat buffer.  Three modes:
+  - shared: reuse the DataParallel wrapper's bucket buffer
+    (``ZeroRedundancyAdamW.from_ddp``) — zero copies, grads arrive
+    already averaged by the bucketed RCCL all-reduce;
+  - owned: no DDP — the optimizer allocates the buffer and re-points
+    ``p.grad`` into it, so autograd accumulates in place;
+  - copy: fallback when grads are externally managed.
 * each DP rank owns a 1/dp_size shard and keeps ONLY its shard's fp32
   master weights + Adam m/v (the ZeRO-1 memory win);
-* the update itself is the fused multi-tensor AdamW HIP kernel
-  (csrc/adamw.hip) over the flat shard;
-* gradients arrive already averaged by DataParallel's bucketed
-  all-reduce; ``step()`` copies them into the flat grad layout, updates
-  the owned shard, and all-gathers the new params.
+* the update is the fused AdamW HIP kernel (csrc/adamw.hip) over the
+  flat shard; grad-norm clipping is two kernels on the flat buffer.
 """
 
 from __future__ import annotations
@@ -38,6 +42,7 @@ class ZeroRedundancyAdamW:
         weight_decay: float = 0.01,
         dp_group=None,
         max_grad_norm: Optional[float] = None,
+        grad_buffer: Optional[torch.Tensor] = None,
     ):
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
@@ -61,17 +66,35 @@ class ZeroRedundancyAdamW:
         self.padded = ((total + self.dp_size - 1) // self.dp_size) * self.dp_size
         self.shard_size = self.padded // self.dp_size
 
+        uniform_dtype = all(p.dtype == self.dtype for p in self.params)
+
         # one flat param buffer; params become views into it
         self.flat_param = torch.zeros(self.padded, dtype=self.dtype, device=device)
         off = 0
         self._offsets: List[int] = []
         for p in self.params:
             n = p.numel()
-            self.flat_param[off : off + n].copy_(p.data.reshape(-1))
-            p.data = self.flat_param[off : off + n].view_as(p.data)
+            self.flat_param[off : off + n].copy_(p.data.reshape(-1).to(self.dtype))
+            if p.dtype == self.dtype:
+                p.data = self.flat_param[off : off + n].view_as(p.data)
             self._offsets.append(off)
             off += n
-        self.flat_grad = torch.zeros_like(self.flat_param)
+
+        # grad buffer: shared (DDP) / owned / copy
+        if grad_buffer is not None:
+            assert grad_buffer.numel() >= self.padded, "shared grad buffer too small"
+            self.flat_grad = grad_buffer
+            self._grad_mode = "shared"
+        elif uniform_dtype and all(p.grad is None for p in self.params):
+            self.flat_grad = torch.zeros(self.padded, dtype=self.dtype, device=device)
+            off = 0
+            for p in self.params:
+                p.grad = self.flat_grad[off : off + p.numel()].view_as(p.data)
+                off += p.numel()
+            self._grad_mode = "owned"
+        else:
+            self.flat_grad = torch.zeros(self.padded, dtype=self.dtype, device=device)
+            self._grad_mode = "copy"
 
         s = self.dp_rank * self.shard_size
         e = s + self.shard_size
@@ -80,27 +103,40 @@ class ZeroRedundancyAdamW:
         self.exp_avg = torch.zeros_like(self.master)
         self.exp_avg_sq = torch.zeros_like(self.master)
 
+    @classmethod
+    def from_ddp(cls, ddp, **kw):
+        """Build over a DataParallel wrapper sharing its grad buffer
+        (bucket order == flat order, zero gather copies)."""
+        buf = ddp.grad_buffer()
+        params = ddp.grad_buffer_params()
+        if buf is None or params is None:
+            return cls(ddp.parameters(), **kw)
+        return cls(params, grad_buffer=buf, **kw)
+
     # ------------------------------------------------------------------
     def _gather_grads(self) -> None:
+        if self._grad_mode != "copy":
+            return
         self.flat_grad.zero_()
         for p, off in zip(self.params, self._offsets):
             if p.grad is not None:
                 self.flat_grad[off : off + p.numel()].copy_(p.grad.reshape(-1))
 
-    def grad_global_norm(self) -> torch.Tensor:
-        """Global L2 grad norm (exact across TP/PP when those grads are
-        local — caller reduces if needed)."""
-        return l2_norm([p.grad for p in self.params])
+    def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
+        """Two-kernel clip on the flat grad buffer (local norm)."""
+        self._gather_grads()
+        norm = l2_norm([self.flat_grad])
+        scale = max_norm / (norm + 1e-6)
+        if float(scale) < 1.0:
+            self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
+        return norm
 
     @torch.no_grad()
     def step(self) -> None:
         self.step_count += 1
         self._gather_grads()
         if self.max_grad_norm:
-            norm = l2_norm([self.flat_grad])
-            scale = self.max_grad_norm / (float(norm) + 1e-6)
-            if scale < 1.0:
-                self.flat_grad.mul_(scale)
+            self.clip_grad_norm_(self.max_grad_norm)
         shard_param = self.flat_param[self._shard_slice]
         shard_grad = self.flat_grad[self._shard_slice]
         adamw_step_flat(
@@ -120,8 +156,20 @@ class ZeroRedundancyAdamW:
             dist.all_gather_into_tensor(
                 self.flat_param, shard_param.contiguous(), group=self.dp_group
             )
+        # params whose dtype differs from the flat buffer keep their own
+        # storage — copy their slice back
+        for p, off in zip(self.params, self._offsets):
+            if p.dtype != self.dtype:
+                p.data.copy_(self.flat_param[off : off + p.numel()].view_as(p).to(p.dtype))
 
     def zero_grad(self, set_to_none: bool = False) -> None:
+        if self._grad_mode == "owned":
+            self.flat_grad.zero_()
+            return
+        if self._grad_mode == "shared":
+            # the DataParallel wrapper zeroes its own buffer
+            self.flat_grad.zero_()
+            return
         for p in self.params:
             if p.grad is not None:
                 if set_to_none:

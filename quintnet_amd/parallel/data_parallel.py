@@ -50,12 +50,16 @@ class DistributedConfig:
 
 
 class GradientBucket:
-    """A contiguous flat grad buffer covering a param group."""
+    """A contiguous flat grad segment covering a param group.
 
-    def __init__(self, params: List[nn.Parameter], dtype: torch.dtype, device: torch.device):
+    ``flat`` is a slice of the wrapper's single grad buffer, so the
+    ZeRO-1 optimizer can share the same storage (no gather copies).
+    """
+
+    def __init__(self, params: List[nn.Parameter], flat: torch.Tensor):
         self.params = params
         self.numel = sum(p.numel() for p in params)
-        self.flat = torch.zeros(self.numel, dtype=dtype, device=device)
+        self.flat = flat
         self.views: List[torch.Tensor] = []
         off = 0
         for p in params:
@@ -69,7 +73,6 @@ class GradientBucket:
             p.grad = v
 
     def reset(self) -> None:
-        self.flat.zero_()
         self.ready_count = 0
         self.work = None
 
@@ -95,6 +98,8 @@ class DataParallel(nn.Module):
             self.backend = LocalBackend()
         self.world_size = self.backend.get_world_size()
         self.require_backward_grad_sync = True
+        self.flat_grads: Dict[torch.dtype, torch.Tensor] = {}
+        self.flat_params_order: Dict[torch.dtype, List[nn.Parameter]] = {}
         self.buckets: List[GradientBucket] = []
         self._param_to_bucket: Dict[int, GradientBucket] = {}
         self._hooks = []
@@ -118,32 +123,47 @@ class DataParallel(nn.Module):
         params = [p for p in self.module.parameters() if p.requires_grad]
         if not params:
             return
-        # reverse order ≈ backward completion order
+        # one flat grad buffer per dtype; buckets are consecutive slices.
+        # reverse order ≈ backward completion order.  Buffer padded to a
+        # world_size multiple so a ZeRO-1 optimizer can shard it evenly.
         params = list(reversed(params))
-        cap = int(self.bucket_config.capacity_mb * 1024 * 1024)
-        cur: List[nn.Parameter] = []
-        cur_bytes = 0
-        groups: List[List[nn.Parameter]] = []
+        by_dtype: Dict[torch.dtype, List[nn.Parameter]] = {}
         for p in params:
-            sz = p.numel() * p.element_size()
-            if cur and cur_bytes + sz > cap:
-                groups.append(cur)
-                cur, cur_bytes = [], 0
-            cur.append(p)
-            cur_bytes += sz
-        if cur:
-            groups.append(cur)
-        for g in groups:
-            # one buffer per dtype within the group
-            by_dtype: Dict[torch.dtype, List[nn.Parameter]] = {}
-            for p in g:
-                by_dtype.setdefault(p.dtype, []).append(p)
-            for dt, ps in by_dtype.items():
-                bucket = GradientBucket(ps, dt, ps[0].device)
+            by_dtype.setdefault(p.dtype, []).append(p)
+        cap = int(self.bucket_config.capacity_mb * 1024 * 1024)
+        self.flat_grads: Dict[torch.dtype, torch.Tensor] = {}
+        self.flat_params_order: Dict[torch.dtype, List[nn.Parameter]] = {}
+        for dt, ps in by_dtype.items():
+            total = sum(p.numel() for p in ps)
+            pad = (self.world_size - total % self.world_size) % self.world_size
+            buf = torch.zeros(total + pad, dtype=dt, device=ps[0].device)
+            self.flat_grads[dt] = buf
+            self.flat_params_order[dt] = ps
+            # carve buckets
+            cur: List[nn.Parameter] = []
+            cur_bytes = 0
+            start = 0
+            off = 0
+            esz = ps[0].element_size()
+
+            def flush(cur, start, off):
+                if not cur:
+                    return
+                bucket = GradientBucket(cur, buf[start:off])
                 bucket.attach_grads()
                 self.buckets.append(bucket)
-                for p in ps:
+                for p in cur:
                     self._param_to_bucket[id(p)] = bucket
+
+            for p in ps:
+                sz = p.numel() * esz
+                if cur and cur_bytes + sz > cap:
+                    flush(cur, start, off)
+                    cur, cur_bytes, start = [], 0, off
+                cur.append(p)
+                cur_bytes += sz
+                off += p.numel()
+            flush(cur, start, off)
 
     def _register_hooks(self) -> None:
         if self.world_size <= 1:
@@ -199,7 +219,24 @@ class DataParallel(nn.Module):
             bucket.ready_count = 0
 
     def zero_grad(self, set_to_none: bool = False):  # noqa: ARG002 — views must persist
+        for buf in self.flat_grads.values():
+            buf.zero_()  # one fill per dtype instead of one per param
         for bucket in self.buckets:
             bucket.reset()
             bucket.attach_grads()
-        # params outside buckets (requires_grad=False) have no grads
+
+    def grad_buffer(self, dtype: torch.dtype = None):
+        """The single flat grad buffer (for the ZeRO-1 optimizer) or None
+        when params span multiple dtypes."""
+        if len(self.flat_grads) != 1:
+            if dtype is not None:
+                return self.flat_grads.get(dtype)
+            return None
+        return next(iter(self.flat_grads.values()))
+
+    def grad_buffer_params(self, dtype: torch.dtype = None):
+        if len(self.flat_params_order) != 1:
+            if dtype is not None:
+                return self.flat_params_order.get(dtype)
+            return None
+        return next(iter(self.flat_params_order.values()))

@@ -144,9 +144,13 @@ class PipelineSchedule(abc.ABC):
         t = self.trainer
         self._finalize_grads()
         if t.max_grad_norm is not None and t.max_grad_norm > 0:
-            clip_grad_norm_local(
-                [p for p in self.model.parameters() if p.requires_grad], t.max_grad_norm
-            )
+            if hasattr(t.optimizer, "clip_grad_norm_"):
+                t.optimizer.clip_grad_norm_(t.max_grad_norm)  # flat-buffer clip
+            else:
+                clip_grad_norm_local(
+                    [p for p in self.model.parameters() if p.requires_grad],
+                    t.max_grad_norm,
+                )
         t.optimizer.step()
         if hasattr(self.model, "zero_grad"):
             self.model.zero_grad()

@@ -23,6 +23,7 @@ sources = [
     os.path.join(CSRC, "softmax.hip"),
     os.path.join(CSRC, "cross_entropy.hip"),
     os.path.join(CSRC, "adamw.hip"),
+    os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "gemm.hip"),
 ]
 
