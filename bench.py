@@ -184,7 +184,7 @@ def main():
         model_name = "vit-mnist"
 
     loader = _ListLoader(batches, grad_acc)
-    from quintnet_amd.ops import cross_entropy
+    from quintnet_amd.ops import causal_lm_loss
     from quintnet_amd.parallel import DataParallel
 
     zero_kw = dict(lr=1e-4, weight_decay=0.01,
@@ -220,7 +220,7 @@ def main():
                     model.require_backward_grad_sync = True
                 if task == "clm":
                     logits = model(b["input_ids"])
-                    loss = cross_entropy(logits[:, :-1, :], b["labels"][:, 1:], -100)
+                    loss = causal_lm_loss(logits, b["labels"], -100)
                 else:
                     loss = criterion(model(b["images"]), b["labels"])
                 (loss / grad_acc).backward()

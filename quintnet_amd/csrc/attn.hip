@@ -1,0 +1,490 @@
+// Fused flash-style attention (fwd + bwd) for gfx950, head_dim = 64.
+//
+// Design (guide: cdna_hip_programming.md §B "Fused attention prefill"):
+//  * swapped-operand QK^T — compute mfma(K, Q) so each lane owns ONE
+//    query row (col = lane&31) and its softmax state (m, l) is
+//    lane-local: the row reduce is a per-reg max/sum + one
+//    __shfl_xor(32), no LDS round trip;
+//  * v_mfma_f32_32x32x16_bf16 tiles; K/Q/V/dO fragments load DIRECTLY
+//    from global memory (A/B fragment layout = 8 contiguous d-elements
+//    per lane, which is exactly a row-major [T, D] slice) — strides are
+//    passed in, so the q/k/v views of the fused QKV projection are
+//    consumed with ZERO transpose/contiguous copies;
+//  * P relayout for the PV mfma via v_cvt_pk_bf16_f32 + permlane32_swap
+//    (guide T12): converts the 16 f32 score regs into the bf16 A/B
+//    fragment in-register;
+//  * O is accumulated TRANSPOSED (O^T[d][q], col = lane&31 = q) so the
+//    online rescale multiplies lane-local registers;
+//  * per block: 4 waves × 32 query rows; V^T (and in backward dO^T/Q^T/
+//    K^T) tiles staged transposed in LDS (row stride 40 elems — bank-
+//    conflict-free ds_read_b128);
+//  * softmax in base-2 (v_exp_f32 is exp2): saved "lse" is
+//    lse2 = m2 + log2(l); backward recomputes P = exp2(s*scale*log2e - lse2).
+//
+// Backward = 3 small kernels (delta, dq, dkv): dq re-derives dS with
+// the fwd (lane-owns-q) orientation; dkv uses the mirrored
+// (lane-owns-key) orientation so dK^T/dV^T accumulate lane-locally.
+//
+// Replaces reference F.scaled_dot_product_attention
+// (utils/GPT2/gpt2_attention.py:156).
+#include "common.h"
+
+typedef __attribute__((ext_vector_type(8))) __bf16 bf16x8;
+
+#define LOG2E 1.4426950408889634f
+#define TPAD 40  // LDS transpose-tile row stride (elems): conflict-free
+
+// ---- cvt_pk + permlane relayout (guide T12) --------------------------------
+// 8 f32 regs (this lane's rows r..r+7 of a 32-col MFMA D tile) -> one
+// bf16x8 fragment whose 8 elements are rows (lane>>5)*8..+7 at this
+// lane's column: the A/B operand layout of mfma_f32_32x32x16_bf16.
+__device__ __forceinline__ unsigned cvt_pk_bf16(float lo, float hi) {
+  unsigned r;
+  // s_nop 1 = the 2 VALU->v_permlane wait states (guide §5.5 T21 hazard;
+  // hipcc pads nothing across an asm boundary for the following builtin)
+  asm volatile("v_cvt_pk_bf16_f32 %0, %1, %2\n\ts_nop 1" : "=v"(r) : "v"(lo), "v"(hi));
+  return r;
+}
+
+__device__ __forceinline__ bf16x8 relayout8(const float* p) {
+  unsigned a0 = cvt_pk_bf16(p[0], p[1]);
+  unsigned a1 = cvt_pk_bf16(p[2], p[3]);
+  unsigned a2 = cvt_pk_bf16(p[4], p[5]);
+  unsigned a3 = cvt_pk_bf16(p[6], p[7]);
+  {
+    auto r = __builtin_amdgcn_permlane32_swap(a0, a2, false, false);
+    a0 = r[0]; a2 = r[1];
+  }
+  {
+    auto r = __builtin_amdgcn_permlane32_swap(a1, a3, false, false);
+    a1 = r[0]; a3 = r[1];
+  }
+  union { unsigned u[4]; bf16x8 v; } out;
+  out.u[0] = a0; out.u[1] = a1; out.u[2] = a2; out.u[3] = a3;
+  return out.v;
+}
+
+// load one 32x32x16 A/B fragment straight from a strided [T, 64] slice:
+// lane row = base_row + (l&31), elems d0 + (l>>5)*8 .. +8 (bf16, 16B).
+__device__ __forceinline__ bf16x8 frag_ld(const unsigned short* base,
+                                          long long row_stride, int row0,
+                                          int d0, int lane) {
+  const unsigned short* p = base + (long long)(row0 + (lane & 31)) * row_stride +
+                            d0 + ((lane >> 5) << 3);
+  return *reinterpret_cast<const bf16x8*>(p);
+}
+
+// MFMA D-tile row of register r for this lane (32x32 layout)
+__device__ __forceinline__ int drow(int r, int lane) {
+  return (r & 3) + ((r >> 2) << 3) + ((lane >> 5) << 2);
+}
+
+// cooperative transpose-stage of a [rows<=32, 64] strided tile into
+// LDS t[64][TPAD] (t[d][r]); 256 threads, 8 contiguous elems each.
+__device__ __forceinline__ void stage_transpose(
+    unsigned short* t, const unsigned short* src, long long row_stride,
+    int row0, int nrows) {
+  int r = threadIdx.x >> 3;
+  int d0 = (threadIdx.x & 7) << 3;
+  if (r < nrows) {
+    const unsigned short* p = src + (long long)(row0 + r) * row_stride + d0;
+    s16x8 v = *reinterpret_cast<const s16x8*>(p);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) t[(d0 + j) * TPAD + r] = (unsigned short)v[j];
+  }
+}
+
+// ===========================================================================
+// forward
+// grid: (T/128, B*H); block 256.  Each wave owns 32 q rows.
+// q/k/v strided [.., T, 64] slices; out written via its own strides
+// (so [B, T, H*64] layout comes out directly); lse2 [BH, T] f32.
+// ===========================================================================
+__global__ __launch_bounds__(256) void attn_fwd_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, unsigned short* __restrict__ out,
+    float* __restrict__ lse2, int T, int H, float scale, int causal,
+    long long qsB, long long qsH, long long qsT,
+    long long ksB, long long ksH, long long ksT,
+    long long vsB, long long vsH, long long vsT,
+    long long osB, long long osH, long long osT) {
+  __shared__ unsigned short vt[64 * TPAD];
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = blockIdx.x * 128;        // block q range [q0, q0+128)
+  const int qw = q0 + wave * 32;          // wave q range  [qw, qw+32)
+  const int myq = qw + (lane & 31);       // this lane's q row
+
+  const unsigned short* qp = q + b * qsB + h * qsH;
+  const unsigned short* kp = k + b * ksB + h * ksH;
+  const unsigned short* vp = v + b * vsB + h * vsH;
+  unsigned short* op = out + b * osB + h * osH;
+
+  // Q fragments (B-operand): held in regs for the whole kernel
+  bf16x8 qf[4];
+#pragma unroll
+  for (int kt = 0; kt < 4; ++kt) qf[kt] = frag_ld(qp, qsT, qw, kt * 16, lane);
+
+  float m2 = -INFINITY, l = 0.f;
+  f32x16 o[2];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { o[0][i] = 0.f; o[1][i] = 0.f; }
+
+  const float s2scale = scale * LOG2E;
+  const int kv_end = causal ? min(q0 + 128, T) : T;
+
+  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+    // cooperative V^T staging
+    __syncthreads();
+    stage_transpose(vt, vp, vsT, kv0, min(32, T - kv0));
+    __syncthreads();
+
+    if (!causal || kv0 <= qw + 31) {  // wave has at least one valid pair
+      // S^T[key][q] = sum_d K[key][d] Q[q][d]
+      f32x16 s;
+#pragma unroll
+      for (int i = 0; i < 16; ++i) s[i] = 0.f;
+#pragma unroll
+      for (int kt = 0; kt < 4; ++kt) {
+        bf16x8 kf = frag_ld(kp, ksT, kv0, kt * 16, lane);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kt], s, 0, 0, 0);
+      }
+      // scale to base-2, causal mask; per-lane row stats over 16 regs
+      float ps[16];
+      float pmax = -INFINITY;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int key = kv0 + drow(r, lane);
+        float x = s[r] * s2scale;
+        if (causal && key > myq) x = -INFINITY;
+        ps[r] = x;
+        pmax = fmaxf(pmax, x);
+      }
+      pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
+      float mnew = fmaxf(m2, pmax);
+      float alpha = (m2 == -INFINITY) ? 0.f : exp2f(m2 - mnew);
+      float psum = 0.f;
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        ps[r] = (ps[r] == -INFINITY) ? 0.f : exp2f(ps[r] - mnew);
+        psum += ps[r];
+      }
+      psum += __shfl_xor(psum, 32, 64);
+      l = l * alpha + psum;
+      m2 = mnew;
+      // rescale O^T
+#pragma unroll
+      for (int i = 0; i < 16; ++i) { o[0][i] *= alpha; o[1][i] *= alpha; }
+      // P fragments (B-operand, k = key): two 16-key blocks
+      bf16x8 pf0 = relayout8(ps);
+      bf16x8 pf1 = relayout8(ps + 8);
+      // O^T[d][q] += V^T · P : A from vt
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt) {
+        const unsigned short* av = &vt[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        bf16x8 a0 = *reinterpret_cast<const bf16x8*>(av);
+        bf16x8 a1 = *reinterpret_cast<const bf16x8*>(av + 16);
+        o[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, pf0, o[mt], 0, 0, 0);
+        o[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, pf1, o[mt], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: normalize and store out[q][d] (scattered: lane owns col q)
+  float inv = (l > 0.f) ? 1.0f / l : 0.f;
+  if (myq < T) {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = mt * 32 + drow(r, lane);
+        op[(long long)myq * osT + d] = f32_to_bf16(o[mt][r] * inv);
+      }
+    }
+    if (lane < 32) lse2[(long long)bh * T + myq] = m2 + log2f(l);
+  }
+}
+
+// ===========================================================================
+// delta[row] = sum_d dout[row][d] * out[row][d]   (one wave per row)
+// ===========================================================================
+__global__ void attn_delta_kernel(
+    const unsigned short* __restrict__ dout, const unsigned short* __restrict__ out,
+    float* __restrict__ delta, long long nrows, int T, int H,
+    long long dsB, long long dsH, long long dsT,
+    long long osB, long long osH, long long osT) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const long long row = (long long)blockIdx.x * 4 + wave;  // (b*H + h)*T + t
+  if (row >= nrows) return;
+  const int t = (int)(row % T);
+  const int bh = (int)(row / T);
+  const int b = bh / H, h = bh % H;
+  const unsigned short* dp = dout + b * dsB + h * dsH + (long long)t * dsT;
+  const unsigned short* op = out + b * osB + h * osH + (long long)t * osT;
+  float acc = bf16_to_f32(dp[lane]) * bf16_to_f32(op[lane]);
+  acc = wave_reduce_sum(acc);
+  if (lane == 0) delta[row] = acc;
+}
+
+// ===========================================================================
+// backward dQ: block owns a 128-row q tile (wave per 32 rows), loops kv.
+// dQ^T[d][q] += K^T[d][key] · g^T[key][q],
+//   g^T = scale * P^T ⊙ (dP^T - delta[q]),  P^T = exp2(s2 - lse2[q])
+// ===========================================================================
+__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse2, const float* __restrict__ delta,
+    unsigned short* __restrict__ dq, int T, int H, float scale, int causal,
+    long long qsB, long long qsH, long long qsT,
+    long long ksB, long long ksH, long long ksT,
+    long long vsB, long long vsH, long long vsT,
+    long long dsB, long long dsH, long long dsT,
+    long long dqsB, long long dqsH, long long dqsT) {
+  __shared__ unsigned short kt_lds[64 * TPAD];
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = blockIdx.x * 128;
+  const int qw = q0 + wave * 32;
+  const int myq = qw + (lane & 31);
+
+  const unsigned short* qp = q + b * qsB + h * qsH;
+  const unsigned short* kp = k + b * ksB + h * ksH;
+  const unsigned short* vp = v + b * vsB + h * vsH;
+  const unsigned short* dop = dout + b * dsB + h * dsH;
+  unsigned short* dqp = dq + b * dqsB + h * dqsH;
+
+  bf16x8 qf[4], dof[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    qf[t] = frag_ld(qp, qsT, qw, t * 16, lane);
+    dof[t] = frag_ld(dop, dsT, qw, t * 16, lane);
+  }
+  const float my_lse = lse2[(long long)bh * T + min(myq, T - 1)];
+  const float my_delta = delta[(long long)bh * T + min(myq, T - 1)];
+  const float s2scale = scale * LOG2E;
+
+  f32x16 dqa[2];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; }
+
+  const int kv_end = causal ? min(q0 + 128, T) : T;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+    __syncthreads();
+    stage_transpose(kt_lds, kp, ksT, kv0, min(32, T - kv0));
+    __syncthreads();
+    if (causal && kv0 > qw + 31) continue;
+
+    f32x16 s, dp_;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) { s[i] = 0.f; dp_[i] = 0.f; }
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      bf16x8 kf = frag_ld(kp, ksT, kv0, t * 16, lane);
+      bf16x8 vf = frag_ld(vp, vsT, kv0, t * 16, lane);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[t], dp_, 0, 0, 0);
+    }
+    float g[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int key = kv0 + drow(r, lane);
+      float p = (causal && key > myq) ? 0.f : exp2f(s[r] * s2scale - my_lse);
+      g[r] = scale * p * (dp_[r] - my_delta);
+    }
+    bf16x8 gf0 = relayout8(g);
+    bf16x8 gf1 = relayout8(g + 8);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      const unsigned short* ak = &kt_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(ak);
+      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(ak + 16);
+      dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, gf0, dqa[mt], 0, 0, 0);
+      dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, gf1, dqa[mt], 0, 0, 0);
+    }
+  }
+
+  if (myq < T) {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = mt * 32 + drow(r, lane);
+        dqp[(long long)myq * dqsT + d] = f32_to_bf16(dqa[mt][r]);
+      }
+  }
+}
+
+// ===========================================================================
+// backward dK/dV: block owns a 128-key kv tile (wave per 32 keys),
+// loops q tiles.  Mirrored orientation: lane owns a KEY column.
+//   S[q][key], P = exp2(s2 - lse2[q]); dP[q][key] = dO·V^T
+//   dV^T[d][key] += dO^T[d][q] · P[q][key]
+//   dK^T[d][key] += Q^T[d][q] · g[q][key]
+// ===========================================================================
+__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse2, const float* __restrict__ delta,
+    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
+    int T, int H, float scale, int causal,
+    long long qsB, long long qsH, long long qsT,
+    long long ksB, long long ksH, long long ksT,
+    long long vsB, long long vsH, long long vsT,
+    long long dsB, long long dsH, long long dsT,
+    long long dksB, long long dksH, long long dksT,
+    long long dvsB, long long dvsH, long long dvsT) {
+  __shared__ unsigned short dot_lds[64 * TPAD];
+  __shared__ unsigned short qt_lds[64 * TPAD];
+  __shared__ float lse_t[32];
+  __shared__ float del_t[32];
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int kv0b = blockIdx.x * 128;
+  const int kw = kv0b + wave * 32;
+  const int mykey = kw + (lane & 31);
+
+  const unsigned short* qp = q + b * qsB + h * qsH;
+  const unsigned short* kp = k + b * ksB + h * ksH;
+  const unsigned short* vp = v + b * vsB + h * vsH;
+  const unsigned short* dop = dout + b * dsB + h * dsH;
+  unsigned short* dkp = dk + b * dksB + h * dksH;
+  unsigned short* dvp = dv + b * dvsB + h * dvsH;
+
+  bf16x8 kf[4], vf[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    kf[t] = frag_ld(kp, ksT, kw, t * 16, lane);
+    vf[t] = frag_ld(vp, vsT, kw, t * 16, lane);
+  }
+  const float s2scale = scale * LOG2E;
+
+  f32x16 dka[2], dva[2];
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; }
+
+  const int q_start = causal ? (kv0b / 32) * 32 : 0;
+  for (int qt0 = q_start; qt0 < T; qt0 += 32) {
+    __syncthreads();
+    stage_transpose(dot_lds, dop, dsT, qt0, min(32, T - qt0));
+    stage_transpose(qt_lds, qp, qsT, qt0, min(32, T - qt0));
+    if (threadIdx.x < 32 && qt0 + threadIdx.x < T) {
+      lse_t[threadIdx.x] = lse2[(long long)bh * T + qt0 + threadIdx.x];
+      del_t[threadIdx.x] = delta[(long long)bh * T + qt0 + threadIdx.x];
+    }
+    __syncthreads();
+    if (causal && qt0 + 31 < kw) continue;  // whole tile above diagonal
+
+    // S[q][key]: A = Q (i=q), B = K (j=key); dP[q][key]: A = dO, B = V
+    f32x16 s, dp_;
+#pragma unroll
+    for (int i = 0; i < 16; ++i) { s[i] = 0.f; dp_[i] = 0.f; }
+#pragma unroll
+    for (int t = 0; t < 4; ++t) {
+      bf16x8 qfr = frag_ld(qp, qsT, qt0, t * 16, lane);
+      bf16x8 dofr = frag_ld(dop, dsT, qt0, t * 16, lane);
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr, vf[t], dp_, 0, 0, 0);
+    }
+    float pv[16], gv[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      int qrow = drow(r, lane);
+      int qg = qt0 + qrow;
+      float p = (qg >= T || (causal && mykey > qg))
+                    ? 0.f
+                    : exp2f(s[r] * s2scale - lse_t[qrow]);
+      pv[r] = p;
+      gv[r] = scale * p * (dp_[r] - del_t[qrow]);
+    }
+    bf16x8 pf0 = relayout8(pv), pf1 = relayout8(pv + 8);
+    bf16x8 gf0 = relayout8(gv), gf1 = relayout8(gv + 8);
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt) {
+      const unsigned short* adot = &dot_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+      const unsigned short* aqt = &qt_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+      bf16x8 d0 = *reinterpret_cast<const bf16x8*>(adot);
+      bf16x8 d1 = *reinterpret_cast<const bf16x8*>(adot + 16);
+      bf16x8 q0f = *reinterpret_cast<const bf16x8*>(aqt);
+      bf16x8 q1f = *reinterpret_cast<const bf16x8*>(aqt + 16);
+      dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d0, pf0, dva[mt], 0, 0, 0);
+      dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d1, pf1, dva[mt], 0, 0, 0);
+      dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q0f, gf0, dka[mt], 0, 0, 0);
+      dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q1f, gf1, dka[mt], 0, 0, 0);
+    }
+  }
+
+  if (mykey < T) {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = mt * 32 + drow(r, lane);
+        dkp[(long long)mykey * dksT + d] = f32_to_bf16(dka[mt][r]);
+        dvp[(long long)mykey * dvsT + d] = f32_to_bf16(dva[mt][r]);
+      }
+  }
+}
+
+// ---- launchers -------------------------------------------------------------
+struct AttnStrides {
+  long long qB, qH, qT, kB, kH, kT, vB, vH, vT, oB, oH, oT;
+};
+
+void attn_fwd_launch(const unsigned short* q, const unsigned short* k,
+                     const unsigned short* v, unsigned short* out, float* lse2,
+                     int B, int H, int T, float scale, int causal,
+                     const AttnStrides& st, hipStream_t stream) {
+  dim3 grid(T / 128, B * H);
+  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream, q, k, v, out,
+                     lse2, T, H, scale, causal, st.qB, st.qH, st.qT, st.kB,
+                     st.kH, st.kT, st.vB, st.vH, st.vT, st.oB, st.oH, st.oT);
+}
+
+void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
+                       float* delta, int B, int H, int T,
+                       long long dsB, long long dsH, long long dsT,
+                       long long osB, long long osH, long long osT,
+                       hipStream_t stream) {
+  long long rows = (long long)B * H * T;
+  hipLaunchKernelGGL(attn_delta_kernel, dim3((unsigned)((rows + 3) / 4)),
+                     dim3(256), 0, stream, dout, out, delta, rows, T, H, dsB,
+                     dsH, dsT, osB, osH, osT);
+}
+
+void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
+                        const unsigned short* v, const unsigned short* dout,
+                        const float* lse2, const float* delta,
+                        unsigned short* dq, int B, int H, int T, float scale,
+                        int causal, const AttnStrides& st,
+                        long long dsB, long long dsH, long long dsT,
+                        hipStream_t stream) {
+  dim3 grid(T / 128, B * H);
+  hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream, q, k, v,
+                     dout, lse2, delta, dq, T, H, scale, causal, st.qB, st.qH,
+                     st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, dsB, dsH,
+                     dsT, st.oB, st.oH, st.oT);
+}
+
+void attn_bwd_dkv_launch(const unsigned short* q, const unsigned short* k,
+                         const unsigned short* v, const unsigned short* dout,
+                         const float* lse2, const float* delta,
+                         unsigned short* dk, unsigned short* dv, int B, int H,
+                         int T, float scale, int causal, const AttnStrides& st,
+                         long long dsB, long long dsH, long long dsT,
+                         long long dkB, long long dkH, long long dkT,
+                         long long dvB, long long dvH, long long dvT,
+                         hipStream_t stream) {
+  dim3 grid(T / 128, B * H);
+  hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0, stream, q, k, v,
+                     dout, lse2, delta, dk, dv, T, H, scale, causal, st.qB,
+                     st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT,
+                     dsB, dsH, dsT, dkB, dkH, dkT, dvB, dvH, dvT);
+}

@@ -26,6 +26,24 @@ template <typename T>
 void sumsq_launch(const T*, long long, float*, hipStream_t);
 void gemm_nt_launch(const unsigned short*, const unsigned short*, const unsigned short*,
                     unsigned short*, unsigned short*, int, int, int, int, hipStream_t);
+struct AttnStrides {
+  long long qB, qH, qT, kB, kH, kT, vB, vH, vT, oB, oH, oT;
+};
+void attn_fwd_launch(const unsigned short*, const unsigned short*, const unsigned short*,
+                     unsigned short*, float*, int, int, int, float, int,
+                     const AttnStrides&, hipStream_t);
+void attn_delta_launch(const unsigned short*, const unsigned short*, float*, int, int, int,
+                       long long, long long, long long, long long, long long, long long,
+                       hipStream_t);
+void attn_bwd_dq_launch(const unsigned short*, const unsigned short*, const unsigned short*,
+                        const unsigned short*, const float*, const float*, unsigned short*,
+                        int, int, int, float, int, const AttnStrides&,
+                        long long, long long, long long, hipStream_t);
+void attn_bwd_dkv_launch(const unsigned short*, const unsigned short*, const unsigned short*,
+                         const unsigned short*, const float*, const float*, unsigned short*,
+                         unsigned short*, int, int, int, float, int, const AttnStrides&,
+                         long long, long long, long long, long long, long long, long long,
+                         long long, long long, long long, hipStream_t);
 
 namespace {
 
@@ -273,6 +291,68 @@ torch::Tensor multi_tensor_sumsq(std::vector<torch::Tensor> tensors) {
   return out;
 }
 
+
+// ---------------------------------------------------------------------------
+// fused flash attention (head_dim 64, bf16, strided [B,H,T,D] views)
+namespace {
+void check_attn_view(const torch::Tensor& t, int64_t B, int64_t H, int64_t T) {
+  TORCH_CHECK(t.dim() == 4 && t.size(0) == B && t.size(1) == H && t.size(2) == T &&
+              t.size(3) == 64, "attn: bad view shape");
+  TORCH_CHECK(t.stride(3) == 1, "attn: head_dim must be contiguous");
+  TORCH_CHECK(t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0 && t.stride(0) % 8 == 0,
+              "attn: strides must be 16B-aligned");
+  TORCH_CHECK(t.dtype() == torch::kBFloat16, "attn: bf16 only");
+}
+const unsigned short* ap(const torch::Tensor& t) {
+  return reinterpret_cast<const unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+unsigned short* ap_mut(torch::Tensor& t) {
+  return reinterpret_cast<unsigned short*>(t.data_ptr<at::BFloat16>());
+}
+}  // namespace
+
+torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                       torch::Tensor out, double scale, bool causal) {
+  int64_t B = q.size(0), H = q.size(1), T = q.size(2);
+  TORCH_CHECK(T % 128 == 0, "attn: T must be a multiple of 128");
+  check_attn_view(q, B, H, T); check_attn_view(k, B, H, T);
+  check_attn_view(v, B, H, T); check_attn_view(out, B, H, T);
+  auto lse2 = torch::empty({B * H, T}, q.options().dtype(torch::kFloat32));
+  AttnStrides st{q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+                 k.stride(2), v.stride(0), v.stride(1), v.stride(2), out.stride(0),
+                 out.stride(1), out.stride(2)};
+  attn_fwd_launch(ap(q), ap(k), ap(v), ap_mut(out), lse2.data_ptr<float>(),
+                  (int)B, (int)H, (int)T, (float)scale, causal ? 1 : 0, st,
+                  cur_stream());
+  return lse2;
+}
+
+void attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+              torch::Tensor out, torch::Tensor dout, torch::Tensor lse2,
+              torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
+              double scale, bool causal) {
+  int64_t B = q.size(0), H = q.size(1), T = q.size(2);
+  check_attn_view(dout, B, H, T); check_attn_view(dq, B, H, T);
+  check_attn_view(dk, B, H, T); check_attn_view(dv, B, H, T);
+  auto delta = torch::empty({B * H, T}, q.options().dtype(torch::kFloat32));
+  attn_delta_launch(ap(dout), ap(out), delta.data_ptr<float>(), (int)B, (int)H,
+                    (int)T, dout.stride(0), dout.stride(1), dout.stride(2),
+                    out.stride(0), out.stride(1), out.stride(2), cur_stream());
+  AttnStrides st{q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
+                 k.stride(2), v.stride(0), v.stride(1), v.stride(2), dq.stride(0),
+                 dq.stride(1), dq.stride(2)};
+  attn_bwd_dq_launch(ap(q), ap(k), ap(v), ap(dout), lse2.data_ptr<float>(),
+                     delta.data_ptr<float>(), ap_mut(dq), (int)B, (int)H, (int)T,
+                     (float)scale, causal ? 1 : 0, st, dout.stride(0),
+                     dout.stride(1), dout.stride(2), cur_stream());
+  attn_bwd_dkv_launch(ap(q), ap(k), ap(v), ap(dout), lse2.data_ptr<float>(),
+                      delta.data_ptr<float>(), ap_mut(dk), ap_mut(dv), (int)B,
+                      (int)H, (int)T, (float)scale, causal ? 1 : 0, st,
+                      dout.stride(0), dout.stride(1), dout.stride(2),
+                      dk.stride(0), dk.stride(1), dk.stride(2), dv.stride(0),
+                      dv.stride(1), dv.stride(2), cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -284,6 +364,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd);
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
   m.def("act_bwd", &act_bwd, "fused activation backward");
+  m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)");
+  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)");
   m.def("adamw_step", &adamw_step);
   m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
   m.attr("gfx") = "gfx950";

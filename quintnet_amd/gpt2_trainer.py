@@ -64,13 +64,13 @@ class GPT2Trainer(Trainer):
     def _train_epoch_plain(self) -> Dict[str, float]:
         total_loss, total_tokens, steps = 0.0, 0, 0
         accum = 0
-        from .ops import cross_entropy
+        from .ops import causal_lm_loss
 
         for batch in self.train_loader:
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             labels = batch["labels"].to(self.device, non_blocking=True)
             logits = self.model(ids)
-            loss = cross_entropy(logits[:, :-1, :], labels[:, 1:], ignore_index=-100)
+            loss = causal_lm_loss(logits, labels, ignore_index=-100)
             (loss / self.grad_acc_steps).backward()
             accum += 1
             total_loss += float(loss.detach())
@@ -105,16 +105,14 @@ class GPT2Trainer(Trainer):
             return self.pipeline_trainer.evaluate(
                 self.val_loader, self.tensor_shapes, self.device, self._dtype
             )
-        from .ops import cross_entropy
+        from .ops import causal_lm_loss
 
         total_loss, steps = 0.0, 0
         for batch in self.val_loader:
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             labels = batch["labels"].to(self.device, non_blocking=True)
             logits = self.model(ids)
-            total_loss += float(
-                cross_entropy(logits[:, :-1, :], labels[:, 1:], ignore_index=-100)
-            )
+            total_loss += float(causal_lm_loss(logits, labels, ignore_index=-100))
             steps += 1
         avg = total_loss / max(steps, 1)
         return {"loss": avg, "ppl": math.exp(min(avg, 20.0))}

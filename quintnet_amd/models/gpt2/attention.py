@@ -18,7 +18,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ...ops import attention as fused_attention
+from ...ops import attention_qkv
 from ...parallel.tensor_parallel import ColumnParallelLinear, RowParallelLinear
 from .config import GPT2Config
 
@@ -53,13 +53,6 @@ class GPT2Attention(nn.Module):
         self.resid_dropout = nn.Dropout(config.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        B, T, _ = x.shape
         qkv = self.c_attn(x)  # [B, T, 3*n_embd/tp] = [q_loc | k_loc | v_loc]
-        q, k, v = qkv.split(self.hidden_local, dim=-1)
-
-        def split_heads(t):
-            return t.view(B, T, self.n_head_local, self.head_dim).transpose(1, 2)
-
-        out = fused_attention(split_heads(q), split_heads(k), split_heads(v), causal=True)
-        out = out.transpose(1, 2).reshape(B, T, self.hidden_local)
+        out = attention_qkv(qkv, self.n_head_local, causal=True)
         return self.resid_dropout(self.c_proj(out))

@@ -9,8 +9,15 @@ kernels against.
 from ._backend import ext, has_ext, use_native, force_eager
 from .linear import linear, LinearFunction
 from .layernorm import layer_norm, FusedLayerNorm, LayerNormFunction
-from .attention import attention, AttentionFunction, causal_softmax, softmax_bwd
-from .cross_entropy import cross_entropy, CrossEntropyFunction
+from .attention import (
+    attention,
+    attention_qkv,
+    AttentionFunction,
+    FlashAttentionFunction,
+    causal_softmax,
+    softmax_bwd,
+)
+from .cross_entropy import cross_entropy, causal_lm_loss, shift_labels, CrossEntropyFunction
 from .adamw import adamw_step_flat, clip_grad_norm_local, l2_norm
 
 __all__ = [
@@ -24,10 +31,14 @@ __all__ = [
     "FusedLayerNorm",
     "LayerNormFunction",
     "attention",
+    "attention_qkv",
     "AttentionFunction",
+    "FlashAttentionFunction",
     "causal_softmax",
     "softmax_bwd",
     "cross_entropy",
+    "causal_lm_loss",
+    "shift_labels",
     "CrossEntropyFunction",
     "adamw_step_flat",
     "clip_grad_norm_local",

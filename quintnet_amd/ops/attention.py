@@ -69,9 +69,133 @@ class AttentionFunction(torch.autograd.Function):
         return dq, dk, dv, None
 
 
+def _flash_ok(q: torch.Tensor) -> bool:
+    return (
+        _backend.has_ext()
+        and not _backend.force_eager()
+        and q.is_cuda
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] == 64
+        and q.shape[-2] % 128 == 0
+    )
+
+
+class FlashAttentionFunction(torch.autograd.Function):
+    """Fully-fused flash-style attention (csrc/attn.hip, D=64 bf16)."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal):
+        scale = 1.0 / math.sqrt(q.shape[-1])
+        out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        lse2 = _backend.ext().attn_fwd(q, k, v, out, scale, causal)
+        ctx.save_for_backward(q, k, v, out, lse2)
+        ctx.scale = scale
+        ctx.causal = causal
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse2 = ctx.saved_tensors
+        if dout.stride(-1) != 1 or dout.stride(-2) % 8 != 0:
+            dout = dout.contiguous()
+        dq = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        dk = torch.empty(k.shape, dtype=k.dtype, device=k.device)
+        dv = torch.empty(v.shape, dtype=v.dtype, device=v.device)
+        _backend.ext().attn_bwd(q, k, v, out, dout, lse2, dq, dk, dv, ctx.scale, ctx.causal)
+        return dq, dk, dv, None
+
+
+class FlashAttentionQKV(torch.autograd.Function):
+    """Fused attention directly on the packed [B, T, 3*H*D] QKV tensor.
+
+    Consumes the column-parallel c_attn output without any
+    transpose/contiguous copies (strided fragment loads) and writes the
+    [B, T, H*D] output layout the row-parallel c_proj wants; backward
+    fills one packed dqkv the same way.
+    """
+
+    @staticmethod
+    def forward(ctx, qkv, n_heads, causal):
+        B, T, three_hl = qkv.shape
+        hl = three_hl // 3
+        D = hl // n_heads
+        scale = 1.0 / math.sqrt(D)
+
+        def split_view(t, off, width):
+            return t[:, :, off : off + width].view(B, T, n_heads, D).permute(0, 2, 1, 3)
+
+        qv = split_view(qkv, 0, hl)
+        kv = split_view(qkv, hl, hl)
+        vv = split_view(qkv, 2 * hl, hl)
+        out = qkv.new_empty(B, T, hl)
+        out_view = out.view(B, T, n_heads, D).permute(0, 2, 1, 3)
+        lse2 = _backend.ext().attn_fwd(qv, kv, vv, out_view, scale, causal)
+        ctx.save_for_backward(qkv, out, lse2)
+        ctx.meta = (n_heads, D, hl, causal, scale)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, out, lse2 = ctx.saved_tensors
+        n_heads, D, hl, causal, scale = ctx.meta
+        B, T, _ = qkv.shape
+        dout = dout.contiguous()
+
+        def split_view(t, off, width):
+            return t[:, :, off : off + width].view(B, T, n_heads, D).permute(0, 2, 1, 3)
+
+        qv = split_view(qkv, 0, hl)
+        kv = split_view(qkv, hl, hl)
+        vv = split_view(qkv, 2 * hl, hl)
+        out_view = out.view(B, T, n_heads, D).permute(0, 2, 1, 3)
+        dout_view = dout.view(B, T, n_heads, D).permute(0, 2, 1, 3)
+        dqkv = torch.empty_like(qkv)
+        dqv = split_view(dqkv, 0, hl)
+        dkv = split_view(dqkv, hl, hl)
+        dvv = split_view(dqkv, 2 * hl, hl)
+        _backend.ext().attn_bwd(
+            qv, kv, vv, out_view, dout_view, lse2, dqv, dkv, dvv, scale, causal
+        )
+        return dqkv, None, None
+
+
 def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool = False) -> torch.Tensor:
     """Multi-head attention core: softmax(QK^T/sqrt(d) [+mask]) V.
 
     q, k, v: [B, H, T, D] (same head count — TP shards heads upstream).
+    Dispatches to the fully-fused flash kernel on gfx950 (D=64,
+    T%128==0); otherwise the composed GEMM+fused-softmax path.
     """
+    if _flash_ok(q) and q.shape == k.shape == v.shape:
+        def ok(t):
+            return t.stride(-1) == 1 and all(s % 8 == 0 for s in t.stride()[:-1])
+
+        qc = q if ok(q) else q.contiguous()
+        kc = k if ok(k) else k.contiguous()
+        vc = v if ok(v) else v.contiguous()
+        return FlashAttentionFunction.apply(qc, kc, vc, causal)
     return AttentionFunction.apply(q, k, v, causal)
+
+
+def attention_qkv(qkv: torch.Tensor, n_heads: int, causal: bool = True) -> torch.Tensor:
+    """Attention on the packed QKV projection output [B, T, 3*H*D] ->
+    [B, T, H*D] (local layout ``[q | k | v]``, per-head sharded)."""
+    B, T, three_hl = qkv.shape
+    hl = three_hl // 3
+    D = hl // n_heads
+    if (
+        _backend.has_ext()
+        and not _backend.force_eager()
+        and qkv.is_cuda
+        and qkv.dtype == torch.bfloat16
+        and D == 64
+        and T % 128 == 0
+    ):
+        return FlashAttentionQKV.apply(qkv, n_heads, causal)
+    q, k, v = qkv.split(hl, dim=-1)
+
+    def heads(t):
+        return t.view(B, T, n_heads, D).transpose(1, 2)
+
+    out = attention(heads(q), heads(k), heads(v), causal=causal)
+    return out.transpose(1, 2).reshape(B, T, hl)

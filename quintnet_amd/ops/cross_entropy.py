@@ -68,3 +68,20 @@ def cross_entropy(
     return CrossEntropyFunction.apply(
         logits.reshape(-1, logits.shape[-1]), target.reshape(-1), ignore_index
     )
+
+
+def shift_labels(labels: torch.Tensor, ignore_index: int = -100) -> torch.Tensor:
+    """Next-token targets aligned with FULL-length logits: position t is
+    labeled with token t+1, last position ignored.  Lets causal-LM loss
+    run on the unsliced logits tensor — ``logits[:, :-1]`` forces an
+    800 MB copy of the GPT-2 logits at reshape time."""
+    tg = torch.full_like(labels, ignore_index)
+    tg[:, :-1] = labels[:, 1:]
+    return tg
+
+
+def causal_lm_loss(
+    logits: torch.Tensor, labels: torch.Tensor, ignore_index: int = -100
+) -> torch.Tensor:
+    """Mean next-token cross-entropy on full-length logits/labels [B,T(,V)]."""
+    return cross_entropy(logits, shift_labels(labels, ignore_index), ignore_index)

@@ -23,7 +23,7 @@ from ...core.comm import (
     bidirectional_pipeline_communicate,
     pipeline_communicate,
 )
-from ...ops import cross_entropy, clip_grad_norm_local
+from ...ops import causal_lm_loss, clip_grad_norm_local, shift_labels
 
 __all__ = ["PipelineSchedule", "AllFwdAllBwdSchedule", "OneFOneBSchedule", "get_schedule"]
 
@@ -91,9 +91,8 @@ class PipelineSchedule(abc.ABC):
     ) -> torch.Tensor:
         labels = batch["labels"].to(device, non_blocking=True)
         if self.task_type == "clm":
-            logits = output[:, :-1, :]
-            targets = labels[:, 1:]
-            loss = cross_entropy(logits, targets, ignore_index=-100)
+            targets = shift_labels(labels, -100)
+            loss = causal_lm_loss(output, labels, ignore_index=-100)
             with torch.no_grad():
                 n_tok = int((targets != -100).sum())
                 metrics["loss"] = metrics.get("loss", 0.0) + float(loss.detach())

@@ -24,6 +24,7 @@ sources = [
     os.path.join(CSRC, "cross_entropy.hip"),
     os.path.join(CSRC, "adamw.hip"),
     os.path.join(CSRC, "elementwise.hip"),
+    os.path.join(CSRC, "attn.hip"),
     os.path.join(CSRC, "gemm.hip"),
 ]
 

@@ -115,7 +115,7 @@ def test_cross_entropy_fwd_bwd(dt):
     tol = 2e-2 if dt == torch.bfloat16 else 1e-4
     assert abs(float(loss) - float(ref)) < tol, (float(loss), float(ref))
 
-    dl = _ext().cross_entropy_bwd(logits, tgt, lse, int(n_valid), -100)
+    dl = _ext().cross_entropy_bwd(logits, tgt, lse, int(n_valid), -100, None)
     lf = logits.float().requires_grad_(True)
     rr = torch.nn.functional.cross_entropy(lf, tgt, ignore_index=-100)
     (g,) = torch.autograd.grad(rr, lf)
@@ -179,3 +179,74 @@ def test_model_linear_uses_native_gemm():
     y.sum().backward()
     for t in (x, w, b):
         assert torch.isfinite(t.grad.float()).all()
+
+
+@pytest.mark.parametrize("causal", [True, False])
+def test_flash_attention_vs_fp32(causal):
+    """Fully-fused flash kernel vs fp32 SDPA (fwd + all three grads)."""
+    from quintnet_amd.ops.attention import FlashAttentionFunction, _flash_ok
+
+    torch.manual_seed(7)
+    B, H, T, D = 2, 3, 256, 64
+    q = torch.randn(B, H, T, D, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn(B, H, T, D, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    v = torch.randn(B, H, T, D, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    assert _flash_ok(q)
+    out = FlashAttentionFunction.apply(q, k, v, causal)
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.scaled_dot_product_attention(qf, kf, vf, is_causal=causal)
+    assert (out.float() - ref).abs().max() < 3e-2, (out.float() - ref).abs().max()
+
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref.backward(dout.float())
+    for g, rg, nm in ((q.grad, qf.grad, "dq"), (k.grad, kf.grad, "dk"), (v.grad, vf.grad, "dv")):
+        err = (g.float() - rg).abs().max()
+        scale = max(float(rg.abs().max()), 1.0)
+        assert err / scale < 4e-2, (nm, err, scale)
+
+
+def test_flash_attention_qkv_packed():
+    """Packed-QKV fused path == composed reference, incl. dqkv."""
+    from quintnet_amd.ops import attention_qkv
+
+    torch.manual_seed(8)
+    B, T, Hh, D = 2, 128, 4, 64
+    hl = Hh * D
+    qkv = torch.randn(B, T, 3 * hl, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    out = attention_qkv(qkv, Hh, causal=True)
+
+    qkv2 = qkv.detach().float().requires_grad_(True)
+    q, k, v = qkv2.split(hl, dim=-1)
+
+    def heads(t):
+        return t.view(B, T, Hh, D).transpose(1, 2)
+
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        heads(q), heads(k), heads(v), is_causal=True
+    ).transpose(1, 2).reshape(B, T, hl)
+    assert (out.float() - ref).abs().max() < 3e-2
+
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    ref.backward(dout.float())
+    err = (qkv.grad.float() - qkv2.grad).abs().max()
+    scale = max(float(qkv2.grad.abs().max()), 1.0)
+    assert err / scale < 4e-2, err
+
+
+def test_flash_attention_long_seq():
+    """T=1024 (the bench shape): finite + row-sum sanity vs SDPA."""
+    from quintnet_amd.ops.attention import FlashAttentionFunction
+
+    torch.manual_seed(9)
+    q = torch.randn(1, 2, 1024, 64, device=DEV, dtype=torch.bfloat16)
+    k = torch.randn(1, 2, 1024, 64, device=DEV, dtype=torch.bfloat16)
+    v = torch.randn(1, 2, 1024, 64, device=DEV, dtype=torch.bfloat16)
+    out = FlashAttentionFunction.apply(q, k, v, True)
+    ref = torch.nn.functional.scaled_dot_product_attention(
+        q.float(), k.float(), v.float(), is_causal=True
+    )
+    assert (out.float() - ref).abs().max() < 3e-2
