@@ -20,6 +20,8 @@ template <typename T>
 void ce_bwd_launch(const T*, const long long*, const float*, const float*, T*, long long, int, long long, float, hipStream_t);
 template <typename T>
 void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
+template <typename T>
+void colsum_launch(const T*, float*, long long, int, hipStream_t);
 template <typename TP, typename TG>
 void adamw_launch(TP*, float*, const TG*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
@@ -234,6 +236,19 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   return dl;
 }
 
+torch::Tensor colsum(torch::Tensor x) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  int64_t N = xc.size(-1);
+  long long rows = xc.numel() / N;
+  auto out = torch::zeros({N}, xc.options().dtype(torch::kFloat32));
+  if (xc.dtype() == torch::kBFloat16)
+    colsum_launch<unsigned short>(bf16p(xc), out.data_ptr<float>(), rows, (int)N, cur_stream());
+  else
+    colsum_launch<float>(xc.data_ptr<float>(), out.data_ptr<float>(), rows, (int)N, cur_stream());
+  return out.to(x.dtype());
+}
+
 torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor pre, int64_t act) {
   CHECK_GPU(dy);
   auto dyc = dy.contiguous();
@@ -364,6 +379,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd);
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
   m.def("act_bwd", &act_bwd, "fused activation backward");
+  m.def("colsum", &colsum, "column sum (bias grad)");
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)");
   m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)");
   m.def("adamw_step", &adamw_step);

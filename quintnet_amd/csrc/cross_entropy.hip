@@ -22,14 +22,45 @@ __global__ void ce_fwd_kernel(
   __shared__ float sm[8], ss[8];
   const int lane = threadIdx.x & (QN_WAVE - 1);
   const int wave = threadIdx.x / QN_WAVE;
+  const int V8 = V >> 3;
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* lr = logits + row * V;
-    float m = -INFINITY, s = 0.f;
-    for (int i = threadIdx.x; i < V; i += blockDim.x) {
-      float x = ld_as_f32(lr + i);
-      if (x > m) { s = s * __expf(m - x) + 1.f; m = x; }
-      else s += __expf(x - m);
+    // 8-wide loads (G13) + 4 independent online partials for ILP
+    float m4[4] = {-INFINITY, -INFINITY, -INFINITY, -INFINITY};
+    float s4[4] = {0.f, 0.f, 0.f, 0.f};
+    if constexpr (sizeof(T) == 2) {
+      for (int gI = threadIdx.x; gI < V8; gI += blockDim.x) {
+        s16x8 vv = *reinterpret_cast<const s16x8*>(lr + gI * 8);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float x = bf16_to_f32((unsigned short)vv[j]);
+          int a = j & 3;
+          if (x > m4[a]) { s4[a] = s4[a] * __expf(m4[a] - x) + 1.f; m4[a] = x; }
+          else s4[a] += __expf(x - m4[a]);
+        }
+      }
+    } else {
+      for (int gI = threadIdx.x; gI < V8; gI += blockDim.x) {
+        f32x4 v0 = *reinterpret_cast<const f32x4*>(lr + gI * 8);
+        f32x4 v1 = *reinterpret_cast<const f32x4*>(lr + gI * 8 + 4);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          float x = (j < 4) ? v0[j & 3] : v1[j & 3];
+          int a = j & 3;
+          if (x > m4[a]) { s4[a] = s4[a] * __expf(m4[a] - x) + 1.f; m4[a] = x; }
+          else s4[a] += __expf(x - m4[a]);
+        }
+      }
     }
+    // vocab tail
+    for (int i = V8 * 8 + threadIdx.x; i < V; i += blockDim.x) {
+      float x = ld_as_f32(lr + i);
+      if (x > m4[0]) { s4[0] = s4[0] * __expf(m4[0] - x) + 1.f; m4[0] = x; }
+      else s4[0] += __expf(x - m4[0]);
+    }
+    float m = m4[0], s = s4[0];
+#pragma unroll
+    for (int a = 1; a < 4; ++a) online_combine(m, s, m4[a], s4[a]);
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
       online_combine(m, s, __shfl_xor(m, off, QN_WAVE), __shfl_xor(s, off, QN_WAVE));
@@ -60,11 +91,51 @@ __global__ void ce_bwd_kernel(
     const T* lr = logits + row * V;
     T* dr = dlogits + row * V;
     const long long tgt = target[row];
+    const int V8 = V >> 3;
     if (tgt == ignore_index) {
-      for (int i = threadIdx.x; i < V; i += blockDim.x) st_from_f32(dr + i, 0.f);
+      if constexpr (sizeof(T) == 2) {
+        s16x8 z = {0, 0, 0, 0, 0, 0, 0, 0};
+        for (int gI = threadIdx.x; gI < V8; gI += blockDim.x)
+          *reinterpret_cast<s16x8*>(dr + gI * 8) = z;
+      } else {
+        for (int gI = threadIdx.x; gI < V8 * 2; gI += blockDim.x) {
+          f32x4 z = {0.f, 0.f, 0.f, 0.f};
+          *reinterpret_cast<f32x4*>(dr + gI * 4) = z;
+        }
+      }
+      for (int i = V8 * 8 + threadIdx.x; i < V; i += blockDim.x)
+        st_from_f32(dr + i, 0.f);
     } else {
       const float l = lse[row];
-      for (int i = threadIdx.x; i < V; i += blockDim.x) {
+      if constexpr (sizeof(T) == 2) {
+        for (int gI = threadIdx.x; gI < V8; gI += blockDim.x) {
+          s16x8 vv = *reinterpret_cast<const s16x8*>(lr + gI * 8);
+          s16x8 ov;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            float p = __expf(bf16_to_f32((unsigned short)vv[j]) - l);
+            if ((long long)(gI * 8 + j) == tgt) p -= 1.0f;
+            ov[j] = (short)f32_to_bf16(p * gs);
+          }
+          *reinterpret_cast<s16x8*>(dr + gI * 8) = ov;
+        }
+      } else {
+        for (int gI = threadIdx.x; gI < V8; gI += blockDim.x) {
+#pragma unroll
+          for (int half = 0; half < 2; ++half) {
+            f32x4 vv = *reinterpret_cast<const f32x4*>(lr + gI * 8 + half * 4);
+            f32x4 ov;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              float p = __expf(vv[j] - l);
+              if ((long long)(gI * 8 + half * 4 + j) == tgt) p -= 1.0f;
+              ov[j] = p * gs;
+            }
+            *reinterpret_cast<f32x4*>(dr + gI * 8 + half * 4) = ov;
+          }
+        }
+      }
+      for (int i = V8 * 8 + threadIdx.x; i < V; i += blockDim.x) {
         float p = __expf(ld_as_f32(lr + i) - l);
         if ((long long)i == tgt) p -= 1.0f;
         st_from_f32(dr + i, p * gs);

@@ -37,3 +37,29 @@ void act_bwd_launch(const T* dy, const T* pre, T* dx, long long n, int act,
 
 template void act_bwd_launch<float>(const float*, const float*, float*, long long, int, hipStream_t);
 template void act_bwd_launch<unsigned short>(const unsigned short*, const unsigned short*, unsigned short*, long long, int, hipStream_t);
+
+// column sum: out[n] = sum_m x[m][n]  (bias gradient; fp32 atomics per
+// column-chunk — replaces torch's generic reduce at ~4x the time)
+template <typename T>
+__global__ void colsum_kernel(const T* __restrict__ x, float* __restrict__ out,
+                              long long rows, int N) {
+  const int col = blockIdx.x * blockDim.x + threadIdx.x;
+  if (col >= N) return;
+  const int r0 = blockIdx.y;
+  const int nchunks = gridDim.y;
+  float acc = 0.f;
+  for (long long r = r0; r < rows; r += nchunks)
+    acc += ld_as_f32(x + r * N + col);
+  atomicAdd(out + col, acc);
+}
+
+template <typename T>
+void colsum_launch(const T* x, float* out, long long rows, int N, hipStream_t stream) {
+  int colblocks = (N + 255) / 256;
+  int chunks = (int)min(max(rows / 64, (long long)1), (long long)256);
+  hipLaunchKernelGGL((colsum_kernel<T>), dim3(colblocks, chunks), dim3(256), 0,
+                     stream, x, out, rows, N);
+}
+
+template void colsum_launch<float>(const float*, float*, long long, int, hipStream_t);
+template void colsum_launch<unsigned short>(const unsigned short*, float*, long long, int, hipStream_t);

@@ -5,15 +5,18 @@
 // Design (guide: cdna_hip_programming.md §5):
 //  * 128×128 block tile, BK=64; 4 waves (2×2), each computing a 64×64
 //    sub-tile as 4×4 fragments of v_mfma_f32_16x16x32_bf16 (fp32 accum).
-//  * A/B tiles staged through LDS in [row][k] layout with the T2 XOR
-//    swizzle on 8-element (16 B) groups — ds_read_b128 fragment reads are
-//    bank-conflict-free; both MFMA operands read identically since the
-//    NT layout makes B's fragment a row of Bs.
-//  * T14 async-stage split: next tile's global loads issue right after
-//    the barrier, before the MFMA cluster, so HBM latency hides under
-//    compute; bounds-checked staging handles arbitrary M/N (K%8==0).
-//  * Bias add + GELU/ReLU fused into the epilogue (optionally emitting
-//    the pre-activation for backward).
+//  * DOUBLE-BUFFERED LDS with ONE barrier per K-tile: next tile stages
+//    into the other buffer while the current one feeds the MFMAs.
+//  * interior tiles stage via async `global_load_lds` (dwordx4, 16 B per
+//    lane — guide Common-mistake #1: width 16 is the 1.7× lever); the
+//    LDS image is lane-linear, so the T2 XOR bank-swizzle moves to the
+//    per-lane SOURCE address (guide §5.4 rule 21) with the matching XOR
+//    on the fragment reads;
+//  * edge tiles (M/N/K remainders) fall back to bounds-checked
+//    register staging producing the SAME LDS image (T14 split: loads
+//    issue before the MFMA cluster, ds_writes after);
+//  * bias add + GELU/ReLU fused into the epilogue (optionally emitting
+//    the pre-activation for backward);
 //  * XCD-aware bijective blockIdx swizzle (T1) for L2 affinity.
 //
 // Replaces the implicit cuBLAS GEMMs of reference tensor_parallel/layers.py
@@ -34,8 +37,34 @@ __device__ __forceinline__ s16x8 zero8() {
   return z;
 }
 
-// staging: 4 passes × 256 threads × 8 bf16 cover a 128×64 tile
-// pass p, thread t -> lin = p*256+t; row = lin>>3, group g = lin&7
+// async 16B global->LDS (lane-linear dest). lds generic pointer cast per
+// the CK idiom; gfx950 supports 16 B per lane.
+__device__ __forceinline__ void glds16(const void* g, void* lds) {
+  __builtin_amdgcn_global_load_lds(
+      (const __attribute__((address_space(1))) unsigned int*)g,
+      (__attribute__((address_space(3))) unsigned int*)(uintptr_t)(lds), 16, 0, 0);
+}
+
+// ---------------------------------------------------------------------------
+// Staging.  LDS image (both paths identical): linear [128][64] tile where
+// LDS(row, g) holds GLOBAL column group g ^ (row & 7)  (g = 16B group 0..7).
+// Fragment reads therefore XOR their group index with (row & 7).
+// ---------------------------------------------------------------------------
+// fast path: one glds instruction stages 8 rows (64 lanes x 16B);
+// each wave covers 32 rows of the 128-row tile with 4 calls.
+__device__ __forceinline__ void stage_glds(
+    unsigned short* lds, const unsigned short* src, long long ld, int row0,
+    int k0, int wave, int lane) {
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int row = wave * 32 + j * 8 + (lane >> 3);
+    int g = (lane & 7) ^ (row & 7);  // source-side swizzle (rule 21)
+    const unsigned short* gp = src + (long long)(row0 + row) * ld + k0 + g * 8;
+    glds16(gp, lds + (wave * 32 + j * 8) * BK);
+  }
+}
+
+// slow path (edges): bounds-checked loads to regs ...
 template <int PASSES>
 __device__ __forceinline__ void load_tile_regs(
     const unsigned short* __restrict__ src, long long ld, int row0, int rows,
@@ -44,7 +73,7 @@ __device__ __forceinline__ void load_tile_regs(
   for (int p = 0; p < PASSES; ++p) {
     int lin = p * 256 + threadIdx.x;
     int row = lin >> 3;
-    int g = lin & 7;
+    int g = (lin & 7) ^ (row & 7);  // read the swizzled source group
     int k = k0 + g * 8;
     bool ok = (row0 + row < rows) && (k < K);
     regs[p] = ok ? *reinterpret_cast<const s16x8*>(src + (long long)(row0 + row) * ld + k)
@@ -52,6 +81,7 @@ __device__ __forceinline__ void load_tile_regs(
   }
 }
 
+// ... then linear ds_writes (same image as glds)
 template <int PASSES>
 __device__ __forceinline__ void write_tile_lds(unsigned short* lds, const s16x8 (&regs)[PASSES]) {
 #pragma unroll
@@ -59,8 +89,7 @@ __device__ __forceinline__ void write_tile_lds(unsigned short* lds, const s16x8 
     int lin = p * 256 + threadIdx.x;
     int row = lin >> 3;
     int g = lin & 7;
-    int gs = g ^ (row & 7);  // T2 XOR swizzle on 16B groups
-    *reinterpret_cast<s16x8*>(lds + row * BK + gs * 8) = regs[p];
+    *reinterpret_cast<s16x8*>(lds + row * BK + g * 8) = regs[p];
   }
 }
 
@@ -72,8 +101,10 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
     unsigned short* __restrict__ C,        // [M,N]
     unsigned short* __restrict__ Cpre,     // [M,N] pre-activation (SAVE_PRE)
     int M, int N, int K) {
-  __shared__ unsigned short As[BM * BK];
-  __shared__ unsigned short Bs[BN * BK];
+  // two double-buffered tiles: buffer b: A at b*2*BM*BK, B at +BM*BK
+  __shared__ unsigned short smem[2 * 2 * BM * BK];
+#define As(b) (smem + (b) * 2 * BM * BK)
+#define Bs(b) (smem + (b) * 2 * BM * BK + BM * BK)
 
   // XCD-aware bijective block swizzle (T1)
   const int nbm = (M + BM - 1) / BM;
@@ -82,9 +113,9 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   int bid = blockIdx.x;
   if (nwg >= 16) {
     const int nx = 8;
-    int q = nwg / nx, r = nwg % nx;
+    int qq = nwg / nx, rr = nwg % nx;
     int xcd = bid % nx, idx = bid / nx;
-    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
   }
   const int bm = bid / nbn;
   const int bn = bid % nbn;
@@ -96,31 +127,49 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int wr = wave >> 1;   // 0..1
   const int wc = wave & 1;    // 0..1
 
+  const bool interior_mn = (m0 + BM <= M) && (n0 + BN <= N);
+
   f32x4 acc[4][4];
 #pragma unroll
   for (int i = 0; i < 4; ++i)
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  s16x8 ra[4], rb[4];
-  load_tile_regs<4>(A, K, m0, M, 0, K, ra);
-  load_tile_regs<4>(B, K, n0, N, 0, K, rb);
-
   const int frow = lane & 15;       // fragment row/col within 16
   const int kq = lane >> 4;         // k quarter (0..3) of the 32-K step
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    __syncthreads();  // previous tile's LDS reads done
-    write_tile_lds<4>(As, ra);
-    write_tile_lds<4>(Bs, rb);
-    __syncthreads();  // tile visible
+  // ---- prologue: stage tile 0 into buffer 0 -------------------------------
+  s16x8 ra[4], rb[4];
+  bool fast0 = interior_mn && (BK <= K);
+  if (fast0) {
+    stage_glds(As(0), A, K, m0, 0, wave, lane);
+    stage_glds(Bs(0), B, K, n0, 0, wave, lane);
+  } else {
+    load_tile_regs<4>(A, K, m0, M, 0, K, ra);
+    load_tile_regs<4>(B, K, n0, N, 0, K, rb);
+    write_tile_lds<4>(As(0), ra);
+    write_tile_lds<4>(Bs(0), rb);
+  }
 
-    // T14: issue next tile's loads before the MFMA cluster
-    if (k0 + BK < K) {
-      load_tile_regs<4>(A, K, m0, M, k0 + BK, K, ra);
-      load_tile_regs<4>(B, K, n0, N, k0 + BK, K, rb);
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();  // buf[cur] complete (drains in-flight glds too)
+
+    const int kn = k0 + BK;
+    const bool have_next = kn < K;
+    const bool fast_next = interior_mn && (kn + BK <= K);
+    if (have_next) {
+      if (fast_next) {
+        stage_glds(As(cur ^ 1), A, K, m0, kn, wave, lane);
+        stage_glds(Bs(cur ^ 1), B, K, n0, kn, wave, lane);
+      } else {
+        load_tile_regs<4>(A, K, m0, M, kn, K, ra);  // issue loads now,
+        load_tile_regs<4>(B, K, n0, N, kn, K, rb);  // write after compute (T14)
+      }
     }
 
+    const unsigned short* at = As(cur);
+    const unsigned short* bt = Bs(cur);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
       bf16x8 af[4], bf[4];
@@ -128,10 +177,10 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       for (int f = 0; f < 4; ++f) {
         int arow = wr * 64 + f * 16 + frow;
         int ag = (ks * 4 + kq) ^ (arow & 7);
-        af[f] = *reinterpret_cast<const bf16x8*>(&As[arow * BK + ag * 8]);
+        af[f] = *reinterpret_cast<const bf16x8*>(&at[arow * BK + ag * 8]);
         int brow = wc * 64 + f * 16 + frow;
         int bg = (ks * 4 + kq) ^ (brow & 7);
-        bf[f] = *reinterpret_cast<const bf16x8*>(&Bs[brow * BK + bg * 8]);
+        bf[f] = *reinterpret_cast<const bf16x8*>(&bt[brow * BK + bg * 8]);
       }
 #pragma unroll
       for (int i = 0; i < 4; ++i)
@@ -140,6 +189,12 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[i], bf[j], acc[i][j], 0, 0, 0);
     }
+
+    if (have_next && !fast_next) {
+      write_tile_lds<4>(As(cur ^ 1), ra);
+      write_tile_lds<4>(Bs(cur ^ 1), rb);
+    }
+    cur ^= 1;
   }
 
   // epilogue: D[i][j] lane map col = lane&15, row = (lane>>4)*4 + reg

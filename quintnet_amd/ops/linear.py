@@ -96,7 +96,10 @@ class LinearFunction(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             grad_w = g.t() @ x2d
         if ctx.has_bias and ctx.needs_input_grad[2]:
-            grad_b = g.sum(dim=0)
+            if _backend.use_native(g) and _backend.has_ext():
+                grad_b = _backend.ext().colsum(g)
+            else:
+                grad_b = g.sum(dim=0)
         return grad_x, grad_w, grad_b, None
 
 

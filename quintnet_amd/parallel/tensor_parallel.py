@@ -175,11 +175,13 @@ class RowParallelLinear(nn.Module):
         return m
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        if not self.input_is_parallel and self.tp_size > 1:
+        if self.tp_size == 1:
+            # bias fuses into the GEMM epilogue when there is no reduction
+            return fused_linear(x, self.weight, self.bias, None)
+        if not self.input_is_parallel:
             x = x.chunk(self.tp_size, dim=-1)[self.tp_rank].contiguous()
         out = fused_linear(x, self.weight, None, None)
-        if self.tp_size > 1:
-            out = All_Reduce.apply(out, self.tp_group)
+        out = All_Reduce.apply(out, self.tp_group)
         if self.bias is not None:
             out = out + self.bias
         return out
