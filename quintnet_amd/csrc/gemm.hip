@@ -93,7 +93,7 @@ __device__ __forceinline__ void write_tile_lds(unsigned short* lds, const s16x8 
   }
 }
 
-template <int ACT, bool SAVE_PRE>
+template <int ACT, bool SAVE_PRE, bool USE_GLDS>
 __global__ __launch_bounds__(256) void gemm_nt_kernel(
     const unsigned short* __restrict__ A,  // [M,K]
     const unsigned short* __restrict__ B,  // [N,K]
@@ -140,7 +140,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
   // ---- prologue: stage tile 0 into buffer 0 -------------------------------
   s16x8 ra[4], rb[4];
-  bool fast0 = interior_mn && (BK <= K);
+  bool fast0 = USE_GLDS && interior_mn && (BK <= K);
   if (fast0) {
     stage_glds(As(0), A, K, m0, 0, wave, lane);
     stage_glds(Bs(0), B, K, n0, 0, wave, lane);
@@ -157,7 +157,7 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 
     const int kn = k0 + BK;
     const bool have_next = kn < K;
-    const bool fast_next = interior_mn && (kn + BK <= K);
+    const bool fast_next = USE_GLDS && interior_mn && (kn + BK <= K);
     if (have_next) {
       if (fast_next) {
         stage_glds(As(cur ^ 1), A, K, m0, kn, wave, lane);
@@ -222,6 +222,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   }
 }
 
+#include <cstdlib>
+
 void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
                     const unsigned short* bias, unsigned short* C,
                     unsigned short* Cpre, int M, int N, int K, int act,
@@ -230,9 +232,20 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
   const int nbn = (N + BN - 1) / BN;
   dim3 grid(nbm * nbn);
   dim3 block(256);
+  static int use_glds = -1;
+  if (use_glds < 0) {
+    const char* e = getenv("QN_GEMM_GLDS");
+    use_glds = (e && e[0] == '0') ? 0 : 1;  // default: glds fast path
+  }
 #define QN_GEMM_CASE(A_, S_)                                                   \
-  hipLaunchKernelGGL((gemm_nt_kernel<A_, S_>), grid, block, 0, stream, A, B,   \
-                     bias, C, Cpre, M, N, K)
+  do {                                                                         \
+    if (use_glds)                                                              \
+      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, true>), grid, block, 0,       \
+                         stream, A, B, bias, C, Cpre, M, N, K);                \
+    else                                                                       \
+      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, false>), grid, block, 0,      \
+                         stream, A, B, bias, C, Cpre, M, N, K);                \
+  } while (0)
   if (act == QN_ACT_GELU) {
     if (Cpre) QN_GEMM_CASE(QN_ACT_GELU, true); else QN_GEMM_CASE(QN_ACT_GELU, false);
   } else if (act == QN_ACT_RELU) {

@@ -78,7 +78,7 @@ class GPT2Stage(nn.Module):
         if self.is_last_stage:
             x = self.ln_f(x)
             w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
-            x = fused_linear(x, w, None, None)  # logits = h @ wte^T (tied)
+            x = fused_linear(x, w, None, None, prefer_library=True)  # plain GEMM: hipBLASLt
         return x
 
     # ------------------------------------------------------------------

@@ -46,13 +46,17 @@ def _native_ok(x: torch.Tensor, weight: torch.Tensor) -> bool:
 
 class LinearFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, activation):
+    def forward(ctx, x, weight, bias, activation, prefer_library=False):
         act = _ACT_MAP[activation]
         x2d = x.reshape(-1, x.shape[-1])
         ctx.x_shape = x.shape
         ctx.act = act
         pre_act = None
-        if _backend.use_native(x) and _native_ok(x, weight):
+        if (
+            _backend.use_native(x)
+            and _native_ok(x, weight)
+            and not (prefer_library and bias is None and act == _ACT_NONE)
+        ):
             res = _backend.ext().gemm_nt(
                 x2d.contiguous(), weight.contiguous(),
                 bias if bias is not None else None, act,
@@ -100,7 +104,7 @@ class LinearFunction(torch.autograd.Function):
                 grad_b = _backend.ext().colsum(g)
             else:
                 grad_b = g.sum(dim=0)
-        return grad_x, grad_w, grad_b, None
+        return grad_x, grad_w, grad_b, None, None
 
 
 def linear(
@@ -108,6 +112,12 @@ def linear(
     weight: torch.Tensor,
     bias: Optional[torch.Tensor] = None,
     activation: Optional[str] = None,
+    prefer_library: bool = False,
 ) -> torch.Tensor:
-    """y = activation(x @ weight.T + bias), fused on gfx950."""
-    return LinearFunction.apply(x, weight, bias, activation)
+    """y = activation(x @ weight.T + bias), fused on gfx950.
+
+    ``prefer_library=True`` sends a PLAIN (no bias, no activation) GEMM
+    through hipBLASLt instead — per the MI355X design split, the vendor
+    library serves plain GEMMs (e.g. the tied LM head) while fused ones
+    run the hand-written MFMA kernel."""
+    return LinearFunction.apply(x, weight, bias, activation, prefer_library)
