@@ -1,0 +1,51 @@
+"""Shared helpers for the example launchers."""
+
+from __future__ import annotations
+
+import argparse
+
+import torch
+from torch.utils.data import DataLoader
+
+from quintnet_amd.models import Model
+from quintnet_amd.utils.data import CustomDataset, SyntheticMNIST
+
+
+def parse_args(default_config: str = "examples/config.yaml"):
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--config", default=default_config)
+    ap.add_argument("--synthetic", action="store_true", default=False)
+    ap.add_argument("--n-train", type=int, default=2048)
+    ap.add_argument("--n-val", type=int, default=256)
+    return ap.parse_args()
+
+
+def build_model(cfg) -> Model:
+    return Model(
+        img_size=cfg.get("img_size", 28),
+        patch_size=cfg.get("patch_size", 4),
+        in_channels=cfg.get("in_channels", 1),
+        hidden_dim=cfg.get("hidden_dim", 64),
+        n_heads=cfg.get("n_heads", 4),
+        depth=cfg.get("depth", 8),
+        n_classes=cfg.get("n_classes", 10),
+    )
+
+
+def build_loaders(cfg, args):
+    if cfg.get("dataset_path"):
+        train = CustomDataset(cfg["dataset_path"], "train")
+        val = CustomDataset(cfg["dataset_path"], "test")
+    else:
+        train = SyntheticMNIST(n=args.n_train, seed=0)
+        val = SyntheticMNIST(n=args.n_val, seed=1)
+    bs = cfg.get("batch_size", 8)
+    nw = cfg.get("num_workers", 0)
+    return (
+        DataLoader(train, batch_size=bs, shuffle=False, num_workers=nw),
+        DataLoader(val, batch_size=bs, shuffle=False, num_workers=nw),
+    )
+
+
+def device_type() -> str:
+    return "cuda" if torch.cuda.is_available() else "cpu"

@@ -55,10 +55,15 @@ class ZeroRedundancyAdamW:
         self.max_grad_norm = max_grad_norm
         self.step_count = 0
 
-        self.dp_size = (
-            dist.get_world_size(group=dp_group) if dist.is_initialized() else 1
-        )
-        self.dp_rank = dist.get_rank(group=dp_group) if dist.is_initialized() else 0
+        # dp_group=None means NO sharding (a PP/TP-only rank must not
+        # shard over the world group — ranks hold different params).
+        # Pass dist.group.WORLD explicitly for world-wide ZeRO.
+        if dp_group is None or not dist.is_initialized():
+            self.dp_size, self.dp_rank = 1, 0
+            self.dp_group = None
+        else:
+            self.dp_size = dist.get_world_size(group=dp_group)
+            self.dp_rank = dist.get_rank(group=dp_group)
 
         device = self.params[0].device
         self.dtype = self.params[0].dtype

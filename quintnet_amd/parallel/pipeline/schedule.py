@@ -76,10 +76,19 @@ class PipelineSchedule(abc.ABC):
         m = self.model
         if hasattr(m, "finalize_gradients"):
             m.finalize_gradients()
-        if hasattr(m, "module") and hasattr(m.module, "sync_tied_weights_grad"):
-            m.module.sync_tied_weights_grad()
-        elif hasattr(m, "sync_tied_weights_grad"):
-            m.sync_tied_weights_grad()
+        # find the tied-weight owner through DataParallel(.module) /
+        # PipelineParallelWrapper(.local_module) nesting
+        obj = m
+        for _ in range(4):
+            if hasattr(obj, "sync_tied_weights_grad"):
+                obj.sync_tied_weights_grad()
+                return
+            if hasattr(obj, "module"):
+                obj = obj.module
+            elif hasattr(obj, "local_module"):
+                obj = obj.local_module
+            else:
+                return
 
     def _stage_input(self, batch: Dict[str, Any], device) -> torch.Tensor:
         if self.task_type == "clm":

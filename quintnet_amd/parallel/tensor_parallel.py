@@ -34,13 +34,17 @@ def ensure_divisibility(numerator: int, denominator: int) -> None:
 
 
 def _group_size(group) -> int:
-    if group is None and not dist.is_initialized():
+    # None = NO tensor parallelism (size 1).  A PP/DP-only rank must
+    # never fall back to the default world group — that silently turns
+    # replicated layers into world-wide TP.  Pass dist.group.WORLD
+    # explicitly to shard over the whole world.
+    if group is None or not dist.is_initialized():
         return 1
     return dist.get_world_size(group=group)
 
 
 def _group_rank(group) -> int:
-    if group is None and not dist.is_initialized():
+    if group is None or not dist.is_initialized():
         return 0
     return dist.get_rank(group=group)
 
@@ -105,7 +109,8 @@ class ColumnParallelLinear(nn.Module):
         return m
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = copy_to_group(x, self.tp_group)  # identity fwd, all-reduce grad bwd
+        if self.tp_size > 1:
+            x = copy_to_group(x, self.tp_group)  # identity fwd, all-reduce grad bwd
         out = fused_linear(x, self.weight, self.bias, self.activation)
         if self.gather_output and self.tp_size > 1:
             out = All_Gather.apply(out, self.tp_group, -1, "slice")

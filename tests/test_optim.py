@@ -42,7 +42,9 @@ def _zero_dp2(rank, world):
     ref = copy.deepcopy(model)
 
     ddp = DataParallel(model)
-    opt = ZeroRedundancyAdamW(ddp.parameters(), lr=1e-2, weight_decay=0.01)
+    opt = ZeroRedundancyAdamW(
+        ddp.parameters(), lr=1e-2, weight_decay=0.01, dp_group=dist.group.WORLD
+    )
 
     ref_opt = torch.optim.AdamW(ref.parameters(), lr=1e-2, weight_decay=0.01)
 

@@ -17,7 +17,7 @@ def _column_parallel_equiv(rank, world):
     # broadcast the full weights so every rank shards the same linear
     for t in (lin.weight, lin.bias):
         dist.broadcast(t.data, src=0)
-    col = ColumnParallelLinear.from_linear(lin, tp_group=None, gather_output=True)
+    col = ColumnParallelLinear.from_linear(lin, tp_group=dist.group.WORLD, gather_output=True)
     x = torch.randn(4, 16)
     dist.broadcast(x, src=0)
     x1 = x.clone().requires_grad_(True)
@@ -44,7 +44,7 @@ def _row_parallel_equiv(rank, world):
     lin = nn.Linear(16, 8)
     for t in (lin.weight, lin.bias):
         dist.broadcast(t.data, src=0)
-    row = RowParallelLinear.from_linear(lin, tp_group=None, input_is_parallel=False)
+    row = RowParallelLinear.from_linear(lin, tp_group=dist.group.WORLD, input_is_parallel=False)
     x = torch.randn(4, 16)
     dist.broadcast(x, src=0)
     y_ref = lin(x)
@@ -61,7 +61,7 @@ def _vocab_parallel_equiv(rank, world):
     torch.manual_seed(2)
     full = nn.Embedding(32, 8)
     dist.broadcast(full.weight.data, src=0)
-    vp = VocabParallelEmbedding(32, 8)
+    vp = VocabParallelEmbedding(32, 8, tp_group=dist.group.WORLD)
     with torch.no_grad():
         vp.weight.copy_(full.weight[vp.vocab_start : vp.vocab_end])
     ids = torch.randint(0, 32, (3, 5))
@@ -82,7 +82,7 @@ def _rewriter(rank, world):
     import copy
 
     ref = copy.deepcopy(model)
-    apply_tensor_parallel(model, tp_size=world, tp_rank=rank, tp_group=None)
+    apply_tensor_parallel(model, tp_size=world, tp_rank=rank, tp_group=dist.group.WORLD)
     # divisible layers replaced, non-divisible (out=3) left alone
     assert isinstance(model[0], ColumnParallelLinear)
     assert isinstance(model[3], nn.Linear)
