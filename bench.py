@@ -170,8 +170,8 @@ def main():
         seq, hidden = args.seq_len, cfg.n_embd
         model_name = "gpt2-124M" if not args.tiny else "gpt2-tiny"
     else:
-        micro_b = args.micro_batch or 8
-        grad_acc = args.grad_acc or 8
+        micro_b = args.micro_batch or 16
+        grad_acc = args.grad_acc or 4
         model, _ = build_vit(args, pg, device, dtype)
         g = torch.Generator(device="cpu").manual_seed(1234 + pg.dp_rank)
         batches = []
