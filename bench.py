@@ -157,8 +157,11 @@ def main():
     rank = pg.rank
 
     if args.model == "gpt2":
-        micro_b = args.micro_batch or 8
-        grad_acc = args.grad_acc or 4
+        # pp>1 wants >=4 micro-batches for 1F1B overlap; pp=1 prefers
+        # bigger GEMMs (same 32 seqs/replica/step either way)
+        pp = 2 if n == 8 else 1
+        micro_b = args.micro_batch or (8 if pp > 1 else 16)
+        grad_acc = args.grad_acc or (4 if pp > 1 else 2)
         model, cfg = build_gpt2(args, pg, device, dtype)
         vocab = cfg.vocab_size
         g = torch.Generator(device="cpu").manual_seed(1234 + pg.dp_rank)

@@ -81,17 +81,28 @@ __device__ __forceinline__ int drow(int r, int lane) {
 
 // cooperative transpose-stage of a [rows<=32, 64] strided tile into
 // LDS t[64][TPAD] (t[d][r]); 256 threads, 8 contiguous elems each.
+// Split into load (issue early, overlap with compute — T14) and write.
+// thread -> (row, d-chunk) map r = t&31, d0 = (t>>5)*8: the ds_write_u16
+// bank pattern is then 4-way (r spreads banks) instead of 16-way.
+__device__ __forceinline__ s16x8 stage_ld(const unsigned short* src,
+                                          long long row_stride, int row0) {
+  int r = threadIdx.x & 31;
+  int d0 = (threadIdx.x >> 5) << 3;
+  const unsigned short* p = src + (long long)(row0 + r) * row_stride + d0;
+  return *reinterpret_cast<const s16x8*>(p);
+}
+
+__device__ __forceinline__ void stage_wr(unsigned short* t, s16x8 v) {
+  int r = threadIdx.x & 31;
+  int d0 = (threadIdx.x >> 5) << 3;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) t[(d0 + j) * TPAD + r] = (unsigned short)v[j];
+}
+
 __device__ __forceinline__ void stage_transpose(
     unsigned short* t, const unsigned short* src, long long row_stride,
     int row0, int nrows) {
-  int r = threadIdx.x >> 3;
-  int d0 = (threadIdx.x & 7) << 3;
-  if (r < nrows) {
-    const unsigned short* p = src + (long long)(row0 + r) * row_stride + d0;
-    s16x8 v = *reinterpret_cast<const s16x8*>(p);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) t[(d0 + j) * TPAD + r] = (unsigned short)v[j];
-  }
+  if ((threadIdx.x & 31) < nrows) stage_wr(t, stage_ld(src, row_stride, row0));
 }
 
 // ===========================================================================
@@ -131,52 +142,88 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   f32x16 o[2];
 #pragma unroll
   for (int i = 0; i < 16; ++i) { o[0][i] = 0.f; o[1][i] = 0.f; }
+  f32x16 zc;  // persistent zero C operand (keeps per-tile init out of the loop)
+#pragma unroll
+  for (int i = 0; i < 16; ++i) zc[i] = 0.f;
 
   const float s2scale = scale * LOG2E;
   const int kv_end = causal ? min(q0 + 128, T) : T;
 
+  // prefetch pipeline: tile t's K fragments + V staging rows load during
+  // tile t-1's MFMA cluster (T14) — the per-iteration global latency was
+  // the dominant cost of the unpipelined version.
+  bf16x8 kf_n[4];
+  s16x8 v_n;
+  v_n = stage_ld(vp, vsT, 0);
+#pragma unroll
+  for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_ld(kp, ksT, 0, kt * 16, lane);
+
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    // cooperative V^T staging
+    // cooperative V^T staging (write the prefetched rows)
     __syncthreads();
-    stage_transpose(vt, vp, vsT, kv0, min(32, T - kv0));
+    stage_wr(vt, v_n);
     __syncthreads();
 
+    bf16x8 kf_c[4];
+#pragma unroll
+    for (int kt = 0; kt < 4; ++kt) kf_c[kt] = kf_n[kt];
+    if (kv0 + 32 < kv_end) {
+      v_n = stage_ld(vp, vsT, kv0 + 32);
+#pragma unroll
+      for (int kt = 0; kt < 4; ++kt)
+        kf_n[kt] = frag_ld(kp, ksT, kv0 + 32, kt * 16, lane);
+    }
+
     if (!causal || kv0 <= qw + 31) {  // wave has at least one valid pair
-      // S^T[key][q] = sum_d K[key][d] Q[q][d]
-      f32x16 s;
+      // S^T[key][q] = sum_d K[key][d] Q[q][d]  (first mfma takes the
+      // persistent zero C — no per-tile accumulator re-init)
+      f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[0], qf[0], zc, 0, 0, 0);
 #pragma unroll
-      for (int i = 0; i < 16; ++i) s[i] = 0.f;
-#pragma unroll
-      for (int kt = 0; kt < 4; ++kt) {
-        bf16x8 kf = frag_ld(kp, ksT, kv0, kt * 16, lane);
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[kt], s, 0, 0, 0);
+      for (int kt = 1; kt < 4; ++kt) {
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[kt], qf[kt], s, 0, 0, 0);
       }
-      // scale to base-2, causal mask; per-lane row stats over 16 regs
+      // scale to base-2; causal mask only on the diagonal tile
+      const bool diag = causal && (kv0 + 31 > qw);
       float ps[16];
       float pmax = -INFINITY;
+      if (diag) {
 #pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int key = kv0 + drow(r, lane);
-        float x = s[r] * s2scale;
-        if (causal && key > myq) x = -INFINITY;
-        ps[r] = x;
-        pmax = fmaxf(pmax, x);
+        for (int r = 0; r < 16; ++r) {
+          int key = kv0 + drow(r, lane);
+          float x = s[r] * s2scale;
+          if (key > myq) x = -INFINITY;
+          ps[r] = x;
+          pmax = fmaxf(pmax, x);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          ps[r] = s[r] * s2scale;
+          pmax = fmaxf(pmax, ps[r]);
+        }
       }
       pmax = fmaxf(pmax, __shfl_xor(pmax, 32, 64));
-      float mnew = fmaxf(m2, pmax);
-      float alpha = (m2 == -INFINITY) ? 0.f : exp2f(m2 - mnew);
+      // defer-max (guide T13, THR=8 in base-2): skip the O/l rescale
+      // when no row max grew past the threshold; P <= 2^8 which bf16
+      // accumulation tolerates.  The rescale path is also what forces
+      // the O accumulator out of AGPRs — skipping it most tiles is the
+      // main VALU saving.
+      if (!__all(pmax - m2 <= 8.0f)) {
+        float mnew = fmaxf(m2, pmax);
+        float alpha = __builtin_amdgcn_exp2f(m2 - mnew);  // exp2(-inf)=0 first tile
+        l *= alpha;
+        m2 = mnew;
+#pragma unroll
+        for (int i = 0; i < 16; ++i) { o[0][i] *= alpha; o[1][i] *= alpha; }
+      }
       float psum = 0.f;
 #pragma unroll
       for (int r = 0; r < 16; ++r) {
-        ps[r] = (ps[r] == -INFINITY) ? 0.f : exp2f(ps[r] - mnew);
+        ps[r] = __builtin_amdgcn_exp2f(ps[r] - m2);  // exp2(-inf - m2) = 0
         psum += ps[r];
       }
       psum += __shfl_xor(psum, 32, 64);
-      l = l * alpha + psum;
-      m2 = mnew;
-      // rescale O^T
-#pragma unroll
-      for (int i = 0; i < 16; ++i) { o[0][i] *= alpha; o[1][i] *= alpha; }
+      l += psum;
       // P fragments (B-operand, k = key): two 16-key blocks
       bf16x8 pf0 = relayout8(ps);
       bf16x8 pf1 = relayout8(ps + 8);
@@ -203,7 +250,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         op[(long long)myq * osT + d] = f32_to_bf16(o[mt][r] * inv);
       }
     }
-    if (lane < 32) lse2[(long long)bh * T + myq] = m2 + log2f(l);
+    if (lane < 32) lse2[(long long)bh * T + myq] = m2 + __log2f(l);
   }
 }
 
@@ -269,33 +316,59 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const float my_delta = delta[(long long)bh * T + min(myq, T - 1)];
   const float s2scale = scale * LOG2E;
 
-  f32x16 dqa[2];
+  f32x16 dqa[2], zc;
 #pragma unroll
-  for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; }
+  for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
 
   const int kv_end = causal ? min(q0 + 128, T) : T;
+  // prefetch pipeline (T14): next tile's K/V fragments + Kt staging rows
+  bf16x8 kf_n[4], vf_n[4];
+  s16x8 kst_n = stage_ld(kp, ksT, 0);
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    kf_n[t] = frag_ld(kp, ksT, 0, t * 16, lane);
+    vf_n[t] = frag_ld(vp, vsT, 0, t * 16, lane);
+  }
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     __syncthreads();
-    stage_transpose(kt_lds, kp, ksT, kv0, min(32, T - kv0));
+    stage_wr(kt_lds, kst_n);
     __syncthreads();
+
+    bf16x8 kf_c[4], vf_c[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) { kf_c[t] = kf_n[t]; vf_c[t] = vf_n[t]; }
+    if (kv0 + 32 < kv_end) {
+      kst_n = stage_ld(kp, ksT, kv0 + 32);
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        kf_n[t] = frag_ld(kp, ksT, kv0 + 32, t * 16, lane);
+        vf_n[t] = frag_ld(vp, vsT, kv0 + 32, t * 16, lane);
+      }
+    }
     if (causal && kv0 > qw + 31) continue;
 
-    f32x16 s, dp_;
+    f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[0], qf[0], zc, 0, 0, 0);
+    f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_c[0], dof[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int i = 0; i < 16; ++i) { s[i] = 0.f; dp_[i] = 0.f; }
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      bf16x8 kf = frag_ld(kp, ksT, kv0, t * 16, lane);
-      bf16x8 vf = frag_ld(vp, vsT, kv0, t * 16, lane);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qf[t], s, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dof[t], dp_, 0, 0, 0);
+    for (int t = 1; t < 4; ++t) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[t], qf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_c[t], dof[t], dp_, 0, 0, 0);
     }
+    const bool diag = causal && (kv0 + 31 > qw);
     float g[16];
+    if (diag) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      int key = kv0 + drow(r, lane);
-      float p = (causal && key > myq) ? 0.f : exp2f(s[r] * s2scale - my_lse);
-      g[r] = scale * p * (dp_[r] - my_delta);
+      for (int r = 0; r < 16; ++r) {
+        int key = kv0 + drow(r, lane);
+        float p = (key > myq) ? 0.f : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+        g[r] = scale * p * (dp_[r] - my_delta);
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        float p = __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+        g[r] = scale * p * (dp_[r] - my_delta);
+      }
     }
     bf16x8 gf0 = relayout8(g);
     bf16x8 gf1 = relayout8(g + 8);
@@ -366,43 +439,82 @@ __global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
   }
   const float s2scale = scale * LOG2E;
 
-  f32x16 dka[2], dva[2];
+  f32x16 dka[2], dva[2], zc;
 #pragma unroll
-  for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; }
+  for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; zc[i] = 0.f; }
 
   const int q_start = causal ? (kv0b / 32) * 32 : 0;
+  // prefetch pipeline (T14): next q-tile's Q/dO fragments, staging rows,
+  // lse/delta — the unpipelined version was per-iteration latency-bound.
+  bf16x8 qf_n[4], dof_n[4];
+  s16x8 dost_n = stage_ld(dop, dsT, q_start);
+  s16x8 qst_n = stage_ld(qp, qsT, q_start);
+  float lse_n = 0.f, del_n = 0.f;
+  if (threadIdx.x < 32) {
+    lse_n = lse2[(long long)bh * T + q_start + threadIdx.x];
+    del_n = delta[(long long)bh * T + q_start + threadIdx.x];
+  }
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    qf_n[t] = frag_ld(qp, qsT, q_start, t * 16, lane);
+    dof_n[t] = frag_ld(dop, dsT, q_start, t * 16, lane);
+  }
   for (int qt0 = q_start; qt0 < T; qt0 += 32) {
     __syncthreads();
-    stage_transpose(dot_lds, dop, dsT, qt0, min(32, T - qt0));
-    stage_transpose(qt_lds, qp, qsT, qt0, min(32, T - qt0));
-    if (threadIdx.x < 32 && qt0 + threadIdx.x < T) {
-      lse_t[threadIdx.x] = lse2[(long long)bh * T + qt0 + threadIdx.x];
-      del_t[threadIdx.x] = delta[(long long)bh * T + qt0 + threadIdx.x];
+    stage_wr(dot_lds, dost_n);
+    stage_wr(qt_lds, qst_n);
+    if (threadIdx.x < 32) {
+      lse_t[threadIdx.x] = lse_n;
+      del_t[threadIdx.x] = del_n;
     }
     __syncthreads();
+
+    bf16x8 qf_c[4], dof_c[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) { qf_c[t] = qf_n[t]; dof_c[t] = dof_n[t]; }
+    if (qt0 + 32 < T) {
+      dost_n = stage_ld(dop, dsT, qt0 + 32);
+      qst_n = stage_ld(qp, qsT, qt0 + 32);
+      if (threadIdx.x < 32) {
+        lse_n = lse2[(long long)bh * T + qt0 + 32 + threadIdx.x];
+        del_n = delta[(long long)bh * T + qt0 + 32 + threadIdx.x];
+      }
+#pragma unroll
+      for (int t = 0; t < 4; ++t) {
+        qf_n[t] = frag_ld(qp, qsT, qt0 + 32, t * 16, lane);
+        dof_n[t] = frag_ld(dop, dsT, qt0 + 32, t * 16, lane);
+      }
+    }
     if (causal && qt0 + 31 < kw) continue;  // whole tile above diagonal
 
     // S[q][key]: A = Q (i=q), B = K (j=key); dP[q][key]: A = dO, B = V
-    f32x16 s, dp_;
+    f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_c[0], kf[0], zc, 0, 0, 0);
+    f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_c[0], vf[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int i = 0; i < 16; ++i) { s[i] = 0.f; dp_[i] = 0.f; }
-#pragma unroll
-    for (int t = 0; t < 4; ++t) {
-      bf16x8 qfr = frag_ld(qp, qsT, qt0, t * 16, lane);
-      bf16x8 dofr = frag_ld(dop, dsT, qt0, t * 16, lane);
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qfr, kf[t], s, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dofr, vf[t], dp_, 0, 0, 0);
+    for (int t = 1; t < 4; ++t) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_c[t], kf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_c[t], vf[t], dp_, 0, 0, 0);
     }
+    const bool diag = causal && (qt0 < kw + 31);  // tile crosses the diagonal
     float pv[16], gv[16];
+    if (diag) {
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      int qrow = drow(r, lane);
-      int qg = qt0 + qrow;
-      float p = (qg >= T || (causal && mykey > qg))
-                    ? 0.f
-                    : exp2f(s[r] * s2scale - lse_t[qrow]);
-      pv[r] = p;
-      gv[r] = scale * p * (dp_[r] - del_t[qrow]);
+      for (int r = 0; r < 16; ++r) {
+        int qrow = drow(r, lane);
+        float p = (mykey > qt0 + qrow)
+                      ? 0.f
+                      : __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[qrow]);
+        pv[r] = p;
+        gv[r] = scale * p * (dp_[r] - del_t[qrow]);
+      }
+    } else {
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int qrow = drow(r, lane);
+        float p = __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[qrow]);
+        pv[r] = p;
+        gv[r] = scale * p * (dp_[r] - del_t[qrow]);
+      }
     }
     bf16x8 pf0 = relayout8(pv), pf1 = relayout8(pv + 8);
     bf16x8 gf0 = relayout8(gv), gf1 = relayout8(gv + 8);

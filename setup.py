@@ -38,7 +38,8 @@ setup(
             sources=sources,
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
-                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950"],
+                "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950",
+                         "-mllvm", "-amdgpu-mfma-vgpr-form"],
             },
         )
     ],
