@@ -51,17 +51,16 @@ __device__ __forceinline__ void glds16(const void* g, void* lds) {
 // Fragment reads therefore XOR their group index with (row & 7).
 // ---------------------------------------------------------------------------
 // fast path: one glds instruction stages 8 rows (64 lanes x 16B);
-// each wave covers 32 rows of the 128-row tile with 4 calls.
-__device__ __forceinline__ void stage_glds(
-    unsigned short* lds, const unsigned short* src, long long ld, int row0,
-    int k0, int wave, int lane) {
+// each wave covers 32 rows of the 128-row tile with 4 calls.  The
+// per-lane source pointers are precomputed once and bumped by BK per
+// tile — the 64-bit address rebuild per tile was ~40% of the kernel's
+// VALU issue (PMC: 6.2 VALU/MFMA).
+__device__ __forceinline__ void stage_glds_pre(
+    unsigned short* lds, const unsigned short* const (&gp)[4], int k_elems,
+    int wave) {
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int row = wave * 32 + j * 8 + (lane >> 3);
-    int g = (lane & 7) ^ (row & 7);  // source-side swizzle (rule 21)
-    const unsigned short* gp = src + (long long)(row0 + row) * ld + k0 + g * 8;
-    glds16(gp, lds + (wave * 32 + j * 8) * BK);
-  }
+  for (int j = 0; j < 4; ++j)
+    glds16(gp[j] + k_elems, lds + (wave * 32 + j * 8) * BK);
 }
 
 // slow path (edges): bounds-checked loads to regs ...
@@ -138,64 +137,94 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int frow = lane & 15;       // fragment row/col within 16
   const int kq = lane >> 4;         // k quarter (0..3) of the 32-K step
 
+  // precomputed per-lane glds source pointers (k advances via offset)
+  const unsigned short* gpa[4];
+  const unsigned short* gpb[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    int row = wave * 32 + j * 8 + (lane >> 3);
+    int g = (lane & 7) ^ (row & 7);  // source-side swizzle (rule 21)
+    gpa[j] = A + (long long)(m0 + row) * K + g * 8;
+    gpb[j] = B + (long long)(n0 + row) * K + g * 8;
+  }
+  // precomputed LDS fragment offsets (elements); ks=1 toggles bit 5 (^32)
+  int aoff[4], boff[4];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    int arow = wr * 64 + f * 16 + frow;
+    aoff[f] = arow * BK + ((kq ^ (arow & 7)) << 3);
+    int brow = wc * 64 + f * 16 + frow;
+    boff[f] = brow * BK + ((kq ^ (brow & 7)) << 3);
+  }
+
+#define QN_MFMA_TILE(at, bt)                                                     _Pragma("unroll")                                                              for (int ks = 0; ks < 2; ++ks) {                                                 bf16x8 af[4], bf[4];                                                           _Pragma("unroll")                                                              for (int f = 0; f < 4; ++f) {                                                    af[f] = *reinterpret_cast<const bf16x8*>(&(at)[aoff[f] ^ (ks << 5)]);          bf[f] = *reinterpret_cast<const bf16x8*>(&(bt)[boff[f] ^ (ks << 5)]);        }                                                                              _Pragma("unroll")                                                              for (int i = 0; i < 4; ++i)                                                      _Pragma("unroll")                                                              for (int j = 0; j < 4; ++j)                                                      acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                               af[i], bf[j], acc[i][j], 0, 0, 0);                                   }
+
   // ---- prologue: stage tile 0 into buffer 0 -------------------------------
   s16x8 ra[4], rb[4];
-  bool fast0 = USE_GLDS && interior_mn && (BK <= K);
-  if (fast0) {
-    stage_glds(As(0), A, K, m0, 0, wave, lane);
-    stage_glds(Bs(0), B, K, n0, 0, wave, lane);
+  const bool all_fast = USE_GLDS && interior_mn && (K % BK == 0);
+  if (all_fast) {
+    // fully-interior fast loop: unrolled x2 so buffer pointers and LDS
+    // offsets stay loop-invariant; glds sources advance by += k
+    stage_glds_pre(As(0), gpa, 0, wave);
+    stage_glds_pre(Bs(0), gpb, 0, wave);
+    int k0 = 0;
+    while (true) {
+      // even tile in buf0
+      __syncthreads();
+      if (k0 + BK < K) {
+        stage_glds_pre(As(1), gpa, k0 + BK, wave);
+        stage_glds_pre(Bs(1), gpb, k0 + BK, wave);
+      }
+      QN_MFMA_TILE(As(0), Bs(0));
+      k0 += BK;
+      if (k0 >= K) break;
+      // odd tile in buf1
+      __syncthreads();
+      if (k0 + BK < K) {
+        stage_glds_pre(As(0), gpa, k0 + BK, wave);
+        stage_glds_pre(Bs(0), gpb, k0 + BK, wave);
+      }
+      QN_MFMA_TILE(As(1), Bs(1));
+      k0 += BK;
+      if (k0 >= K) break;
+    }
   } else {
-    load_tile_regs<4>(A, K, m0, M, 0, K, ra);
-    load_tile_regs<4>(B, K, n0, N, 0, K, rb);
-    write_tile_lds<4>(As(0), ra);
-    write_tile_lds<4>(Bs(0), rb);
-  }
-
-  int cur = 0;
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    __syncthreads();  // buf[cur] complete (drains in-flight glds too)
-
-    const int kn = k0 + BK;
-    const bool have_next = kn < K;
-    const bool fast_next = USE_GLDS && interior_mn && (kn + BK <= K);
-    if (have_next) {
-      if (fast_next) {
-        stage_glds(As(cur ^ 1), A, K, m0, kn, wave, lane);
-        stage_glds(Bs(cur ^ 1), B, K, n0, kn, wave, lane);
-      } else {
-        load_tile_regs<4>(A, K, m0, M, kn, K, ra);  // issue loads now,
-        load_tile_regs<4>(B, K, n0, N, kn, K, rb);  // write after compute (T14)
+    bool fast0 = USE_GLDS && interior_mn && (BK <= K);
+    if (fast0) {
+      stage_glds_pre(As(0), gpa, 0, wave);
+      stage_glds_pre(Bs(0), gpb, 0, wave);
+    } else {
+      load_tile_regs<4>(A, K, m0, M, 0, K, ra);
+      load_tile_regs<4>(B, K, n0, N, 0, K, rb);
+      write_tile_lds<4>(As(0), ra);
+      write_tile_lds<4>(Bs(0), rb);
+    }
+    int cur = 0;
+    for (int k0 = 0; k0 < K; k0 += BK) {
+      __syncthreads();  // buf[cur] complete (drains in-flight glds too)
+      const int kn = k0 + BK;
+      const bool have_next = kn < K;
+      const bool fast_next = USE_GLDS && interior_mn && (kn + BK <= K);
+      if (have_next) {
+        if (fast_next) {
+          stage_glds_pre(As(cur ^ 1), gpa, kn, wave);
+          stage_glds_pre(Bs(cur ^ 1), gpb, kn, wave);
+        } else {
+          load_tile_regs<4>(A, K, m0, M, kn, K, ra);  // issue loads now,
+          load_tile_regs<4>(B, K, n0, N, kn, K, rb);  // write after compute (T14)
+        }
       }
-    }
-
-    const unsigned short* at = As(cur);
-    const unsigned short* bt = Bs(cur);
-#pragma unroll
-    for (int ks = 0; ks < BK / 32; ++ks) {
-      bf16x8 af[4], bf[4];
-#pragma unroll
-      for (int f = 0; f < 4; ++f) {
-        int arow = wr * 64 + f * 16 + frow;
-        int ag = (ks * 4 + kq) ^ (arow & 7);
-        af[f] = *reinterpret_cast<const bf16x8*>(&at[arow * BK + ag * 8]);
-        int brow = wc * 64 + f * 16 + frow;
-        int bg = (ks * 4 + kq) ^ (brow & 7);
-        bf[f] = *reinterpret_cast<const bf16x8*>(&bt[brow * BK + bg * 8]);
+      const unsigned short* at = As(cur);
+      const unsigned short* bt = Bs(cur);
+      QN_MFMA_TILE(at, bt);
+      if (have_next && !fast_next) {
+        write_tile_lds<4>(As(cur ^ 1), ra);
+        write_tile_lds<4>(Bs(cur ^ 1), rb);
       }
-#pragma unroll
-      for (int i = 0; i < 4; ++i)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              af[i], bf[j], acc[i][j], 0, 0, 0);
+      cur ^= 1;
     }
-
-    if (have_next && !fast_next) {
-      write_tile_lds<4>(As(cur ^ 1), ra);
-      write_tile_lds<4>(Bs(cur ^ 1), rb);
-    }
-    cur ^= 1;
   }
+#undef QN_MFMA_TILE
 
   // epilogue: D[i][j] lane map col = lane&15, row = (lane>>4)*4 + reg
   const int erow = (lane >> 4) * 4;
