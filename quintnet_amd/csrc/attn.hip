@@ -281,7 +281,7 @@ __global__ void attn_delta_kernel(
 // dQ^T[d][q] += K^T[d][key] · g^T[key][q],
 //   g^T = scale * P^T ⊙ (dP^T - delta[q]),  P^T = exp2(s2 - lse2[q])
 // ===========================================================================
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse2, const float* __restrict__ delta,
@@ -400,7 +400,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 //   dV^T[d][key] += dO^T[d][q] · P[q][key]
 //   dK^T[d][key] += Q^T[d][q] · g[q][key]
 // ===========================================================================
-__global__ __launch_bounds__(256) void attn_bwd_dkv_kernel(
+__global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse2, const float* __restrict__ delta,
