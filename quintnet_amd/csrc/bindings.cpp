@@ -22,6 +22,10 @@ template <typename T>
 void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
 template <typename T>
 void colsum_launch(const T*, float*, long long, int, hipStream_t);
+template <typename T>
+void dropout_fwd_launch(const T*, T*, unsigned char*, long long, float, unsigned, hipStream_t);
+template <typename T>
+void dropout_bwd_launch(const T*, T*, const unsigned char*, long long, float, hipStream_t);
 template <typename TP, typename TG>
 void adamw_launch(TP*, float*, const TG*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
@@ -236,6 +240,33 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   return dl;
 }
 
+std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed) {
+  CHECK_GPU(x);
+  auto xc = x.contiguous();
+  auto y = torch::empty_like(xc);
+  auto mask = torch::empty({xc.numel()}, xc.options().dtype(torch::kUInt8));
+  if (xc.dtype() == torch::kBFloat16)
+    dropout_fwd_launch<unsigned short>(bf16p(xc), bf16p_mut(y), mask.data_ptr<unsigned char>(),
+                                       xc.numel(), (float)p, (unsigned)seed, cur_stream());
+  else
+    dropout_fwd_launch<float>(xc.data_ptr<float>(), y.data_ptr<float>(), mask.data_ptr<unsigned char>(),
+                              xc.numel(), (float)p, (unsigned)seed, cur_stream());
+  return {y, mask};
+}
+
+torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
+  CHECK_GPU(dy);
+  auto dyc = dy.contiguous();
+  auto dx = torch::empty_like(dyc);
+  if (dyc.dtype() == torch::kBFloat16)
+    dropout_bwd_launch<unsigned short>(bf16p(dyc), bf16p_mut(dx), mask.data_ptr<unsigned char>(),
+                                       dyc.numel(), (float)p, cur_stream());
+  else
+    dropout_bwd_launch<float>(dyc.data_ptr<float>(), dx.data_ptr<float>(), mask.data_ptr<unsigned char>(),
+                              dyc.numel(), (float)p, cur_stream());
+  return dx;
+}
+
 torch::Tensor colsum(torch::Tensor x) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
@@ -380,6 +411,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
   m.def("act_bwd", &act_bwd, "fused activation backward");
   m.def("colsum", &colsum, "column sum (bias grad)");
+  m.def("dropout_fwd", &dropout_fwd);
+  m.def("dropout_bwd", &dropout_bwd);
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)");
   m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)");
   m.def("adamw_step", &adamw_step);

@@ -63,3 +63,63 @@ void colsum_launch(const T* x, float* out, long long rows, int N, hipStream_t st
 
 template void colsum_launch<float>(const float*, float*, long long, int, hipStream_t);
 template void colsum_launch<unsigned short>(const unsigned short*, float*, long long, int, hipStream_t);
+
+// ---------------------------------------------------------------------------
+// fused dropout (counter-based PRNG: mask recomputable, stored as u8 for
+// the backward pass).  Replaces eager nn.Dropout (reference
+// gpt2_attention.py:108, gpt2_mlp.py:125, gpt2_embeddings.py:59).
+__device__ __forceinline__ unsigned wang_hash(unsigned s) {
+  s = (s ^ 61u) ^ (s >> 16);
+  s *= 9u;
+  s = s ^ (s >> 4);
+  s *= 0x27d4eb2du;
+  s = s ^ (s >> 15);
+  return s;
+}
+
+template <typename T>
+__global__ void dropout_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                                   unsigned char* __restrict__ mask,
+                                   long long n, float p, float scale,
+                                   unsigned seed) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  const unsigned thresh = (unsigned)(p * 4294967296.0f);
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n; i += stride) {
+    unsigned r = wang_hash(seed ^ (unsigned)(i & 0xffffffffu)) ^ wang_hash((unsigned)(i >> 32) + seed * 2654435761u);
+    unsigned char keep = r >= thresh;
+    mask[i] = keep;
+    st_from_f32(y + i, keep ? ld_as_f32(x + i) * scale : 0.f);
+  }
+}
+
+template <typename T>
+__global__ void dropout_bwd_kernel(const T* __restrict__ dy, T* __restrict__ dx,
+                                   const unsigned char* __restrict__ mask,
+                                   long long n, float scale) {
+  const long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n; i += stride)
+    st_from_f32(dx + i, mask[i] ? ld_as_f32(dy + i) * scale : 0.f);
+}
+
+template <typename T>
+void dropout_fwd_launch(const T* x, T* y, unsigned char* mask, long long n,
+                        float p, unsigned seed, hipStream_t stream) {
+  long long blocks = min((n + 255) / 256, (long long)2048);
+  float scale = 1.0f / (1.0f - p);
+  hipLaunchKernelGGL((dropout_fwd_kernel<T>), dim3((unsigned)blocks), dim3(256),
+                     0, stream, x, y, mask, n, p, scale, seed);
+}
+
+template <typename T>
+void dropout_bwd_launch(const T* dy, T* dx, const unsigned char* mask,
+                        long long n, float p, hipStream_t stream) {
+  long long blocks = min((n + 255) / 256, (long long)2048);
+  float scale = 1.0f / (1.0f - p);
+  hipLaunchKernelGGL((dropout_bwd_kernel<T>), dim3((unsigned)blocks), dim3(256),
+                     0, stream, dy, dx, mask, n, scale);
+}
+
+template void dropout_fwd_launch<float>(const float*, float*, unsigned char*, long long, float, unsigned, hipStream_t);
+template void dropout_fwd_launch<unsigned short>(const unsigned short*, unsigned short*, unsigned char*, long long, float, unsigned, hipStream_t);
+template void dropout_bwd_launch<float>(const float*, float*, const unsigned char*, long long, float, hipStream_t);
+template void dropout_bwd_launch<unsigned short>(const unsigned short*, unsigned short*, const unsigned char*, long long, float, hipStream_t);

@@ -18,7 +18,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ...ops import attention_qkv
+from ...ops import FusedDropout, attention_qkv
 from ...parallel.tensor_parallel import ColumnParallelLinear, RowParallelLinear
 from .config import GPT2Config
 
@@ -50,7 +50,7 @@ class GPT2Attention(nn.Module):
         self.n_head_local = config.n_head // self.tp_size
         self.hidden_local = config.n_embd // self.tp_size
         self.head_dim = config.head_dim
-        self.resid_dropout = nn.Dropout(config.dropout)
+        self.resid_dropout = FusedDropout(config.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         qkv = self.c_attn(x)  # [B, T, 3*n_embd/tp] = [q_loc | k_loc | v_loc]

@@ -9,6 +9,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ...ops import FusedDropout
 from .config import GPT2Config
 
 __all__ = ["GPT2Embedding"]
@@ -21,7 +22,7 @@ class GPT2Embedding(nn.Module):
         self.config = config
         self.wte = nn.Embedding(config.vocab_size, config.n_embd, **kw)
         self.wpe = nn.Embedding(config.n_positions, config.n_embd, **kw)
-        self.drop = nn.Dropout(config.dropout)
+        self.drop = FusedDropout(config.dropout)
         nn.init.normal_(self.wte.weight, std=config.initializer_range)
         nn.init.normal_(self.wpe.weight, std=config.initializer_range)
 

@@ -10,6 +10,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
+from ...ops import FusedDropout
 from ...parallel.tensor_parallel import ColumnParallelLinear, RowParallelLinear
 from .config import GPT2Config
 
@@ -36,7 +37,7 @@ class GPT2MLP(nn.Module):
             device=device,
             dtype=dtype,
         )
-        self.dropout = nn.Dropout(config.dropout)
+        self.dropout = FusedDropout(config.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.dropout(self.c_proj(self.c_fc(x)))

@@ -19,6 +19,7 @@ from .attention import (
 )
 from .cross_entropy import cross_entropy, causal_lm_loss, shift_labels, CrossEntropyFunction
 from .adamw import adamw_step_flat, clip_grad_norm_local, l2_norm
+from .dropout import fused_dropout, FusedDropout
 
 __all__ = [
     "ext",
@@ -43,4 +44,6 @@ __all__ = [
     "adamw_step_flat",
     "clip_grad_norm_local",
     "l2_norm",
+    "fused_dropout",
+    "FusedDropout",
 ]

@@ -140,16 +140,31 @@ class Trainer:
         return self._train_epoch_plain()
 
     def _train_epoch_pipeline(self) -> Dict[str, float]:
+        from .utils.profiling import StepTimer
+
         loader = PipelineDataLoader(self.train_loader, self.grad_acc_steps, self.task_type)
         num_steps = max(len(self.train_loader) // self.grad_acc_steps, 1)
         agg: Dict[str, float] = {}
+        timer = StepTimer() if self.config.get("profile") else None
         for _ in range(num_steps):
+            if timer:
+                timer.start()
             m = self.pipeline_trainer.train_step(
                 loader, self.tensor_shapes, self.device, self._dtype
             )
+            if timer:
+                timer.stop()
             for k, v in m.items():
                 agg[k] = agg.get(k, 0.0) + float(v)
-        return {k: v / num_steps for k, v in agg.items()}
+        out = {k: v / num_steps for k, v in agg.items()}
+        if timer and self._is_rank0():
+            bs = self._infer_micro_batch() * self.grad_acc_steps
+            print(
+                f"[profile] {timer.mean_ms:.1f} ms/step, "
+                f"{1000.0 * bs / max(timer.mean_ms, 1e-9):.1f} samples/s/replica",
+                flush=True,
+            )
+        return out
 
     def _train_epoch_plain(self) -> Dict[str, float]:
         total_loss, correct, total, steps = 0.0, 0, 0, 0

@@ -250,3 +250,20 @@ def test_flash_attention_long_seq():
         q.float(), k.float(), v.float(), is_causal=True
     )
     assert (out.float() - ref).abs().max() < 3e-2
+
+
+def test_fused_dropout():
+    from quintnet_amd.ops import fused_dropout
+
+    x = torch.ones(100000, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y = fused_dropout(x, 0.3, training=True)
+    kept = (y != 0).float().mean()
+    assert abs(float(kept) - 0.7) < 0.02, float(kept)
+    # kept values scaled by 1/(1-p)
+    nz = y[y != 0].float()
+    assert torch.allclose(nz, torch.full_like(nz, 1.0 / 0.7), atol=1e-2)
+    y.sum().backward()
+    g = x.grad.float()
+    assert torch.allclose((g != 0).float().mean(), kept, atol=1e-3)
+    # eval mode / p=0: identity
+    assert fused_dropout(x, 0.3, training=False) is x
