@@ -7,8 +7,14 @@ template <typename TP, typename TG>
 __global__ void adamw_kernel(
     TP* __restrict__ param, float* __restrict__ master, const TG* __restrict__ grad,
     float* __restrict__ m, float* __restrict__ v,
+    const long long* __restrict__ step_dev,
     long long n, float lr, float beta1, float beta2, float eps, float wd,
     float bc1, float bc2) {
+  if (step_dev) {  // hipGraph-replay path: bias correction from device step
+    float st = (float)*step_dev;
+    bc1 = 1.f - __powf(beta1, st);
+    bc2 = 1.f - __powf(beta2, st);
+  }
   const long long stride = (long long)gridDim.x * blockDim.x;
   for (long long i = blockIdx.x * (long long)blockDim.x + threadIdx.x; i < n; i += stride) {
     float g = ld_as_f32(grad + i);
@@ -38,14 +44,14 @@ __global__ void sumsq_kernel(const T* __restrict__ x, long long n, float* __rest
 
 template <typename TP, typename TG>
 void adamw_launch(TP* param, float* master, const TG* grad, float* m, float* v,
-                  long long n, float lr, float beta1, float beta2, float eps,
-                  float wd, int step, hipStream_t stream) {
+                  const long long* step_dev, long long n, float lr, float beta1,
+                  float beta2, float eps, float wd, int step, hipStream_t stream) {
   float bc1 = 1.f - powf(beta1, (float)step);
   float bc2 = 1.f - powf(beta2, (float)step);
   long long blocks = min((n + 255) / 256, (long long)2048);
   hipLaunchKernelGGL((adamw_kernel<TP, TG>), dim3((unsigned)blocks), dim3(256), 0,
-                     stream, param, master, grad, m, v, n, lr, beta1, beta2, eps,
-                     wd, bc1, bc2);
+                     stream, param, master, grad, m, v, step_dev, n, lr, beta1,
+                     beta2, eps, wd, bc1, bc2);
 }
 
 template <typename T>
@@ -54,9 +60,9 @@ void sumsq_launch(const T* x, long long n, float* out, hipStream_t stream) {
   hipLaunchKernelGGL((sumsq_kernel<T>), dim3((unsigned)blocks), dim3(256), 0, stream, x, n, out);
 }
 
-template void adamw_launch<float, float>(float*, float*, const float*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
-template void adamw_launch<unsigned short, unsigned short>(unsigned short*, float*, const unsigned short*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
-template void adamw_launch<unsigned short, float>(unsigned short*, float*, const float*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
-template void adamw_launch<float, unsigned short>(float*, float*, const unsigned short*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
+template void adamw_launch<float, float>(float*, float*, const float*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
+template void adamw_launch<unsigned short, unsigned short>(unsigned short*, float*, const unsigned short*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
+template void adamw_launch<unsigned short, float>(unsigned short*, float*, const float*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
+template void adamw_launch<float, unsigned short>(float*, float*, const unsigned short*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
 template void sumsq_launch<float>(const float*, long long, float*, hipStream_t);
 template void sumsq_launch<unsigned short>(const unsigned short*, long long, float*, hipStream_t);

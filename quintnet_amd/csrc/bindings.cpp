@@ -17,7 +17,7 @@ void softmax_bwd_launch(const T*, const T*, T*, long long, int, float, hipStream
 template <typename T>
 void ce_fwd_launch(const T*, const long long*, float*, float*, long long, int, long long, hipStream_t);
 template <typename T>
-void ce_bwd_launch(const T*, const long long*, const float*, const float*, T*, long long, int, long long, float, hipStream_t);
+void ce_bwd_launch(const T*, const long long*, const float*, const float*, const long long*, T*, long long, int, long long, float, hipStream_t);
 template <typename T>
 void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
 template <typename T>
@@ -27,7 +27,7 @@ void dropout_fwd_launch(const T*, T*, unsigned char*, long long, float, unsigned
 template <typename T>
 void dropout_bwd_launch(const T*, T*, const unsigned char*, long long, float, hipStream_t);
 template <typename TP, typename TG>
-void adamw_launch(TP*, float*, const TG*, float*, float*, long long, float, float, float, float, float, int, hipStream_t);
+void adamw_launch(TP*, float*, const TG*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
 void sumsq_launch(const T*, long long, float*, hipStream_t);
 void gemm_nt_launch(const unsigned short*, const unsigned short*, const unsigned short*,
@@ -211,7 +211,7 @@ std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits,
 }
 
 torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
-                                torch::Tensor lse, int64_t n_valid,
+                                torch::Tensor lse, torch::Tensor n_valid,
                                 int64_t ignore_index,
                                 c10::optional<torch::Tensor> grad_scale) {
   CHECK_GPU(logits);
@@ -220,7 +220,10 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   int64_t V = lc.size(-1);
   long long rows = lc.numel() / V;
   auto dl = torch::empty_like(lc);
-  float inv_n = 1.0f / (float)std::max<int64_t>(n_valid, 1);
+  auto nv = n_valid.to(torch::kInt64).contiguous();
+  TORCH_CHECK(nv.is_cuda() && nv.numel() == 1, "n_valid must be a GPU scalar");
+  const long long* nvp = (const long long*)nv.data_ptr<int64_t>();
+  float inv_n = 1.0f;
   const float* gp = nullptr;
   torch::Tensor gs;
   if (grad_scale.has_value() && grad_scale->defined()) {
@@ -230,12 +233,12 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   }
   if (lc.dtype() == torch::kBFloat16) {
     ce_bwd_launch<unsigned short>(bf16p(lc), (const long long*)tc.data_ptr<int64_t>(),
-                                  lse.data_ptr<float>(), gp, bf16p_mut(dl), rows,
-                                  (int)V, ignore_index, inv_n, cur_stream());
+                                  lse.data_ptr<float>(), gp, nvp, bf16p_mut(dl),
+                                  rows, (int)V, ignore_index, inv_n, cur_stream());
   } else {
     ce_bwd_launch<float>(lc.data_ptr<float>(), (const long long*)tc.data_ptr<int64_t>(),
-                         lse.data_ptr<float>(), gp, dl.data_ptr<float>(), rows,
-                         (int)V, ignore_index, inv_n, cur_stream());
+                         lse.data_ptr<float>(), gp, nvp, dl.data_ptr<float>(),
+                         rows, (int)V, ignore_index, inv_n, cur_stream());
   }
   return dl;
 }
@@ -298,7 +301,8 @@ torch::Tensor act_bwd(torch::Tensor dy, torch::Tensor pre, int64_t act) {
 // ---------------------------------------------------------------------------
 void adamw_step(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
                 torch::Tensor m, torch::Tensor v, int64_t step, double lr,
-                double beta1, double beta2, double eps, double wd) {
+                double beta1, double beta2, double eps, double wd,
+                c10::optional<torch::Tensor> step_dev) {
   CHECK_GPU(param);
   long long n = param.numel();
   TORCH_CHECK(master.numel() == n && grad.numel() == n && m.numel() == n && v.numel() == n,
@@ -307,19 +311,22 @@ void adamw_step(torch::Tensor param, torch::Tensor master, torch::Tensor grad,
   float* mp = master.data_ptr<float>();
   float* m1 = m.data_ptr<float>();
   float* m2 = v.data_ptr<float>();
+  const long long* sd = nullptr;
+  if (step_dev.has_value() && step_dev->defined())
+    sd = (const long long*)step_dev->data_ptr<int64_t>();
   bool pbf = param.dtype() == torch::kBFloat16;
   bool gbf = grad.dtype() == torch::kBFloat16;
   if (pbf && gbf)
-    adamw_launch<unsigned short, unsigned short>(bf16p_mut(param), mp, bf16p(grad), m1, m2, n,
+    adamw_launch<unsigned short, unsigned short>(bf16p_mut(param), mp, bf16p(grad), m1, m2, sd, n,
         (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
   else if (pbf && !gbf)
-    adamw_launch<unsigned short, float>(bf16p_mut(param), mp, grad.data_ptr<float>(), m1, m2, n,
+    adamw_launch<unsigned short, float>(bf16p_mut(param), mp, grad.data_ptr<float>(), m1, m2, sd, n,
         (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
   else if (!pbf && gbf)
-    adamw_launch<float, unsigned short>(param.data_ptr<float>(), mp, bf16p(grad), m1, m2, n,
+    adamw_launch<float, unsigned short>(param.data_ptr<float>(), mp, bf16p(grad), m1, m2, sd, n,
         (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
   else
-    adamw_launch<float, float>(param.data_ptr<float>(), mp, grad.data_ptr<float>(), m1, m2, n,
+    adamw_launch<float, float>(param.data_ptr<float>(), mp, grad.data_ptr<float>(), m1, m2, sd, n,
         (float)lr, (float)beta1, (float)beta2, (float)eps, (float)wd, (int)step, st);
 }
 

@@ -84,9 +84,11 @@ template <typename T>
 __global__ void ce_bwd_kernel(
     const T* __restrict__ logits, const long long* __restrict__ target,
     const float* __restrict__ lse, const float* __restrict__ grad_scale,
-    T* __restrict__ dlogits, long long rows, int V, long long ignore_index,
-    float inv_n) {
-  const float gs = (grad_scale ? *grad_scale : 1.f) * inv_n;
+    const long long* __restrict__ n_valid, T* __restrict__ dlogits,
+    long long rows, int V, long long ignore_index, float inv_n) {
+  // n_valid read on-device (no host sync; graph-capture safe)
+  const float gs = (grad_scale ? *grad_scale : 1.f) *
+                   (n_valid ? 1.f / (float)max(*n_valid, 1ll) : inv_n);
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     const T* lr = logits + row * V;
     T* dr = dlogits + row * V;
@@ -155,15 +157,16 @@ void ce_fwd_launch(const T* logits, const long long* target, float* lse,
 
 template <typename T>
 void ce_bwd_launch(const T* logits, const long long* target, const float* lse,
-                   const float* grad_scale, T* dlogits, long long rows, int V,
-                   long long ignore_index, float inv_n, hipStream_t stream) {
+                   const float* grad_scale, const long long* n_valid,
+                   T* dlogits, long long rows, int V, long long ignore_index,
+                   float inv_n, hipStream_t stream) {
   int grid = (int)min((long long)2048, rows);
   hipLaunchKernelGGL((ce_bwd_kernel<T>), dim3(grid), dim3(256), 0, stream,
-                     logits, target, lse, grad_scale, dlogits, rows, V,
-                     ignore_index, inv_n);
+                     logits, target, lse, grad_scale, n_valid, dlogits, rows,
+                     V, ignore_index, inv_n);
 }
 
 template void ce_fwd_launch<float>(const float*, const long long*, float*, float*, long long, int, long long, hipStream_t);
 template void ce_fwd_launch<unsigned short>(const unsigned short*, const long long*, float*, float*, long long, int, long long, hipStream_t);
-template void ce_bwd_launch<float>(const float*, const long long*, const float*, const float*, float*, long long, int, long long, float, hipStream_t);
-template void ce_bwd_launch<unsigned short>(const unsigned short*, const long long*, const float*, const float*, unsigned short*, long long, int, long long, float, hipStream_t);
+template void ce_bwd_launch<float>(const float*, const long long*, const float*, const float*, const long long*, float*, long long, int, long long, float, hipStream_t);
+template void ce_bwd_launch<unsigned short>(const unsigned short*, const long long*, const float*, const float*, const long long*, unsigned short*, long long, int, long long, float, hipStream_t);

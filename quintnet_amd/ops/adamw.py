@@ -30,12 +30,16 @@ def adamw_step_flat(
     beta2: float,
     eps: float,
     weight_decay: float,
+    step_dev: Optional[torch.Tensor] = None,  # int64 GPU scalar (graph-safe)
 ) -> None:
-    """One fused AdamW step over a flat shard. Decoupled weight decay."""
+    """One fused AdamW step over a flat shard. Decoupled weight decay.
+
+    When ``step_dev`` is given the kernel reads the step count from the
+    device (bias correction stays correct under hipGraph replay)."""
     if _backend.use_native(param_out) and _backend.has_ext():
         _backend.ext().adamw_step(
             param_out, master, grad, exp_avg, exp_avg_sq,
-            step, lr, beta1, beta2, eps, weight_decay,
+            step, lr, beta1, beta2, eps, weight_decay, step_dev,
         )
         return
     g = grad.float()

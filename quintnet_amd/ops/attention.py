@@ -70,6 +70,10 @@ class AttentionFunction(torch.autograd.Function):
 
 
 def _flash_ok(q: torch.Tensor) -> bool:
+    import os
+
+    if os.environ.get("QN_NO_FLASH") == "1":
+        return False
     return (
         _backend.has_ext()
         and not _backend.force_eager()
@@ -183,8 +187,11 @@ def attention_qkv(qkv: torch.Tensor, n_heads: int, causal: bool = True) -> torch
     B, T, three_hl = qkv.shape
     hl = three_hl // 3
     D = hl // n_heads
+    import os
+
     if (
-        _backend.has_ext()
+        os.environ.get("QN_NO_FLASH") != "1"
+        and _backend.has_ext()
         and not _backend.force_eager()
         and qkv.is_cuda
         and qkv.dtype == torch.bfloat16

@@ -44,7 +44,7 @@ class CrossEntropyFunction(torch.autograd.Function):
         if _backend.use_native(logits) and _backend.has_ext():
             # grad scale fused into the kernel (no extra 800 MB pass)
             dl = _backend.ext().cross_entropy_bwd(
-                logits.contiguous(), target.contiguous(), lse, int(n_valid.item()),
+                logits.contiguous(), target.contiguous(), lse, n_valid,
                 ctx.ignore_index, grad_output,
             )
         else:

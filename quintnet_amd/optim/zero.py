@@ -107,6 +107,11 @@ class ZeroRedundancyAdamW:
         self.master = self.flat_param[s:e].detach().float().clone()
         self.exp_avg = torch.zeros_like(self.master)
         self.exp_avg_sq = torch.zeros_like(self.master)
+        # device-side step counter: advanced ON DEVICE inside step() so
+        # bias correction stays correct under hipGraph replay
+        self.step_dev = (
+            torch.zeros((), dtype=torch.int64, device=device) if device.type == "cuda" else None
+        )
 
     @classmethod
     def from_ddp(cls, ddp, **kw):
@@ -128,17 +133,20 @@ class ZeroRedundancyAdamW:
                 self.flat_grad[off : off + p.numel()].copy_(p.grad.reshape(-1))
 
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
-        """Two-kernel clip on the flat grad buffer (local norm)."""
+        """Clip on the flat grad buffer (local norm).  Fully device-side
+        (no host sync — hipGraph-capture safe): always multiplies by
+        min(max/norm, 1)."""
         self._gather_grads()
         norm = l2_norm([self.flat_grad])
-        scale = max_norm / (norm + 1e-6)
-        if float(scale) < 1.0:
-            self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
+        scale = (max_norm / (norm + 1e-6)).clamp_(max=1.0)
+        self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
         return norm
 
     @torch.no_grad()
     def step(self) -> None:
         self.step_count += 1
+        if self.step_dev is not None:
+            self.step_dev += 1  # device op: correct under graph replay
         self._gather_grads()
         if self.max_grad_norm:
             self.clip_grad_norm_(self.max_grad_norm)
@@ -156,6 +164,7 @@ class ZeroRedundancyAdamW:
             self.beta2,
             self.eps,
             self.weight_decay,
+            step_dev=self.step_dev,
         )
         if self.dp_size > 1:
             dist.all_gather_into_tensor(
@@ -200,6 +209,8 @@ class ZeroRedundancyAdamW:
 
     def load_state_dict(self, sd):
         self.step_count = sd["step"]
+        if self.step_dev is not None:
+            self.step_dev.fill_(sd["step"])
         self.master.copy_(sd["master"])
         self.exp_avg.copy_(sd["exp_avg"])
         self.exp_avg_sq.copy_(sd["exp_avg_sq"])

@@ -115,7 +115,7 @@ def test_cross_entropy_fwd_bwd(dt):
     tol = 2e-2 if dt == torch.bfloat16 else 1e-4
     assert abs(float(loss) - float(ref)) < tol, (float(loss), float(ref))
 
-    dl = _ext().cross_entropy_bwd(logits, tgt, lse, int(n_valid), -100, None)
+    dl = _ext().cross_entropy_bwd(logits, tgt, lse, n_valid, -100, None)
     lf = logits.float().requires_grad_(True)
     rr = torch.nn.functional.cross_entropy(lf, tgt, ignore_index=-100)
     (g,) = torch.autograd.grad(rr, lf)
@@ -134,7 +134,7 @@ def test_adamw_step_vs_torch():
     ref = master.clone().requires_grad_(True)
     opt = torch.optim.AdamW([ref], lr=1e-2, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.01)
     for step in range(1, 4):
-        _ext().adamw_step(param, master, grad, m, v, step, 1e-2, 0.9, 0.999, 1e-8, 0.01)
+        _ext().adamw_step(param, master, grad, m, v, step, 1e-2, 0.9, 0.999, 1e-8, 0.01, None)
         ref.grad = grad.float()
         opt.step()
     assert (master - ref.detach()).abs().max() < 1e-4
