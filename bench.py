@@ -245,32 +245,27 @@ def main():
         if (use_cuda and pg.world_size == 1 and want_graphs
                 and os.environ.get("QN_NO_GRAPHS") != "1"):
             try:
+                from quintnet_amd.utils.graphs import CapturedStep
+
                 static_batches = [
                     {k: v.clone() for k, v in batches[i % len(batches)].items()}
                     for i in range(grad_acc)
                 ]
                 loader.batches = static_batches
-                loader.i = 0
-                side = torch.cuda.Stream()
-                side.wait_stream(torch.cuda.current_stream())
-                with torch.cuda.stream(side):
-                    for _ in range(2):
-                        step()
-                torch.cuda.current_stream().wait_stream(side)
-                graph = torch.cuda.CUDAGraph()
-                loader.i = 0
-                with torch.cuda.graph(graph):
+
+                def reset_and_step():
+                    loader.i = 0
                     step()
 
+                captured = CapturedStep(reset_and_step, static_batches)
                 state = {"i": 0}
 
                 def graph_step():
-                    for j in range(grad_acc):
-                        src = batches[(state["i"] + j) % len(batches)]
-                        for k2, v2 in static_batches[j].items():
-                            v2.copy_(src[k2], non_blocking=True)
+                    feed = [
+                        batches[(state["i"] + j) % len(batches)] for j in range(grad_acc)
+                    ]
                     state["i"] += grad_acc
-                    graph.replay()
+                    captured(feed)
 
                 step = graph_step
                 if rank == 0:

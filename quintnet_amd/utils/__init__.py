@@ -1,6 +1,7 @@
 from .logger import setup_rank_logger, print_rank_0
 from .memory import memory_stats, print_memory_stats
 from .profiling import StepTimer, PhaseTimer
+from .graphs import CapturedStep
 from .metrics import (
     rouge_n,
     rouge_l,
@@ -26,6 +27,7 @@ __all__ = [
     "print_memory_stats",
     "StepTimer",
     "PhaseTimer",
+    "CapturedStep",
     "rouge_n",
     "rouge_l",
     "bleu",
