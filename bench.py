@@ -151,11 +151,11 @@ def main():
     if world != n and world > 1:
         n = world
     backend = "nccl" if use_cuda else "gloo"
+    if use_cuda:
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
     if n > 1 and not dist.is_initialized():
         dist.init_process_group(backend=backend,
                                 timeout=datetime.timedelta(seconds=600))
-    if use_cuda:
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", 0)))
 
     from quintnet_amd import init_process_groups
     from quintnet_amd.optim import ZeroRedundancyAdamW

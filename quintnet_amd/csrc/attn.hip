@@ -74,6 +74,29 @@ __device__ __forceinline__ bf16x8 frag_ld(const unsigned short* base,
   return *reinterpret_cast<const bf16x8*>(p);
 }
 
+// per-lane fragment base pointer: subsequent tiles advance it by
+// 32*row_stride and the 4 kt-fragments sit at +16 elem immediates — the
+// per-tile 64-bit address rebuild was a major VALU cost (cf. gemm.hip).
+__device__ __forceinline__ const unsigned short* frag_base(
+    const unsigned short* base, long long row_stride, int row0, int lane) {
+  return base + (long long)(row0 + (lane & 31)) * row_stride + ((lane >> 5) << 3);
+}
+
+__device__ __forceinline__ bf16x8 frag_at(const unsigned short* p, int kt) {
+  return *reinterpret_cast<const bf16x8*>(p + kt * 16);
+}
+
+__device__ __forceinline__ const unsigned short* stage_base(
+    const unsigned short* src, long long row_stride, int row0) {
+  int r = threadIdx.x & 31;
+  int d0 = (threadIdx.x >> 5) << 3;
+  return src + (long long)(row0 + r) * row_stride + d0;
+}
+
+__device__ __forceinline__ s16x8 stage_at(const unsigned short* p) {
+  return *reinterpret_cast<const s16x8*>(p);
+}
+
 // MFMA D-tile row of register r for this lane (32x32 layout)
 __device__ __forceinline__ int drow(int r, int lane) {
   return (r & 3) + ((r >> 2) << 3) + ((lane >> 5) << 2);
@@ -151,12 +174,15 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   // prefetch pipeline: tile t's K fragments + V staging rows load during
   // tile t-1's MFMA cluster (T14) — the per-iteration global latency was
-  // the dominant cost of the unpipelined version.
+  // the dominant cost of the unpipelined version.  Source addresses are
+  // pointer-bumped (+= 32 rows per tile), never rebuilt.
+  const unsigned short* kfp = frag_base(kp, ksT, 0, lane);
+  const unsigned short* vsp = stage_base(vp, vsT, 0);
+  const long long kstep = 32 * ksT, vstep = 32 * vsT;
   bf16x8 kf_n[4];
-  s16x8 v_n;
-  v_n = stage_ld(vp, vsT, 0);
+  s16x8 v_n = stage_at(vsp);
 #pragma unroll
-  for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_ld(kp, ksT, 0, kt * 16, lane);
+  for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     // cooperative V^T staging (write the prefetched rows)
@@ -168,10 +194,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
     for (int kt = 0; kt < 4; ++kt) kf_c[kt] = kf_n[kt];
     if (kv0 + 32 < kv_end) {
-      v_n = stage_ld(vp, vsT, kv0 + 32);
+      vsp += vstep;
+      kfp += kstep;
+      v_n = stage_at(vsp);
 #pragma unroll
-      for (int kt = 0; kt < 4; ++kt)
-        kf_n[kt] = frag_ld(kp, ksT, kv0 + 32, kt * 16, lane);
+      for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
     }
 
     if (!causal || kv0 <= qw + 31) {  // wave has at least one valid pair
@@ -321,13 +348,17 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
 
   const int kv_end = causal ? min(q0 + 128, T) : T;
-  // prefetch pipeline (T14): next tile's K/V fragments + Kt staging rows
+  // prefetch pipeline (T14), pointer-bumped sources
+  const unsigned short* kfp = frag_base(kp, ksT, 0, lane);
+  const unsigned short* vfp = frag_base(vp, vsT, 0, lane);
+  const unsigned short* ksp = stage_base(kp, ksT, 0);
+  const long long kstep = 32 * ksT, vstep = 32 * vsT;
   bf16x8 kf_n[4], vf_n[4];
-  s16x8 kst_n = stage_ld(kp, ksT, 0);
+  s16x8 kst_n = stage_at(ksp);
 #pragma unroll
   for (int t = 0; t < 4; ++t) {
-    kf_n[t] = frag_ld(kp, ksT, 0, t * 16, lane);
-    vf_n[t] = frag_ld(vp, vsT, 0, t * 16, lane);
+    kf_n[t] = frag_at(kfp, t);
+    vf_n[t] = frag_at(vfp, t);
   }
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     __syncthreads();
@@ -338,11 +369,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
 #pragma unroll
     for (int t = 0; t < 4; ++t) { kf_c[t] = kf_n[t]; vf_c[t] = vf_n[t]; }
     if (kv0 + 32 < kv_end) {
-      kst_n = stage_ld(kp, ksT, kv0 + 32);
+      kfp += kstep;
+      vfp += vstep;
+      ksp += kstep;
+      kst_n = stage_at(ksp);
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        kf_n[t] = frag_ld(kp, ksT, kv0 + 32, t * 16, lane);
-        vf_n[t] = frag_ld(vp, vsT, kv0 + 32, t * 16, lane);
+        kf_n[t] = frag_at(kfp, t);
+        vf_n[t] = frag_at(vfp, t);
       }
     }
     if (causal && kv0 > qw + 31) continue;
@@ -444,20 +478,26 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; zc[i] = 0.f; }
 
   const int q_start = causal ? (kv0b / 32) * 32 : 0;
-  // prefetch pipeline (T14): next q-tile's Q/dO fragments, staging rows,
-  // lse/delta — the unpipelined version was per-iteration latency-bound.
+  // prefetch pipeline (T14), pointer-bumped sources
+  const unsigned short* qfp = frag_base(qp, qsT, q_start, lane);
+  const unsigned short* dofp = frag_base(dop, dsT, q_start, lane);
+  const unsigned short* qsp = stage_base(qp, qsT, q_start);
+  const unsigned short* dosp = stage_base(dop, dsT, q_start);
+  const float* lsep = lse2 + (long long)bh * T + q_start + threadIdx.x;
+  const float* delp = delta + (long long)bh * T + q_start + threadIdx.x;
+  const long long qstep = 32 * qsT, dstep = 32 * dsT;
   bf16x8 qf_n[4], dof_n[4];
-  s16x8 dost_n = stage_ld(dop, dsT, q_start);
-  s16x8 qst_n = stage_ld(qp, qsT, q_start);
+  s16x8 dost_n = stage_at(dosp);
+  s16x8 qst_n = stage_at(qsp);
   float lse_n = 0.f, del_n = 0.f;
   if (threadIdx.x < 32) {
-    lse_n = lse2[(long long)bh * T + q_start + threadIdx.x];
-    del_n = delta[(long long)bh * T + q_start + threadIdx.x];
+    lse_n = *lsep;
+    del_n = *delp;
   }
 #pragma unroll
   for (int t = 0; t < 4; ++t) {
-    qf_n[t] = frag_ld(qp, qsT, q_start, t * 16, lane);
-    dof_n[t] = frag_ld(dop, dsT, q_start, t * 16, lane);
+    qf_n[t] = frag_at(qfp, t);
+    dof_n[t] = frag_at(dofp, t);
   }
   for (int qt0 = q_start; qt0 < T; qt0 += 32) {
     __syncthreads();
@@ -473,16 +513,22 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
     for (int t = 0; t < 4; ++t) { qf_c[t] = qf_n[t]; dof_c[t] = dof_n[t]; }
     if (qt0 + 32 < T) {
-      dost_n = stage_ld(dop, dsT, qt0 + 32);
-      qst_n = stage_ld(qp, qsT, qt0 + 32);
+      qfp += qstep;
+      dofp += dstep;
+      qsp += qstep;
+      dosp += dstep;
+      lsep += 32;
+      delp += 32;
+      dost_n = stage_at(dosp);
+      qst_n = stage_at(qsp);
       if (threadIdx.x < 32) {
-        lse_n = lse2[(long long)bh * T + qt0 + 32 + threadIdx.x];
-        del_n = delta[(long long)bh * T + qt0 + 32 + threadIdx.x];
+        lse_n = *lsep;
+        del_n = *delp;
       }
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
-        qf_n[t] = frag_ld(qp, qsT, qt0 + 32, t * 16, lane);
-        dof_n[t] = frag_ld(dop, dsT, qt0 + 32, t * 16, lane);
+        qf_n[t] = frag_at(qfp, t);
+        dof_n[t] = frag_at(dofp, t);
       }
     }
     if (causal && qt0 + 31 < kw) continue;  // whole tile above diagonal

@@ -127,3 +127,12 @@ def test_1f1b_matches_single_process():
 
 def test_afab_matches_single_process():
     run_distributed(_run_afab, 2)
+
+
+def _run_1f1b_pp4(rank, world):
+    _run_schedule(rank, world, "1f1b")
+
+
+def test_1f1b_pp4_matches_single_process():
+    """The BASELINE 'ViT PP=4 1F1B' config shape, on gloo."""
+    run_distributed(_run_1f1b_pp4, 4, timeout=300)
