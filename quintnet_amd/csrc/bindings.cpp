@@ -7,9 +7,9 @@
 
 // launcher decls (defined in the .hip TUs)
 template <typename T>
-void layernorm_fwd_launch(const T*, const T*, const T*, T*, float*, float*, int, int, float, hipStream_t);
+void layernorm_fwd_launch(const T*, const T*, const T*, const T*, T*, T*, float*, float*, int, int, float, hipStream_t);
 template <typename T>
-void layernorm_bwd_launch(const T*, const T*, const T*, const float*, const float*, T*, float*, float*, int, int, hipStream_t);
+void layernorm_bwd_launch(const T*, const T*, const T*, const float*, const float*, const T*, T*, float*, float*, int, int, hipStream_t);
 template <typename T>
 void softmax_fwd_launch(const T*, T*, long long, int, int, float, int, hipStream_t);
 template <typename T>
@@ -100,7 +100,8 @@ std::vector<torch::Tensor> gemm_nt(torch::Tensor x, torch::Tensor w,
 
 // ---------------------------------------------------------------------------
 std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
-                                         torch::Tensor b, double eps) {
+                                         torch::Tensor b, double eps,
+                                         c10::optional<torch::Tensor> residual) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
   int64_t H = xc.size(-1);
@@ -109,23 +110,33 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, torch::Tensor w,
   auto mean = torch::empty({rows}, xc.options().dtype(torch::kFloat32));
   auto rstd = torch::empty({rows}, xc.options().dtype(torch::kFloat32));
   auto wc = w.contiguous(); auto bc = b.contiguous();
+  torch::Tensor rc, sum_out;
+  bool has_res = residual.has_value() && residual->defined();
+  if (has_res) {
+    rc = residual->contiguous();
+    sum_out = torch::empty_like(xc);
+  }
   if (xc.dtype() == torch::kBFloat16) {
     layernorm_fwd_launch<unsigned short>(
-        bf16p(xc), bf16p(wc), bf16p(bc), bf16p_mut(y),
+        bf16p(xc), has_res ? bf16p(rc) : nullptr, bf16p(wc), bf16p(bc),
+        bf16p_mut(y), has_res ? bf16p_mut(sum_out) : nullptr,
         mean.data_ptr<float>(), rstd.data_ptr<float>(), (int)rows, (int)H,
         (float)eps, cur_stream());
   } else {
     layernorm_fwd_launch<float>(
-        xc.data_ptr<float>(), wc.data_ptr<float>(), bc.data_ptr<float>(),
-        y.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-        (int)rows, (int)H, (float)eps, cur_stream());
+        xc.data_ptr<float>(), has_res ? rc.data_ptr<float>() : nullptr,
+        wc.data_ptr<float>(), bc.data_ptr<float>(), y.data_ptr<float>(),
+        has_res ? sum_out.data_ptr<float>() : nullptr, mean.data_ptr<float>(),
+        rstd.data_ptr<float>(), (int)rows, (int)H, (float)eps, cur_stream());
   }
+  if (has_res) return {y, mean, rstd, sum_out};
   return {y, mean, rstd};
 }
 
 std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
                                          torch::Tensor w, torch::Tensor mean,
-                                         torch::Tensor rstd) {
+                                         torch::Tensor rstd,
+                                         c10::optional<torch::Tensor> dsum) {
   CHECK_GPU(x);
   auto dyc = dy.contiguous(); auto xc = x.contiguous(); auto wc = w.contiguous();
   int64_t H = xc.size(-1);
@@ -133,15 +144,20 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor x,
   auto dx = torch::empty_like(xc);
   auto dwf = torch::zeros({H}, xc.options().dtype(torch::kFloat32));
   auto dbf = torch::zeros({H}, xc.options().dtype(torch::kFloat32));
+  torch::Tensor dsc;
+  bool has_ds = dsum.has_value() && dsum->defined();
+  if (has_ds) dsc = dsum->contiguous();
   if (xc.dtype() == torch::kBFloat16) {
     layernorm_bwd_launch<unsigned short>(
         bf16p(dyc), bf16p(xc), bf16p(wc), mean.data_ptr<float>(),
-        rstd.data_ptr<float>(), bf16p_mut(dx), dwf.data_ptr<float>(),
-        dbf.data_ptr<float>(), (int)rows, (int)H, cur_stream());
+        rstd.data_ptr<float>(), has_ds ? bf16p(dsc) : nullptr, bf16p_mut(dx),
+        dwf.data_ptr<float>(), dbf.data_ptr<float>(), (int)rows, (int)H,
+        cur_stream());
   } else {
     layernorm_bwd_launch<float>(
         dyc.data_ptr<float>(), xc.data_ptr<float>(), wc.data_ptr<float>(),
-        mean.data_ptr<float>(), rstd.data_ptr<float>(), dx.data_ptr<float>(),
+        mean.data_ptr<float>(), rstd.data_ptr<float>(),
+        has_ds ? dsc.data_ptr<float>() : nullptr, dx.data_ptr<float>(),
         dwf.data_ptr<float>(), dbf.data_ptr<float>(), (int)rows, (int)H,
         cur_stream());
   }

@@ -14,15 +14,19 @@
 // ---------------------------------------------------------------------------
 template <typename T, int WPB>
 __global__ void layernorm_fwd_kernel(
-    const T* __restrict__ x, const T* __restrict__ w, const T* __restrict__ b,
-    T* __restrict__ y, float* __restrict__ mean_out, float* __restrict__ rstd_out,
+    const T* __restrict__ x, const T* __restrict__ res,
+    const T* __restrict__ w, const T* __restrict__ b,
+    T* __restrict__ y, T* __restrict__ sum_out,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out,
     int rows, int H, float eps) {
   const int lane = threadIdx.x & (QN_WAVE - 1);
   const int wave = threadIdx.x / QN_WAVE;
   const long long row = (long long)blockIdx.x * WPB + wave;
   if (row >= rows) return;
   const T* xr = x + row * H;
+  const T* rr = res ? res + row * H : nullptr;
   T* yr = y + row * H;
+  T* sr = sum_out ? sum_out + row * H : nullptr;
 
   float sum = 0.f, sq = 0.f;
   // cache up to 16 elems/lane in registers (H <= 1024) to avoid re-reads
@@ -32,12 +36,20 @@ __global__ void layernorm_fwd_kernel(
     int n = 0;
     for (int i = lane; i < H; i += QN_WAVE, ++n) {
       float f = ld_as_f32(xr + i);
+      if (rr) {  // fused residual add (SURVEY §2.4 "fused residual-add")
+        f += ld_as_f32(rr + i);
+        if (sr) st_from_f32(sr + i, f);
+      }
       cache[n] = f;
       sum += f; sq += f * f;
     }
   } else {
     for (int i = lane; i < H; i += QN_WAVE) {
       float f = ld_as_f32(xr + i);
+      if (rr) {
+        f += ld_as_f32(rr + i);
+        if (sr) st_from_f32(sr + i, f);
+      }
       sum += f; sq += f * f;
     }
   }
@@ -58,6 +70,7 @@ __global__ void layernorm_fwd_kernel(
   } else {
     for (int i = lane; i < H; i += QN_WAVE) {
       float f = ld_as_f32(xr + i);
+      if (rr) f += ld_as_f32(rr + i);
       float wi = ld_as_f32(w + i);
       float bi = ld_as_f32(b + i);
       st_from_f32(yr + i, (f - mean) * rstd * wi + bi);
@@ -74,13 +87,14 @@ template <typename T, int WPB>
 __global__ void layernorm_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, const T* __restrict__ w,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
-    T* __restrict__ dx, int rows, int H) {
+    const T* __restrict__ dsum, T* __restrict__ dx, int rows, int H) {
   const int lane = threadIdx.x & (QN_WAVE - 1);
   const int wave = threadIdx.x / QN_WAVE;
   const long long row = (long long)blockIdx.x * WPB + wave;
   if (row >= rows) return;
   const T* xr = x + row * H;
   const T* dyr = dy + row * H;
+  const T* dsr = dsum ? dsum + row * H : nullptr;
   T* dxr = dx + row * H;
   const float mean = mean_in[row], rstd = rstd_in[row];
 
@@ -107,13 +121,18 @@ __global__ void layernorm_bwd_dx_kernel(
 
   if (cached) {
     int n = 0;
-    for (int i = lane; i < H; i += QN_WAVE, ++n)
-      st_from_f32(dxr + i, (cwdy[n] - c1 - cx[n] * c2) * rstd);
+    for (int i = lane; i < H; i += QN_WAVE, ++n) {
+      float v = (cwdy[n] - c1 - cx[n] * c2) * rstd;
+      if (dsr) v += ld_as_f32(dsr + i);  // residual-branch grad fused in
+      st_from_f32(dxr + i, v);
+    }
   } else {
     for (int i = lane; i < H; i += QN_WAVE) {
       float xhat = (ld_as_f32(xr + i) - mean) * rstd;
       float wdy = ld_as_f32(w + i) * ld_as_f32(dyr + i);
-      st_from_f32(dxr + i, (wdy - c1 - xhat * c2) * rstd);
+      float v = (wdy - c1 - xhat * c2) * rstd;
+      if (dsr) v += ld_as_f32(dsr + i);
+      st_from_f32(dxr + i, v);
     }
   }
 }
@@ -145,23 +164,23 @@ __global__ void layernorm_bwd_dwdb_kernel(
 
 // ---- launchers -------------------------------------------------------------
 template <typename T>
-void layernorm_fwd_launch(const T* x, const T* w, const T* b, T* y,
-                          float* mean, float* rstd, int rows, int H, float eps,
-                          hipStream_t stream) {
+void layernorm_fwd_launch(const T* x, const T* res, const T* w, const T* b,
+                          T* y, T* sum_out, float* mean, float* rstd, int rows,
+                          int H, float eps, hipStream_t stream) {
   constexpr int WPB = 4;
   dim3 grid((rows + WPB - 1) / WPB);
   hipLaunchKernelGGL((layernorm_fwd_kernel<T, WPB>), grid, dim3(WPB * QN_WAVE), 0,
-                     stream, x, w, b, y, mean, rstd, rows, H, eps);
+                     stream, x, res, w, b, y, sum_out, mean, rstd, rows, H, eps);
 }
 
 template <typename T>
 void layernorm_bwd_launch(const T* dy, const T* x, const T* w, const float* mean,
-                          const float* rstd, T* dx, float* dw, float* db,
-                          int rows, int H, hipStream_t stream) {
+                          const float* rstd, const T* dsum, T* dx, float* dw,
+                          float* db, int rows, int H, hipStream_t stream) {
   constexpr int WPB = 4;
   dim3 grid((rows + WPB - 1) / WPB);
   hipLaunchKernelGGL((layernorm_bwd_dx_kernel<T, WPB>), grid, dim3(WPB * QN_WAVE),
-                     0, stream, dy, x, w, mean, rstd, dx, rows, H);
+                     0, stream, dy, x, w, mean, rstd, dsum, dx, rows, H);
   int colblocks = (H + 255) / 256;
   int chunks = min(max(rows / 64, 1), 256);
   hipLaunchKernelGGL((layernorm_bwd_dwdb_kernel<T>), dim3(colblocks, chunks),
@@ -169,13 +188,14 @@ void layernorm_bwd_launch(const T* dy, const T* x, const T* w, const float* mean
 }
 
 // explicit instantiations
-template void layernorm_fwd_launch<float>(const float*, const float*, const float*,
-                                          float*, float*, float*, int, int, float, hipStream_t);
+template void layernorm_fwd_launch<float>(const float*, const float*, const float*, const float*,
+                                          float*, float*, float*, float*, int, int, float, hipStream_t);
 template void layernorm_fwd_launch<unsigned short>(const unsigned short*, const unsigned short*,
-                                                   const unsigned short*, unsigned short*, float*,
+                                                   const unsigned short*, const unsigned short*,
+                                                   unsigned short*, unsigned short*, float*,
                                                    float*, int, int, float, hipStream_t);
 template void layernorm_bwd_launch<float>(const float*, const float*, const float*, const float*,
-                                          const float*, float*, float*, float*, int, int, hipStream_t);
+                                          const float*, const float*, float*, float*, float*, int, int, hipStream_t);
 template void layernorm_bwd_launch<unsigned short>(const unsigned short*, const unsigned short*,
                                                    const unsigned short*, const float*, const float*,
-                                                   unsigned short*, float*, float*, int, int, hipStream_t);
+                                                   const unsigned short*, unsigned short*, float*, float*, int, int, hipStream_t);

@@ -26,6 +26,20 @@ class GPT2Block(nn.Module):
         self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = x + self.attn(self.ln_1(x))
-        x = x + self.mlp(self.ln_2(x))
-        return x
+        m, s2 = self.forward_fused(x, None)
+        return s2 + m
+
+    def forward_fused(self, x: torch.Tensor, pending):
+        """Residual-fused form: the incoming pending residual (previous
+        block's MLP output) is added INSIDE ln_1's kernel; this block's
+        MLP output is returned as the next pending residual.
+
+        Returns (mlp_out, post_attention_sum)."""
+        if pending is None:
+            n1, s1 = self.ln_1(x), x
+        else:
+            n1, s1 = self.ln_1(x, residual=pending)
+        a = self.attn(n1)
+        n2, s2 = self.ln_2(s1, residual=a)
+        m = self.mlp(n2)
+        return m, s2

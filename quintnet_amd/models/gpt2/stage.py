@@ -73,12 +73,20 @@ class GPT2Stage(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.is_first_stage:
             x = self.embedding(x)
+        # residual-fused block chain: each block's trailing residual add
+        # rides the NEXT LayerNorm kernel (ops/layernorm.py)
+        pending = None
         for blk in self.blocks:
-            x = blk(x)
+            pending, x = blk.forward_fused(x, pending)
         if self.is_last_stage:
-            x = self.ln_f(x)
+            if pending is None:
+                x = self.ln_f(x)
+            else:
+                x, _ = self.ln_f(x, residual=pending)
             w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
             x = fused_linear(x, w, None, None, prefer_library=True)  # plain GEMM: hipBLASLt
+        elif pending is not None:
+            x = x + pending  # fold before the PP send
         return x
 
     # ------------------------------------------------------------------

@@ -8,7 +8,7 @@ kernels against.
 
 from ._backend import ext, has_ext, use_native, force_eager
 from .linear import linear, LinearFunction
-from .layernorm import layer_norm, FusedLayerNorm, LayerNormFunction
+from .layernorm import layer_norm, layer_norm_residual, FusedLayerNorm, LayerNormFunction
 from .attention import (
     attention,
     attention_qkv,
@@ -29,6 +29,7 @@ __all__ = [
     "linear",
     "LinearFunction",
     "layer_norm",
+    "layer_norm_residual",
     "FusedLayerNorm",
     "LayerNormFunction",
     "attention",

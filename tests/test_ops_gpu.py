@@ -58,13 +58,13 @@ def test_layernorm_fwd_bwd(rows, h, dt):
     x = torch.randn(rows, h, device=DEV, dtype=dt)
     w = torch.randn(h, device=DEV, dtype=dt)
     b = torch.randn(h, device=DEV, dtype=dt)
-    y, mean, rstd = _ext().layernorm_fwd(x, w, b, 1e-5)
+    y, mean, rstd = _ext().layernorm_fwd(x, w, b, 1e-5, None)
     ref = torch.nn.functional.layer_norm(x.float(), (h,), w.float(), b.float(), 1e-5)
     tol = 2e-2 if dt == torch.bfloat16 else 1e-5
     assert (y.float() - ref).abs().max() < tol * max(float(ref.abs().max()), 1.0)
 
     dy = torch.randn(rows, h, device=DEV, dtype=dt)
-    dx, dw, db = _ext().layernorm_bwd(dy, x, w, mean, rstd)
+    dx, dw, db = _ext().layernorm_bwd(dy, x, w, mean, rstd, None)
     xf = x.float().requires_grad_(True)
     wf = w.float().requires_grad_(True)
     bf = b.float().requires_grad_(True)
@@ -267,3 +267,29 @@ def test_fused_dropout():
     assert torch.allclose((g != 0).float().mean(), kept, atol=1e-3)
     # eval mode / p=0: identity
     assert fused_dropout(x, 0.3, training=False) is x
+
+
+def test_layernorm_residual_fused():
+    from quintnet_amd.ops import layer_norm_residual
+
+    torch.manual_seed(11)
+    x = torch.randn(64, 768, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    r = torch.randn(64, 768, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(768, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    b = torch.randn(768, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    y, s = layer_norm_residual(x, r, w, b, 1e-5)
+    xf = x.detach().float().requires_grad_(True)
+    rf = r.detach().float().requires_grad_(True)
+    wf = w.detach().float().requires_grad_(True)
+    bf = b.detach().float().requires_grad_(True)
+    sf = xf + rf
+    yf = torch.nn.functional.layer_norm(sf, (768,), wf, bf, 1e-5)
+    assert (s.float() - sf).abs().max() < 2e-2
+    assert (y.float() - yf).abs().max() < 3e-2
+    dy = torch.randn_like(y)
+    ds = torch.randn_like(s)
+    (y * dy.detach() + s * ds.detach()).sum().backward()
+    (yf * dy.detach().float() + sf * ds.detach().float()).sum().backward()
+    for g, rg in ((x.grad, xf.grad), (r.grad, rf.grad)):
+        assert (g.float() - rg).abs().max() / max(float(rg.abs().max()), 1.0) < 4e-2
+    assert (w.grad.float() - wf.grad).abs().max() / max(float(wf.grad.abs().max()), 1.0) < 4e-2
