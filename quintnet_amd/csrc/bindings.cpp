@@ -26,6 +26,10 @@ template <typename T>
 void dropout_fwd_launch(const T*, T*, unsigned char*, long long, float, unsigned, hipStream_t);
 template <typename T>
 void dropout_bwd_launch(const T*, T*, const unsigned char*, long long, float, hipStream_t);
+template <typename T>
+void embedding_pair_fwd_launch(const long long*, const T*, const T*, T*, long long, int, int, hipStream_t);
+template <typename T>
+void embedding_pair_bwd_launch(const long long*, const T*, float*, float*, long long, int, int, hipStream_t);
 template <typename TP, typename TG>
 void adamw_launch(TP*, float*, const TG*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
@@ -259,6 +263,50 @@ torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor target,
   return dl;
 }
 
+torch::Tensor embedding_pair_fwd(torch::Tensor ids, torch::Tensor wte,
+                                 torch::Tensor wpe) {
+  CHECK_GPU(ids);
+  auto ic = ids.contiguous();
+  TORCH_CHECK(ic.dtype() == torch::kInt64, "ids must be int64");
+  int64_t Tlen = ic.size(-1);
+  long long rows = ic.numel();
+  int64_t H = wte.size(1);
+  auto out = torch::empty({ic.size(0), Tlen, H}, wte.options());
+  if (wte.dtype() == torch::kBFloat16)
+    embedding_pair_fwd_launch<unsigned short>(
+        (const long long*)ic.data_ptr<int64_t>(), bf16p(wte), bf16p(wpe),
+        bf16p_mut(out), rows, (int)Tlen, (int)H, cur_stream());
+  else
+    embedding_pair_fwd_launch<float>(
+        (const long long*)ic.data_ptr<int64_t>(), wte.data_ptr<float>(),
+        wpe.data_ptr<float>(), out.data_ptr<float>(), rows, (int)Tlen, (int)H,
+        cur_stream());
+  return out;
+}
+
+std::vector<torch::Tensor> embedding_pair_bwd(torch::Tensor ids, torch::Tensor dout,
+                                              int64_t vocab, int64_t n_pos) {
+  CHECK_GPU(ids);
+  auto ic = ids.contiguous();
+  auto dc = dout.contiguous();
+  int64_t Tlen = ic.size(-1);
+  long long rows = ic.numel();
+  int64_t H = dc.size(-1);
+  auto dwte = torch::zeros({vocab, H}, dc.options().dtype(torch::kFloat32));
+  auto dwpe = torch::zeros({n_pos, H}, dc.options().dtype(torch::kFloat32));
+  if (dc.dtype() == torch::kBFloat16)
+    embedding_pair_bwd_launch<unsigned short>(
+        (const long long*)ic.data_ptr<int64_t>(), bf16p(dc),
+        dwte.data_ptr<float>(), dwpe.data_ptr<float>(), rows, (int)Tlen, (int)H,
+        cur_stream());
+  else
+    embedding_pair_bwd_launch<float>(
+        (const long long*)ic.data_ptr<int64_t>(), dc.data_ptr<float>(),
+        dwte.data_ptr<float>(), dwpe.data_ptr<float>(), rows, (int)Tlen, (int)H,
+        cur_stream());
+  return {dwte.to(dout.dtype()), dwpe.to(dout.dtype())};
+}
+
 std::vector<torch::Tensor> dropout_fwd(torch::Tensor x, double p, int64_t seed) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
@@ -435,6 +483,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("act_bwd", &act_bwd, "fused activation backward");
   m.def("colsum", &colsum, "column sum (bias grad)");
   m.def("dropout_fwd", &dropout_fwd);
+  m.def("embedding_pair_fwd", &embedding_pair_fwd);
+  m.def("embedding_pair_bwd", &embedding_pair_bwd);
   m.def("dropout_bwd", &dropout_bwd);
   m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)");
   m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)");

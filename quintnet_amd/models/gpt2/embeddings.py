@@ -9,7 +9,7 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ...ops import FusedDropout
+from ...ops import FusedDropout, embedding_pair
 from .config import GPT2Config
 
 __all__ = ["GPT2Embedding"]
@@ -27,7 +27,5 @@ class GPT2Embedding(nn.Module):
         nn.init.normal_(self.wpe.weight, std=config.initializer_range)
 
     def forward(self, input_ids: torch.Tensor) -> torch.Tensor:
-        T = input_ids.shape[1]
-        pos = torch.arange(T, device=input_ids.device).unsqueeze(0)
-        x = self.wte(input_ids) + self.wpe(pos)
+        x = embedding_pair(input_ids, self.wte.weight, self.wpe.weight)
         return self.drop(x)

@@ -20,6 +20,7 @@ from .attention import (
 from .cross_entropy import cross_entropy, causal_lm_loss, shift_labels, CrossEntropyFunction
 from .adamw import adamw_step_flat, clip_grad_norm_local, l2_norm
 from .dropout import fused_dropout, FusedDropout
+from .embedding import embedding_pair
 
 __all__ = [
     "ext",
@@ -47,4 +48,5 @@ __all__ = [
     "l2_norm",
     "fused_dropout",
     "FusedDropout",
+    "embedding_pair",
 ]

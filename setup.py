@@ -25,6 +25,7 @@ sources = [
     os.path.join(CSRC, "adamw.hip"),
     os.path.join(CSRC, "elementwise.hip"),
     os.path.join(CSRC, "attn.hip"),
+    os.path.join(CSRC, "embedding.hip"),
     os.path.join(CSRC, "gemm.hip"),
 ]
 

@@ -293,3 +293,25 @@ def test_layernorm_residual_fused():
     for g, rg in ((x.grad, xf.grad), (r.grad, rf.grad)):
         assert (g.float() - rg).abs().max() / max(float(rg.abs().max()), 1.0) < 4e-2
     assert (w.grad.float() - wf.grad).abs().max() / max(float(wf.grad.abs().max()), 1.0) < 4e-2
+
+
+def test_embedding_pair_fused():
+    from quintnet_amd.ops import embedding_pair
+
+    torch.manual_seed(12)
+    V, P, H = 512, 64, 256
+    wte = torch.randn(V, H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    wpe = torch.randn(P, H, device=DEV, dtype=torch.bfloat16, requires_grad=True)
+    ids = torch.randint(0, V, (4, 32), device=DEV)
+    out = embedding_pair(ids, wte, wpe)
+    pos = torch.arange(32, device=DEV)
+    ref = torch.nn.functional.embedding(ids, wte.float()) + torch.nn.functional.embedding(pos, wpe.float())
+    assert (out.float() - ref).abs().max() < 2e-2
+    dout = torch.randn_like(out)
+    out.backward(dout)
+    wtef = wte.detach().float().requires_grad_(True)
+    wpef = wpe.detach().float().requires_grad_(True)
+    (torch.nn.functional.embedding(ids, wtef) +
+     torch.nn.functional.embedding(pos, wpef)).backward(dout.float())
+    assert (wte.grad.float() - wtef.grad).abs().max() / max(float(wtef.grad.abs().max()), 1.0) < 3e-2
+    assert (wpe.grad.float() - wpef.grad).abs().max() / max(float(wpef.grad.abs().max()), 1.0) < 3e-2
