@@ -446,10 +446,14 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     long long dsB, long long dsH, long long dsT,
     long long dksB, long long dksH, long long dksT,
     long long dvsB, long long dvsH, long long dvsT) {
-  __shared__ unsigned short dot_lds[64 * TPAD];
-  __shared__ unsigned short qt_lds[64 * TPAD];
-  __shared__ float lse_t[32];
-  __shared__ float del_t[32];
+  // double-buffered transpose tiles + per-tile lse/delta rows: the loop
+  // is software-pipelined across q-tiles — S/dP MFMAs of tile i+1
+  // interleave with the dV/dK MFMAs of tile i (all 6 accumulator chains
+  // independent), ONE barrier per tile.
+  __shared__ unsigned short dot_lds[2][64 * TPAD];
+  __shared__ unsigned short qt_lds[2][64 * TPAD];
+  __shared__ float lse_t[2][32];
+  __shared__ float del_t[2][32];
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & 63;
@@ -477,8 +481,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
   for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; zc[i] = 0.f; }
 
-  const int q_start = causal ? (kv0b / 32) * 32 : 0;
-  // prefetch pipeline (T14), pointer-bumped sources
+  const int q_start = causal ? kv0b : 0;
+  // pointer-bumped sources
   const unsigned short* qfp = frag_base(qp, qsT, q_start, lane);
   const unsigned short* dofp = frag_base(dop, dsT, q_start, lane);
   const unsigned short* qsp = stage_base(qp, qsT, q_start);
@@ -486,44 +490,96 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const float* lsep = lse2 + (long long)bh * T + q_start + threadIdx.x;
   const float* delp = delta + (long long)bh * T + q_start + threadIdx.x;
   const long long qstep = 32 * qsT, dstep = 32 * dsT;
-  bf16x8 qf_n[4], dof_n[4];
-  s16x8 dost_n = stage_at(dosp);
-  s16x8 qst_n = stage_at(qsp);
-  float lse_n = 0.f, del_n = 0.f;
-  if (threadIdx.x < 32) {
-    lse_n = *lsep;
-    del_n = *delp;
+
+  // ---- prologue: stage tile 0 into buf0, prime s/dp for tile 0 ------------
+  {
+    s16x8 d0 = stage_at(dosp);
+    s16x8 q0 = stage_at(qsp);
+    stage_wr(dot_lds[0], d0);
+    stage_wr(qt_lds[0], q0);
+    if (threadIdx.x < 32) {
+      lse_t[0][threadIdx.x] = *lsep;
+      del_t[0][threadIdx.x] = *delp;
+    }
   }
+  bf16x8 qf_n[4], dof_n[4];
 #pragma unroll
   for (int t = 0; t < 4; ++t) {
     qf_n[t] = frag_at(qfp, t);
     dof_n[t] = frag_at(dofp, t);
   }
-  for (int qt0 = q_start; qt0 < T; qt0 += 32) {
-    __syncthreads();
-    stage_wr(dot_lds, dost_n);
-    stage_wr(qt_lds, qst_n);
-    if (threadIdx.x < 32) {
-      lse_t[threadIdx.x] = lse_n;
-      del_t[threadIdx.x] = del_n;
-    }
-    __syncthreads();
-
-    bf16x8 qf_c[4], dof_c[4];
+  f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[0], kf[0], zc, 0, 0, 0);
+  f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[0], vf[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int t = 0; t < 4; ++t) { qf_c[t] = qf_n[t]; dof_c[t] = dof_n[t]; }
-    if (qt0 + 32 < T) {
-      qfp += qstep;
-      dofp += dstep;
-      qsp += qstep;
-      dosp += dstep;
-      lsep += 32;
-      delp += 32;
-      dost_n = stage_at(dosp);
-      qst_n = stage_at(qsp);
+  for (int t = 1; t < 4; ++t) {
+    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[t], kf[t], s, 0, 0, 0);
+    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[t], vf[t], dp_, 0, 0, 0);
+  }
+  // prefetch tile 1's staging rows + lse/del
+  s16x8 dost_n = {0, 0, 0, 0, 0, 0, 0, 0}, qst_n = dost_n;
+  float lse_n = 0.f, del_n = 0.f;
+  if (q_start + 32 < T) {
+    dost_n = stage_at(dosp + dstep);
+    qst_n = stage_at(qsp + qstep);
+    if (threadIdx.x < 32) {
+      lse_n = lsep[32];
+      del_n = delp[32];
+    }
+  }
+  __syncthreads();  // buf0 visible
+
+  int cur = 0;
+  for (int qt0 = q_start; qt0 < T; qt0 += 32) {
+    const bool have_next = qt0 + 32 < T;
+    const bool active = !(causal && qt0 + 31 < kw);
+
+    // softmax + relayout for tile i (s/dp computed last iteration)
+    bf16x8 pf0, pf1, gf0, gf1;
+    if (active) {
+      const bool diag = causal && (qt0 < kw + 31);
+      float pv[16], gv[16];
+      if (diag) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qrow = drow(r, lane);
+          float p = (mykey > qt0 + qrow)
+                        ? 0.f
+                        : __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[cur][qrow]);
+          pv[r] = p;
+          gv[r] = scale * p * (dp_[r] - del_t[cur][qrow]);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int qrow = drow(r, lane);
+          float p = __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[cur][qrow]);
+          pv[r] = p;
+          gv[r] = scale * p * (dp_[r] - del_t[cur][qrow]);
+        }
+      }
+      pf0 = relayout8(pv); pf1 = relayout8(pv + 8);
+      gf0 = relayout8(gv); gf1 = relayout8(gv + 8);
+    }
+
+    // stage tile i+1 into the other buffer (its last readers finished
+    // before the barrier at the end of the previous iteration)
+    if (have_next) {
+      stage_wr(dot_lds[cur ^ 1], dost_n);
+      stage_wr(qt_lds[cur ^ 1], qst_n);
       if (threadIdx.x < 32) {
-        lse_n = *lsep;
-        del_n = *delp;
+        lse_t[cur ^ 1][threadIdx.x] = lse_n;
+        del_t[cur ^ 1][threadIdx.x] = del_n;
+      }
+      // issue tile i+2 staging prefetch + tile i+1 fragment loads
+      qfp += qstep; dofp += dstep; qsp += qstep; dosp += dstep;
+      lsep += 32; delp += 32;
+      if (qt0 + 64 < T) {
+        dost_n = stage_at(dosp + dstep);
+        qst_n = stage_at(qsp + qstep);
+        if (threadIdx.x < 32) {
+          lse_n = lsep[32];
+          del_n = delp[32];
+        }
       }
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
@@ -531,52 +587,35 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         dof_n[t] = frag_at(dofp, t);
       }
     }
-    if (causal && qt0 + 31 < kw) continue;  // whole tile above diagonal
 
-    // S[q][key]: A = Q (i=q), B = K (j=key); dP[q][key]: A = dO, B = V
-    f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_c[0], kf[0], zc, 0, 0, 0);
-    f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_c[0], vf[0], zc, 0, 0, 0);
+    // dV/dK MFMAs for tile i (LDS buf[cur]) — interleaves with the
+    // S/dP MFMAs for tile i+1 below (independent accumulators)
+    if (active) {
 #pragma unroll
-    for (int t = 1; t < 4; ++t) {
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_c[t], kf[t], s, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_c[t], vf[t], dp_, 0, 0, 0);
-    }
-    const bool diag = causal && (qt0 < kw + 31);  // tile crosses the diagonal
-    float pv[16], gv[16];
-    if (diag) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int qrow = drow(r, lane);
-        float p = (mykey > qt0 + qrow)
-                      ? 0.f
-                      : __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[qrow]);
-        pv[r] = p;
-        gv[r] = scale * p * (dp_[r] - del_t[qrow]);
-      }
-    } else {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int qrow = drow(r, lane);
-        float p = __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[qrow]);
-        pv[r] = p;
-        gv[r] = scale * p * (dp_[r] - del_t[qrow]);
+      for (int mt = 0; mt < 2; ++mt) {
+        const unsigned short* adot = &dot_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        const unsigned short* aqt = &qt_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        bf16x8 d0 = *reinterpret_cast<const bf16x8*>(adot);
+        bf16x8 d1 = *reinterpret_cast<const bf16x8*>(adot + 16);
+        bf16x8 q0f = *reinterpret_cast<const bf16x8*>(aqt);
+        bf16x8 q1f = *reinterpret_cast<const bf16x8*>(aqt + 16);
+        dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d0, pf0, dva[mt], 0, 0, 0);
+        dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d1, pf1, dva[mt], 0, 0, 0);
+        dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q0f, gf0, dka[mt], 0, 0, 0);
+        dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q1f, gf1, dka[mt], 0, 0, 0);
       }
     }
-    bf16x8 pf0 = relayout8(pv), pf1 = relayout8(pv + 8);
-    bf16x8 gf0 = relayout8(gv), gf1 = relayout8(gv + 8);
+    if (have_next) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[0], kf[0], zc, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[0], vf[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
-      const unsigned short* adot = &dot_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
-      const unsigned short* aqt = &qt_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
-      bf16x8 d0 = *reinterpret_cast<const bf16x8*>(adot);
-      bf16x8 d1 = *reinterpret_cast<const bf16x8*>(adot + 16);
-      bf16x8 q0f = *reinterpret_cast<const bf16x8*>(aqt);
-      bf16x8 q1f = *reinterpret_cast<const bf16x8*>(aqt + 16);
-      dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d0, pf0, dva[mt], 0, 0, 0);
-      dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d1, pf1, dva[mt], 0, 0, 0);
-      dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q0f, gf0, dka[mt], 0, 0, 0);
-      dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q1f, gf1, dka[mt], 0, 0, 0);
+      for (int t = 1; t < 4; ++t) {
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[t], kf[t], s, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[t], vf[t], dp_, 0, 0, 0);
+      }
     }
+    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
+    cur ^= 1;
   }
 
   if (mykey < T) {
