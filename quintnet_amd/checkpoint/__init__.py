@@ -56,21 +56,50 @@ def save_sharded_checkpoint(
             },
             "config": config or {},
         }
-        if optimizer is not None and hasattr(optimizer, "state_dict"):
-            try:
-                payload["optimizer_state_dict"] = optimizer.state_dict()
-            except Exception:  # noqa: BLE001 — optimizer state is best-effort
-                pass
         torch.save(payload, path)
+    # ZeRO-1 optimizer state is sharded over the DP axis too: every dp
+    # rank writes its own shard (mid-training resume — a capability the
+    # reference never finished).
+    if optimizer is not None and hasattr(optimizer, "state_dict"):
+        opath = os.path.join(
+            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}_dp{dp_rank}.pt"
+        )
+        try:
+            sd = optimizer.state_dict()
+            sd = {
+                k: (v.cpu() if isinstance(v, torch.Tensor) else v) for k, v in sd.items()
+            }
+            torch.save(sd, opath)
+        except Exception:  # noqa: BLE001 — optimizer state is best-effort
+            pass
     if dist.is_initialized():
         dist.barrier()
     return path
 
 
-def load_sharded_checkpoint(model, out_dir: str, name: str, pg_manager=None, strict=True):
+def load_sharded_checkpoint(
+    model, out_dir: str, name: str, pg_manager=None, optimizer=None, strict=True
+):
+    """Reload this rank's model shard (and, for mid-training resume, its
+    ZeRO optimizer shard when ``optimizer`` is given)."""
     pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
     tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
+    dp_rank = pg_manager.dp_rank if pg_manager is not None else 0
     path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}.pt")
     ckpt = torch.load(path, map_location="cpu", weights_only=False)
     model.load_state_dict(ckpt["model_state_dict"], strict=strict)
+    if optimizer is not None:
+        if hasattr(optimizer, "refresh_master_"):
+            optimizer.refresh_master_()
+        opath = os.path.join(
+            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}_dp{dp_rank}.pt"
+        )
+        if os.path.exists(opath):
+            osd = torch.load(opath, map_location="cpu", weights_only=False)
+            dev = next(model.parameters()).device
+            osd = {
+                k: (v.to(dev) if isinstance(v, torch.Tensor) else v)
+                for k, v in osd.items()
+            }
+            optimizer.load_state_dict(osd)
     return ckpt

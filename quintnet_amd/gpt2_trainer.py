@@ -52,6 +52,15 @@ class GPT2Trainer(Trainer):
             # re-bind the pipeline trainer to the new optimizer/criterion
             self.pipeline_trainer.optimizer = self.optimizer
             self.pipeline_trainer.criterion = self.criterion
+        resume = config.get("resume_from")
+        if resume:
+            from .checkpoint import load_sharded_checkpoint
+
+            load_sharded_checkpoint(
+                _unwrap(self.model), resume,
+                name=config.get("checkpoint_name", "final_model"),
+                pg_manager=pg_manager, optimizer=self.optimizer,
+            )
 
     # ------------------------------------------------------------------
     def _infer_seq_hidden(self, inner) -> tuple:

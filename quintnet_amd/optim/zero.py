@@ -176,6 +176,12 @@ class ZeroRedundancyAdamW:
             if p.dtype != self.dtype:
                 p.data.copy_(self.flat_param[off : off + p.numel()].view_as(p).to(p.dtype))
 
+    def refresh_master_(self) -> None:
+        """Re-sync the fp32 master shard from the (possibly just-loaded)
+        model params.  Call after load_state_dict on the MODEL when no
+        optimizer checkpoint is restored."""
+        self.master.copy_(self.flat_param[self._shard_slice].float())
+
     def zero_grad(self, set_to_none: bool = False) -> None:
         if self._grad_mode == "owned":
             self.flat_grad.zero_()

@@ -94,3 +94,43 @@ def test_staged_load_matches_full_model(tmp_path):
     qkv_ref = ln @ hf["h.0.attn.c_attn.weight"] + hf["h.0.attn.c_attn.bias"]
     qkv_ours = stage.blocks[0].attn.c_attn(stage.blocks[0].ln_1(x))
     assert torch.allclose(qkv_ours, qkv_ref, atol=1e-4), (qkv_ours - qkv_ref).abs().max()
+
+
+def test_resume_roundtrip(tmp_path):
+    """Mid-training resume: save model+optimizer, reload, training
+    continues with identical state (beyond-reference capability —
+    SURVEY.md §5.4 notes the reference never loads optimizer state)."""
+    import copy
+
+    from quintnet_amd.checkpoint import load_sharded_checkpoint, save_sharded_checkpoint
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.optim import ZeroRedundancyAdamW
+
+    torch.manual_seed(0)
+    cfg = GPT2Config(n_embd=32, n_layer=2, n_head=2, vocab_size=64, n_positions=32, dropout=0.0)
+    stage = GPT2Stage(cfg)
+    opt = ZeroRedundancyAdamW(stage.parameters(), lr=1e-3)
+    ids = torch.randint(0, 64, (2, 16))
+    for _ in range(3):
+        loss = causal_lm_loss(stage(ids), ids)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+    save_sharded_checkpoint(stage, str(tmp_path), name="ck", optimizer=opt)
+
+    stage2 = GPT2Stage(cfg)  # fresh random init
+    opt2 = ZeroRedundancyAdamW(stage2.parameters(), lr=1e-3)
+    load_sharded_checkpoint(stage2, str(tmp_path), name="ck", optimizer=opt2)
+    assert opt2.step_count == 3
+    assert torch.allclose(opt2.master, opt.master)
+    assert torch.allclose(opt2.exp_avg, opt.exp_avg)
+
+    # both continue identically
+    for m, o in ((stage, opt), (stage2, opt2)):
+        loss = causal_lm_loss(m(ids), ids)
+        loss.backward()
+        o.step()
+        o.zero_grad()
+    for p1, p2 in zip(stage.parameters(), stage2.parameters()):
+        assert torch.allclose(p1, p2, atol=1e-6)
