@@ -49,6 +49,8 @@ def parse_args():
     p.add_argument("--grad-acc", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
     p.add_argument("--tiny", action="store_true", help="tiny model for CPU smoke runs")
+    p.add_argument("--sequence-parallel", action="store_true",
+                   help="Megatron-SP over the TP axis (GPT-2)")
     return p.parse_args()
 
 
@@ -66,11 +68,14 @@ def build_gpt2(args, pg, device, dtype):
         PipelineParallelWrapper,
     )
 
+    sp = args.sequence_parallel and pg.tp_size > 1
     if args.tiny:
         cfg = GPT2Config(n_embd=64, n_layer=2, n_head=2, vocab_size=512,
-                         n_positions=args.seq_len, dropout=0.0)
+                         n_positions=args.seq_len, dropout=0.0,
+                         sequence_parallel=sp)
     else:
-        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len))
+        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len),
+                         sequence_parallel=sp)
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     stage = GPT2Stage(
         cfg,
@@ -81,14 +86,17 @@ def build_gpt2(args, pg, device, dtype):
         device=device,
         dtype=dtype,
     )
-    stage.seq_len = args.seq_len
+    # under SP the inter-stage activation is a sequence SHARD
+    args.pipe_seq = args.seq_len // pg.tp_size if sp else args.seq_len
+    pipe_seq = args.pipe_seq
+    stage.seq_len = pipe_seq
     stage.hidden_dim = cfg.n_embd
     model = stage
     if pg.pp_size > 1:
         model = PipelineParallelWrapper(stage_module=stage, pp_rank=pg.pp_rank,
                                         pp_group=pg.get_group("pp"), pp_size=pg.pp_size,
                                         device=device)
-        model.seq_len, model.hidden_dim = args.seq_len, cfg.n_embd
+        model.seq_len, model.hidden_dim = pipe_seq, cfg.n_embd
     if pg.dp_size > 1:
         model = DataParallel(model, DistributedConfig(
             pg.dp_rank, pg.dp_size, pg.get_group("dp")))
@@ -181,6 +189,7 @@ def main():
             batches.append({"input_ids": ids, "labels": ids.clone()})
         task = "clm"
         seq, hidden = args.seq_len, cfg.n_embd
+        pipe_shape_seq = getattr(args, "pipe_seq", seq)
         model_name = "gpt2-124M" if not args.tiny else "gpt2-tiny"
     else:
         micro_b = args.micro_batch or 16
@@ -216,7 +225,7 @@ def main():
             pp_group=pg.get_group("pp"), pp_group_ranks=pg.get_group_ranks("pp"),
             schedule="1f1b", task_type=task, max_grad_norm=1.0,
         )
-        shapes = (micro_b, seq, hidden)
+        shapes = (micro_b, pipe_shape_seq if task == "clm" else seq, hidden)
 
         def step():
             ptrainer.train_step(loader, shapes, device, dtype)

@@ -10,6 +10,7 @@ from .comm import (
     All_Reduce,
     ReduceScatter,
     copy_to_group,
+    scatter_to_sequence,
     pipeline_communicate,
     bidirectional_pipeline_communicate,
 )
@@ -37,6 +38,7 @@ __all__ = [
     "All_Reduce",
     "ReduceScatter",
     "copy_to_group",
+    "scatter_to_sequence",
     "pipeline_communicate",
     "bidirectional_pipeline_communicate",
     "get_rank",

@@ -186,6 +186,37 @@ def copy_to_group(x: torch.Tensor, group=None) -> torch.Tensor:
     return _CopyToGroup.apply(x, group)
 
 
+class _ScatterToSequence(torch.autograd.Function):
+    """fwd: keep this rank's sequence shard; bwd: all-gather the grads.
+
+    The entry operator of Megatron sequence parallelism (splits the
+    replicated embedding output across the TP group).
+    """
+
+    @staticmethod
+    def forward(ctx, x: torch.Tensor, group=None, dim: int = 1):
+        world = _ws(group)
+        ctx.group = group
+        ctx.dim = dim
+        ctx.world = world
+        if world == 1:
+            return x
+        rank = dist.get_rank(group=group)
+        return x.chunk(world, dim=dim)[rank].contiguous()
+
+    @staticmethod
+    def backward(ctx, grad_output):
+        if ctx.world == 1:
+            return grad_output, None, None
+        gathered = [torch.empty_like(grad_output) for _ in range(ctx.world)]
+        dist.all_gather(gathered, grad_output.contiguous(), group=ctx.group)
+        return torch.cat(gathered, dim=ctx.dim), None, None
+
+
+def scatter_to_sequence(x: torch.Tensor, group=None, dim: int = 1) -> torch.Tensor:
+    return _ScatterToSequence.apply(x, group, dim)
+
+
 class ReduceScatter(torch.autograd.Function):
     """forward: reduce_scatter(SUM) along dim; backward: all_gather.
 

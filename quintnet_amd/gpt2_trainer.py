@@ -67,6 +67,8 @@ class GPT2Trainer(Trainer):
         mc = self.config.get("model_config", {})
         seq = int(self.config.get("max_seq_length", mc.get("n_positions", 1024)))
         hidden = int(mc.get("n_embd", 768))
+        if mc.get("sequence_parallel") and self.pg is not None and self.pg.tp_size > 1:
+            seq //= self.pg.tp_size  # inter-stage activations are seq shards
         return seq, hidden
 
     # ------------------------------------------------------------------

@@ -29,11 +29,13 @@ class GPT2Attention(nn.Module):
     def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
         super().__init__()
         self.config = config
+        sp = config.sequence_parallel
         self.c_attn = ColumnParallelLinear(
             config.n_embd,
             3 * config.n_embd,
             tp_group=tp_group,
             gather_output=False,
+            sequence_parallel=sp,
             device=device,
             dtype=dtype,
         )
@@ -42,6 +44,7 @@ class GPT2Attention(nn.Module):
             config.n_embd,
             tp_group=tp_group,
             input_is_parallel=True,
+            sequence_parallel=sp,
             device=device,
             dtype=dtype,
         )

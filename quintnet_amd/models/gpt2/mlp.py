@@ -20,12 +20,14 @@ __all__ = ["GPT2MLP"]
 class GPT2MLP(nn.Module):
     def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
         super().__init__()
+        sp = config.sequence_parallel
         self.c_fc = ColumnParallelLinear(
             config.n_embd,
             config.n_inner,
             tp_group=tp_group,
             gather_output=False,
             activation="gelu",  # fused into the MFMA GEMM epilogue
+            sequence_parallel=sp,
             device=device,
             dtype=dtype,
         )
@@ -34,6 +36,7 @@ class GPT2MLP(nn.Module):
             config.n_embd,
             tp_group=tp_group,
             input_is_parallel=True,
+            sequence_parallel=sp,
             device=device,
             dtype=dtype,
         )

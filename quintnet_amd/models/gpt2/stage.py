@@ -21,6 +21,7 @@ import torch
 import torch.distributed as dist
 import torch.nn as nn
 
+from ...core.comm import All_Gather, scatter_to_sequence
 from ...ops import FusedLayerNorm, linear as fused_linear
 from ...parallel.pipeline.wrapper import distribute_layers
 from .block import GPT2Block
@@ -45,7 +46,9 @@ class GPT2Stage(nn.Module):
         self.config = config
         self.pp_rank = pp_rank
         self.pp_size = pp_size
+        self.tp_group = tp_group
         self.tied_group = tied_group
+        self.sequence_parallel = config.sequence_parallel
         self.is_first_stage = pp_rank == 0
         self.is_last_stage = pp_rank == pp_size - 1
         kw = {"device": device, "dtype": dtype}
@@ -73,6 +76,9 @@ class GPT2Stage(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.is_first_stage:
             x = self.embedding(x)
+            if self.sequence_parallel:
+                # enter SP: keep only this rank's sequence shard
+                x = scatter_to_sequence(x, self.tp_group, 1)
         # residual-fused block chain: each block's trailing residual add
         # rides the NEXT LayerNorm kernel (ops/layernorm.py)
         pending = None
@@ -83,6 +89,12 @@ class GPT2Stage(nn.Module):
                 x = self.ln_f(x)
             else:
                 x, _ = self.ln_f(x, residual=pending)
+            if self.sequence_parallel:
+                # leave SP: full sequence for the LM head.  Backward mode
+                # is "slice": everything downstream (lm_head + loss) is
+                # REPLICATED across TP ranks, so each rank already holds
+                # the complete grad — reduce-scatter would double-count.
+                x = All_Gather.apply(x, self.tp_group, 1, "slice")
             w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
             x = fused_linear(x, w, None, None, prefer_library=True)  # plain GEMM: hipBLASLt
         elif pending is not None:
@@ -90,8 +102,28 @@ class GPT2Stage(nn.Module):
         return x
 
     # ------------------------------------------------------------------
+    def sync_sequence_parallel_grads(self) -> None:
+        """Under SP the LayerNorms and row-parallel biases compute their
+        grads from sequence SHARDS — all-reduce them over the TP group
+        (the Megatron "sequence-parallel params" sync)."""
+        if not self.sequence_parallel or self.tp_group is None:
+            return
+        if not dist.is_initialized() or dist.get_world_size(group=self.tp_group) == 1:
+            return
+        params = []
+        for blk in self.blocks:
+            params += [blk.ln_1.weight, blk.ln_1.bias, blk.ln_2.weight, blk.ln_2.bias]
+            params += [blk.attn.c_proj.bias, blk.mlp.c_proj.bias]
+        if self.is_last_stage:
+            params += [self.ln_f.weight, self.ln_f.bias]
+        for p in params:
+            if p is not None and p.grad is not None:
+                dist.all_reduce(p.grad, op=dist.ReduceOp.SUM, group=self.tp_group)
+
     def sync_tied_weights_grad(self) -> None:
-        """Average the tied wte/lm_head gradient between first & last stage."""
+        """Average the tied wte/lm_head gradient between first & last
+        stage (and run the SP grad sync on every stage)."""
+        self.sync_sequence_parallel_grads()
         if self.pp_size == 1 or not (self.is_first_stage or self.is_last_stage):
             return
         if not dist.is_initialized():

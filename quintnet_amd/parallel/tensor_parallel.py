@@ -16,7 +16,7 @@ import torch
 import torch.distributed as dist
 import torch.nn as nn
 
-from ..core.comm import All_Gather, All_Reduce, copy_to_group
+from ..core.comm import All_Gather, All_Reduce, ReduceScatter, copy_to_group
 from ..ops import linear as fused_linear
 
 __all__ = [
@@ -66,6 +66,7 @@ class ColumnParallelLinear(nn.Module):
         bias: bool = True,
         gather_output: bool = True,
         activation: Optional[str] = None,
+        sequence_parallel: bool = False,
         device=None,
         dtype=None,
     ):
@@ -73,6 +74,7 @@ class ColumnParallelLinear(nn.Module):
         self.in_features = in_features
         self.out_features = out_features
         self.tp_group = tp_group
+        self.sequence_parallel = sequence_parallel
         self.tp_size = _group_size(tp_group)
         self.tp_rank = _group_rank(tp_group)
         ensure_divisibility(out_features, self.tp_size)
@@ -110,7 +112,12 @@ class ColumnParallelLinear(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.tp_size > 1:
-            x = copy_to_group(x, self.tp_group)  # identity fwd, all-reduce grad bwd
+            if self.sequence_parallel:
+                # Megatron-SP g operator: all-gather the sequence shards
+                # fwd, reduce-scatter grads bwd
+                x = All_Gather.apply(x, self.tp_group, 1, "reduce_scatter")
+            else:
+                x = copy_to_group(x, self.tp_group)  # identity fwd, all-reduce bwd
         out = fused_linear(x, self.weight, self.bias, self.activation)
         if self.gather_output and self.tp_size > 1:
             out = All_Gather.apply(out, self.tp_group, -1, "slice")
@@ -138,6 +145,7 @@ class RowParallelLinear(nn.Module):
         tp_group=None,
         bias: bool = True,
         input_is_parallel: bool = True,
+        sequence_parallel: bool = False,
         device=None,
         dtype=None,
     ):
@@ -145,6 +153,7 @@ class RowParallelLinear(nn.Module):
         self.in_features = in_features
         self.out_features = out_features
         self.tp_group = tp_group
+        self.sequence_parallel = sequence_parallel
         self.tp_size = _group_size(tp_group)
         self.tp_rank = _group_rank(tp_group)
         ensure_divisibility(in_features, self.tp_size)
@@ -186,7 +195,12 @@ class RowParallelLinear(nn.Module):
         if not self.input_is_parallel:
             x = x.chunk(self.tp_size, dim=-1)[self.tp_rank].contiguous()
         out = fused_linear(x, self.weight, None, None)
-        out = All_Reduce.apply(out, self.tp_group)
+        if self.sequence_parallel:
+            # Megatron-SP ḡ operator: reduce-scatter the partial sums
+            # over the sequence dim (grads all-gather back)
+            out = ReduceScatter.apply(out, self.tp_group, 1)
+        else:
+            out = All_Reduce.apply(out, self.tp_group)
         if self.bias is not None:
             out = out + self.bias
         return out
