@@ -51,26 +51,27 @@ __device__ __forceinline__ void glds16(const void* g, void* lds) {
 // Fragment reads therefore XOR their group index with (row & 7).
 // ---------------------------------------------------------------------------
 // fast path: one glds instruction stages 8 rows (64 lanes x 16B);
-// each wave covers 32 rows of the 128-row tile with 4 calls.  The
+// NWAVES waves cover ROWS rows with ROWS/(8*NWAVES) calls each.  The
 // per-lane source pointers are precomputed once and bumped by BK per
 // tile — the 64-bit address rebuild per tile was ~40% of the kernel's
 // VALU issue (PMC: 6.2 VALU/MFMA).
+template <int ROWS, int NWAVES, int CALLS = ROWS / (8 * NWAVES)>
 __device__ __forceinline__ void stage_glds_pre(
-    unsigned short* lds, const unsigned short* const (&gp)[4], int k_elems,
+    unsigned short* lds, const unsigned short* const (&gp)[CALLS], int k_elems,
     int wave) {
 #pragma unroll
-  for (int j = 0; j < 4; ++j)
-    glds16(gp[j] + k_elems, lds + (wave * 32 + j * 8) * BK);
+  for (int j = 0; j < CALLS; ++j)
+    glds16(gp[j] + k_elems, lds + (wave * (ROWS / NWAVES) + j * 8) * BK);
 }
 
 // slow path (edges): bounds-checked loads to regs ...
-template <int PASSES>
+template <int NTHREADS, int PASSES>
 __device__ __forceinline__ void load_tile_regs(
     const unsigned short* __restrict__ src, long long ld, int row0, int rows,
     int k0, int K, s16x8 (&regs)[PASSES]) {
 #pragma unroll
   for (int p = 0; p < PASSES; ++p) {
-    int lin = p * 256 + threadIdx.x;
+    int lin = p * NTHREADS + threadIdx.x;
     int row = lin >> 3;
     int g = (lin & 7) ^ (row & 7);  // read the swizzled source group
     int k = k0 + g * 8;
@@ -81,33 +82,35 @@ __device__ __forceinline__ void load_tile_regs(
 }
 
 // ... then linear ds_writes (same image as glds)
-template <int PASSES>
+template <int NTHREADS, int PASSES>
 __device__ __forceinline__ void write_tile_lds(unsigned short* lds, const s16x8 (&regs)[PASSES]) {
 #pragma unroll
   for (int p = 0; p < PASSES; ++p) {
-    int lin = p * 256 + threadIdx.x;
+    int lin = p * NTHREADS + threadIdx.x;
     int row = lin >> 3;
     int g = lin & 7;
     *reinterpret_cast<s16x8*>(lds + row * BK + g * 8) = regs[p];
   }
 }
 
-template <int ACT, bool SAVE_PRE, bool USE_GLDS>
-__global__ __launch_bounds__(256) void gemm_nt_kernel(
+template <int ACT, bool SAVE_PRE, bool USE_GLDS, int TBM, int TBN, int WR, int WC>
+__global__ __launch_bounds__(WR * WC * 64) void gemm_nt_kernel(
     const unsigned short* __restrict__ A,  // [M,K]
     const unsigned short* __restrict__ B,  // [N,K]
     const unsigned short* __restrict__ bias,  // [N] or nullptr
     unsigned short* __restrict__ C,        // [M,N]
     unsigned short* __restrict__ Cpre,     // [M,N] pre-activation (SAVE_PRE)
     int M, int N, int K) {
-  // two double-buffered tiles: buffer b: A at b*2*BM*BK, B at +BM*BK
-  __shared__ unsigned short smem[2 * 2 * BM * BK];
-#define As(b) (smem + (b) * 2 * BM * BK)
-#define Bs(b) (smem + (b) * 2 * BM * BK + BM * BK)
+  constexpr int NWAVES = WR * WC;
+  constexpr int NTHREADS = NWAVES * 64;
+  // two double-buffered tiles: buffer b: A at b*(TBM+TBN)*BK, B after A
+  __shared__ unsigned short smem[2 * (TBM + TBN) * BK];
+#define As(b) (smem + (b) * (TBM + TBN) * BK)
+#define Bs(b) (smem + (b) * (TBM + TBN) * BK + TBM * BK)
 
   // XCD-aware bijective block swizzle (T1)
-  const int nbm = (M + BM - 1) / BM;
-  const int nbn = (N + BN - 1) / BN;
+  const int nbm = (M + TBM - 1) / TBM;
+  const int nbn = (N + TBN - 1) / TBN;
   const int nwg = nbm * nbn;
   int bid = blockIdx.x;
   if (nwg >= 16) {
@@ -118,15 +121,15 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   }
   const int bm = bid / nbn;
   const int bn = bid % nbn;
-  const int m0 = bm * BM;
-  const int n0 = bn * BN;
+  const int m0 = bm * TBM;
+  const int n0 = bn * TBN;
 
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const int wr = wave >> 1;   // 0..1
-  const int wc = wave & 1;    // 0..1
+  const int wr = wave / WC;
+  const int wc = wave % WC;
 
-  const bool interior_mn = (m0 + BM <= M) && (n0 + BN <= N);
+  const bool interior_mn = (m0 + TBM <= M) && (n0 + TBN <= N);
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -138,13 +141,20 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   const int kq = lane >> 4;         // k quarter (0..3) of the 32-K step
 
   // precomputed per-lane glds source pointers (k advances via offset)
-  const unsigned short* gpa[4];
-  const unsigned short* gpb[4];
+  constexpr int ACALLS = TBM / (8 * NWAVES);
+  constexpr int BCALLS = TBN / (8 * NWAVES);
+  const unsigned short* gpa[ACALLS];
+  const unsigned short* gpb[BCALLS];
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    int row = wave * 32 + j * 8 + (lane >> 3);
+  for (int j = 0; j < ACALLS; ++j) {
+    int row = wave * (TBM / NWAVES) + j * 8 + (lane >> 3);
     int g = (lane & 7) ^ (row & 7);  // source-side swizzle (rule 21)
     gpa[j] = A + (long long)(m0 + row) * K + g * 8;
+  }
+#pragma unroll
+  for (int j = 0; j < BCALLS; ++j) {
+    int row = wave * (TBN / NWAVES) + j * 8 + (lane >> 3);
+    int g = (lane & 7) ^ (row & 7);
     gpb[j] = B + (long long)(n0 + row) * K + g * 8;
   }
   // precomputed LDS fragment offsets (elements); ks=1 toggles bit 5 (^32)
@@ -160,20 +170,22 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
 #define QN_MFMA_TILE(at, bt)                                                     _Pragma("unroll")                                                              for (int ks = 0; ks < 2; ++ks) {                                                 bf16x8 af[4], bf[4];                                                           _Pragma("unroll")                                                              for (int f = 0; f < 4; ++f) {                                                    af[f] = *reinterpret_cast<const bf16x8*>(&(at)[aoff[f] ^ (ks << 5)]);          bf[f] = *reinterpret_cast<const bf16x8*>(&(bt)[boff[f] ^ (ks << 5)]);        }                                                                              _Pragma("unroll")                                                              for (int i = 0; i < 4; ++i)                                                      _Pragma("unroll")                                                              for (int j = 0; j < 4; ++j)                                                      acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                               af[i], bf[j], acc[i][j], 0, 0, 0);                                   }
 
   // ---- prologue: stage tile 0 into buffer 0 -------------------------------
-  s16x8 ra[4], rb[4];
+  constexpr int APASS = TBM * 8 / NTHREADS;
+  constexpr int BPASS = TBN * 8 / NTHREADS;
+  s16x8 ra[APASS], rb[BPASS];
   const bool all_fast = USE_GLDS && interior_mn && (K % BK == 0);
   if (all_fast) {
     // fully-interior fast loop: unrolled x2 so buffer pointers and LDS
     // offsets stay loop-invariant; glds sources advance by += k
-    stage_glds_pre(As(0), gpa, 0, wave);
-    stage_glds_pre(Bs(0), gpb, 0, wave);
+    stage_glds_pre<TBM, NWAVES>(As(0), gpa, 0, wave);
+    stage_glds_pre<TBN, NWAVES>(Bs(0), gpb, 0, wave);
     int k0 = 0;
     while (true) {
       // even tile in buf0
       __syncthreads();
       if (k0 + BK < K) {
-        stage_glds_pre(As(1), gpa, k0 + BK, wave);
-        stage_glds_pre(Bs(1), gpb, k0 + BK, wave);
+        stage_glds_pre<TBM, NWAVES>(As(1), gpa, k0 + BK, wave);
+        stage_glds_pre<TBN, NWAVES>(Bs(1), gpb, k0 + BK, wave);
       }
       QN_MFMA_TILE(As(0), Bs(0));
       k0 += BK;
@@ -181,8 +193,8 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       // odd tile in buf1
       __syncthreads();
       if (k0 + BK < K) {
-        stage_glds_pre(As(0), gpa, k0 + BK, wave);
-        stage_glds_pre(Bs(0), gpb, k0 + BK, wave);
+        stage_glds_pre<TBM, NWAVES>(As(0), gpa, k0 + BK, wave);
+        stage_glds_pre<TBN, NWAVES>(Bs(0), gpb, k0 + BK, wave);
       }
       QN_MFMA_TILE(As(1), Bs(1));
       k0 += BK;
@@ -191,13 +203,13 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
   } else {
     bool fast0 = USE_GLDS && interior_mn && (BK <= K);
     if (fast0) {
-      stage_glds_pre(As(0), gpa, 0, wave);
-      stage_glds_pre(Bs(0), gpb, 0, wave);
+      stage_glds_pre<TBM, NWAVES>(As(0), gpa, 0, wave);
+      stage_glds_pre<TBN, NWAVES>(Bs(0), gpb, 0, wave);
     } else {
-      load_tile_regs<4>(A, K, m0, M, 0, K, ra);
-      load_tile_regs<4>(B, K, n0, N, 0, K, rb);
-      write_tile_lds<4>(As(0), ra);
-      write_tile_lds<4>(Bs(0), rb);
+      load_tile_regs<NTHREADS, APASS>(A, K, m0, M, 0, K, ra);
+      load_tile_regs<NTHREADS, BPASS>(B, K, n0, N, 0, K, rb);
+      write_tile_lds<NTHREADS, APASS>(As(0), ra);
+      write_tile_lds<NTHREADS, BPASS>(Bs(0), rb);
     }
     int cur = 0;
     for (int k0 = 0; k0 < K; k0 += BK) {
@@ -207,19 +219,19 @@ __global__ __launch_bounds__(256) void gemm_nt_kernel(
       const bool fast_next = USE_GLDS && interior_mn && (kn + BK <= K);
       if (have_next) {
         if (fast_next) {
-          stage_glds_pre(As(cur ^ 1), gpa, kn, wave);
-          stage_glds_pre(Bs(cur ^ 1), gpb, kn, wave);
+          stage_glds_pre<TBM, NWAVES>(As(cur ^ 1), gpa, kn, wave);
+          stage_glds_pre<TBN, NWAVES>(Bs(cur ^ 1), gpb, kn, wave);
         } else {
-          load_tile_regs<4>(A, K, m0, M, kn, K, ra);  // issue loads now,
-          load_tile_regs<4>(B, K, n0, N, kn, K, rb);  // write after compute (T14)
+          load_tile_regs<NTHREADS, APASS>(A, K, m0, M, kn, K, ra);  // issue now,
+          load_tile_regs<NTHREADS, BPASS>(B, K, n0, N, kn, K, rb);  // write later (T14)
         }
       }
       const unsigned short* at = As(cur);
       const unsigned short* bt = Bs(cur);
       QN_MFMA_TILE(at, bt);
       if (have_next && !fast_next) {
-        write_tile_lds<4>(As(cur ^ 1), ra);
-        write_tile_lds<4>(Bs(cur ^ 1), rb);
+        write_tile_lds<NTHREADS, APASS>(As(cur ^ 1), ra);
+        write_tile_lds<NTHREADS, BPASS>(Bs(cur ^ 1), rb);
       }
       cur ^= 1;
     }
@@ -257,23 +269,32 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
                     const unsigned short* bias, unsigned short* C,
                     unsigned short* Cpre, int M, int N, int K, int act,
                     hipStream_t stream) {
-  const int nbm = (M + BM - 1) / BM;
-  const int nbn = (N + BN - 1) / BN;
-  dim3 grid(nbm * nbn);
-  dim3 block(256);
   static int use_glds = -1;
   if (use_glds < 0) {
     const char* e = getenv("QN_GEMM_GLDS");
     use_glds = (e && e[0] == '0') ? 0 : 1;  // default: glds fast path
   }
+  static int big_tile = -1;
+  if (big_tile < 0) {
+    const char* e = getenv("QN_GEMM_BIG");
+    big_tile = (e && e[0] == '1') ? 1 : 0;  // 128x128 default (A/B: big tile is a wash at K=768)
+  }
+  // 256x128 (8 waves) when the problem fills the chip with it; else 128x128
+  const bool big = big_tile && (M % 256 == 0) && ((long long)(M / 256) * ((N + 127) / 128) >= 256);
+#define QN_GEMM_LAUNCH(A_, S_, G_, TBM_, TBN_, WR_, WC_)                       \
+  hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, G_, TBM_, TBN_, WR_, WC_>),       \
+                     dim3(((M + TBM_ - 1) / TBM_) * ((N + TBN_ - 1) / TBN_)),  \
+                     dim3(WR_ * WC_ * 64), 0, stream, A, B, bias, C, Cpre, M,  \
+                     N, K)
 #define QN_GEMM_CASE(A_, S_)                                                   \
   do {                                                                         \
-    if (use_glds)                                                              \
-      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, true>), grid, block, 0,       \
-                         stream, A, B, bias, C, Cpre, M, N, K);                \
-    else                                                                       \
-      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, false>), grid, block, 0,      \
-                         stream, A, B, bias, C, Cpre, M, N, K);                \
+    if (big) {                                                                 \
+      if (use_glds) QN_GEMM_LAUNCH(A_, S_, true, 256, 128, 4, 2);              \
+      else QN_GEMM_LAUNCH(A_, S_, false, 256, 128, 4, 2);                      \
+    } else {                                                                   \
+      if (use_glds) QN_GEMM_LAUNCH(A_, S_, true, 128, 128, 2, 2);              \
+      else QN_GEMM_LAUNCH(A_, S_, false, 128, 128, 2, 2);                      \
+    }                                                                          \
   } while (0)
   if (act == QN_ACT_GELU) {
     if (Cpre) QN_GEMM_CASE(QN_ACT_GELU, true); else QN_GEMM_CASE(QN_ACT_GELU, false);
@@ -283,4 +304,5 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
     QN_GEMM_CASE(QN_ACT_NONE, false);
   }
 #undef QN_GEMM_CASE
+#undef QN_GEMM_LAUNCH
 }
