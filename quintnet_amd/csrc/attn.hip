@@ -289,18 +289,29 @@ __global__ void attn_delta_kernel(
     float* __restrict__ delta, long long nrows, int T, int H,
     long long dsB, long long dsH, long long dsT,
     long long osB, long long osH, long long osT) {
+  // 4 rows per wave, s16x4 (8B) loads per lane: lane l covers row l>>4,
+  // elems (l&15)*4 .. +4 (D = 64)
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
-  const long long row = (long long)blockIdx.x * 4 + wave;  // (b*H + h)*T + t
+  const int sub = lane >> 4;       // row within the wave's group of 4
+  const int d0 = (lane & 15) * 4;
+  const long long row = ((long long)blockIdx.x * 4 + wave) * 4 + sub;
   if (row >= nrows) return;
   const int t = (int)(row % T);
   const int bh = (int)(row / T);
   const int b = bh / H, h = bh % H;
-  const unsigned short* dp = dout + b * dsB + h * dsH + (long long)t * dsT;
-  const unsigned short* op = out + b * osB + h * osH + (long long)t * osT;
-  float acc = bf16_to_f32(dp[lane]) * bf16_to_f32(op[lane]);
-  acc = wave_reduce_sum(acc);
-  if (lane == 0) delta[row] = acc;
+  const unsigned short* dp = dout + b * dsB + h * dsH + (long long)t * dsT + d0;
+  const unsigned short* op = out + b * osB + h * osH + (long long)t * osT + d0;
+  s16x4 dv = *reinterpret_cast<const s16x4*>(dp);
+  s16x4 ov = *reinterpret_cast<const s16x4*>(op);
+  float acc = 0.f;
+#pragma unroll
+  for (int j = 0; j < 4; ++j)
+    acc += bf16_to_f32((unsigned short)dv[j]) * bf16_to_f32((unsigned short)ov[j]);
+  // reduce within each 16-lane group
+#pragma unroll
+  for (int off = 1; off < 16; off <<= 1) acc += __shfl_xor(acc, off, 64);
+  if ((lane & 15) == 0) delta[row] = acc;
 }
 
 // ===========================================================================
@@ -651,7 +662,7 @@ void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
                        long long osB, long long osH, long long osT,
                        hipStream_t stream) {
   long long rows = (long long)B * H * T;
-  hipLaunchKernelGGL(attn_delta_kernel, dim3((unsigned)((rows + 3) / 4)),
+  hipLaunchKernelGGL(attn_delta_kernel, dim3((unsigned)((rows + 15) / 16)),
                      dim3(256), 0, stream, dout, out, delta, rows, T, H, dsB,
                      dsH, dsT, osB, osH, osT);
 }

@@ -42,6 +42,20 @@ __device__ __forceinline__ unsigned short f32_to_bf16(float f) {
   return (unsigned short)(x >> 16);
 }
 
+// 8-wide bf16 load/store helpers (guide G13: scalar bf16 is ~2-2.5x)
+__device__ __forceinline__ void ld8_f32(const unsigned short* p, float* out) {
+  s16x8 v = *reinterpret_cast<const s16x8*>(p);
+#pragma unroll
+  for (int j = 0; j < 8; ++j) out[j] = bf16_to_f32((unsigned short)v[j]);
+}
+
+__device__ __forceinline__ void st8_f32(unsigned short* p, const float* in) {
+  s16x8 v;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) v[j] = (short)f32_to_bf16(in[j]);
+  *reinterpret_cast<s16x8*>(p) = v;
+}
+
 // generic scalar load/store as float, templated on element type
 template <typename T> __device__ __forceinline__ float ld_as_f32(const T* p);
 template <> __device__ __forceinline__ float ld_as_f32<float>(const float* p) { return *p; }

@@ -32,7 +32,26 @@ __global__ void layernorm_fwd_kernel(
   // cache up to 16 elems/lane in registers (H <= 1024) to avoid re-reads
   float cache[16];
   const bool cached = H <= QN_WAVE * 16;
-  if (cached) {
+  const bool vec8 = sizeof(T) == 2 && (H & 7) == 0 && H >= 512;
+  if (cached && vec8) {
+    // 8-wide bf16 loads (G13); lanes own disjoint 8-elem groups
+    int n = 0;
+    for (int i = lane * 8; i < H; i += QN_WAVE * 8, n += 8) {
+      ld8_f32(reinterpret_cast<const unsigned short*>(xr) + i, cache + n);
+      if (rr) {
+        float rv[8];
+        ld8_f32(reinterpret_cast<const unsigned short*>(rr) + i, rv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) cache[n + j] += rv[j];
+        if (sr) st8_f32(reinterpret_cast<unsigned short*>(sr) + i, cache + n);
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sum += cache[n + j];
+        sq += cache[n + j] * cache[n + j];
+      }
+    }
+  } else if (cached) {
     int n = 0;
     for (int i = lane; i < H; i += QN_WAVE, ++n) {
       float f = ld_as_f32(xr + i);
@@ -60,7 +79,18 @@ __global__ void layernorm_fwd_kernel(
   const float rstd = rsqrtf(var + eps);
   if (lane == 0) { mean_out[row] = mean; rstd_out[row] = rstd; }
 
-  if (cached) {
+  if (cached && vec8) {
+    int n = 0;
+    for (int i = lane * 8; i < H; i += QN_WAVE * 8, n += 8) {
+      float wv[8], bv[8], ov[8];
+      ld8_f32(reinterpret_cast<const unsigned short*>(w) + i, wv);
+      ld8_f32(reinterpret_cast<const unsigned short*>(b) + i, bv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        ov[j] = (cache[n + j] - mean) * rstd * wv[j] + bv[j];
+      st8_f32(reinterpret_cast<unsigned short*>(yr) + i, ov);
+    }
+  } else if (cached) {
     int n = 0;
     for (int i = lane; i < H; i += QN_WAVE, ++n) {
       float wi = ld_as_f32(w + i);
@@ -100,8 +130,24 @@ __global__ void layernorm_bwd_dx_kernel(
 
   float cx[16], cwdy[16];
   const bool cached = H <= QN_WAVE * 16;
+  const bool vec8 = sizeof(T) == 2 && (H & 7) == 0 && H >= 512;
   float c1 = 0.f, c2 = 0.f;
-  if (cached) {
+  if (cached && vec8) {
+    int n = 0;
+    for (int i = lane * 8; i < H; i += QN_WAVE * 8, n += 8) {
+      float xv[8], wv[8], dv[8];
+      ld8_f32(reinterpret_cast<const unsigned short*>(xr) + i, xv);
+      ld8_f32(reinterpret_cast<const unsigned short*>(w) + i, wv);
+      ld8_f32(reinterpret_cast<const unsigned short*>(dyr) + i, dv);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        float xhat = (xv[j] - mean) * rstd;
+        float wdy = wv[j] * dv[j];
+        cx[n + j] = xhat; cwdy[n + j] = wdy;
+        c1 += wdy; c2 += wdy * xhat;
+      }
+    }
+  } else if (cached) {
     int n = 0;
     for (int i = lane; i < H; i += QN_WAVE, ++n) {
       float xhat = (ld_as_f32(xr + i) - mean) * rstd;
@@ -119,7 +165,21 @@ __global__ void layernorm_bwd_dx_kernel(
   c1 = wave_reduce_sum(c1) / H;
   c2 = wave_reduce_sum(c2) / H;
 
-  if (cached) {
+  if (cached && vec8) {
+    int n = 0;
+    for (int i = lane * 8; i < H; i += QN_WAVE * 8, n += 8) {
+      float ov[8];
+#pragma unroll
+      for (int j = 0; j < 8; ++j) ov[j] = (cwdy[n + j] - c1 - cx[n + j] * c2) * rstd;
+      if (dsr) {
+        float dsv[8];
+        ld8_f32(reinterpret_cast<const unsigned short*>(dsr) + i, dsv);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) ov[j] += dsv[j];
+      }
+      st8_f32(reinterpret_cast<unsigned short*>(dxr) + i, ov);
+    }
+  } else if (cached) {
     int n = 0;
     for (int i = lane; i < H; i += QN_WAVE, ++n) {
       float v = (cwdy[n] - c1 - cx[n] * c2) * rstd;
