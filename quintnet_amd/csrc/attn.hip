@@ -329,7 +329,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     long long vsB, long long vsH, long long vsT,
     long long dsB, long long dsH, long long dsT,
     long long dqsB, long long dqsH, long long dqsT) {
-  __shared__ unsigned short kt_lds[64 * TPAD];
+  __shared__ unsigned short kt_lds[2][64 * TPAD];
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & 63;
@@ -359,72 +359,94 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
 
   const int kv_end = causal ? min(q0 + 128, T) : T;
-  // prefetch pipeline (T14), pointer-bumped sources
+  // cross-tile software pipeline (like attn_bwd_dkv): S/dP MFMAs for
+  // tile i+1 interleave with the dQ accumulation of tile i; ONE barrier
+  // per tile; double-buffered Kt.
   const unsigned short* kfp = frag_base(kp, ksT, 0, lane);
   const unsigned short* vfp = frag_base(vp, vsT, 0, lane);
   const unsigned short* ksp = stage_base(kp, ksT, 0);
   const long long kstep = 32 * ksT, vstep = 32 * vsT;
+
+  // prologue: stage Kt tile 0, prime s/dp for tile 0
+  stage_wr(kt_lds[0], stage_at(ksp));
   bf16x8 kf_n[4], vf_n[4];
-  s16x8 kst_n = stage_at(ksp);
 #pragma unroll
   for (int t = 0; t < 4; ++t) {
     kf_n[t] = frag_at(kfp, t);
     vf_n[t] = frag_at(vfp, t);
   }
-  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    __syncthreads();
-    stage_wr(kt_lds, kst_n);
-    __syncthreads();
-
-    bf16x8 kf_c[4], vf_c[4];
+  f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[0], qf[0], zc, 0, 0, 0);
+  f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[0], dof[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int t = 0; t < 4; ++t) { kf_c[t] = kf_n[t]; vf_c[t] = vf_n[t]; }
-    if (kv0 + 32 < kv_end) {
+  for (int t = 1; t < 4; ++t) {
+    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[t], qf[t], s, 0, 0, 0);
+    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[t], dof[t], dp_, 0, 0, 0);
+  }
+  s16x8 kst_n = {0, 0, 0, 0, 0, 0, 0, 0};
+  if (32 < kv_end) kst_n = stage_at(ksp + kstep);
+  __syncthreads();  // Kt buf0 visible
+
+  int cur = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+    const bool have_next = kv0 + 32 < kv_end;
+    const bool active = !(causal && kv0 > qw + 31);
+
+    bf16x8 gf0, gf1;
+    if (active) {
+      const bool diag = causal && (kv0 + 31 > qw);
+      float g[16];
+      if (diag) {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          int key = kv0 + drow(r, lane);
+          float p = (key > myq) ? 0.f : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+          g[r] = scale * p * (dp_[r] - my_delta);
+        }
+      } else {
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float p = __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+          g[r] = scale * p * (dp_[r] - my_delta);
+        }
+      }
+      gf0 = relayout8(g);
+      gf1 = relayout8(g + 8);
+    }
+
+    if (have_next) {
+      stage_wr(kt_lds[cur ^ 1], kst_n);
       kfp += kstep;
       vfp += vstep;
       ksp += kstep;
-      kst_n = stage_at(ksp);
+      if (kv0 + 64 < kv_end) kst_n = stage_at(ksp + kstep);
 #pragma unroll
       for (int t = 0; t < 4; ++t) {
         kf_n[t] = frag_at(kfp, t);
         vf_n[t] = frag_at(vfp, t);
       }
     }
-    if (causal && kv0 > qw + 31) continue;
 
-    f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[0], qf[0], zc, 0, 0, 0);
-    f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_c[0], dof[0], zc, 0, 0, 0);
+    if (active) {
 #pragma unroll
-    for (int t = 1; t < 4; ++t) {
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[t], qf[t], s, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_c[t], dof[t], dp_, 0, 0, 0);
-    }
-    const bool diag = causal && (kv0 + 31 > qw);
-    float g[16];
-    if (diag) {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        int key = kv0 + drow(r, lane);
-        float p = (key > myq) ? 0.f : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
-        g[r] = scale * p * (dp_[r] - my_delta);
-      }
-    } else {
-#pragma unroll
-      for (int r = 0; r < 16; ++r) {
-        float p = __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
-        g[r] = scale * p * (dp_[r] - my_delta);
+      for (int mt = 0; mt < 2; ++mt) {
+        const unsigned short* ak = &kt_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        bf16x8 a0 = *reinterpret_cast<const bf16x8*>(ak);
+        bf16x8 a1 = *reinterpret_cast<const bf16x8*>(ak + 16);
+        dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, gf0, dqa[mt], 0, 0, 0);
+        dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, gf1, dqa[mt], 0, 0, 0);
       }
     }
-    bf16x8 gf0 = relayout8(g);
-    bf16x8 gf1 = relayout8(g + 8);
+    if (have_next) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[0], qf[0], zc, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[0], dof[0], zc, 0, 0, 0);
 #pragma unroll
-    for (int mt = 0; mt < 2; ++mt) {
-      const unsigned short* ak = &kt_lds[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
-      bf16x8 a0 = *reinterpret_cast<const bf16x8*>(ak);
-      bf16x8 a1 = *reinterpret_cast<const bf16x8*>(ak + 16);
-      dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, gf0, dqa[mt], 0, 0, 0);
-      dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, gf1, dqa[mt], 0, 0, 0);
+      for (int t = 1; t < 4; ++t) {
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[t], qf[t], s, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[t], dof[t], dp_, 0, 0, 0);
+      }
     }
+    __syncthreads();
+    cur ^= 1;
   }
 
   if (myq < T) {
