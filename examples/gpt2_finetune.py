@@ -39,7 +39,8 @@ def build_loaders(cfg, pg):
             None,
         )
     vocab = cfg.get("model_config", {}).get("vocab_size", 50257)
-    train = SyntheticCLM(n=512, seq_len=seq, vocab_size=vocab, seed=pg.dp_rank)
+    n_train = int(cfg.get("n_train", 512))
+    train = SyntheticCLM(n=n_train, seq_len=seq, vocab_size=vocab, seed=pg.dp_rank)
     val = SyntheticCLM(n=64, seq_len=seq, vocab_size=vocab, seed=1000)
     return (
         DataLoader(train, batch_size=bs, shuffle=False),

@@ -23,9 +23,18 @@ __all__ = ["Trainer"]
 
 
 def _unwrap(model):
+    """Peel DataParallel(.module) and PipelineParallelWrapper(.local_module)
+    down to the stage/model that owns the canonical state-dict names."""
     m = model
-    while hasattr(m, "module"):
-        m = m.module
+    for _ in range(4):
+        if hasattr(m, "module"):
+            m = m.module
+        elif hasattr(m, "local_module") and not isinstance(
+            m.local_module, torch.nn.Sequential
+        ):
+            m = m.local_module
+        else:
+            break
     return m
 
 

@@ -134,3 +134,31 @@ def test_resume_roundtrip(tmp_path):
         o.zero_grad()
     for p1, p2 in zip(stage.parameters(), stage2.parameters()):
         assert torch.allclose(p1, p2, atol=1e-6)
+
+
+def test_unwrap_peels_pipeline_wrapper():
+    """Trainer._save_checkpoint must save stage-canonical keys even when the
+    stage is nested in PipelineParallelWrapper (regression: shards were saved
+    with a ``local_module.`` prefix, which the merge CLI silently passed
+    through unmapped)."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.trainer import _unwrap
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=2, n_head=2)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+
+    class FakePPWrapper(torch.nn.Module):
+        def __init__(self, m):
+            super().__init__()
+            self.local_module = m
+
+    class FakeDP(torch.nn.Module):
+        def __init__(self, m):
+            super().__init__()
+            self.module = m
+
+    wrapped = FakeDP(FakePPWrapper(stage))
+    inner = _unwrap(wrapped)
+    assert inner is stage
+    assert all(not k.startswith("local_module.") for k in inner.state_dict())
+    assert any(k.startswith("embedding.wte") for k in inner.state_dict())
