@@ -22,6 +22,9 @@ template <typename T>
 void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
 template <typename T>
 void colsum_launch(const T*, float*, long long, int, hipStream_t);
+long long colsum_bf16_chunks(long long rows, int N);
+bool colsum_bf16_launch(const unsigned short*, float*, unsigned short*,
+                        long long, int, hipStream_t);
 template <typename T>
 void dropout_fwd_launch(const T*, T*, unsigned char*, long long, float, unsigned, hipStream_t);
 template <typename T>
@@ -339,6 +342,14 @@ torch::Tensor colsum(torch::Tensor x) {
   auto xc = x.contiguous();
   int64_t N = xc.size(-1);
   long long rows = xc.numel() / N;
+  if (xc.dtype() == torch::kBFloat16 && (N & 7) == 0) {
+    long long chunks = colsum_bf16_chunks(rows, (int)N);
+    auto part = torch::empty({chunks, N}, xc.options().dtype(torch::kFloat32));
+    auto out = torch::empty({N}, xc.options());
+    if (colsum_bf16_launch(bf16p(xc), part.data_ptr<float>(), bf16p_mut(out),
+                           rows, (int)N, cur_stream()))
+      return out;
+  }
   auto out = torch::zeros({N}, xc.options().dtype(torch::kFloat32));
   if (xc.dtype() == torch::kBFloat16)
     colsum_launch<unsigned short>(bf16p(xc), out.data_ptr<float>(), rows, (int)N, cur_stream());
