@@ -263,6 +263,223 @@ __global__ __launch_bounds__(WR * WC * 64) void gemm_nt_kernel(
   }
 }
 
+// ===========================================================================
+// 256x256 8-phase deep-pipelined kernel (guide §"The 256² 8-phase template"):
+// 8 waves (2M x 4N), per-wave C = 128x64 (128 acc VGPRs); the K-step (BK=64)
+// is split into 4 quadrant phases of 16 MFMAs each; fragment ds_reads run one
+// phase ahead and BOTH A quadrant-sets and B quadrant-sets stay register-
+// resident, so every LDS region is read exactly once per K-tile.  Staging is
+// glds-only (2 x global_load_lds_dwordx4 per wave per phase = one "half-tile"),
+// with RAW s_barrier + lgkmcnt(0) (never __syncthreads: its fence would drain
+// the in-flight glds queue, the ~20% stall of the 2-barrier structure) and a
+// counted s_waitcnt vmcnt(6) only at the two K-tile boundaries (3 half-tiles
+// stay in flight across barriers).  LDS = 2 buffers x (A+B) 256x64 = 128 KiB,
+// st_16x32 XOR-swizzled via the per-lane SOURCE address (the glds LDS dest is
+// lane-linear).  One block/CU, 512 threads, ~250 VGPRs.
+//
+// Derived half-tile schedule (phase -> {MFMA quadrant | prep reads | stage}):
+//   p0: Q00(T) | BO(T)->B1        | AE(T+2)        Q00=(QM0,QN0) uses A0,B0
+//   p1: Q01(T) | AO(T)->A1        | BE(T+2)        Q01 uses A0,B1
+//   p2: Q10(T) | -                | BO(T+2)  [vmcnt(6) before trailing barrier]
+//   p3: Q11(T) | AE,BE(T+1)->A0,B0| AO(T+2)
+//   p4..p7: same over tile T+1, staging T+3.
+// Landing/overwrite safety for every phase was checked against the retire
+// points (each region's single read) — see the schedule table in the commit.
+// ===========================================================================
+
+#define P8_LDSEL (2 * 512 * 64)  // elements: 2 buf x (256 A + 256 B) x 64
+
+// per-lane source column group for a glds whose 64 lanes fill one 8-row x
+// 128B LDS subtile: st_16x32 swizzle = flip granule bit1 when row bit2 set
+// (row bit2 == lane bit5 within a subtile)
+__device__ __forceinline__ int p8_src_koct(int lane) {
+  return (lane & 7) ^ (((lane >> 5) & 1) << 1);
+}
+
+template <int ACT, bool SAVE_PRE>
+__global__ __launch_bounds__(512, 1) void gemm_nt_8p_kernel(
+    const unsigned short* __restrict__ A,  // [M,K]
+    const unsigned short* __restrict__ B,  // [N,K]
+    const unsigned short* __restrict__ bias,
+    unsigned short* __restrict__ C, unsigned short* __restrict__ Cpre,
+    int M, int N, int K) {
+  __shared__ unsigned short smem[P8_LDSEL];
+
+  const int nbm = M >> 8, nbn = N >> 8;
+  const int nwg = nbm * nbn;
+  int bid = blockIdx.x;
+  if (nwg >= 16) {
+    int qq = nwg / 8, rr = nwg % 8;
+    int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < rr ? xcd * (qq + 1) : rr * (qq + 1) + (xcd - rr) * qq) + idx;
+  }
+  const int m0 = (bid / nbn) << 8;
+  const int n0 = (bid % nbn) << 8;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 2;  // 0..1 (M)
+  const int wc = wave & 3;   // 0..3 (N)
+
+  // ---- per-lane glds source pointers + wave-uniform LDS dests -------------
+  // half-tiles: 0=AE (A rows {0-63,128-191}), 1=BE (B rows {64wc..+32}),
+  //             2=BO (BE+32), 3=AO (AE+64); each = 16 subtiles of 8 rows,
+  //             wave w stages subtiles 2w, 2w+1.
+  // wave w stages subtiles 2w and 2w+1 of each half; the pair is always 8
+  // rows apart (pairs never straddle a stripe boundary), so only the first
+  // subtile's pointer is kept (+8*K / +512 for the second) — 8 fewer VGPRs,
+  // which is what keeps this kernel spill-free at the 256-VGPR cap.
+  const unsigned short* gsrc[4];
+  int ldst[4];
+  const int skoct = p8_src_koct(lane);
+  const int srow = lane >> 3;
+  {
+    int st = 2 * wave;
+    int ra = ((st & 8) << 4) + ((st & 7) << 3);        // AE subtile row0
+    int rb = ((st >> 2) << 6) + ((st & 3) << 3);       // BE subtile row0
+    gsrc[0] = A + (long long)(m0 + ra + srow) * K + skoct * 8;
+    gsrc[3] = A + (long long)(m0 + ra + 64 + srow) * K + skoct * 8;
+    gsrc[1] = B + (long long)(n0 + rb + srow) * K + skoct * 8;
+    gsrc[2] = B + (long long)(n0 + rb + 32 + srow) * K + skoct * 8;
+    ldst[0] = ra * 64;
+    ldst[3] = (ra + 64) * 64;
+    ldst[1] = (256 + rb) * 64;
+    ldst[2] = (256 + rb + 32) * 64;
+  }
+  const long long row8 = (long long)8 * K;
+
+#define P8_STAGE(h, buf, kel)                                                  \
+  do {                                                                         \
+    glds16(gsrc[h] + (kel), smem + (buf) * (512 * 64) + ldst[h]);              \
+    glds16(gsrc[h] + row8 + (kel), smem + (buf) * (512 * 64) + ldst[h] + 512); \
+  } while (0)
+
+  // ---- fragment LDS offsets (elements), ks=1 toggles +32 ------------------
+  const int frow = lane & 15;
+  const int kq = lane >> 4;
+  int aoffE[4], boffE[2];
+#pragma unroll
+  for (int f = 0; f < 4; ++f) {
+    int row = wr * 128 + f * 16 + frow;
+    aoffE[f] = row * 64 + ((kq ^ (((row >> 2) & 1) << 1)) << 3);
+  }
+#pragma unroll
+  for (int f = 0; f < 2; ++f) {
+    int row = wc * 64 + f * 16 + frow;
+    boffE[f] = (256 + row) * 64 + ((kq ^ (((row >> 2) & 1) << 1)) << 3);
+  }
+
+  bf16x8 Ar[2][8];  // [QM set][f*2+ks]
+  bf16x8 Br[2][4];  // [QN set][f*2+ks]
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // extra: AO rows are +64 (+4096 el), BO rows +32 (+2048 el); row bit2 —
+  // and so the swizzle XOR — is unchanged by either shift.
+#define P8_READ_A(set, buf, extra)                                            \
+  _Pragma("unroll") for (int f = 0; f < 4; ++f) _Pragma("unroll")             \
+      for (int ks = 0; ks < 2; ++ks)                                          \
+          Ar[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
+              &smem[(buf) * (512 * 64) + ((aoffE[f] + (extra)) ^ (ks << 5))]);
+#define P8_READ_B(set, buf, extra)                                            \
+  _Pragma("unroll") for (int f = 0; f < 2; ++f) _Pragma("unroll")             \
+      for (int ks = 0; ks < 2; ++ks)                                          \
+          Br[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
+              &smem[(buf) * (512 * 64) + ((boffE[f] + (extra)) ^ (ks << 5))]);
+
+#define P8_MFMA(QM, QN)                                                       \
+  _Pragma("unroll") for (int ks = 0; ks < 2; ++ks) _Pragma("unroll")          \
+      for (int i = 0; i < 4; ++i) _Pragma("unroll") for (int j = 0; j < 2;    \
+                                                         ++j)                 \
+          acc[(QM) * 4 + i][(QN) * 2 + j] =                                   \
+              __builtin_amdgcn_mfma_f32_16x16x32_bf16(                        \
+                  Ar[QM][i * 2 + ks], Br[QN][j * 2 + ks],                     \
+                  acc[(QM) * 4 + i][(QN) * 2 + j], 0, 0, 0);
+
+#define P8_BAR() __builtin_amdgcn_s_barrier()
+#define P8_LGKM0() asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory")
+#define P8_VM(n) asm volatile("s_waitcnt vmcnt(" #n ")" ::: "memory")
+
+// one phase: prep reads + one half-tile stage, raw barrier, drained LDS,
+// prioritized MFMA cluster; VMW expands to the counted vmcnt before the
+// trailing barrier on the two boundary phases (cross-wave: every wave waits
+// its own count BEFORE the barrier that precedes the dependent reads).
+#define P8_PHASE(PREP, STAGE, QM, QN, VMW)                                    \
+  PREP;                                                                       \
+  STAGE;                                                                      \
+  P8_BAR();                                                                   \
+  P8_LGKM0();                                                                 \
+  __builtin_amdgcn_s_setprio(1);                                              \
+  P8_MFMA(QM, QN);                                                            \
+  __builtin_amdgcn_s_setprio(0);                                              \
+  VMW;                                                                        \
+  P8_BAR();
+
+  // ---- prologue: stage tiles 0 (buf0) and 1 (buf1) ------------------------
+  P8_STAGE(0, 0, 0);
+  P8_STAGE(1, 0, 0);
+  P8_STAGE(2, 0, 0);
+  P8_STAGE(3, 0, 0);
+  P8_STAGE(0, 1, 64);
+  P8_STAGE(1, 1, 64);
+  P8_STAGE(2, 1, 64);
+  P8_STAGE(3, 1, 64);
+  P8_VM(8);  // tile 0 landed (per wave; barrier below makes it block-wide)
+  P8_BAR();
+  P8_READ_A(0, 0, 0);  // AE(0) -> A0
+  P8_READ_B(0, 0, 0);  // BE(0) -> B0
+
+  const int niter = K >> 7;  // K-tile pairs
+  int k2 = 128, k3 = 192;    // k offsets (elements) of tiles T+2, T+3
+  for (int it = 0; it < niter; ++it) {
+    P8_PHASE(P8_READ_B(1, 0, 2048), P8_STAGE(0, 0, k2), 0, 0, );  // p0
+    P8_PHASE(P8_READ_A(1, 0, 4096), P8_STAGE(1, 0, k2), 0, 1, );  // p1
+    P8_PHASE(, P8_STAGE(2, 0, k2), 1, 0, P8_VM(6));               // p2
+    P8_PHASE(P8_READ_A(0, 1, 0) P8_READ_B(0, 1, 0), P8_STAGE(3, 0, k2), 1, 1, );  // p3
+    P8_PHASE(P8_READ_B(1, 1, 2048), P8_STAGE(0, 1, k3), 0, 0, );  // p4
+    P8_PHASE(P8_READ_A(1, 1, 4096), P8_STAGE(1, 1, k3), 0, 1, );  // p5
+    P8_PHASE(, P8_STAGE(2, 1, k3), 1, 0, P8_VM(6));               // p6
+    P8_PHASE(P8_READ_A(0, 0, 0) P8_READ_B(0, 0, 0), P8_STAGE(3, 1, k3), 1, 1, );  // p7
+    k2 += 128;
+    if (k2 >= K) k2 -= K;  // tail stagings wrap to valid (unused) addresses
+    k3 += 128;
+    if (k3 >= K) k3 -= K;
+  }
+
+  // ---- epilogue: bias + activation, all-interior --------------------------
+  const int erow = (lane >> 4) * 4;
+  const int ecol = lane & 15;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int col = n0 + wc * 64 + j * 16 + ecol;
+    const float bv = (bias != nullptr) ? bf16_to_f32(bias[col]) : 0.f;
+#pragma unroll
+    for (int i = 0; i < 8; ++i) {
+      const long long base = (long long)(m0 + wr * 128 + i * 16 + erow) * N + col;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float v = acc[i][j][r] + bv;
+        const long long idx = base + (long long)r * N;
+        if constexpr (SAVE_PRE) Cpre[idx] = f32_to_bf16(v);
+        if constexpr (ACT == QN_ACT_GELU) v = gelu_tanh(v);
+        else if constexpr (ACT == QN_ACT_RELU) v = fmaxf(v, 0.f);
+        C[idx] = f32_to_bf16(v);
+      }
+    }
+  }
+#undef P8_PHASE
+#undef P8_MFMA
+#undef P8_READ_A
+#undef P8_READ_B
+#undef P8_STAGE
+#undef P8_BAR
+#undef P8_LGKM0
+#undef P8_VM
+}
+
 #include <cstdlib>
 
 void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
@@ -278,6 +495,32 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
   if (big_tile < 0) {
     const char* e = getenv("QN_GEMM_BIG");
     big_tile = (e && e[0] == '1') ? 1 : 0;  // 128x128 default (A/B: big tile is a wash at K=768)
+  }
+  static int use_8p = -1;
+  if (use_8p < 0) {
+    const char* e = getenv("QN_GEMM_8P");
+    use_8p = (e && e[0] == '0') ? 0 : 1;  // default: 8-phase for eligible shapes
+  }
+  // 8-phase wins only where its 1-block/CU deep pipeline can fill the chip
+  // and amortize the 2-tile prologue: ≥256 workgroups and deep K (measured:
+  // +22% vs the 128² kernel at 8192³, +5% at 4096³, −20..−40% on the skinny
+  // K=768 / N=768 GPT-2 shapes — see profiles/README.md r04).
+  const bool p8_shape = (M % 256 == 0) && (N % 256 == 0) && (K % 128 == 0) &&
+                        ((long long)(M >> 8) * (N >> 8) >= 256) && (K >= 2048);
+  if (use_8p && p8_shape) {
+    dim3 grid((M >> 8) * (N >> 8)), blk(512);
+#define QN_P8_LAUNCH(A_, S_)                                                   \
+  hipLaunchKernelGGL((gemm_nt_8p_kernel<A_, S_>), grid, blk, 0, stream, A, B,  \
+                     bias, C, Cpre, M, N, K)
+    if (act == QN_ACT_GELU) {
+      if (Cpre) QN_P8_LAUNCH(QN_ACT_GELU, true); else QN_P8_LAUNCH(QN_ACT_GELU, false);
+    } else if (act == QN_ACT_RELU) {
+      if (Cpre) QN_P8_LAUNCH(QN_ACT_RELU, true); else QN_P8_LAUNCH(QN_ACT_RELU, false);
+    } else {
+      QN_P8_LAUNCH(QN_ACT_NONE, false);
+    }
+#undef QN_P8_LAUNCH
+    return;
   }
   // 256x128 (8 waves) when the problem fills the chip with it; else 128x128
   const bool big = big_tile && (M % 256 == 0) && ((long long)(M / 256) * ((N + 127) / 128) >= 256);

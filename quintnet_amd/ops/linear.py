@@ -32,6 +32,26 @@ def _gelu_tanh(x: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.gelu(x, approximate="tanh")
 
 
+import os
+
+# Forward dispatch policy (A/B-able, profiles/README.md r04): hipBLASLt via
+# F.linear wins the GPT-2-sized forwards outright (770-997 TF vs 595-713
+# custom; even GELU-fused, library + a separate act pass measured 58.4 vs
+# 59.6 ms/step) — the MI355X split is "vendor library for plain GEMMs".  The
+# hand-written 8-phase 256² kernel wins once the shape fills the chip at one
+# block/CU with deep K (+22% vs library-class at 8192³), so "auto" routes
+# activation-fused linears of that size to the fused custom epilogue and
+# everything else to the library.  "custom"/"library" force one side.
+_FWD_MODE = os.environ.get("QN_GEMM_FWD", "auto")
+
+
+def _big_fused_shape(m: int, n: int, k: int) -> bool:
+    return (
+        m % 256 == 0 and n % 256 == 0 and k % 128 == 0
+        and (m // 256) * (n // 256) >= 256 and k >= 2048
+    )
+
+
 def _native_ok(x: torch.Tensor, weight: torch.Tensor) -> bool:
     # The MFMA kernel covers bf16 with K a multiple of 8 (staging width);
     # anything else takes the library GEMM path.
@@ -52,11 +72,12 @@ class LinearFunction(torch.autograd.Function):
         ctx.x_shape = x.shape
         ctx.act = act
         pre_act = None
-        if (
-            _backend.use_native(x)
-            and _native_ok(x, weight)
-            and not (prefer_library and bias is None and act == _ACT_NONE)
-        ):
+        want_native = (
+            _FWD_MODE == "custom"
+            or (_FWD_MODE == "auto" and act != _ACT_NONE
+                and _big_fused_shape(x2d.shape[0], weight.shape[0], weight.shape[1]))
+        ) and not prefer_library
+        if want_native and _backend.use_native(x) and _native_ok(x, weight):
             res = _backend.ext().gemm_nt(
                 x2d.contiguous(), weight.contiguous(),
                 bias if bias is not None else None, act,
