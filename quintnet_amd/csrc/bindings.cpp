@@ -23,6 +23,9 @@ void act_bwd_launch(const T*, const T*, T*, long long, int, hipStream_t);
 template <typename T>
 void colsum_launch(const T*, float*, long long, int, hipStream_t);
 long long colsum_bf16_chunks(long long rows, int N);
+int wgrad_tn_splits(int M, int N, int K);
+void wgrad_tn_launch(const unsigned short*, const unsigned short*, float*,
+                     unsigned short*, int, int, int, int, hipStream_t);
 bool colsum_bf16_launch(const unsigned short*, float*, unsigned short*,
                         long long, int, hipStream_t);
 template <typename T>
@@ -337,6 +340,24 @@ torch::Tensor dropout_bwd(torch::Tensor dy, torch::Tensor mask, double p) {
   return dx;
 }
 
+torch::Tensor wgrad_tn(torch::Tensor dy, torch::Tensor x) {
+  CHECK_GPU(dy); CHECK_GPU(x);
+  TORCH_CHECK(dy.dtype() == torch::kBFloat16 && x.dtype() == torch::kBFloat16,
+              "wgrad_tn: bf16 only");
+  auto dyc = dy.contiguous(); auto xc = x.contiguous();
+  TORCH_CHECK(dyc.dim() == 2 && xc.dim() == 2 && dyc.size(0) == xc.size(0),
+              "wgrad_tn: [M,N],[M,K] expected");
+  int64_t M = dyc.size(0), N = dyc.size(1), K = xc.size(1);
+  TORCH_CHECK(N % 128 == 0 && K % 128 == 0 && M % 64 == 0,
+              "wgrad_tn: shape not eligible");
+  int splits = wgrad_tn_splits((int)M, (int)N, (int)K);
+  auto slab = torch::empty({splits, N, K}, dyc.options().dtype(torch::kFloat32));
+  auto out = torch::empty({N, K}, dyc.options());
+  wgrad_tn_launch(bf16p(dyc), bf16p(xc), slab.data_ptr<float>(), bf16p_mut(out),
+                  (int)M, (int)N, (int)K, splits, cur_stream());
+  return out;
+}
+
 torch::Tensor colsum(torch::Tensor x) {
   CHECK_GPU(x);
   auto xc = x.contiguous();
@@ -493,6 +514,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_bwd", &cross_entropy_bwd);
   m.def("act_bwd", &act_bwd, "fused activation backward");
   m.def("colsum", &colsum, "column sum (bias grad)");
+  m.def("wgrad_tn", &wgrad_tn, "split-K TN weight gradient (tr-read MFMA)");
   m.def("dropout_fwd", &dropout_fwd);
   m.def("embedding_pair_fwd", &embedding_pair_fwd);
   m.def("embedding_pair_bwd", &embedding_pair_bwd);

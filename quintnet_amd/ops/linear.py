@@ -43,6 +43,7 @@ import os
 # activation-fused linears of that size to the fused custom epilogue and
 # everything else to the library.  "custom"/"library" force one side.
 _FWD_MODE = os.environ.get("QN_GEMM_FWD", "auto")
+_WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
 
 
 def _big_fused_shape(m: int, n: int, k: int) -> bool:
@@ -119,7 +120,21 @@ class LinearFunction(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             grad_x = (g @ weight).reshape(ctx.x_shape)
         if ctx.needs_input_grad[1]:
-            grad_w = g.t() @ x2d
+            # hand-written split-K TN kernel: hipBLASLt runs these deep-
+            # contraction small-output shapes at 208-513 TF (profiles r04)
+            if (
+                _backend.use_native(g)
+                and _backend.has_ext()
+                and g.dtype == torch.bfloat16
+                and g.shape[1] % 128 == 0
+                and x2d.shape[1] % 128 == 0
+                and g.shape[0] % 64 == 0
+                and (g.shape[1] // 128) * (x2d.shape[1] // 128) < 128
+                and _WGRAD_MODE != "library"
+            ):
+                grad_w = _backend.ext().wgrad_tn(g.contiguous(), x2d.contiguous())
+            else:
+                grad_w = g.t() @ x2d
         if ctx.has_bias and ctx.needs_input_grad[2]:
             if _backend.use_native(g) and _backend.has_ext():
                 grad_b = _backend.ext().colsum(g)

@@ -27,6 +27,7 @@ sources = [
     os.path.join(CSRC, "attn.hip"),
     os.path.join(CSRC, "embedding.hip"),
     os.path.join(CSRC, "gemm.hip"),
+    os.path.join(CSRC, "wgrad.hip"),
 ]
 
 setup(
