@@ -30,6 +30,8 @@ __all__ = [
     "ReduceScatter",
     "pipeline_communicate",
     "bidirectional_pipeline_communicate",
+    "ring_send",
+    "ring_recv",
     "send_tensor",
     "recv_tensor",
 ]
@@ -341,3 +343,43 @@ def bidirectional_pipeline_communicate(
     for r in reqs:
         r.wait()
     return recv_buf
+
+
+# ---------------------------------------------------------------------------
+# Ring P2P for the interleaved-1F1B schedule: peers wrap around (rank p-1
+# sends chunk boundaries forward to rank 0).  Sends are NON-blocking (the
+# caller keeps the tensor alive and waits the returned reqs at step end) —
+# with sends posted before the next blocking recv, the schedule is
+# deadlock-free: the topologically-earliest unsatisfied recv's producer has
+# already posted its isend (program order: recv -> compute -> isend).
+# ---------------------------------------------------------------------------
+def ring_send(
+    tensor: torch.Tensor,
+    pp_rank: int,
+    pp_size: int,
+    group_ranks: List[int],
+    delta: int,
+    group=None,
+):
+    dst = group_ranks[(pp_rank + delta) % pp_size]
+    op = dist.P2POp(dist.isend, tensor.contiguous(), peer=dst, group=group)
+    return dist.batch_isend_irecv([op])
+
+
+def ring_recv(
+    pp_rank: int,
+    pp_size: int,
+    group_ranks: List[int],
+    delta: int,
+    shapes,
+    dtype,
+    device,
+    group=None,
+    requires_grad: bool = False,
+) -> torch.Tensor:
+    src = group_ranks[(pp_rank + delta) % pp_size]
+    buf = torch.empty(shapes, dtype=dtype, device=device, requires_grad=requires_grad)
+    reqs = dist.batch_isend_irecv([dist.P2POp(dist.irecv, buf, peer=src, group=group)])
+    for r in reqs:
+        r.wait()
+    return buf

@@ -112,6 +112,27 @@ class MeshGenerator:
             self.groups[name] = my_group
             self.group_ranks[name] = my_ranks
 
+        # Dedicated duplicate pp communicators for the interleaved-1F1B
+        # schedule: its ring P2P wraps around (rank p-1 -> 0 between model
+        # chunks) and at pp=2 the forward and backward directions share a
+        # rank pair, so each direction gets its OWN communicator to keep
+        # per-pair FIFO message matching independent per direction.
+        if "pp" in self.mesh_dim_names:
+            dim = self.mesh_dim_names.index("pp")
+            size = self.mesh.shape[dim]
+            rows = self.mesh.swapdims(-1, dim).reshape(-1, size)
+            for tag in ("pp_fwd", "pp_bwd"):
+                my_group = None
+                my_ranks = None
+                for row in rows:
+                    ranks = row.tolist()
+                    grp = dist.new_group(ranks=ranks, timeout=self.timeout)
+                    if self.rank in ranks:
+                        my_group = grp
+                        my_ranks = ranks
+                self.groups[tag] = my_group
+                self.group_ranks[tag] = my_ranks
+
         # Tied-embedding subgroup: first+last rank of each pp row (used
         # for GPT-2 wte/lm_head grad sync; restricting the collective to
         # the two stages that own the weight avoids the pp>2 deadlock of

@@ -8,6 +8,7 @@ from .tensor_parallel import (
     ensure_divisibility,
 )
 from .pipeline import (
+    InterleavedPipelineWrapper,
     PipelineParallelWrapper,
     PipelineSchedule,
     AllFwdAllBwdSchedule,
@@ -21,6 +22,7 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "InterleavedPipelineWrapper",
     "DataParallel",
     "BucketConfig",
     "DistributedConfig",

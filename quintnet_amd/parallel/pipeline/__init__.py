@@ -1,5 +1,6 @@
-from .wrapper import PipelineParallelWrapper, distribute_layers
+from .wrapper import InterleavedPipelineWrapper, PipelineParallelWrapper, distribute_layers
 from .schedule import (
+    InterleavedOneFOneBSchedule,
     PipelineSchedule,
     AllFwdAllBwdSchedule,
     OneFOneBSchedule,
@@ -9,6 +10,8 @@ from .trainer import PipelineTrainer
 from .dataloader import PipelineDataLoader
 
 __all__ = [
+    "InterleavedPipelineWrapper",
+    "InterleavedOneFOneBSchedule",
     "PipelineParallelWrapper",
     "distribute_layers",
     "PipelineSchedule",

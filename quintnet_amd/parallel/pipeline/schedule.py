@@ -25,7 +25,13 @@ from ...core.comm import (
 )
 from ...ops import causal_lm_loss, clip_grad_norm_local, shift_labels
 
-__all__ = ["PipelineSchedule", "AllFwdAllBwdSchedule", "OneFOneBSchedule", "get_schedule"]
+__all__ = [
+    "PipelineSchedule",
+    "AllFwdAllBwdSchedule",
+    "OneFOneBSchedule",
+    "InterleavedOneFOneBSchedule",
+    "get_schedule",
+]
 
 
 class PipelineSchedule(abc.ABC):
@@ -328,10 +334,150 @@ class OneFOneBSchedule(PipelineSchedule):
         return self._final_metrics(metrics, num_micro)
 
 
+class InterleavedOneFOneBSchedule(PipelineSchedule):
+    """Interleaved (virtual-pipeline) 1F1B — Megatron-style, beyond
+    reference parity.
+
+    Each rank holds ``v`` model chunks (``InterleavedPipelineWrapper``);
+    global stage ``c*p + r`` lives on rank r, so every inter-stage edge is
+    rank r -> r+1 with a ring wrap between chunks.  Bubble fraction drops
+    from (p-1)/(m+p-1) to ~(p-1)/(v*m).  Template (rank-independent):
+    forward step k runs chunk ``(k % (p*v)) // p`` on micro-batch
+    ``(k // (p*v))*p + k % p``; backward mirrors with chunks reversed.
+    Identical templates on every rank make the per-edge message order of
+    producer and consumer provably equal, so FIFO P2P matching is safe;
+    the forward and backward directions use DEDICATED duplicate
+    communicators (mesh "pp_fwd"/"pp_bwd") because at p=2 both directions
+    share a rank pair.
+
+    Requires ``num_micro % p == 0``.  DDP bucket overlap is disabled
+    during the step (per-param hooks would over-count across the out-of-
+    order chunk backwards and fire on partial sums); the full flat-buffer
+    reduction happens in ``finalize_gradients``.
+    """
+
+    def train_step(self, data_loader, tensor_shapes, device, dtype) -> Dict[str, float]:
+        from collections import deque
+
+        from ...core.comm import ring_recv, ring_send
+
+        owner = self._chunks_owner()
+        v = int(owner.num_chunks)
+        p, r = self.pp_size, self.pp_rank
+        num_micro = data_loader.grad_acc_steps
+        if p < 2:
+            raise ValueError("interleaved schedule needs pp_size >= 2")
+        if num_micro % p != 0:
+            raise ValueError(
+                f"interleaved 1F1B needs grad_acc_steps ({num_micro}) divisible "
+                f"by pp_size ({p})"
+            )
+        fwd_group = getattr(self.trainer, "pp_fwd_group", None) or self.pp_group
+        bwd_group = getattr(self.trainer, "pp_bwd_group", None) or self.pp_group
+
+        pv = p * v
+        total = num_micro * v
+        warmup = min((p - r - 1) * 2 + (v - 1) * p, total)
+        last_g = pv - 1
+
+        metrics: Dict[str, float] = {}
+        inputs = [deque() for _ in range(v)]
+        outputs = [deque() for _ in range(v)]
+        batch_cache: Dict[int, Any] = {}
+        send_reqs: List[Any] = []
+        send_keep: List[torch.Tensor] = []
+        self._set_grad_sync(False)
+
+        def forward(k: int) -> None:
+            c = (k % pv) // p
+            mu = (k // pv) * p + (k % p)
+            g = c * p + r
+            if c == 0:
+                # every rank consumes the loader once per micro-batch, in
+                # order, to stay lockstep; the batch is cached for the
+                # label consumer (last global stage)
+                batch_cache[mu] = next(data_loader)
+                if r != 0 and r != p - 1:
+                    batch = batch_cache.pop(mu)
+            if g == 0:
+                x_data = self._stage_input(batch_cache[mu], device)
+                if r != p - 1:
+                    batch_cache.pop(mu)
+                x_in = None
+                out = owner.forward(x_data, chunk_id=c)
+            else:
+                x_in = ring_recv(
+                    r, p, self.group_ranks, -1, tensor_shapes, dtype, device,
+                    group=fwd_group, requires_grad=True,
+                )
+                out = owner.forward(x_in, chunk_id=c)
+            if g == last_g:
+                loss = self._loss_and_metrics(
+                    out, batch_cache.pop(mu), device, metrics
+                ) / num_micro
+                inputs[c].append(x_in)
+                outputs[c].append(loss)
+            else:
+                send_reqs.extend(
+                    ring_send(out, r, p, self.group_ranks, +1, group=fwd_group)
+                )
+                inputs[c].append(x_in)
+                outputs[c].append(out)
+
+        def backward(k: int) -> None:
+            c = v - 1 - (k % pv) // p
+            g = c * p + r
+            b_in = inputs[c].popleft()
+            b_out = outputs[c].popleft()
+            if g == last_g:
+                grad = None
+            else:
+                grad = ring_recv(
+                    r, p, self.group_ranks, +1, tensor_shapes, dtype, device,
+                    group=bwd_group,
+                )
+            in_grad = self.model.backward(b_in, b_out, grad)
+            if g != 0 and in_grad is not None:
+                send_keep.append(in_grad)
+                send_reqs.extend(
+                    ring_send(in_grad, r, p, self.group_ranks, -1, group=bwd_group)
+                )
+
+        for k in range(warmup):
+            forward(k)
+        for i in range(total - warmup):
+            forward(warmup + i)
+            backward(i)
+        for i in range(total - warmup, total):
+            backward(i)
+        for rq in send_reqs:
+            rq.wait()
+
+        self._set_grad_sync(True)
+        self._optimizer_step()
+        return self._final_metrics(metrics, num_micro)
+
+    def _chunks_owner(self):
+        m = self.model
+        for _ in range(3):
+            if hasattr(m, "num_chunks"):
+                return m
+            if hasattr(m, "module"):
+                m = m.module
+            else:
+                break
+        raise ValueError(
+            "interleaved schedule requires the model to be (or wrap) an "
+            "InterleavedPipelineWrapper"
+        )
+
+
 def get_schedule(name: str, trainer, task_type: str) -> PipelineSchedule:
     name = (name or "1f1b").lower()
     if name in ("1f1b", "one_f_one_b", "onefoneb"):
         return OneFOneBSchedule(trainer, task_type)
     if name in ("afab", "all_forward_all_backward", "gpipe"):
         return AllFwdAllBwdSchedule(trainer, task_type)
+    if name in ("interleaved", "interleaved_1f1b", "vpp"):
+        return InterleavedOneFOneBSchedule(trainer, task_type)
     raise ValueError(f"unknown schedule {name!r}")

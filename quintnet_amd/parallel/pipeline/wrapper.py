@@ -16,7 +16,7 @@ from typing import List, Optional
 import torch
 import torch.nn as nn
 
-__all__ = ["PipelineParallelWrapper", "distribute_layers"]
+__all__ = ["PipelineParallelWrapper", "InterleavedPipelineWrapper", "distribute_layers"]
 
 
 def distribute_layers(depth: int, pp_size: int) -> List[List[int]]:
@@ -103,3 +103,64 @@ class PipelineParallelWrapper(nn.Module):
             input_tensor.retain_grad()
         torch.autograd.backward(output_tensor, grad_tensors=output_tensor_grad)
         return input_tensor.grad if input_tensor is not None else None
+
+
+class InterleavedPipelineWrapper(nn.Module):
+    """Virtual-pipeline stage holder for the interleaved-1F1B schedule.
+
+    The model's blocks are split into ``pp_size * num_chunks`` global
+    stages; this rank holds stages ``{c * pp_size + pp_rank}`` for chunk
+    c in [0, num_chunks) (Megatron-style assignment: inter-stage edges
+    are always rank r -> r+1 with a ring wrap between chunks).  Global
+    stage 0 gets the embedding, the last global stage the head.  Beyond
+    reference parity (the reference has no interleaved schedule).
+    """
+
+    def __init__(
+        self,
+        model: nn.Module,
+        pp_rank: int = 0,
+        pp_group=None,
+        pp_size: int = 1,
+        num_chunks: int = 2,
+        device: Optional[torch.device] = None,
+    ):
+        super().__init__()
+        if not hasattr(model, "blocks"):
+            raise ValueError("InterleavedPipelineWrapper needs a model with .blocks")
+        self.pp_rank = pp_rank
+        self.pp_size = pp_size
+        self.pp_group = pp_group
+        self.num_chunks = num_chunks
+        self.device = device
+        self.seq_len = getattr(model, "seq_len", None)
+        self.hidden_dim = getattr(model, "hidden_dim", None)
+
+        n_stages = pp_size * num_chunks
+        depth = len(model.blocks)
+        if depth < n_stages:
+            raise ValueError(
+                f"{depth} blocks cannot fill {n_stages} virtual stages"
+            )
+        self.layer_distribution = distribute_layers(depth, n_stages)
+        chunks: List[nn.Module] = []
+        for c in range(num_chunks):
+            g = c * pp_size + pp_rank
+            mods: List[nn.Module] = []
+            if g == 0 and hasattr(model, "embedding"):
+                mods.append(model.embedding)
+            mods.extend(model.blocks[i] for i in self.layer_distribution[g])
+            if g == n_stages - 1:
+                head = _find_head(model)
+                if head is not None:
+                    mods.append(head)
+            chunks.append(nn.Sequential(*mods))
+        self.chunks = nn.ModuleList(chunks)
+        if device is not None:
+            self.chunks.to(device)
+
+    def forward(self, x, chunk_id: int = 0):
+        return self.chunks[chunk_id](x)
+
+    # same manual-backward contract as PipelineParallelWrapper
+    backward = PipelineParallelWrapper.backward

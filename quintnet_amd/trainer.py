@@ -68,6 +68,7 @@ class Trainer:
         inner = _unwrap(self.model)
         seq, hidden = self._infer_seq_hidden(inner)
         self.tensor_shapes = (micro_b, seq, hidden)
+        groups = self.pg.get_all_groups()
         self.pipeline_trainer = PipelineTrainer(
             model=self.model,
             optimizer=self.optimizer,
@@ -79,6 +80,8 @@ class Trainer:
             schedule=self.config.get("schedule", "1f1b"),
             task_type=self.task_type,
             max_grad_norm=self.max_grad_norm,
+            pp_fwd_group=groups.get("pp_fwd"),
+            pp_bwd_group=groups.get("pp_bwd"),
         )
 
     def _infer_micro_batch(self) -> int:
