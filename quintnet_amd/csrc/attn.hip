@@ -137,7 +137,8 @@ __device__ __forceinline__ void stage_transpose(
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, unsigned short* __restrict__ out,
-    float* __restrict__ lse2, int T, int H, float scale, int causal,
+    float* __restrict__ lse2, int Tq, int Tk, int qoff, int H, float scale,
+    int causal,
     long long qsB, long long qsH, long long qsT,
     long long ksB, long long ksH, long long ksT,
     long long vsB, long long vsH, long long vsT,
@@ -170,7 +171,9 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   for (int i = 0; i < 16; ++i) zc[i] = 0.f;
 
   const float s2scale = scale * LOG2E;
-  const int kv_end = causal ? min(q0 + 128, T) : T;
+  // q rows are local [0,Tq); their GLOBAL positions are +qoff (context
+  // parallelism: K/V cover the full sequence [0,Tk), Q is this rank's shard)
+  const int kv_end = causal ? min(q0 + qoff + 128, Tk) : Tk;
 
   // prefetch pipeline: tile t's K fragments + V staging rows load during
   // tile t-1's MFMA cluster (T14) — the per-iteration global latency was
@@ -201,7 +204,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
     }
 
-    if (!causal || kv0 <= qw + 31) {  // wave has at least one valid pair
+    if (!causal || kv0 <= qw + qoff + 31) {  // wave has >= one valid pair
       // S^T[key][q] = sum_d K[key][d] Q[q][d]  (first mfma takes the
       // persistent zero C — no per-tile accumulator re-init)
       f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[0], qf[0], zc, 0, 0, 0);
@@ -210,7 +213,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_c[kt], qf[kt], s, 0, 0, 0);
       }
       // scale to base-2; causal mask only on the diagonal tile
-      const bool diag = causal && (kv0 + 31 > qw);
+      const bool diag = causal && (kv0 + 31 > qw + qoff);
       float ps[16];
       float pmax = -INFINITY;
       if (diag) {
@@ -218,7 +221,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         for (int r = 0; r < 16; ++r) {
           int key = kv0 + drow(r, lane);
           float x = s[r] * s2scale;
-          if (key > myq) x = -INFINITY;
+          if (key > myq + qoff) x = -INFINITY;
           ps[r] = x;
           pmax = fmaxf(pmax, x);
         }
@@ -268,7 +271,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   // epilogue: normalize and store out[q][d] (scattered: lane owns col q)
   float inv = (l > 0.f) ? 1.0f / l : 0.f;
-  if (myq < T) {
+  if (myq < Tq) {
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt) {
 #pragma unroll
@@ -277,7 +280,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
         op[(long long)myq * osT + d] = f32_to_bf16(o[mt][r] * inv);
       }
     }
-    if (lane < 32) lse2[(long long)bh * T + myq] = m2 + __log2f(l);
+    if (lane < 32) lse2[(long long)bh * Tq + myq] = m2 + __log2f(l);
   }
 }
 
@@ -323,7 +326,8 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse2, const float* __restrict__ delta,
-    unsigned short* __restrict__ dq, int T, int H, float scale, int causal,
+    unsigned short* __restrict__ dq, int Tq, int Tk, int qoff, int H,
+    float scale, int causal,
     long long qsB, long long qsH, long long qsT,
     long long ksB, long long ksH, long long ksT,
     long long vsB, long long vsH, long long vsT,
@@ -350,15 +354,15 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     qf[t] = frag_ld(qp, qsT, qw, t * 16, lane);
     dof[t] = frag_ld(dop, dsT, qw, t * 16, lane);
   }
-  const float my_lse = lse2[(long long)bh * T + min(myq, T - 1)];
-  const float my_delta = delta[(long long)bh * T + min(myq, T - 1)];
+  const float my_lse = lse2[(long long)bh * Tq + min(myq, Tq - 1)];
+  const float my_delta = delta[(long long)bh * Tq + min(myq, Tq - 1)];
   const float s2scale = scale * LOG2E;
 
   f32x16 dqa[2], zc;
 #pragma unroll
   for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
 
-  const int kv_end = causal ? min(q0 + 128, T) : T;
+  const int kv_end = causal ? min(q0 + qoff + 128, Tk) : Tk;
   // cross-tile software pipeline (like attn_bwd_dkv): S/dP MFMAs for
   // tile i+1 interleave with the dQ accumulation of tile i; ONE barrier
   // per tile; double-buffered Kt.
@@ -389,17 +393,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
   int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
     const bool have_next = kv0 + 32 < kv_end;
-    const bool active = !(causal && kv0 > qw + 31);
+    const bool active = !(causal && kv0 > qw + qoff + 31);
 
     bf16x8 gf0, gf1;
     if (active) {
-      const bool diag = causal && (kv0 + 31 > qw);
+      const bool diag = causal && (kv0 + 31 > qw + qoff);
       float g[16];
       if (diag) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int key = kv0 + drow(r, lane);
-          float p = (key > myq) ? 0.f : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+          float p = (key > myq + qoff) ? 0.f
+                                       : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
           g[r] = scale * p * (dp_[r] - my_delta);
         }
       } else {
@@ -449,7 +454,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
     cur ^= 1;
   }
 
-  if (myq < T) {
+  if (myq < Tq) {
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -472,7 +477,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse2, const float* __restrict__ delta,
     unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
-    int T, int H, float scale, int causal,
+    int Tq, int Tk, int qoff, int H, float scale, int causal,
     long long qsB, long long qsH, long long qsT,
     long long ksB, long long ksH, long long ksT,
     long long vsB, long long vsH, long long vsT,
@@ -514,14 +519,16 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
 #pragma unroll
   for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; zc[i] = 0.f; }
 
-  const int q_start = causal ? kv0b : 0;
+  // q loop is in LOCAL q space; keys at kv0b are causally visible to local
+  // q rows >= kv0b - qoff (qoff and kv0b are both multiples of 128)
+  const int q_start = causal ? max(kv0b - qoff, 0) : 0;
   // pointer-bumped sources
   const unsigned short* qfp = frag_base(qp, qsT, q_start, lane);
   const unsigned short* dofp = frag_base(dop, dsT, q_start, lane);
   const unsigned short* qsp = stage_base(qp, qsT, q_start);
   const unsigned short* dosp = stage_base(dop, dsT, q_start);
-  const float* lsep = lse2 + (long long)bh * T + q_start + threadIdx.x;
-  const float* delp = delta + (long long)bh * T + q_start + threadIdx.x;
+  const float* lsep = lse2 + (long long)bh * Tq + q_start + threadIdx.x;
+  const float* delp = delta + (long long)bh * Tq + q_start + threadIdx.x;
   const long long qstep = 32 * qsT, dstep = 32 * dsT;
 
   // ---- prologue: stage tile 0 into buf0, prime s/dp for tile 0 ------------
@@ -551,7 +558,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // prefetch tile 1's staging rows + lse/del
   s16x8 dost_n = {0, 0, 0, 0, 0, 0, 0, 0}, qst_n = dost_n;
   float lse_n = 0.f, del_n = 0.f;
-  if (q_start + 32 < T) {
+  if (q_start + 32 < Tq) {
     dost_n = stage_at(dosp + dstep);
     qst_n = stage_at(qsp + qstep);
     if (threadIdx.x < 32) {
@@ -562,20 +569,20 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   __syncthreads();  // buf0 visible
 
   int cur = 0;
-  for (int qt0 = q_start; qt0 < T; qt0 += 32) {
-    const bool have_next = qt0 + 32 < T;
-    const bool active = !(causal && qt0 + 31 < kw);
+  for (int qt0 = q_start; qt0 < Tq; qt0 += 32) {
+    const bool have_next = qt0 + 32 < Tq;
+    const bool active = !(causal && qt0 + qoff + 31 < kw);
 
     // softmax + relayout for tile i (s/dp computed last iteration)
     bf16x8 pf0, pf1, gf0, gf1;
     if (active) {
-      const bool diag = causal && (qt0 < kw + 31);
+      const bool diag = causal && (qt0 + qoff < kw + 31);
       float pv[16], gv[16];
       if (diag) {
 #pragma unroll
         for (int r = 0; r < 16; ++r) {
           int qrow = drow(r, lane);
-          float p = (mykey > qt0 + qrow)
+          float p = (mykey > qt0 + qoff + qrow)
                         ? 0.f
                         : __builtin_amdgcn_exp2f(s[r] * s2scale - lse_t[cur][qrow]);
           pv[r] = p;
@@ -606,7 +613,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
       // issue tile i+2 staging prefetch + tile i+1 fragment loads
       qfp += qstep; dofp += dstep; qsp += qstep; dosp += dstep;
       lsep += 32; delp += 32;
-      if (qt0 + 64 < T) {
+      if (qt0 + 64 < Tq) {
         dost_n = stage_at(dosp + dstep);
         qst_n = stage_at(qsp + qstep);
         if (threadIdx.x < 32) {
@@ -651,7 +658,7 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     cur ^= 1;
   }
 
-  if (mykey < T) {
+  if (mykey < Tk) {
 #pragma unroll
     for (int mt = 0; mt < 2; ++mt)
 #pragma unroll
@@ -670,12 +677,13 @@ struct AttnStrides {
 
 void attn_fwd_launch(const unsigned short* q, const unsigned short* k,
                      const unsigned short* v, unsigned short* out, float* lse2,
-                     int B, int H, int T, float scale, int causal,
-                     const AttnStrides& st, hipStream_t stream) {
-  dim3 grid(T / 128, B * H);
+                     int B, int H, int Tq, int Tk, int qoff, float scale,
+                     int causal, const AttnStrides& st, hipStream_t stream) {
+  dim3 grid(Tq / 128, B * H);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream, q, k, v, out,
-                     lse2, T, H, scale, causal, st.qB, st.qH, st.qT, st.kB,
-                     st.kH, st.kT, st.vB, st.vH, st.vT, st.oB, st.oH, st.oT);
+                     lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH, st.qT,
+                     st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB, st.oH,
+                     st.oT);
 }
 
 void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
@@ -692,13 +700,14 @@ void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
 void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
                         const unsigned short* v, const unsigned short* dout,
                         const float* lse2, const float* delta,
-                        unsigned short* dq, int B, int H, int T, float scale,
+                        unsigned short* dq, int B, int H, int Tq, int Tk,
+                        int qoff, float scale,
                         int causal, const AttnStrides& st,
                         long long dsB, long long dsH, long long dsT,
                         hipStream_t stream) {
-  dim3 grid(T / 128, B * H);
+  dim3 grid(Tq / 128, B * H);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream, q, k, v,
-                     dout, lse2, delta, dq, T, H, scale, causal, st.qB, st.qH,
+                     dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
                      st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, dsB, dsH,
                      dsT, st.oB, st.oH, st.oT);
 }
@@ -707,14 +716,15 @@ void attn_bwd_dkv_launch(const unsigned short* q, const unsigned short* k,
                          const unsigned short* v, const unsigned short* dout,
                          const float* lse2, const float* delta,
                          unsigned short* dk, unsigned short* dv, int B, int H,
-                         int T, float scale, int causal, const AttnStrides& st,
+                         int Tq, int Tk, int qoff, float scale, int causal,
+                         const AttnStrides& st,
                          long long dsB, long long dsH, long long dsT,
                          long long dkB, long long dkH, long long dkT,
                          long long dvB, long long dvH, long long dvT,
                          hipStream_t stream) {
-  dim3 grid(T / 128, B * H);
+  dim3 grid(Tk / 128, B * H);
   hipLaunchKernelGGL(attn_bwd_dkv_kernel, grid, dim3(256), 0, stream, q, k, v,
-                     dout, lse2, delta, dk, dv, T, H, scale, causal, st.qB,
+                     dout, lse2, delta, dk, dv, Tq, Tk, qoff, H, scale, causal, st.qB,
                      st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT,
                      dsB, dsH, dsT, dkB, dkH, dkT, dvB, dvH, dvT);
 }

@@ -46,18 +46,18 @@ struct AttnStrides {
   long long qB, qH, qT, kB, kH, kT, vB, vH, vT, oB, oH, oT;
 };
 void attn_fwd_launch(const unsigned short*, const unsigned short*, const unsigned short*,
-                     unsigned short*, float*, int, int, int, float, int,
-                     const AttnStrides&, hipStream_t);
+                     unsigned short*, float*, int, int, int, int, int, float,
+                     int, const AttnStrides&, hipStream_t);
 void attn_delta_launch(const unsigned short*, const unsigned short*, float*, int, int, int,
                        long long, long long, long long, long long, long long, long long,
                        hipStream_t);
 void attn_bwd_dq_launch(const unsigned short*, const unsigned short*, const unsigned short*,
                         const unsigned short*, const float*, const float*, unsigned short*,
-                        int, int, int, float, int, const AttnStrides&,
+                        int, int, int, int, int, float, int, const AttnStrides&,
                         long long, long long, long long, hipStream_t);
 void attn_bwd_dkv_launch(const unsigned short*, const unsigned short*, const unsigned short*,
                          const unsigned short*, const float*, const float*, unsigned short*,
-                         unsigned short*, int, int, int, float, int, const AttnStrides&,
+                         unsigned short*, int, int, int, int, int, float, int, const AttnStrides&,
                          long long, long long, long long, long long, long long, long long,
                          long long, long long, long long, hipStream_t);
 
@@ -461,42 +461,48 @@ unsigned short* ap_mut(torch::Tensor& t) {
 }  // namespace
 
 torch::Tensor attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                       torch::Tensor out, double scale, bool causal) {
-  int64_t B = q.size(0), H = q.size(1), T = q.size(2);
-  TORCH_CHECK(T % 128 == 0, "attn: T must be a multiple of 128");
-  check_attn_view(q, B, H, T); check_attn_view(k, B, H, T);
-  check_attn_view(v, B, H, T); check_attn_view(out, B, H, T);
-  auto lse2 = torch::empty({B * H, T}, q.options().dtype(torch::kFloat32));
+                       torch::Tensor out, double scale, bool causal,
+                       int64_t q_offset) {
+  int64_t B = q.size(0), H = q.size(1), Tq = q.size(2), Tk = k.size(2);
+  TORCH_CHECK(Tq % 128 == 0 && Tk % 128 == 0 && q_offset % 128 == 0,
+              "attn: Tq/Tk/q_offset must be multiples of 128");
+  TORCH_CHECK(q_offset >= 0 && q_offset + Tq <= Tk,
+              "attn: q rows [q_offset, q_offset+Tq) must lie within [0, Tk)");
+  check_attn_view(q, B, H, Tq); check_attn_view(k, B, H, Tk);
+  check_attn_view(v, B, H, Tk); check_attn_view(out, B, H, Tq);
+  auto lse2 = torch::empty({B * H, Tq}, q.options().dtype(torch::kFloat32));
   AttnStrides st{q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
                  k.stride(2), v.stride(0), v.stride(1), v.stride(2), out.stride(0),
                  out.stride(1), out.stride(2)};
   attn_fwd_launch(ap(q), ap(k), ap(v), ap_mut(out), lse2.data_ptr<float>(),
-                  (int)B, (int)H, (int)T, (float)scale, causal ? 1 : 0, st,
-                  cur_stream());
+                  (int)B, (int)H, (int)Tq, (int)Tk, (int)q_offset, (float)scale,
+                  causal ? 1 : 0, st, cur_stream());
   return lse2;
 }
 
 void attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
               torch::Tensor out, torch::Tensor dout, torch::Tensor lse2,
               torch::Tensor dq, torch::Tensor dk, torch::Tensor dv,
-              double scale, bool causal) {
-  int64_t B = q.size(0), H = q.size(1), T = q.size(2);
-  check_attn_view(dout, B, H, T); check_attn_view(dq, B, H, T);
-  check_attn_view(dk, B, H, T); check_attn_view(dv, B, H, T);
-  auto delta = torch::empty({B * H, T}, q.options().dtype(torch::kFloat32));
+              double scale, bool causal, int64_t q_offset) {
+  int64_t B = q.size(0), H = q.size(1), Tq = q.size(2), Tk = k.size(2);
+  check_attn_view(dout, B, H, Tq); check_attn_view(dq, B, H, Tq);
+  check_attn_view(dk, B, H, Tk); check_attn_view(dv, B, H, Tk);
+  auto delta = torch::empty({B * H, Tq}, q.options().dtype(torch::kFloat32));
   attn_delta_launch(ap(dout), ap(out), delta.data_ptr<float>(), (int)B, (int)H,
-                    (int)T, dout.stride(0), dout.stride(1), dout.stride(2),
+                    (int)Tq, dout.stride(0), dout.stride(1), dout.stride(2),
                     out.stride(0), out.stride(1), out.stride(2), cur_stream());
   AttnStrides st{q.stride(0), q.stride(1), q.stride(2), k.stride(0), k.stride(1),
                  k.stride(2), v.stride(0), v.stride(1), v.stride(2), dq.stride(0),
                  dq.stride(1), dq.stride(2)};
   attn_bwd_dq_launch(ap(q), ap(k), ap(v), ap(dout), lse2.data_ptr<float>(),
-                     delta.data_ptr<float>(), ap_mut(dq), (int)B, (int)H, (int)T,
+                     delta.data_ptr<float>(), ap_mut(dq), (int)B, (int)H,
+                     (int)Tq, (int)Tk, (int)q_offset,
                      (float)scale, causal ? 1 : 0, st, dout.stride(0),
                      dout.stride(1), dout.stride(2), cur_stream());
   attn_bwd_dkv_launch(ap(q), ap(k), ap(v), ap(dout), lse2.data_ptr<float>(),
                       delta.data_ptr<float>(), ap_mut(dk), ap_mut(dv), (int)B,
-                      (int)H, (int)T, (float)scale, causal ? 1 : 0, st,
+                      (int)H, (int)Tq, (int)Tk, (int)q_offset, (float)scale,
+                      causal ? 1 : 0, st,
                       dout.stride(0), dout.stride(1), dout.stride(2),
                       dk.stride(0), dk.stride(1), dk.stride(2), dv.stride(0),
                       dv.stride(1), dv.stride(2), cur_stream());
@@ -519,8 +525,14 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("embedding_pair_fwd", &embedding_pair_fwd);
   m.def("embedding_pair_bwd", &embedding_pair_bwd);
   m.def("dropout_bwd", &dropout_bwd);
-  m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)");
-  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)");
+  m.def("attn_fwd", &attn_fwd, "fused flash attention forward (D=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"),
+        py::arg("scale"), py::arg("causal"), py::arg("q_offset") = 0);
+  m.def("attn_bwd", &attn_bwd, "fused flash attention backward (D=64)",
+        py::arg("q"), py::arg("k"), py::arg("v"), py::arg("out"),
+        py::arg("dout"), py::arg("lse2"), py::arg("dq"), py::arg("dk"),
+        py::arg("dv"), py::arg("scale"), py::arg("causal"),
+        py::arg("q_offset") = 0);
   m.def("adamw_step", &adamw_step);
   m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
   m.attr("gfx") = "gfx950";

@@ -24,14 +24,25 @@ from . import _backend
 __all__ = ["attention", "AttentionFunction", "causal_softmax", "softmax_bwd"]
 
 
-def causal_softmax(scores: torch.Tensor, scale: float, causal: bool) -> torch.Tensor:
-    """P = softmax(scale * scores [+ causal mask]) along the last dim."""
-    if _backend.use_native(scores) and _backend.has_ext():
+def causal_softmax(
+    scores: torch.Tensor, scale: float, causal: bool, q_offset: int = 0
+) -> torch.Tensor:
+    """P = softmax(scale * scores [+ causal mask]) along the last dim.
+
+    ``q_offset`` shifts the causal diagonal: query row i is globally row
+    ``i + q_offset`` (context parallelism — K covers the full sequence).
+    """
+    if (
+        q_offset == 0
+        and scores.shape[-1] == scores.shape[-2]
+        and _backend.use_native(scores)
+        and _backend.has_ext()
+    ):
         return _backend.ext().softmax_fwd(scores.contiguous(), scale, causal)
     s = scores.float() * scale
     if causal:
         T, S = s.shape[-2], s.shape[-1]
-        mask = torch.ones(T, S, dtype=torch.bool, device=s.device).tril()
+        mask = torch.ones(T, S, dtype=torch.bool, device=s.device).tril(q_offset)
         s = s.masked_fill(~mask, float("-inf"))
     return torch.softmax(s, dim=-1).to(scores.dtype)
 
@@ -47,11 +58,11 @@ def softmax_bwd(p: torch.Tensor, dp: torch.Tensor, scale: float) -> torch.Tensor
 
 class AttentionFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, q, k, v, causal):
-        # q,k,v: [B, H, T, D]
+    def forward(ctx, q, k, v, causal, q_offset=0):
+        # q: [B, H, Tq, D]; k, v: [B, H, Tk, D]
         scale = 1.0 / math.sqrt(q.shape[-1])
         scores = torch.matmul(q, k.transpose(-2, -1))
-        p = causal_softmax(scores, scale, causal)
+        p = causal_softmax(scores, scale, causal, q_offset)
         out = torch.matmul(p, v)
         ctx.save_for_backward(q, k, v, p)
         ctx.scale = scale
@@ -66,7 +77,7 @@ class AttentionFunction(torch.autograd.Function):
         ds = softmax_bwd(p, dp, ctx.scale)
         dq = torch.matmul(ds, k)
         dk = torch.matmul(ds.transpose(-2, -1), q)
-        return dq, dk, dv, None
+        return dq, dk, dv, None, None
 
 
 def _flash_ok(q: torch.Tensor) -> bool:
@@ -88,13 +99,14 @@ class FlashAttentionFunction(torch.autograd.Function):
     """Fully-fused flash-style attention (csrc/attn.hip, D=64 bf16)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, causal):
+    def forward(ctx, q, k, v, causal, q_offset=0):
         scale = 1.0 / math.sqrt(q.shape[-1])
         out = torch.empty(q.shape, dtype=q.dtype, device=q.device)
-        lse2 = _backend.ext().attn_fwd(q, k, v, out, scale, causal)
+        lse2 = _backend.ext().attn_fwd(q, k, v, out, scale, causal, q_offset)
         ctx.save_for_backward(q, k, v, out, lse2)
         ctx.scale = scale
         ctx.causal = causal
+        ctx.q_offset = q_offset
         return out
 
     @staticmethod
@@ -105,8 +117,11 @@ class FlashAttentionFunction(torch.autograd.Function):
         dq = torch.empty(q.shape, dtype=q.dtype, device=q.device)
         dk = torch.empty(k.shape, dtype=k.dtype, device=k.device)
         dv = torch.empty(v.shape, dtype=v.dtype, device=v.device)
-        _backend.ext().attn_bwd(q, k, v, out, dout, lse2, dq, dk, dv, ctx.scale, ctx.causal)
-        return dq, dk, dv, None
+        _backend.ext().attn_bwd(
+            q, k, v, out, dout, lse2, dq, dk, dv, ctx.scale, ctx.causal,
+            ctx.q_offset,
+        )
+        return dq, dk, dv, None, None
 
 
 class FlashAttentionQKV(torch.autograd.Function):
@@ -163,22 +178,37 @@ class FlashAttentionQKV(torch.autograd.Function):
         return dqkv, None, None
 
 
-def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor, causal: bool = False) -> torch.Tensor:
+def attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    causal: bool = False,
+    q_offset: int = 0,
+) -> torch.Tensor:
     """Multi-head attention core: softmax(QK^T/sqrt(d) [+mask]) V.
 
-    q, k, v: [B, H, T, D] (same head count — TP shards heads upstream).
+    q: [B, H, Tq, D]; k, v: [B, H, Tk, D] (same head count — TP shards
+    heads upstream).  ``q_offset`` places the query shard at global rows
+    [q_offset, q_offset+Tq) of the K/V sequence (context parallelism).
     Dispatches to the fully-fused flash kernel on gfx950 (D=64,
-    T%128==0); otherwise the composed GEMM+fused-softmax path.
+    Tq/Tk/q_offset%128==0); otherwise the composed GEMM+softmax path.
     """
-    if _flash_ok(q) and q.shape == k.shape == v.shape:
+    if (
+        _flash_ok(q)
+        and k.shape == v.shape
+        and q.shape[-1] == k.shape[-1]
+        and k.shape[-2] % 128 == 0
+        and q_offset % 128 == 0
+        and q_offset + q.shape[-2] <= k.shape[-2]
+    ):
         def ok(t):
             return t.stride(-1) == 1 and all(s % 8 == 0 for s in t.stride()[:-1])
 
         qc = q if ok(q) else q.contiguous()
         kc = k if ok(k) else k.contiguous()
         vc = v if ok(v) else v.contiguous()
-        return FlashAttentionFunction.apply(qc, kc, vc, causal)
-    return AttentionFunction.apply(q, k, v, causal)
+        return FlashAttentionFunction.apply(qc, kc, vc, causal, q_offset)
+    return AttentionFunction.apply(q, k, v, causal, q_offset)
 
 
 def attention_qkv(qkv: torch.Tensor, n_heads: int, causal: bool = True) -> torch.Tensor:

@@ -7,6 +7,7 @@ from .tensor_parallel import (
     apply_tensor_parallel,
     ensure_divisibility,
 )
+from .context_parallel import context_parallel_attention, scatter_to_context
 from .pipeline import (
     InterleavedPipelineWrapper,
     PipelineParallelWrapper,
@@ -22,6 +23,8 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "context_parallel_attention",
+    "scatter_to_context",
     "InterleavedPipelineWrapper",
     "DataParallel",
     "BucketConfig",
