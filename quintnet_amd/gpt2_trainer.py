@@ -77,23 +77,36 @@ class GPT2Trainer(Trainer):
         accum = 0
         from .ops import causal_lm_loss
 
+        pt = self._phase_timer()
         for batch in self.train_loader:
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             labels = batch["labels"].to(self.device, non_blocking=True)
+            if pt:
+                pt.start("forward")
             logits = self.model(ids)
             loss = causal_lm_loss(logits, labels, ignore_index=-100)
+            if pt:
+                pt.stop("forward")
+                pt.start("backward")
             (loss / self.grad_acc_steps).backward()
+            if pt:
+                pt.stop("backward")
             accum += 1
             total_loss += float(loss.detach())
             total_tokens += int((labels[:, 1:] != -100).sum())
             steps += 1
             if accum == self.grad_acc_steps:
                 accum = 0
+                if pt:
+                    pt.start("grad_comm")
                 if isinstance(self.model, DataParallel):
                     self.model.finalize_gradients()
                 inner = _unwrap(self.model)
                 if hasattr(inner, "sync_tied_weights_grad"):
                     inner.sync_tied_weights_grad()
+                if pt:
+                    pt.stop("grad_comm")
+                    pt.start("optimizer")
                 if self.max_grad_norm:
                     from .ops import clip_grad_norm_local
 
@@ -106,6 +119,9 @@ class GPT2Trainer(Trainer):
                     self.model.zero_grad()
                 else:
                     self.optimizer.zero_grad()
+                if pt:
+                    pt.stop("optimizer")
+        self._report_phases(pt, steps)
         avg = total_loss / max(steps, 1)
         return {"loss": avg, "ppl": math.exp(min(avg, 20.0)), "n_tokens": total_tokens}
 

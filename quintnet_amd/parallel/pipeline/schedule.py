@@ -121,6 +121,10 @@ class PipelineSchedule(abc.ABC):
                 metrics["total"] = metrics.get("total", 0) + labels.numel()
         return loss
 
+    @property
+    def _pt(self):
+        return getattr(self.trainer, "phase_timer", None)
+
     def _forward_step(
         self,
         data_loader,
@@ -135,6 +139,25 @@ class PipelineSchedule(abc.ABC):
         stage consumes the loader to stay in lockstep (labels come from
         the same batch object on the last stage — SURVEY.md §8.6).
         """
+        pt = self._pt
+        if pt:
+            pt.start("forward")
+        try:
+            return self._forward_step_inner(
+                data_loader, input_tensor, device, metrics, num_micro
+            )
+        finally:
+            if pt:
+                pt.stop("forward")
+
+    def _forward_step_inner(
+        self,
+        data_loader,
+        input_tensor: Optional[torch.Tensor],
+        device,
+        metrics: Dict[str, float],
+        num_micro: int,
+    ) -> Tuple[torch.Tensor, Optional[torch.Tensor]]:
         batch = next(data_loader)
         if self.is_first:
             x = self._stage_input(batch, device)
@@ -152,10 +175,20 @@ class PipelineSchedule(abc.ABC):
         output_tensor: torch.Tensor,
         output_grad: Optional[torch.Tensor],
     ) -> Optional[torch.Tensor]:
-        return self.model.backward(input_tensor, output_tensor, output_grad)
+        pt = self._pt
+        if pt:
+            pt.start("backward")
+        try:
+            return self.model.backward(input_tensor, output_tensor, output_grad)
+        finally:
+            if pt:
+                pt.stop("backward")
 
     def _optimizer_step(self) -> None:
         t = self.trainer
+        pt = self._pt
+        if pt:
+            pt.start("optimizer")
         self._finalize_grads()
         if t.max_grad_norm is not None and t.max_grad_norm > 0:
             if hasattr(t.optimizer, "clip_grad_norm_"):
@@ -170,6 +203,8 @@ class PipelineSchedule(abc.ABC):
             self.model.zero_grad()
         else:
             t.optimizer.zero_grad()
+        if pt:
+            pt.stop("optimizer")
 
     @abc.abstractmethod
     def train_step(self, data_loader, tensor_shapes, device, dtype) -> Dict[str, float]:

@@ -128,6 +128,23 @@ class Trainer:
     def _is_rank0(self) -> bool:
         return not dist.is_initialized() or dist.get_rank() == 0
 
+    # -- per-phase tracing (the reference left this as TODO
+    # stubs; here HIP-event timing is wired into the real step loop) ----
+    def _phase_timer(self):
+        if not self.config.get("profile_phases"):
+            return None
+        from .utils.profiling import PhaseTimer
+
+        return PhaseTimer()
+
+    def _report_phases(self, pt, steps: int) -> None:
+        if pt is None or not self._is_rank0():
+            return
+        tot = pt.summary()
+        per = {k: v / max(steps, 1) for k, v in sorted(tot.items())}
+        line = "  ".join(f"{k}={v:.2f}ms" for k, v in per.items())
+        print(f"[phases/step] {line}", flush=True)
+
     def _broadcast_metrics(self, train: Dict[str, float], val: Dict[str, float]) -> Dict[str, float]:
         # world-uniform key schema (non-last PP stages have no metrics;
         # MAX all-reduce propagates the last stage's values — K17 pattern)
@@ -158,6 +175,8 @@ class Trainer:
         num_steps = max(len(self.train_loader) // self.grad_acc_steps, 1)
         agg: Dict[str, float] = {}
         timer = StepTimer() if self.config.get("profile") else None
+        pt = self._phase_timer()
+        self.pipeline_trainer.phase_timer = pt
         for _ in range(num_steps):
             if timer:
                 timer.start()
@@ -169,6 +188,7 @@ class Trainer:
             for k, v in m.items():
                 agg[k] = agg.get(k, 0.0) + float(v)
         out = {k: v / num_steps for k, v in agg.items()}
+        self._report_phases(pt, num_steps)
         if timer and self._is_rank0():
             bs = self._infer_micro_batch() * self.grad_acc_steps
             print(

@@ -37,3 +37,26 @@ def test_public_api():
         "ZeroRedundancyAdamW",
     ]:
         assert hasattr(quintnet_amd, sym), sym
+
+
+def test_phase_timer_integration():
+    """profile_phases config wires HIP-event/CPU phase timing into the
+    real step loop (reference's tracing was TODO stubs)."""
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1, n_head=2)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    dl = DataLoader(ds, batch_size=2)
+    tr = GPT2Trainer(
+        stage, dl, None,
+        {"profile_phases": True, "num_epochs": 1, "grad_acc_steps": 2, "zero1": False},
+        None,
+    )
+    hist = tr.fit()
+    assert "train_loss" in hist
