@@ -55,7 +55,36 @@ class GPT2Attention(nn.Module):
         self.head_dim = config.head_dim
         self.resid_dropout = FusedDropout(config.dropout)
 
-    def forward(self, x: torch.Tensor) -> torch.Tensor:
+    def forward(self, x: torch.Tensor, kv_cache=None) -> torch.Tensor:
         qkv = self.c_attn(x)  # [B, T, 3*n_embd/tp] = [q_loc | k_loc | v_loc]
-        out = attention_qkv(qkv, self.n_head_local, causal=True)
+        if kv_cache is None:
+            out = attention_qkv(qkv, self.n_head_local, causal=True)
+        else:
+            out = self._forward_cached(qkv, kv_cache)
         return self.resid_dropout(self.c_proj(out))
+
+    def _forward_cached(self, qkv: torch.Tensor, kv_cache: dict) -> torch.Tensor:
+        """Incremental decode: append this step's K/V to the cache and
+        attend the new queries against the whole cache (causal diagonal
+        shifted by the cache length — the same q_offset machinery the
+        context-parallel path uses)."""
+        from ...ops.attention import attention
+
+        B, T, _ = qkv.shape
+        H, D = self.n_head_local, self.head_dim
+        hl = self.hidden_local
+
+        def heads(t):
+            return t.view(B, T, H, D).permute(0, 2, 1, 3)
+
+        q = heads(qkv[:, :, :hl])
+        k = heads(qkv[:, :, hl : 2 * hl])
+        v = heads(qkv[:, :, 2 * hl :])
+        if kv_cache.get("k") is None:
+            kv_cache["k"], kv_cache["v"] = k, v
+        else:
+            kv_cache["k"] = torch.cat([kv_cache["k"], k], dim=2)
+            kv_cache["v"] = torch.cat([kv_cache["v"], v], dim=2)
+        past = kv_cache["k"].shape[2] - T
+        out = attention(q, kv_cache["k"], kv_cache["v"], causal=True, q_offset=past)
+        return out.permute(0, 2, 1, 3).reshape(B, T, hl)

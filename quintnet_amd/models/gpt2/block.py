@@ -29,6 +29,13 @@ class GPT2Block(nn.Module):
         m, s2 = self.forward_fused(x, None)
         return s2 + m
 
+    def forward_cached(self, x: torch.Tensor, kv_cache: dict) -> torch.Tensor:
+        """Plain (non-residual-fused) form for incremental decoding."""
+        n1 = self.ln_1(x)
+        s1 = x + self.attn(n1, kv_cache=kv_cache)
+        n2 = self.ln_2(s1)
+        return s1 + self.mlp(n2)
+
     def forward_fused(self, x: torch.Tensor, pending):
         """Residual-fused form: the incoming pending residual (previous
         block's MLP output) is added INSIDE ln_1's kernel; this block's

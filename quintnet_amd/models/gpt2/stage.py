@@ -102,6 +102,62 @@ class GPT2Stage(nn.Module):
         return x
 
     # ------------------------------------------------------------------
+    @torch.no_grad()
+    def generate(
+        self,
+        input_ids: torch.Tensor,
+        max_new_tokens: int = 32,
+        temperature: float = 0.0,
+        top_k: int = 0,
+    ) -> torch.Tensor:
+        """KV-cached autoregressive generation (serving path; pp==1).
+
+        temperature 0 = greedy; otherwise softmax sampling with optional
+        top-k truncation.  Works under TP (all TP ranks call this and run
+        the same collectives; sampling is made rank-consistent by seeding
+        from the tokens).  The cache attends via the q_offset attention
+        path (models/gpt2/attention.py), positions via wpe slicing.
+        """
+        assert self.is_first_stage and self.is_last_stage, "generate needs pp==1"
+        assert not self.sequence_parallel, "generate: disable sequence_parallel"
+        was_training = self.training
+        self.eval()
+        caches = [{"k": None, "v": None} for _ in self.blocks]
+        ids = input_ids
+        out = input_ids
+        past = 0
+        try:
+            for _ in range(max_new_tokens):
+                x = self.embedding(ids, pos_offset=past)
+                for blk, cache in zip(self.blocks, caches):
+                    x = blk.forward_cached(x, cache)
+                x = self.ln_f(x[:, -1:])
+                w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
+                logits = fused_linear(x, w, None, None, prefer_library=True)[:, -1]
+                if temperature and temperature > 0:
+                    logits = logits.float() / temperature
+                    if top_k and top_k > 0:
+                        kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
+                        logits = logits.masked_fill(logits < kth, float("-inf"))
+                    # rank-consistent sampling under TP: derive the RNG from
+                    # the current sequence so every TP rank draws the same
+                    gen = torch.Generator(device="cpu")
+                    gen.manual_seed(int(out.sum().item()) & 0x7FFFFFFF)
+                    probs = torch.softmax(logits, dim=-1).cpu()
+                    nxt = torch.multinomial(probs, 1, generator=gen).to(out.device)
+                else:
+                    nxt = logits.argmax(dim=-1, keepdim=True)
+                past += ids.shape[1]
+                ids = nxt
+                out = torch.cat([out, nxt], dim=1)
+                if past + 1 >= self.config.n_positions:
+                    break
+        finally:
+            if was_training:
+                self.train()
+        return out
+
+    # ------------------------------------------------------------------
     def sync_sequence_parallel_grads(self) -> None:
         """Under SP the LayerNorms and row-parallel biases compute their
         grads from sequence SHARDS — all-reduce them over the TP group

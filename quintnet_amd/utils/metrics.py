@@ -75,6 +75,14 @@ def compute_generation_metrics(candidates: List[str], references: List[str]) -> 
 @torch.no_grad()
 def generate_greedy(model, tokenizer, prompt: str, max_new_tokens: int, device) -> str:
     ids = tokenizer(prompt, return_tensors="pt")["input_ids"].to(device)
+    if hasattr(model, "generate"):  # KV-cached serving path (GPT2Stage)
+        out = model.generate(ids, max_new_tokens=max_new_tokens, temperature=0.0)
+        if tokenizer.eos_token_id is not None:
+            new = out[0, ids.shape[1]:]
+            eos = (new == tokenizer.eos_token_id).nonzero()
+            if eos.numel():
+                out = out[:, : ids.shape[1] + int(eos[0]) + 1]
+        return tokenizer.decode(out[0], skip_special_tokens=True)
     for _ in range(max_new_tokens):
         logits = model(ids)
         nxt = logits[:, -1, :].argmax(dim=-1, keepdim=True)

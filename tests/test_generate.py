@@ -1,0 +1,45 @@
+"""KV-cached generation: equality with full re-forward greedy decoding."""
+
+import torch
+
+
+def _full_greedy(stage, input_ids, n):
+    ids = input_ids
+    for _ in range(n):
+        logits = stage(ids)
+        nxt = logits[:, -1].argmax(dim=-1, keepdim=True)
+        ids = torch.cat([ids, nxt], dim=1)
+    return ids
+
+
+def test_kv_cache_greedy_matches_full_forward():
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    torch.manual_seed(0)
+    cfg = GPT2Config(vocab_size=96, n_positions=64, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    stage.eval()
+    ids = torch.randint(0, 96, (2, 7))
+    with torch.no_grad():
+        ref = _full_greedy(stage, ids, 12)
+    out = stage.generate(ids, max_new_tokens=12, temperature=0.0)
+    assert torch.equal(out, ref), (out, ref)
+
+
+def test_generate_sampling_shapes_and_determinism():
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    torch.manual_seed(1)
+    cfg = GPT2Config(vocab_size=64, n_positions=48, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    ids = torch.randint(0, 64, (1, 5))
+    a = stage.generate(ids, max_new_tokens=8, temperature=0.8, top_k=10)
+    b = stage.generate(ids, max_new_tokens=8, temperature=0.8, top_k=10)
+    assert a.shape == (1, 13)
+    assert torch.equal(a, b)  # tokens-seeded sampling is deterministic
+    # respects the context window
+    long_ids = torch.randint(0, 64, (1, 40))
+    c = stage.generate(long_ids, max_new_tokens=30)
+    assert c.shape[1] <= 48 + 1
