@@ -48,6 +48,7 @@ class GPT2Attention(nn.Module):
             device=device,
             dtype=dtype,
         )
+        self.c_attn.fp8 = self.c_proj.fp8 = config.fp8
         self.tp_size = self.c_attn.tp_size
         assert config.n_head % self.tp_size == 0, "n_head must divide by tp"
         self.n_head_local = config.n_head // self.tp_size

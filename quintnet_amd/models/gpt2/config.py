@@ -22,7 +22,8 @@ class GPT2Config:
     dropout: float = 0.1
     layer_norm_epsilon: float = 1e-5
     initializer_range: float = 0.02
-    sequence_parallel: bool = False  # Megatron-SP over the TP group
+    sequence_parallel: bool = False
+    fp8: bool = False  # experimental: e4m3 forward GEMMs, bf16 backward  # Megatron-SP over the TP group
 
     def __post_init__(self):
         if self.n_inner is None:

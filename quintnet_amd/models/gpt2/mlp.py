@@ -40,6 +40,7 @@ class GPT2MLP(nn.Module):
             device=device,
             dtype=dtype,
         )
+        self.c_fc.fp8 = self.c_proj.fp8 = config.fp8
         self.dropout = FusedDropout(config.dropout)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -65,14 +65,52 @@ def _native_ok(x: torch.Tensor, weight: torch.Tensor) -> bool:
     )
 
 
+def _fp8_ok(x2d: torch.Tensor, weight: torch.Tensor) -> bool:
+    return (
+        x2d.is_cuda
+        and x2d.dtype == torch.bfloat16
+        and hasattr(torch, "float8_e4m3fn")
+        and x2d.shape[0] % 16 == 0
+        and weight.shape[0] % 16 == 0
+        and weight.shape[1] % 32 == 0
+    )
+
+
+def _fp8_linear(x2d: torch.Tensor, weight: torch.Tensor, bias):
+    """Forward-only OCP e4m3 GEMM via hipBLASLt (torch._scaled_mm) with
+    per-tensor dynamic scaling — measured 1287 TF on the GPT-2 c_fc shape
+    vs 833 bf16 (probe: tools/probe_fp8.py).  Backward stays bf16."""
+    E4M3_MAX = 448.0
+    sa = (x2d.abs().amax().float() / E4M3_MAX).clamp(min=1e-12)
+    sb = (weight.abs().amax().float() / E4M3_MAX).clamp(min=1e-12)
+    a8 = (x2d / sa).to(torch.float8_e4m3fn)
+    b8 = (weight / sb).to(torch.float8_e4m3fn).t()
+    return torch._scaled_mm(
+        a8, b8, scale_a=sa, scale_b=sb, bias=bias, out_dtype=torch.bfloat16
+    )
+
+
 class LinearFunction(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, activation, prefer_library=False):
+    def forward(ctx, x, weight, bias, activation, prefer_library=False, fp8=False):
         act = _ACT_MAP[activation]
         x2d = x.reshape(-1, x.shape[-1])
         ctx.x_shape = x.shape
         ctx.act = act
         pre_act = None
+        if fp8 and _fp8_ok(x2d, weight):
+            out = _fp8_linear(x2d.contiguous(), weight.contiguous(), bias)
+            if act == _ACT_GELU:
+                pre_act = out
+                out = _gelu_tanh(out)
+            elif act == _ACT_RELU:
+                pre_act = out
+                out = torch.relu(out)
+            ctx.save_for_backward(
+                x2d, weight, pre_act if pre_act is not None else torch.empty(0)
+            )
+            ctx.has_bias = bias is not None
+            return out.reshape(*x.shape[:-1], weight.shape[0])
         want_native = (
             _FWD_MODE == "custom"
             or (_FWD_MODE == "auto" and act != _ACT_NONE
@@ -140,7 +178,7 @@ class LinearFunction(torch.autograd.Function):
                 grad_b = _backend.ext().colsum(g)
             else:
                 grad_b = g.sum(dim=0)
-        return grad_x, grad_w, grad_b, None, None
+        return grad_x, grad_w, grad_b, None, None, None
 
 
 def linear(
@@ -149,11 +187,14 @@ def linear(
     bias: Optional[torch.Tensor] = None,
     activation: Optional[str] = None,
     prefer_library: bool = False,
+    fp8: bool = False,
 ) -> torch.Tensor:
     """y = activation(x @ weight.T + bias), fused on gfx950.
 
     ``prefer_library=True`` sends a PLAIN (no bias, no activation) GEMM
     through hipBLASLt instead — per the MI355X design split, the vendor
     library serves plain GEMMs (e.g. the tied LM head) while fused ones
-    run the hand-written MFMA kernel."""
-    return LinearFunction.apply(x, weight, bias, activation, prefer_library)
+    run the hand-written MFMA kernel.  ``fp8=True`` (experimental) runs
+    the FORWARD in OCP e4m3 with per-tensor dynamic scaling (~1.5-1.8x
+    GEMM throughput); backward stays bf16."""
+    return LinearFunction.apply(x, weight, bias, activation, prefer_library, fp8)

@@ -118,7 +118,8 @@ class ColumnParallelLinear(nn.Module):
                 x = All_Gather.apply(x, self.tp_group, 1, "reduce_scatter")
             else:
                 x = copy_to_group(x, self.tp_group)  # identity fwd, all-reduce bwd
-        out = fused_linear(x, self.weight, self.bias, self.activation)
+        out = fused_linear(x, self.weight, self.bias, self.activation,
+                           fp8=getattr(self, "fp8", False))
         if self.gather_output and self.tp_size > 1:
             out = All_Gather.apply(out, self.tp_group, -1, "slice")
         return out
@@ -191,10 +192,12 @@ class RowParallelLinear(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.tp_size == 1:
             # bias fuses into the GEMM epilogue when there is no reduction
-            return fused_linear(x, self.weight, self.bias, None)
+            return fused_linear(x, self.weight, self.bias, None,
+                                fp8=getattr(self, "fp8", False))
         if not self.input_is_parallel:
             x = x.chunk(self.tp_size, dim=-1)[self.tp_rank].contiguous()
-        out = fused_linear(x, self.weight, None, None)
+        out = fused_linear(x, self.weight, None, None,
+                           fp8=getattr(self, "fp8", False))
         if self.sequence_parallel:
             # Megatron-SP ḡ operator: reduce-scatter the partial sums
             # over the sequence dim (grads all-gather back)

@@ -355,3 +355,21 @@ def test_flash_attention_q_offset_vs_reference():
         assert (q.grad.float() - q2.grad).abs().max() < 5e-2
         assert (kk.grad.float() - k2.grad).abs().max() < 5e-2
         assert (vv.grad.float() - v2.grad).abs().max() < 5e-2
+
+
+@pytest.mark.gpu
+def test_fp8_linear_forward():
+    """Experimental e4m3 forward: correct within fp8 rounding, bf16 grads."""
+    from quintnet_amd.ops.linear import linear
+
+    torch.manual_seed(5)
+    x = torch.randn(256, 768, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(512, 768, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    b = torch.randn(512, device="cuda", dtype=torch.bfloat16)
+    out = linear(x, w, b, fp8=True)
+    ref = torch.nn.functional.linear(x.float(), w.float(), b.float())
+    rel = (out.float() - ref).abs().mean() / ref.abs().mean()
+    assert rel < 0.08, float(rel)
+    out.sum().backward()
+    assert x.grad is not None and w.grad is not None
+    assert torch.isfinite(x.grad).all() and torch.isfinite(w.grad).all()
