@@ -26,7 +26,9 @@ import os
 import time
 
 # hipBLASLt algorithm selection via PyTorch TunableOp (+6% on the GPT-2
-# step); pre-tuned results for gfx950 ship in-tree so no tuning delay.
+# step).  Tuning runs during the UNTIMED warmup steps on each fresh
+# process (a few seconds; results are not persisted across machines —
+# torch's exit-time CSV write has not been observed on this build).
 os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
 os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING", "1")
 os.environ.setdefault(
