@@ -221,11 +221,14 @@ def main():
     criterion = torch.nn.CrossEntropyLoss(ignore_index=-100)
 
     if pg.pp_size > 1:
+        groups = pg.get_all_groups()
         ptrainer = PipelineTrainer(
             model=model, optimizer=optimizer, criterion=criterion,
             pp_rank=pg.pp_rank, pp_size=pg.pp_size,
             pp_group=pg.get_group("pp"), pp_group_ranks=pg.get_group_ranks("pp"),
-            schedule="1f1b", task_type=task, max_grad_norm=1.0,
+            schedule=os.environ.get("QN_SCHEDULE", "1f1b"),
+            task_type=task, max_grad_norm=1.0,
+            pp_fwd_group=groups.get("pp_fwd"), pp_bwd_group=groups.get("pp_bwd"),
         )
         shapes = (micro_b, pipe_shape_seq if task == "clm" else seq, hidden)
 
