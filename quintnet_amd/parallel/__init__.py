@@ -8,6 +8,7 @@ from .tensor_parallel import (
     ensure_divisibility,
 )
 from .context_parallel import context_parallel_attention, scatter_to_context
+from .zero3 import ZeRO3Block, apply_zero3
 from .pipeline import (
     InterleavedPipelineWrapper,
     PipelineParallelWrapper,
@@ -23,6 +24,8 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "ZeRO3Block",
+    "apply_zero3",
     "context_parallel_attention",
     "scatter_to_context",
     "InterleavedPipelineWrapper",

@@ -1,0 +1,90 @@
+"""ZeRO-3 parameter sharding: trajectory equivalence with replicated DP
+training, shard-only storage, and grad reduce-scatter correctness."""
+
+import copy
+
+import torch
+
+from conftest import run_distributed
+
+
+def test_zero3_single_rank_noop_math():
+    """world=1: wrapped model computes identically and grads land on shard."""
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel.zero3 import apply_zero3
+
+    torch.manual_seed(0)
+    m = Model(hidden_dim=32, n_heads=2, depth=2)
+    ref = copy.deepcopy(m)
+    apply_zero3(m, dp_group=None)
+    x = torch.randn(2, 1, 28, 28)
+    out = m(x)
+    assert torch.allclose(out, ref(x), atol=1e-5)
+    out.sum().backward()
+    for blk in m.blocks:
+        assert blk.shard.grad is not None and torch.isfinite(blk.shard.grad).all()
+
+
+def _run_zero3(rank, world):
+    import torch.distributed as dist
+
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel.zero3 import apply_zero3
+
+    torch.manual_seed(11)
+    model = Model(hidden_dim=32, n_heads=2, depth=4)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    # freeze everything outside the sharded blocks: syncing those is
+    # DataParallel's job, and letting them drift per-rank would
+    # contaminate the block-gradient comparison
+    for name, p in model.named_parameters():
+        if not name.startswith("blocks."):
+            p.requires_grad_(False)
+    ref = copy.deepcopy(model)
+
+    apply_zero3(model, dp_group=dist.group.WORLD)
+
+    # each block's own storage is freed; only the 1/world shard remains
+    for blk in model.blocks:
+        assert sum(p.numel() for p in blk.module.parameters()) == 0
+        assert blk.shard.numel() * world >= sum(n for _, n in blk._metas)
+
+    # DP semantics: different data per rank, grads averaged
+    opt = torch.optim.SGD([p for p in model.parameters() if p.requires_grad], lr=0.1)
+    ref_opt = torch.optim.SGD([p for p in ref.parameters() if p.requires_grad], lr=0.1)
+    for step in range(3):
+        torch.manual_seed(500 + step * world + rank)
+        x = torch.randn(2, 1, 28, 28)
+        y = torch.randint(0, 10, (2,))
+        loss = torch.nn.functional.cross_entropy(model(x), y)
+        (loss / world).backward()  # ZeRO grads reduce-SUM; scale for mean
+        opt.step()
+        opt.zero_grad()
+
+        # reference: replicated model sees ALL ranks' batches (DP mean)
+        ref_loss = 0.0
+        for r in range(world):
+            torch.manual_seed(500 + step * world + r)
+            xr = torch.randn(2, 1, 28, 28)
+            yr = torch.randint(0, 10, (2,))
+            ref_loss = ref_loss + torch.nn.functional.cross_entropy(ref(xr), yr)
+        (ref_loss / world).backward()
+        ref_opt.step()
+        ref_opt.zero_grad()
+
+    # compare: re-gather full params of each block vs reference blocks
+    for blk, rblk in zip(model.blocks, ref.blocks):
+        full = blk.full_state_dict_tensors()
+        rparams = dict(rblk.named_parameters())
+        for name, t in full.items():
+            assert torch.allclose(t, rparams[name], atol=1e-4), (name,)
+
+
+
+def test_zero3_world2():
+    run_distributed(_run_zero3, 2)
+
+
+def test_zero3_world4():
+    run_distributed(_run_zero3, 4)
