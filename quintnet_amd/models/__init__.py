@@ -14,7 +14,7 @@ from .gpt2 import (
     GPT2Attention,
     GPT2MLP,
     GPT2Block,
-    GPT2Stage,
+    GPT2ForInterleaving, GPT2Stage,
 )
 
 __all__ = [
@@ -31,5 +31,6 @@ __all__ = [
     "GPT2Attention",
     "GPT2MLP",
     "GPT2Block",
+    "GPT2ForInterleaving",
     "GPT2Stage",
 ]

@@ -4,6 +4,7 @@ from .attention import GPT2Attention
 from .mlp import GPT2MLP
 from .block import GPT2Block
 from .stage import GPT2Stage
+from .interleaved import GPT2ForInterleaving, TiedLMHead
 
 __all__ = [
     "GPT2Config",
@@ -12,4 +13,6 @@ __all__ = [
     "GPT2MLP",
     "GPT2Block",
     "GPT2Stage",
+    "GPT2ForInterleaving",
+    "TiedLMHead",
 ]

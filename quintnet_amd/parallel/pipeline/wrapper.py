@@ -124,10 +124,12 @@ class InterleavedPipelineWrapper(nn.Module):
         pp_size: int = 1,
         num_chunks: int = 2,
         device: Optional[torch.device] = None,
+        tied_group=None,
     ):
         super().__init__()
         if not hasattr(model, "blocks"):
             raise ValueError("InterleavedPipelineWrapper needs a model with .blocks")
+        self.tied_group = tied_group
         self.pp_rank = pp_rank
         self.pp_size = pp_size
         self.pp_group = pp_group
@@ -158,6 +160,27 @@ class InterleavedPipelineWrapper(nn.Module):
         self.chunks = nn.ModuleList(chunks)
         if device is not None:
             self.chunks.to(device)
+
+        # tied embedding/LM-head: global stage 0 holds the embedding
+        # (module with .wte), the last global stage a TiedLMHead-style
+        # module (.tied_weight); when they live on DIFFERENT ranks their
+        # grads must be summed over the first+last-stage subgroup.
+        self._tied_params: List[torch.nn.Parameter] = []
+        for chunk in self.chunks:
+            for m in chunk.modules():
+                if hasattr(m, "tied_weight"):
+                    self._tied_params.append(m.tied_weight)
+                elif hasattr(m, "wte") and isinstance(getattr(m, "wte"), nn.Embedding):
+                    self._tied_params.append(m.wte.weight)
+
+    def sync_tied_weights_grad(self) -> None:
+        import torch.distributed as dist
+
+        if self.tied_group is None or len(self._tied_params) != 1:
+            return  # both copies local (p==1) or no tied weight
+        p = self._tied_params[0]
+        if p.grad is not None:
+            dist.all_reduce(p.grad, op=dist.ReduceOp.SUM, group=self.tied_group)
 
     def forward(self, x, chunk_id: int = 0):
         return self.chunks[chunk_id](x)
