@@ -110,6 +110,15 @@ class Trainer:
     # ------------------------------------------------------------------
     def fit(self) -> Dict[str, float]:
         history = {}
+        wd = None
+        if self.config.get("watchdog_timeout_s"):
+            from .utils.watchdog import Watchdog
+
+            wd = Watchdog(
+                float(self.config["watchdog_timeout_s"]),
+                kill_on_hang=bool(self.config.get("watchdog_kill", False)),
+            ).start()
+            self._watchdog = wd
         for epoch in range(self.num_epochs):
             t0 = time.time()
             train_metrics = self._train_epoch(epoch)
@@ -122,7 +131,14 @@ class Trainer:
                     msg += f" {k}={v:.4f}"
                 print(msg, flush=True)
             history = metrics
+            if wd:
+                wd.beat()
+            every = int(self.config.get("save_every", 0))
+            if every and (epoch + 1) % every == 0 and epoch + 1 < self.num_epochs:
+                self._save_checkpoint()
         self._save_checkpoint()
+        if wd:
+            wd.stop()
         return history
 
     def _is_rank0(self) -> bool:
@@ -177,7 +193,10 @@ class Trainer:
         timer = StepTimer() if self.config.get("profile") else None
         pt = self._phase_timer()
         self.pipeline_trainer.phase_timer = pt
+        wd = getattr(self, "_watchdog", None)
         for _ in range(num_steps):
+            if wd:
+                wd.beat()
             if timer:
                 timer.start()
             m = self.pipeline_trainer.train_step(

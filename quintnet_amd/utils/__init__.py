@@ -1,6 +1,7 @@
 from .logger import setup_rank_logger, print_rank_0
 from .memory import memory_stats, print_memory_stats
 from .profiling import StepTimer, PhaseTimer
+from .watchdog import Watchdog
 from .graphs import CapturedStep
 from .metrics import (
     rouge_n,
@@ -27,6 +28,7 @@ __all__ = [
     "print_memory_stats",
     "StepTimer",
     "PhaseTimer",
+    "Watchdog",
     "CapturedStep",
     "rouge_n",
     "rouge_l",
