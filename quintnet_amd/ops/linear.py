@@ -77,7 +77,9 @@ def _fp8_ok(x2d: torch.Tensor, weight: torch.Tensor) -> bool:
 
 
 _FP8_MAX = 448.0
-_FP8_STATE: dict = {}  # id(weight) -> cached fp8 weight + delayed act scale
+from torch.utils.weak import WeakTensorKeyDictionary
+
+_FP8_STATE = WeakTensorKeyDictionary()  # weight -> cached fp8 + delayed scale
 
 
 def _fp8_linear(x2d: torch.Tensor, weight: torch.Tensor, bias):
@@ -89,7 +91,7 @@ def _fp8_linear(x2d: torch.Tensor, weight: torch.Tensor, bias):
     scaling (cast with the previous step's amax; this step's amax is
     computed asynchronously and becomes the next scale) so the only
     per-call extra work is the single cast pass the fp8 GEMM requires."""
-    st = _FP8_STATE.setdefault(id(weight), {})
+    st = _FP8_STATE.setdefault(weight, {})
     wver = getattr(weight, "_version", None)
     if st.get("wver") != wver or st.get("w8") is None:
         sb = (weight.detach().abs().amax().float() / _FP8_MAX).clamp(min=1e-12)
