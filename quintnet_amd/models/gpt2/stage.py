@@ -120,6 +120,9 @@ class GPT2Stage(nn.Module):
         """
         assert self.is_first_stage and self.is_last_stage, "generate needs pp==1"
         assert not self.sequence_parallel, "generate: disable sequence_parallel"
+        assert input_ids.shape[1] <= self.config.n_positions, (
+            "prompt longer than the model's context window"
+        )
         was_training = self.training
         self.eval()
         caches = [{"k": None, "v": None} for _ in self.blocks]

@@ -35,6 +35,10 @@ def scatter_to_context(x: torch.Tensor, cp_group, dim: int = 1) -> torch.Tensor:
     world = _ws(cp_group)
     if world == 1:
         return x
+    if x.shape[dim] % world != 0:
+        raise ValueError(
+            f"context parallel: dim {dim} ({x.shape[dim]}) must divide by cp={world}"
+        )
     rank = dist.get_rank(group=cp_group)
     return x.chunk(world, dim=dim)[rank].contiguous()
 

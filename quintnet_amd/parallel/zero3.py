@@ -78,6 +78,17 @@ class ZeRO3Block(nn.Module):
 
     def __init__(self, module: nn.Module, dp_group=None):
         super().__init__()
+        # checkpoint's RNG preservation covers torch RNG only; the fused
+        # counter-based dropout kernel would draw a DIFFERENT mask in the
+        # backward recompute, corrupting gradients silently.
+        for m in module.modules():
+            if type(m).__name__ == "FusedDropout" and getattr(m, "p", 0) > 0:
+                raise ValueError(
+                    "ZeRO3Block: module uses FusedDropout with p>0 — its "
+                    "counter-based mask is not recompute-stable under "
+                    "activation checkpointing; set dropout=0 or use "
+                    "torch.nn.Dropout"
+                )
         self.module = module
         self.dp_group = dp_group
         self.world = _group_size(dp_group)
