@@ -65,8 +65,13 @@ class GPT2Trainer(Trainer):
     # ------------------------------------------------------------------
     def _infer_seq_hidden(self, inner) -> tuple:
         mc = self.config.get("model_config", {})
-        seq = int(self.config.get("max_seq_length", mc.get("n_positions", 1024)))
-        hidden = int(mc.get("n_embd", 768))
+        # prefer explicit config; fall back to the stage's own hints before
+        # assuming GPT-2-small defaults
+        seq = self.config.get("max_seq_length") or mc.get("n_positions") or getattr(
+            inner, "seq_len", None
+        ) or 1024
+        hidden = mc.get("n_embd") or getattr(inner, "hidden_dim", None) or 768
+        seq, hidden = int(seq), int(hidden)
         if mc.get("sequence_parallel") and self.pg is not None and self.pg.tp_size > 1:
             seq //= self.pg.tp_size  # inter-stage activations are seq shards
         return seq, hidden
