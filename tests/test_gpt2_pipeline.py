@@ -243,3 +243,86 @@ def _gpt2_3d(rank, world):
 
 def test_gpt2_3d_zero1_matches_single_process():
     run_distributed(_gpt2_3d, 8, timeout=300)
+
+
+def _gpt2_3d_sp(rank, world):
+    """Same [2,2,2] assembly with Megatron sequence parallelism enabled:
+    inter-stage activations are sequence shards; loss must still match."""
+    _gpt2_3d_impl(rank, world, sequence_parallel=True)
+
+
+def _gpt2_3d_impl(rank, world, sequence_parallel=False):
+    import torch.distributed as dist
+
+    from quintnet_amd import GPT2Trainer, init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        PipelineParallelWrapper,
+    )
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    pg = init_process_groups("cpu", [2, 2, 2], ["dp", "tp", "pp"])
+    torch.manual_seed(78)
+    cfg = GPT2Config(vocab_size=128, n_positions=16, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0, sequence_parallel=sequence_parallel)
+    full_cfg = GPT2Config(vocab_size=128, n_positions=16, n_embd=32, n_layer=2,
+                          n_head=2, dropout=0.0)
+    stage = GPT2Stage(
+        cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+        tp_group=pg.get_group("tp"), tied_group=pg.get_tied_embedding_group(),
+    )
+    master = GPT2Stage(full_cfg, pp_rank=0, pp_size=1, tp_group=None)
+    for p in master.parameters():
+        dist.broadcast(p.data, src=0)
+    stage.load_state_dict(
+        _shard_full_stage_sd(master.state_dict(), cfg, pg.pp_rank, pg.pp_size,
+                             pg.tp_rank, pg.tp_size, stage),
+        strict=False,
+    )
+    seq = 16
+    stage.seq_len, stage.hidden_dim = seq, cfg.n_embd
+    pmodel = PipelineParallelWrapper(
+        stage_module=stage, pp_rank=pg.pp_rank, pp_group=pg.get_group("pp"),
+        pp_size=pg.pp_size,
+    )
+    pmodel.seq_len, pmodel.hidden_dim = seq, cfg.n_embd
+    pmodel = DataParallel(
+        pmodel, DistributedConfig(pg.dp_rank, pg.dp_size, pg.get_group("dp"))
+    )
+    tcfg = {"batch_size": 2, "num_epochs": 1, "learning_rate": 1e-3,
+            "grad_acc_steps": 2, "max_grad_norm": None, "zero1": True,
+            "task_type": "clm", "max_seq_length": seq,
+            "model_config": {"n_embd": cfg.n_embd, "n_positions": seq,
+                             "sequence_parallel": sequence_parallel}}
+    ds = SyntheticCLM(n=8, seq_len=seq, vocab_size=128, seed=9)
+    dl = torch.utils.data.DataLoader(ds, batch_size=2, shuffle=False)
+    trainer = GPT2Trainer(pmodel, dl, None, tcfg, pg)
+    metrics = trainer.fit()
+    assert torch.isfinite(torch.tensor(metrics["train_loss"]))
+
+    if rank == 0:
+        opt = torch.optim.AdamW(master.parameters(), lr=1e-3, weight_decay=0.01)
+        it = iter(dl)
+        step_losses = []
+        for _ in range(2):
+            opt.zero_grad()
+            tot = 0.0
+            for _ in range(2):
+                b = next(it)
+                loss = causal_lm_loss(master(b["input_ids"]), b["labels"], ignore_index=-100)
+                (loss / 2).backward()
+                tot += float(loss.detach())
+            master.sync_tied_weights_grad()
+            opt.step()
+            step_losses.append(tot / 2)
+        ref_mean = sum(step_losses) / len(step_losses)
+        assert abs(metrics["train_loss"] - ref_mean) < 5e-3, (
+            metrics["train_loss"], ref_mean,
+        )
+
+
+def test_gpt2_3d_sequence_parallel():
+    run_distributed(_gpt2_3d_sp, 8, timeout=300)
