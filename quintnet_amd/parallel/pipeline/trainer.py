@@ -59,6 +59,13 @@ class PipelineTrainer:
         sched = self.schedule
         from .dataloader import PipelineDataLoader
 
+        from .schedule import InterleavedOneFOneBSchedule
+
+        if isinstance(sched, InterleavedOneFOneBSchedule):
+            return self._evaluate_interleaved(
+                data_loader, tensor_shapes, device, dtype, max_batches
+            )
+
         for batch in data_loader:
             if max_batches is not None and n >= max_batches:
                 break
@@ -79,4 +86,39 @@ class PipelineTrainer:
                     "send_forward", self.pp_rank, self.pp_size, self.pp_group_ranks,
                     tensor=out, group=self.pp_group,
                 )
+        return sched._final_metrics(metrics, n)
+
+    @torch.no_grad()
+    def _evaluate_interleaved(self, data_loader, tensor_shapes, device, dtype,
+                              max_batches=None) -> Dict[str, float]:
+        """Forward-only ring walk over the v*p virtual stages per batch."""
+        from ...core.comm import ring_recv, ring_send
+        from .dataloader import PipelineDataLoader
+
+        sched = self.schedule
+        owner = sched._chunks_owner()
+        v, p, r = int(owner.num_chunks), self.pp_size, self.pp_rank
+        fwd_group = self.pp_fwd_group or self.pp_group
+        metrics: Dict[str, float] = {}
+        n = 0
+        for batch in data_loader:
+            if max_batches is not None and n >= max_batches:
+                break
+            batch = PipelineDataLoader._normalize(batch)
+            n += 1
+            for g in range(v * p):
+                if g % p != r:
+                    continue
+                if g == 0:
+                    inp = sched._stage_input(batch, device)
+                else:
+                    inp = ring_recv(r, p, self.pp_group_ranks, -1, tensor_shapes,
+                                    dtype, device, group=fwd_group)
+                out = owner.forward(inp, chunk_id=g // p)
+                if g == v * p - 1:
+                    sched._loss_and_metrics(out, batch, device, metrics)
+                else:
+                    for rq in ring_send(out, r, p, self.pp_group_ranks, +1,
+                                        group=fwd_group):
+                        rq.wait()
         return sched._final_metrics(metrics, n)

@@ -260,3 +260,44 @@ def _run_gpt2_interleaved(rank, world):
 
 def test_gpt2_interleaved_tied_weights():
     run_distributed(_run_gpt2_interleaved, 2)
+
+
+def _run_interleaved_eval(rank, world):
+    """Forward-only evaluate() under interleaving matches a single-process
+    evaluation."""
+    import copy
+
+    import torch.distributed as dist
+
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel import InterleavedPipelineWrapper, PipelineTrainer
+
+    fwd_group = dist.new_group(list(range(world)))
+    bwd_group = dist.new_group(list(range(world)))
+    torch.manual_seed(42)
+    model = Model(hidden_dim=32, n_heads=2, depth=8)
+    for p in model.parameters():
+        dist.broadcast(p.data, src=0)
+    ref = copy.deepcopy(model)
+    stage = InterleavedPipelineWrapper(model, pp_rank=rank, pp_size=world, num_chunks=2)
+    pt = PipelineTrainer(
+        model=stage, optimizer=None, criterion=torch.nn.CrossEntropyLoss(),
+        pp_rank=rank, pp_size=world, pp_group=None,
+        pp_group_ranks=list(range(world)), schedule="interleaved",
+        task_type="classification", pp_fwd_group=fwd_group, pp_bwd_group=bwd_group,
+    )
+    from quintnet_amd.utils.data import SyntheticMNIST
+
+    dl = torch.utils.data.DataLoader(SyntheticMNIST(n=8, seed=4), batch_size=2)
+    m = pt.evaluate(dl, (2, 50, 32), torch.device("cpu"), torch.float32)
+    if rank == world - 1:
+        crit = torch.nn.CrossEntropyLoss()
+        tot, steps = 0.0, 0
+        for b in dl:
+            tot += float(crit(ref(b["images"]), b["labels"]))
+            steps += 1
+        assert abs(m["loss"] - tot / steps) < 1e-4, (m, tot / steps)
+
+
+def test_interleaved_evaluate():
+    run_distributed(_run_interleaved_eval, 2)
