@@ -7,7 +7,11 @@ from .tensor_parallel import (
     apply_tensor_parallel,
     ensure_divisibility,
 )
-from .context_parallel import context_parallel_attention, scatter_to_context
+from .context_parallel import (
+    context_parallel_attention,
+    ring_attention,
+    scatter_to_context,
+)
 from .zero3 import ZeRO3Block, apply_zero3
 from .pipeline import (
     InterleavedPipelineWrapper,
@@ -24,6 +28,7 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "ring_attention",
     "ZeRO3Block",
     "apply_zero3",
     "context_parallel_attention",

@@ -21,7 +21,7 @@ import torch.distributed as dist
 from ..core.comm import All_Gather
 from ..ops.attention import attention
 
-__all__ = ["context_parallel_attention", "scatter_to_context"]
+__all__ = ["context_parallel_attention", "ring_attention", "scatter_to_context"]
 
 
 def _ws(group) -> int:
@@ -64,3 +64,137 @@ def context_parallel_attention(
     k_full = All_Gather.apply(k, cp_group, -2, "reduce_scatter")
     v_full = All_Gather.apply(v, cp_group, -2, "reduce_scatter")
     return attention(q, k_full, v_full, causal=causal, q_offset=rank * t_local)
+
+
+# ---------------------------------------------------------------------------
+# Ring attention: the memory-scalable CP flavor — K/V chunks rotate around
+# the CP ring instead of being materialized in full on every rank.  One
+# custom autograd Function owns BOTH ring schedules (forward: cp-1 hops of
+# [K,V]; backward: cp hops of [K,V,dK,dV] with the grad accumulators
+# traveling alongside their chunk until they return to the owner), so the
+# P2P order is deterministic — no reliance on autograd's reverse-order
+# engine to sequence collectives.  Per-chunk math is the composed fp32
+# path (streaming log-sum-exp merge); the fused flash kernel can be slotted
+# per chunk later.
+# ---------------------------------------------------------------------------
+def _ring_swap(tensors, rank, world, group):
+    """Send tensors to rank+1, receive the same shapes from rank-1."""
+    nxt = (rank + 1) % world
+    prv = (rank - 1 + world) % world
+    ops = []
+    recvs = []
+    for t in tensors:
+        ops.append(dist.P2POp(dist.isend, t.contiguous(), peer=nxt, group=group))
+    for t in tensors:
+        buf = torch.empty_like(t)
+        recvs.append(buf)
+        ops.append(dist.P2POp(dist.irecv, buf, peer=prv, group=group))
+    for r in dist.batch_isend_irecv(ops):
+        r.wait()
+    return recvs
+
+
+def _partial(q32, k32, scale, causal_mode, Tl):
+    """scores [B,H,Tl,Tl] for one chunk; causal_mode: 0 full, 1 diagonal."""
+    s = torch.matmul(q32, k32.transpose(-2, -1)) * scale
+    if causal_mode == 1:
+        mask = torch.ones(Tl, Tl, dtype=torch.bool, device=s.device).tril()
+        s = s.masked_fill(~mask, float("-inf"))
+    return s
+
+
+class _RingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, group, causal):
+        import math
+
+        world = _ws(group)
+        rank = dist.get_rank(group=group) if world > 1 else 0
+        B, H, Tl, D = q.shape
+        scale = 1.0 / math.sqrt(D)
+        q32 = q.float()
+        m = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
+        l = torch.zeros(B, H, Tl, 1, device=q.device)
+        acc = torch.zeros(B, H, Tl, D, device=q.device)
+        kc, vc = k, v
+        for s_hop in range(world):
+            j = (rank - s_hop) % world
+            if not causal or j <= rank:
+                sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
+                # every processed chunk has >=1 unmasked key per row (the
+                # diagonal chunk includes key<=query within-chunk), so mj is
+                # finite and m_new is finite from the first hop; the initial
+                # m=-inf gives alpha=exp(-inf)=0 naturally.
+                mj = sc.amax(dim=-1, keepdim=True)
+                m_new = torch.maximum(m, mj)
+                alpha = torch.exp(m - m_new)
+                p = torch.exp(sc - m_new)
+                acc = acc * alpha + torch.matmul(p, vc.float())
+                l = l * alpha + p.sum(dim=-1, keepdim=True)
+                m = m_new
+            if s_hop < world - 1:
+                kc, vc = _ring_swap([kc, vc], rank, world, group)
+        out = (acc / l.clamp(min=1e-30)).to(q.dtype)
+        lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # [B,H,Tl]
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.group = group
+        ctx.causal = causal
+        ctx.world = world
+        ctx.rank = rank
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        group, causal = ctx.group, ctx.causal
+        world, rank, scale = ctx.world, ctx.rank, ctx.scale
+        B, H, Tl, D = q.shape
+        q32, dout32, out32 = q.float(), dout.float(), out.float()
+        delta = (dout32 * out32).sum(dim=-1, keepdim=True)  # [B,H,Tl,1]
+        lse_ = lse.unsqueeze(-1)
+        dq = torch.zeros_like(q32)
+        # grad accumulators travel with their chunk for a full cycle
+        kc, vc = k, v
+        dk_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        dv_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        for s_hop in range(world):
+            j = (rank - s_hop) % world
+            if not causal or j <= rank:
+                sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
+                p = torch.exp(sc - lse_)  # normalized probs of this chunk
+                dp = torch.matmul(dout32, vc.float().transpose(-2, -1))
+                ds = p * (dp - delta) * scale
+                dq = dq + torch.matmul(ds, kc.float())
+                dk_acc = dk_acc + torch.matmul(ds.transpose(-2, -1), q32)
+                dv_acc = dv_acc + torch.matmul(p.transpose(-2, -1), dout32)
+            # rotate kv + their grads one more hop; after `world` hops the
+            # accumulators are back at the chunk owner
+            kc, vc, dk_acc, dv_acc = _ring_swap(
+                [kc, vc, dk_acc, dv_acc], rank, world, group
+            )
+        return (
+            dq.to(q.dtype),
+            dk_acc.to(k.dtype),
+            dv_acc.to(v.dtype),
+            None,
+            None,
+        )
+
+
+def ring_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cp_group,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Ring-form context-parallel attention over [B, H, T/cp, D] shards.
+
+    Peak memory is O(T/cp) per rank for K/V (vs the full-KV all-gather of
+    ``context_parallel_attention``) at the cost of cp-1 (forward) / cp
+    (backward) ring exchanges.  Differentiable; the comm schedule is
+    owned by the op, so pass a DEDICATED cp communicator."""
+    if _ws(cp_group) == 1:
+        return attention(q, k, v, causal=causal)
+    return _RingAttention.apply(q, k, v, cp_group, causal)

@@ -96,3 +96,57 @@ def test_context_parallel_cp2():
 
 def test_context_parallel_cp4():
     run_distributed(_run_cp, 4)
+
+
+def _run_ring(rank, world, causal):
+    import torch.distributed as dist
+
+    from quintnet_amd.parallel import ring_attention, scatter_to_context
+
+    torch.manual_seed(13)
+    B, H, T, D = 2, 2, 64, 16
+    q = torch.randn(B, H, T, D)
+    k = torch.randn(B, H, T, D)
+    v = torch.randn(B, H, T, D)
+    for t in (q, k, v):
+        dist.broadcast(t, src=0)
+
+    qr = q.clone().requires_grad_(True)
+    kr = k.clone().requires_grad_(True)
+    vr = v.clone().requires_grad_(True)
+    ref = _ref_attention(qr, kr, vr, causal)
+    ref.square().sum().backward()
+
+    ql = scatter_to_context(q, dist.group.WORLD, dim=2).requires_grad_(True)
+    kl = scatter_to_context(k, dist.group.WORLD, dim=2).requires_grad_(True)
+    vl = scatter_to_context(v, dist.group.WORLD, dim=2).requires_grad_(True)
+    out = ring_attention(ql, kl, vl, dist.group.WORLD, causal=causal)
+    tl = T // world
+    sl = slice(rank * tl, (rank + 1) * tl)
+    assert torch.allclose(out, ref.detach()[:, :, sl].to(out.dtype), atol=1e-5), (
+        (out - ref.detach()[:, :, sl]).abs().max()
+    )
+    out.square().sum().backward()
+    assert torch.allclose(ql.grad, qr.grad[:, :, sl], atol=1e-5)
+    assert torch.allclose(kl.grad, kr.grad[:, :, sl], atol=1e-5)
+    assert torch.allclose(vl.grad, vr.grad[:, :, sl], atol=1e-5)
+
+
+def _ring_causal(rank, world):
+    _run_ring(rank, world, True)
+
+
+def _ring_full(rank, world):
+    _run_ring(rank, world, False)
+
+
+def test_ring_attention_cp2_causal():
+    run_distributed(_ring_causal, 2)
+
+
+def test_ring_attention_cp4_causal():
+    run_distributed(_ring_causal, 4)
+
+
+def test_ring_attention_cp2_noncausal():
+    run_distributed(_ring_full, 2)
