@@ -51,9 +51,24 @@ def _tp_apply(model: nn.Module, pg, device) -> nn.Module:
     )
 
 
-def _pp_wrap(model: nn.Module, pg, device, stage_module=None) -> nn.Module:
+def _pp_wrap(model: nn.Module, pg, device, stage_module=None, config=None) -> nn.Module:
     if pg.pp_size <= 1 and stage_module is None:
         return model.to(device)
+    sched = str((config or {}).get("schedule", "1f1b")).lower()
+    if sched in ("interleaved", "interleaved_1f1b", "vpp") and stage_module is None:
+        from ..parallel import InterleavedPipelineWrapper
+
+        return InterleavedPipelineWrapper(
+            model,
+            pp_rank=pg.pp_rank,
+            pp_group=pg.get_group("pp") if "pp" in pg.mesh_name else None,
+            pp_size=pg.pp_size,
+            num_chunks=int((config or {}).get("num_chunks", 2)),
+            device=device,
+            tied_group=pg.get_tied_embedding_group()
+            if pg.pp_rank in (0, pg.pp_size - 1)
+            else None,
+        )
     return PipelineParallelWrapper(
         model=model,
         pp_rank=pg.pp_rank,
@@ -77,7 +92,7 @@ class TensorParallelCoordinator(BaseCoordinator):
 
 class PipelineParallelCoordinator(BaseCoordinator):
     def parallelize(self) -> nn.Module:
-        return _pp_wrap(self.model, self.pg, self.device)
+        return _pp_wrap(self.model, self.pg, self.device, config=self.config)
 
 
 class DPTCoordinator(BaseCoordinator):
@@ -90,14 +105,14 @@ class DPTCoordinator(BaseCoordinator):
 
 class DPPCoordinator(BaseCoordinator):
     def parallelize(self) -> nn.Module:
-        m = _pp_wrap(self.model, self.pg, self.device)
+        m = _pp_wrap(self.model, self.pg, self.device, config=self.config)
         return _dp_wrap(m, self.pg)
 
 
 class TPPCoordinator(BaseCoordinator):
     def parallelize(self) -> nn.Module:
         m = _tp_apply(self.model, self.pg, self.device)
-        return _pp_wrap(m, self.pg, self.device)
+        return _pp_wrap(m, self.pg, self.device, config=self.config)
 
 
 class Hybrid3DCoordinator(BaseCoordinator):
@@ -118,7 +133,7 @@ class Hybrid3DCoordinator(BaseCoordinator):
 
     def _parallelize_non_staged(self) -> nn.Module:
         m = _tp_apply(self.model, self.pg, self.device)
-        m = _pp_wrap(m, self.pg, self.device)
+        m = _pp_wrap(m, self.pg, self.device, config=self.config)
         return _dp_wrap(m, self.pg)
 
     def _parallelize_staged(self) -> nn.Module:

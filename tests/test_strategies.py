@@ -74,3 +74,26 @@ def test_tp_pp_composition():
 
 def test_dp_pp_composition():
     run_distributed(_dp_pp, 4)
+
+
+def _pp_interleaved_strategy(rank, world):
+    import torch.distributed as dist
+
+    from quintnet_amd import get_strategy, init_process_groups
+    from quintnet_amd.models import Model
+    from quintnet_amd.parallel import InterleavedPipelineWrapper
+
+    pg = init_process_groups("cpu", [world], ["pp"])
+    torch.manual_seed(5)
+    model = Model(hidden_dim=32, n_heads=2, depth=8)
+    pmodel = get_strategy("pp", pg, {"schedule": "interleaved", "num_chunks": 2}).apply(model)
+    assert isinstance(pmodel, InterleavedPipelineWrapper)
+    assert pmodel.num_chunks == 2
+    # chunked forward works
+    x = torch.randn(2, 50, 32)
+    y = pmodel(x, chunk_id=1)
+    assert y.shape[0] == 2
+
+
+def test_pp_strategy_interleaved_wrapping():
+    run_distributed(_pp_interleaved_strategy, 2)
