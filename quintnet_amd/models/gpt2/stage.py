@@ -109,6 +109,7 @@ class GPT2Stage(nn.Module):
         max_new_tokens: int = 32,
         temperature: float = 0.0,
         top_k: int = 0,
+        eos_token_id: int = None,
     ) -> torch.Tensor:
         """KV-cached autoregressive generation (serving path; pp==1).
 
@@ -153,6 +154,8 @@ class GPT2Stage(nn.Module):
                 past += ids.shape[1]
                 ids = nxt
                 out = torch.cat([out, nxt], dim=1)
+                if eos_token_id is not None and bool((nxt == eos_token_id).all()):
+                    break
                 if past + 1 >= self.config.n_positions:
                     break
         finally:

@@ -109,7 +109,6 @@ class Trainer:
 
     # ------------------------------------------------------------------
     def fit(self) -> Dict[str, float]:
-        history = {}
         wd = None
         if self.config.get("watchdog_timeout_s"):
             from .utils.watchdog import Watchdog
@@ -119,6 +118,14 @@ class Trainer:
                 kill_on_hang=bool(self.config.get("watchdog_kill", False)),
             ).start()
             self._watchdog = wd
+        try:
+            return self._fit_inner(wd)
+        finally:
+            if wd:
+                wd.stop()
+
+    def _fit_inner(self, wd) -> Dict[str, float]:
+        history = {}
         for epoch in range(self.num_epochs):
             t0 = time.time()
             train_metrics = self._train_epoch(epoch)
@@ -137,8 +144,6 @@ class Trainer:
             if every and (epoch + 1) % every == 0 and epoch + 1 < self.num_epochs:
                 self._save_checkpoint()
         self._save_checkpoint()
-        if wd:
-            wd.stop()
         return history
 
     def _is_rank0(self) -> bool:
