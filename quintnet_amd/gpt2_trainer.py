@@ -53,6 +53,13 @@ class GPT2Trainer(Trainer):
             self.pipeline_trainer.optimizer = self.optimizer
             self.pipeline_trainer.criterion = self.criterion
         resume = config.get("resume_from")
+        if resume == "auto":
+            # elastic restart convenience: resume from checkpoint_dir when
+            # its shards exist, start fresh otherwise
+            cdir = config.get("checkpoint_dir") or config.get("output_dir")
+            name = config.get("checkpoint_name", "final_model")
+            probe = os.path.join(cdir or "", f"{name}_pp0_tp0.pt")
+            resume = cdir if cdir and os.path.exists(probe) else None
         if resume:
             from .checkpoint import load_sharded_checkpoint
 

@@ -162,3 +162,28 @@ def test_unwrap_peels_pipeline_wrapper():
     assert inner is stage
     assert all(not k.startswith("local_module.") for k in inner.state_dict())
     assert any(k.startswith("embedding.wte") for k in inner.state_dict())
+
+
+def test_auto_resume(tmp_path):
+    """resume_from='auto': fresh start without shards, resume when the
+    checkpoint_dir holds them (elastic-restart convenience)."""
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=16, n_embd=16, n_layer=1, n_head=2)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    tcfg = {"num_epochs": 1, "grad_acc_steps": 2, "zero1": False,
+            "checkpoint_dir": str(tmp_path), "resume_from": "auto",
+            "task_type": "clm"}
+    # run 1: nothing to resume -> trains and saves
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    GPT2Trainer(stage, DataLoader(ds, batch_size=2), None, tcfg, None).fit()
+    assert (tmp_path / "final_model_pp0_tp0.pt").exists()
+    w0 = stage.state_dict()["embedding.wte.weight"].clone()
+    # run 2: auto-resume loads the saved weights into a fresh model
+    stage2 = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    tr2 = GPT2Trainer(stage2, DataLoader(ds, batch_size=2), None, tcfg, None)
+    assert torch.allclose(stage2.state_dict()["embedding.wte.weight"], w0)
