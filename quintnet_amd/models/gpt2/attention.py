@@ -68,7 +68,9 @@ class GPT2Attention(nn.Module):
         """Incremental decode: append this step's K/V to the cache and
         attend the new queries against the whole cache (causal diagonal
         shifted by the cache length — the same q_offset machinery the
-        context-parallel path uses)."""
+        context-parallel path uses).  ``kv_cache['quant'] == 'int8'``
+        stores the cache as per-token-per-head symmetric int8 (+fp scale),
+        halving KV memory for long-context serving."""
         from ...ops.attention import attention
 
         B, T, _ = qkv.shape
@@ -81,11 +83,34 @@ class GPT2Attention(nn.Module):
         q = heads(qkv[:, :, :hl])
         k = heads(qkv[:, :, hl : 2 * hl])
         v = heads(qkv[:, :, 2 * hl :])
-        if kv_cache.get("k") is None:
-            kv_cache["k"], kv_cache["v"] = k, v
+        if kv_cache.get("quant") == "int8":
+            kf, vf = self._append_int8(kv_cache, k, v)
         else:
-            kv_cache["k"] = torch.cat([kv_cache["k"], k], dim=2)
-            kv_cache["v"] = torch.cat([kv_cache["v"], v], dim=2)
-        past = kv_cache["k"].shape[2] - T
-        out = attention(q, kv_cache["k"], kv_cache["v"], causal=True, q_offset=past)
+            if kv_cache.get("k") is None:
+                kv_cache["k"], kv_cache["v"] = k, v
+            else:
+                kv_cache["k"] = torch.cat([kv_cache["k"], k], dim=2)
+                kv_cache["v"] = torch.cat([kv_cache["v"], v], dim=2)
+            kf, vf = kv_cache["k"], kv_cache["v"]
+        past = kf.shape[2] - T
+        out = attention(q, kf, vf, causal=True, q_offset=past)
         return out.permute(0, 2, 1, 3).reshape(B, T, hl)
+
+    @staticmethod
+    def _append_int8(kv_cache: dict, k: torch.Tensor, v: torch.Tensor):
+        def quant(t):
+            s = t.abs().amax(dim=-1, keepdim=True).float().clamp(min=1e-8) / 127.0
+            return (t.float() / s).round().clamp(-127, 127).to(torch.int8), s
+
+        k8, ks = quant(k)
+        v8, vs = quant(v)
+        if kv_cache.get("k8") is None:
+            kv_cache.update(k8=k8, ks=ks, v8=v8, vs=vs)
+        else:
+            kv_cache["k8"] = torch.cat([kv_cache["k8"], k8], dim=2)
+            kv_cache["ks"] = torch.cat([kv_cache["ks"], ks], dim=2)
+            kv_cache["v8"] = torch.cat([kv_cache["v8"], v8], dim=2)
+            kv_cache["vs"] = torch.cat([kv_cache["vs"], vs], dim=2)
+        kf = (kv_cache["k8"].float() * kv_cache["ks"]).to(k.dtype)
+        vf = (kv_cache["v8"].float() * kv_cache["vs"]).to(v.dtype)
+        return kf, vf

@@ -110,6 +110,7 @@ class GPT2Stage(nn.Module):
         temperature: float = 0.0,
         top_k: int = 0,
         eos_token_id: int = None,
+        cache_dtype: str = None,
     ) -> torch.Tensor:
         """KV-cached autoregressive generation (serving path; pp==1).
 
@@ -126,7 +127,11 @@ class GPT2Stage(nn.Module):
         )
         was_training = self.training
         self.eval()
-        caches = [{"k": None, "v": None} for _ in self.blocks]
+        caches = [
+            {"k": None, "v": None, "k8": None,
+             "quant": "int8" if cache_dtype == "int8" else None}
+            for _ in self.blocks
+        ]
         ids = input_ids
         out = input_ids
         past = 0
