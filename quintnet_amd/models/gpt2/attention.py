@@ -26,9 +26,11 @@ __all__ = ["GPT2Attention"]
 
 
 class GPT2Attention(nn.Module):
-    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
+    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None,
+                 cp_group=None):
         super().__init__()
         self.config = config
+        self.cp_group = cp_group
         sp = config.sequence_parallel
         self.c_attn = ColumnParallelLinear(
             config.n_embd,
@@ -58,11 +60,32 @@ class GPT2Attention(nn.Module):
 
     def forward(self, x: torch.Tensor, kv_cache=None) -> torch.Tensor:
         qkv = self.c_attn(x)  # [B, T, 3*n_embd/tp] = [q_loc | k_loc | v_loc]
-        if kv_cache is None:
-            out = attention_qkv(qkv, self.n_head_local, causal=True)
-        else:
+        if kv_cache is not None:
             out = self._forward_cached(qkv, kv_cache)
+        elif self.cp_group is not None:
+            out = self._forward_cp(qkv)
+        else:
+            out = attention_qkv(qkv, self.n_head_local, causal=True)
         return self.resid_dropout(self.c_proj(out))
+
+    def _forward_cp(self, qkv: torch.Tensor) -> torch.Tensor:
+        """Context parallelism: x is this rank's sequence shard; attention
+        runs against the all-gathered K/V with the shifted causal diagonal
+        (parallel/context_parallel.py)."""
+        from ...parallel.context_parallel import context_parallel_attention
+
+        B, T, _ = qkv.shape
+        H, D = self.n_head_local, self.head_dim
+        hl = self.hidden_local
+
+        def heads(t):
+            return t.reshape(B, T, H, D).permute(0, 2, 1, 3)
+
+        q = heads(qkv[:, :, :hl])
+        k = heads(qkv[:, :, hl : 2 * hl])
+        v = heads(qkv[:, :, 2 * hl :])
+        out = context_parallel_attention(q, k, v, self.cp_group, causal=True)
+        return out.permute(0, 2, 1, 3).reshape(B, T, hl)
 
     def _forward_cached(self, qkv: torch.Tensor, kv_cache: dict) -> torch.Tensor:
         """Incremental decode: append this step's K/V to the cache and

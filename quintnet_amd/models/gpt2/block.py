@@ -18,10 +18,12 @@ __all__ = ["GPT2Block"]
 
 
 class GPT2Block(nn.Module):
-    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
+    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None,
+                 cp_group=None):
         super().__init__()
         self.ln_1 = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, device=device, dtype=dtype)
-        self.attn = GPT2Attention(config, tp_group=tp_group, device=device, dtype=dtype)
+        self.attn = GPT2Attention(config, tp_group=tp_group, device=device, dtype=dtype,
+                                  cp_group=cp_group)
         self.ln_2 = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, device=device, dtype=dtype)
         self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)
 

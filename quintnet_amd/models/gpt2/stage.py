@@ -41,6 +41,7 @@ class GPT2Stage(nn.Module):
         tied_group=None,
         device=None,
         dtype=None,
+        cp_group=None,
     ):
         super().__init__()
         self.config = config
@@ -48,6 +49,10 @@ class GPT2Stage(nn.Module):
         self.pp_size = pp_size
         self.tp_group = tp_group
         self.tied_group = tied_group
+        self.cp_group = cp_group
+        if cp_group is not None:
+            assert pp_size == 1, "context parallelism: pp composition is round-2"
+            assert not config.sequence_parallel, "CP and Megatron-SP are exclusive"
         self.sequence_parallel = config.sequence_parallel
         self.is_first_stage = pp_rank == 0
         self.is_last_stage = pp_rank == pp_size - 1
@@ -59,7 +64,8 @@ class GPT2Stage(nn.Module):
         if self.is_first_stage:
             self.embedding = GPT2Embedding(config, **kw)
         self.blocks = nn.ModuleList(
-            GPT2Block(config, tp_group=tp_group, **kw) for _ in self.my_layers
+            GPT2Block(config, tp_group=tp_group, cp_group=cp_group, **kw)
+            for _ in self.my_layers
         )
         if self.is_last_stage:
             self.ln_f = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, **kw)
@@ -75,7 +81,16 @@ class GPT2Stage(nn.Module):
     # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         if self.is_first_stage:
-            x = self.embedding(x)
+            if self.cp_group is not None:
+                # context parallelism: x is this rank's ids SHARD (caller
+                # scatters, see parallel/context_parallel.py); positions
+                # start at the shard's global offset
+                import torch.distributed as dist
+
+                cp_rank = dist.get_rank(group=self.cp_group)
+                x = self.embedding(x, pos_offset=cp_rank * x.shape[1])
+            else:
+                x = self.embedding(x)
             if self.sequence_parallel:
                 # enter SP: keep only this rank's sequence shard
                 x = scatter_to_sequence(x, self.tp_group, 1)

@@ -9,7 +9,9 @@ from .tensor_parallel import (
 )
 from .context_parallel import (
     context_parallel_attention,
+    cp_causal_lm_loss,
     ring_attention,
+    scatter_clm_targets,
     scatter_to_context,
 )
 from .zero3 import ZeRO3Block, apply_zero3
@@ -28,6 +30,8 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "cp_causal_lm_loss",
+    "scatter_clm_targets",
     "ring_attention",
     "ZeRO3Block",
     "apply_zero3",

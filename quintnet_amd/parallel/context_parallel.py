@@ -21,7 +21,13 @@ import torch.distributed as dist
 from ..core.comm import All_Gather
 from ..ops.attention import attention
 
-__all__ = ["context_parallel_attention", "ring_attention", "scatter_to_context"]
+__all__ = [
+    "context_parallel_attention",
+    "ring_attention",
+    "scatter_to_context",
+    "scatter_clm_targets",
+    "cp_causal_lm_loss",
+]
 
 
 def _ws(group) -> int:
@@ -198,3 +204,52 @@ def ring_attention(
     if _ws(cp_group) == 1:
         return attention(q, k, v, causal=causal)
     return _RingAttention.apply(q, k, v, cp_group, causal)
+
+
+# ---------------------------------------------------------------------------
+# CP training helpers: with every activation sequence-sharded end to end
+# (no gather anywhere), every parameter gradient is a PARTITION of the full
+# gradient — so a DataParallel wrap over the cp group with the loss
+# pre-scaled by cp gives exactly ∇L after its MEAN reduction:
+#   grad = (1/cp) Σ_r ∇(cp · L_r) = Σ_r ∇L_r = ∇L.
+# The local loss L_r is CE summed over this rank's shard divided by the
+# GLOBAL valid-token count (all-reduced, non-differentiable).
+# ---------------------------------------------------------------------------
+def scatter_clm_targets(labels: torch.Tensor, cp_group, ignore_index: int = -100):
+    """Shift the FULL label sequence for causal LM (so cross-shard next
+    tokens are correct), then keep this rank's shard."""
+    from ..ops import shift_labels
+
+    return scatter_to_context(shift_labels(labels, ignore_index), cp_group, dim=1)
+
+
+def cp_causal_lm_loss(
+    logits_shard: torch.Tensor,
+    targets_shard: torch.Tensor,
+    cp_group,
+    ignore_index: int = -100,
+):
+    """Returns (loss_for_backward, true_loss_detached).
+
+    ``loss_for_backward`` is cp * CE_sum(shard) / N_total — backward it on
+    every rank and let a DataParallel(cp_group) MEAN-reduce the gradients.
+    ``true_loss`` is the exact global mean CE for logging.
+    """
+    world = _ws(cp_group)
+    ce_sum = torch.nn.functional.cross_entropy(
+        logits_shard.float().reshape(-1, logits_shard.shape[-1]),
+        targets_shard.reshape(-1),
+        ignore_index=ignore_index,
+        reduction="sum",
+    )
+    n_local = (targets_shard != ignore_index).sum()
+    n_total = n_local.clone()
+    if world > 1:
+        dist.all_reduce(n_total, group=cp_group)
+    loss_bwd = ce_sum / n_total.clamp(min=1) * world
+    with torch.no_grad():
+        true = ce_sum.detach().clone()
+        if world > 1:
+            dist.all_reduce(true, group=cp_group)
+        true = true / n_total.clamp(min=1)
+    return loss_bwd, true
