@@ -220,3 +220,76 @@ def test_gpt2_context_parallel_training_step():
 
 def test_gpt2_context_parallel_cp4():
     run_distributed(_run_gpt2_cp, 4)
+
+
+def _run_gpt2_cp_tp(rank, world):
+    """CP x TP composition: heads sharded by tp, sequence by cp; grads of
+    the TP shards must match slices of the full-model gradients."""
+    import torch.distributed as dist
+
+    from test_gpt2_pipeline import _shard_full_stage_sd
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        cp_causal_lm_loss,
+        scatter_clm_targets,
+        scatter_to_context,
+    )
+
+    pg = init_process_groups("cpu", [2, 2], ["dp", "tp"])  # dp axis = cp here
+    cp_group = pg.get_group("dp")
+    cp_rank, cp_size = pg.dp_rank, pg.dp_size
+
+    torch.manual_seed(41)
+    cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    full = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    for p in full.parameters():
+        dist.broadcast(p.data, src=0)
+
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=pg.get_group("tp"),
+                      cp_group=cp_group)
+    stage.load_state_dict(
+        _shard_full_stage_sd(full.state_dict(), cfg, 0, 1, pg.tp_rank,
+                             pg.tp_size, stage),
+        strict=False,
+    )
+    model = DataParallel(stage, DistributedConfig(cp_rank, cp_size, cp_group))
+
+    ids = torch.randint(0, 96, (2, 32))
+    labels = ids.clone()
+    dist.broadcast(ids, src=0)
+    dist.broadcast(labels, src=0)
+    ids_shard = scatter_to_context(ids, cp_group, dim=1)
+    tgt_shard = scatter_clm_targets(labels, cp_group)
+    logits_shard = model(ids_shard)
+    loss_bwd, true_loss = cp_causal_lm_loss(logits_shard, tgt_shard, cp_group)
+    loss_bwd.backward()
+    model.finalize_gradients()
+
+    out = full(ids)
+    ref_loss = causal_lm_loss(out, labels, ignore_index=-100)
+    ref_loss.backward()
+    assert abs(float(true_loss) - float(ref_loss)) < 1e-5
+
+    # spot-check a TP-sharded grad (c_fc rows) and a replicated one (wte)
+    rp = dict(full.named_parameters())
+    inner_loc = cfg.n_inner // pg.tp_size
+    isl = slice(pg.tp_rank * inner_loc, (pg.tp_rank + 1) * inner_loc)
+    got = dict(stage.named_parameters())
+    assert torch.allclose(
+        got["blocks.0.mlp.c_fc.weight"].grad,
+        rp["blocks.0.mlp.c_fc.weight"].grad[isl], atol=2e-5,
+    )
+    assert torch.allclose(
+        got["embedding.wte.weight"].grad,
+        rp["embedding.wte.weight"].grad, atol=2e-5,
+    )
+
+
+def test_gpt2_cp_tp_composition():
+    run_distributed(_run_gpt2_cp_tp, 4)
