@@ -152,9 +152,20 @@ class ZeRO3Block(nn.Module):
             return {k: v.clone() for k, v in self._param_views(full).items()}
 
 
-def apply_zero3(model: nn.Module, dp_group=None, attr: str = "blocks") -> nn.Module:
-    """Shard every element of ``model.<attr>`` (a ModuleList) in place."""
+def apply_zero3(
+    model: nn.Module,
+    dp_group=None,
+    attr: str = "blocks",
+    extra_attrs: tuple = (),
+) -> nn.Module:
+    """Shard every element of ``model.<attr>`` (a ModuleList) in place;
+    ``extra_attrs`` names additional single modules (e.g. "embedding",
+    "classification_head") to shard the same way — with every trainable
+    module wrapped, NO separate DP sync is needed (each shard's gradient
+    is already reduce-scattered across the group)."""
     blocks = getattr(model, attr)
     wrapped = nn.ModuleList(ZeRO3Block(b, dp_group) for b in blocks)
     setattr(model, attr, wrapped)
+    for name in extra_attrs:
+        setattr(model, name, ZeRO3Block(getattr(model, name), dp_group))
     return model
