@@ -14,6 +14,7 @@ from .context_parallel import (
     scatter_clm_targets,
     scatter_to_context,
 )
+from .expert_parallel import ExpertParallelMLP, all_to_all_var
 from .zero3 import ZeRO3Block, apply_zero3
 from .pipeline import (
     InterleavedPipelineWrapper,
@@ -30,6 +31,8 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "ExpertParallelMLP",
+    "all_to_all_var",
     "cp_causal_lm_loss",
     "scatter_clm_targets",
     "ring_attention",
