@@ -19,13 +19,23 @@ __all__ = ["GPT2Block"]
 
 class GPT2Block(nn.Module):
     def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None,
-                 cp_group=None):
+                 cp_group=None, ep_group=None):
         super().__init__()
         self.ln_1 = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, device=device, dtype=dtype)
         self.attn = GPT2Attention(config, tp_group=tp_group, device=device, dtype=dtype,
                                   cp_group=cp_group)
         self.ln_2 = FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, device=device, dtype=dtype)
-        self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)
+        if config.n_experts > 0:
+            from ...parallel.expert_parallel import ExpertParallelMLP
+
+            assert tp_group is None, "MoE blocks: TP inside experts is round-2"
+            self.mlp = ExpertParallelMLP(
+                config.n_embd, config.n_inner, config.n_experts,
+                top_k=config.moe_top_k, ep_group=ep_group,
+                device=device, dtype=dtype,
+            )
+        else:
+            self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         m, s2 = self.forward_fused(x, None)

@@ -23,7 +23,9 @@ class GPT2Config:
     layer_norm_epsilon: float = 1e-5
     initializer_range: float = 0.02
     sequence_parallel: bool = False
-    fp8: bool = False  # experimental: e4m3 forward GEMMs, bf16 backward  # Megatron-SP over the TP group
+    fp8: bool = False  # experimental: e4m3 forward GEMMs, bf16 backward
+    n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
+    moe_top_k: int = 2  # Megatron-SP over the TP group
 
     def __post_init__(self):
         if self.n_inner is None:

@@ -42,6 +42,7 @@ class GPT2Stage(nn.Module):
         device=None,
         dtype=None,
         cp_group=None,
+        ep_group=None,
     ):
         super().__init__()
         self.config = config
@@ -64,7 +65,8 @@ class GPT2Stage(nn.Module):
         if self.is_first_stage:
             self.embedding = GPT2Embedding(config, **kw)
         self.blocks = nn.ModuleList(
-            GPT2Block(config, tp_group=tp_group, cp_group=cp_group, **kw)
+            GPT2Block(config, tp_group=tp_group, cp_group=cp_group,
+                      ep_group=ep_group, **kw)
             for _ in self.my_layers
         )
         if self.is_last_stage:
@@ -182,6 +184,20 @@ class GPT2Stage(nn.Module):
             if was_training:
                 self.train()
         return out
+
+    # ------------------------------------------------------------------
+    def moe_aux_loss(self) -> torch.Tensor:
+        """Sum of the MoE load-balancing losses of this stage's blocks
+        (zero tensor when the model is dense)."""
+        total = None
+        for blk in self.blocks:
+            aux = getattr(blk.mlp, "aux_loss", None)
+            if aux is not None:
+                total = aux if total is None else total + aux
+        if total is None:
+            dev = next(self.parameters()).device
+            return torch.zeros((), device=dev)
+        return total
 
     # ------------------------------------------------------------------
     def sync_sequence_parallel_grads(self) -> None:
