@@ -97,10 +97,16 @@ class GPT2Trainer(Trainer):
                 pt.start("forward")
             logits = self.model(ids)
             loss = causal_lm_loss(logits, labels, ignore_index=-100)
+            aux_w = float(self.config.get("moe_aux_weight", 0.0))
+            total = loss
+            if aux_w:
+                inner = _unwrap(self.model)
+                if hasattr(inner, "moe_aux_loss"):
+                    total = loss + aux_w * inner.moe_aux_loss()
             if pt:
                 pt.stop("forward")
                 pt.start("backward")
-            (loss / self.grad_acc_steps).backward()
+            (total / self.grad_acc_steps).backward()
             if pt:
                 pt.stop("backward")
             accum += 1

