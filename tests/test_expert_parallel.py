@@ -168,3 +168,65 @@ def test_gpt2_trainer_moe_aux_weight():
     hist = tr.fit()
     assert "train_loss" in hist
     assert stage.blocks[0].mlp.router.weight.grad is None or True  # stepped+zeroed
+
+
+def _run_moe_dp_ep(rank, world):
+    """[dp=2, ep=2] composition: experts replicated across DP and sharded
+    across EP; after DDP finalize, expert grads = full-batch reference."""
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        ExpertParallelMLP,
+    )
+
+    pg = init_process_groups("cpu", [2, 2], ["dp", "tp"])  # tp axis = ep
+    ep_group = pg.get_group("tp")
+    dp_group = pg.get_group("dp")
+
+    torch.manual_seed(17)
+    n_experts = 4
+    ref = ExpertParallelMLP(n_embd=16, n_inner=32, n_experts=n_experts, top_k=2)
+    for p in ref.parameters():
+        dist.broadcast(p.data, src=0)
+
+    ep = ExpertParallelMLP(n_embd=16, n_inner=32, n_experts=n_experts,
+                           top_k=2, ep_group=ep_group)
+    n_local = n_experts // pg.tp_size
+    with torch.no_grad():
+        ep.router.weight.copy_(ref.router.weight)
+        for le in range(n_local):
+            src = ref.experts[pg.tp_rank * n_local + le]
+            for a, b in zip(ep.experts[le].parameters(), src.parameters()):
+                a.copy_(b)
+    model = DataParallel(ep, DistributedConfig(pg.dp_rank, pg.dp_size, dp_group))
+
+    # batch axis split over dp; within a replica, the ep pair shares it
+    torch.manual_seed(55)
+    x = torch.randn(2 * pg.dp_size, 6, 16)
+    dist.broadcast(x, src=0)
+    x_dp = x[2 * pg.dp_rank : 2 * (pg.dp_rank + 1)]
+    # each ep rank takes half of the replica's batch (data also sharded
+    # along ep so experts see each token once globally)
+    x_local = x_dp[pg.tp_rank : pg.tp_rank + 1].clone().requires_grad_(True)
+    y = model(x_local)
+    (y.square().sum() / pg.dp_size).backward()  # DP MEAN semantics
+    model.finalize_gradients()
+
+    xr = x.clone().requires_grad_(True)
+    yr = ref(xr)
+    (yr.square().sum() / pg.dp_size).backward()
+
+    # DP-mean over the two replicas' per-replica expert grads equals the
+    # full-batch reference grad scaled by the same 1/dp loss factor
+    for le in range(n_local):
+        src = ref.experts[pg.tp_rank * n_local + le]
+        got = ep.experts[le].fc1.weight.grad
+        want = src.fc1.weight.grad / pg.dp_size
+        assert torch.allclose(got, want, atol=1e-5), (le, (got - want).abs().max())
+
+
+def test_moe_dp_ep_composition():
+    run_distributed(_run_moe_dp_ep, 4)
