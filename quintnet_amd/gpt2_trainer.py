@@ -84,19 +84,42 @@ class GPT2Trainer(Trainer):
         return seq, hidden
 
     # ------------------------------------------------------------------
+    @property
+    def _cp_group(self):
+        if (
+            self.config.get("context_parallel")
+            and self.pg is not None
+            and "cp" in getattr(self.pg, "mesh_name", ())
+        ):
+            return self.pg.get_group("cp")
+        return None
+
     def _train_epoch_plain(self) -> Dict[str, float]:
         total_loss, total_tokens, steps = 0.0, 0, 0
         accum = 0
         from .ops import causal_lm_loss
 
+        cp_group = self._cp_group
         pt = self._phase_timer()
         for batch in self.train_loader:
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             labels = batch["labels"].to(self.device, non_blocking=True)
             if pt:
                 pt.start("forward")
-            logits = self.model(ids)
-            loss = causal_lm_loss(logits, labels, ignore_index=-100)
+            if cp_group is not None:
+                from .parallel import (
+                    cp_causal_lm_loss,
+                    scatter_clm_targets,
+                    scatter_to_context,
+                )
+
+                ids_shard = scatter_to_context(ids, cp_group, dim=1)
+                tgt_shard = scatter_clm_targets(labels, cp_group)
+                logits = self.model(ids_shard)
+                loss, true_loss = cp_causal_lm_loss(logits, tgt_shard, cp_group)
+            else:
+                logits = self.model(ids)
+                loss = causal_lm_loss(logits, labels, ignore_index=-100)
             aux_w = float(self.config.get("moe_aux_weight", 0.0))
             total = loss
             if aux_w:
@@ -110,7 +133,7 @@ class GPT2Trainer(Trainer):
             if pt:
                 pt.stop("backward")
             accum += 1
-            total_loss += float(loss.detach())
+            total_loss += float(true_loss if cp_group is not None else loss.detach())
             total_tokens += int((labels[:, 1:] != -100).sum())
             steps += 1
             if accum == self.grad_acc_steps:
