@@ -15,7 +15,7 @@ the output; add ``aux_weight * aux`` to the training loss.
 
 from __future__ import annotations
 
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 import torch
 import torch.distributed as dist
