@@ -230,3 +230,104 @@ def _run_moe_dp_ep(rank, world):
 
 def test_moe_dp_ep_composition():
     run_distributed(_run_moe_dp_ep, 4)
+
+
+def _run_moe_pp_ep(rank, world):
+    """[pp=2, ep=2] composition: MoE blocks inside pipeline stages; EP
+    peers share a stage so their collective sequencing is identical.
+    Loss trajectory must match a single-process MoE run."""
+    import copy
+
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel import (
+        PipelineDataLoader,
+        PipelineParallelWrapper,
+        PipelineTrainer,
+    )
+    from quintnet_amd.utils.data import SyntheticCLM
+    from torch.utils.data import DataLoader
+
+    pg = init_process_groups("cpu", [2, 2], ["tp", "pp"])  # tp axis = ep
+    ep_group = pg.get_group("tp")
+    torch.manual_seed(71)
+    cfg = GPT2Config(vocab_size=96, n_positions=16, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0, n_experts=2, moe_top_k=1)
+    full = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    for p in full.parameters():
+        dist.broadcast(p.data, src=0)
+
+    stage = GPT2Stage(cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+                      tp_group=None, ep_group=ep_group,
+                      tied_group=pg.get_tied_embedding_group())
+    # copy weights: blocks (incl. router + this rank's experts), emb, head
+    sd = full.state_dict()
+    off = stage.layer_distribution[pg.pp_rank][0]
+    n_local = cfg.n_experts // pg.tp_size
+    tgt = {}
+    for i, _ in enumerate(stage.my_layers):
+        src = f"blocks.{i + off}"
+        for k in ("ln_1.weight", "ln_1.bias", "ln_2.weight", "ln_2.bias",
+                  "attn.c_attn.weight", "attn.c_attn.bias",
+                  "attn.c_proj.weight", "attn.c_proj.bias",
+                  "mlp.router.weight"):
+            tgt[f"blocks.{i}.{k}"] = sd[f"{src}.{k}"]
+        for le in range(n_local):
+            ge = pg.tp_rank * n_local + le
+            for k in ("fc1.weight", "fc1.bias", "fc2.weight", "fc2.bias"):
+                tgt[f"blocks.{i}.mlp.experts.{le}.{k}"] = sd[f"{src}.mlp.experts.{ge}.{k}"]
+    if stage.is_first_stage:
+        tgt["embedding.wte.weight"] = sd["embedding.wte.weight"]
+        tgt["embedding.wpe.weight"] = sd["embedding.wpe.weight"]
+    if stage.is_last_stage and not stage.is_first_stage:
+        tgt["ln_f.weight"] = sd["ln_f.weight"]
+        tgt["ln_f.bias"] = sd["ln_f.bias"]
+        tgt["lm_head"] = sd["embedding.wte.weight"].clone()
+    stage.load_state_dict(tgt, strict=False)
+    stage.seq_len, stage.hidden_dim = 16, cfg.n_embd
+
+    wrapper = PipelineParallelWrapper(
+        stage_module=stage, pp_rank=pg.pp_rank, pp_group=pg.get_group("pp"),
+        pp_size=pg.pp_size,
+    )
+    wrapper.seq_len, wrapper.hidden_dim = 16, cfg.n_embd
+    opt = torch.optim.Adam(wrapper.parameters(), lr=1e-3)
+    pt = PipelineTrainer(
+        model=wrapper, optimizer=opt, criterion=None,
+        pp_rank=pg.pp_rank, pp_size=pg.pp_size, pp_group=pg.get_group("pp"),
+        pp_group_ranks=pg.get_group_ranks("pp"), schedule="1f1b",
+        task_type="clm", max_grad_norm=None,
+    )
+    ds = SyntheticCLM(n=8, seq_len=16, vocab_size=96, seed=6)
+    dl = DataLoader(ds, batch_size=2, shuffle=False)
+    loader = PipelineDataLoader(dl, grad_acc_steps=2, task_type="clm")
+    losses = []
+    for _ in range(2):
+        m = pt.train_step(loader, (2, 16, 32), torch.device("cpu"), torch.float32)
+        if pg.pp_rank == pg.pp_size - 1:
+            losses.append(m["loss"])
+
+    if rank == world - 1 and pg.pp_rank == pg.pp_size - 1:
+        opt_r = torch.optim.Adam(full.parameters(), lr=1e-3)
+        it = iter(PipelineDataLoader(dl, 2, "clm"))
+        ref_losses = []
+        for _ in range(2):
+            opt_r.zero_grad()
+            tot = 0.0
+            for _ in range(2):
+                b = next(it)
+                loss = causal_lm_loss(full(b["input_ids"]), b["labels"], ignore_index=-100)
+                (loss / 2).backward()
+                tot += float(loss.detach())
+            full.sync_tied_weights_grad()
+            opt_r.step()
+            ref_losses.append(tot / 2)
+        for a, b in zip(losses, ref_losses):
+            assert abs(a - b) < 5e-4, (losses, ref_losses)
+
+
+def test_moe_pp_ep_composition():
+    run_distributed(_run_moe_pp_ep, 4)
