@@ -52,7 +52,6 @@ class GPT2Stage(nn.Module):
         self.tied_group = tied_group
         self.cp_group = cp_group
         if cp_group is not None:
-            assert pp_size == 1, "context parallelism: pp composition is round-2"
             assert not config.sequence_parallel, "CP and Megatron-SP are exclusive"
         self.sequence_parallel = config.sequence_parallel
         self.is_first_stage = pp_rank == 0

@@ -96,16 +96,35 @@ class PipelineSchedule(abc.ABC):
             else:
                 return
 
+    @property
+    def cp_group(self):
+        return getattr(self.trainer, "cp_group", None)
+
     def _stage_input(self, batch: Dict[str, Any], device) -> torch.Tensor:
         if self.task_type == "clm":
-            return batch["input_ids"].to(device, non_blocking=True)
+            ids = batch["input_ids"].to(device, non_blocking=True)
+            if self.cp_group is not None:
+                from ..context_parallel import scatter_to_context
+
+                ids = scatter_to_context(ids, self.cp_group, dim=1)
+            return ids
         return batch["images"].to(device, non_blocking=True)
 
     def _loss_and_metrics(
         self, output: torch.Tensor, batch: Dict[str, Any], device, metrics: Dict[str, float]
     ) -> torch.Tensor:
         labels = batch["labels"].to(device, non_blocking=True)
-        if self.task_type == "clm":
+        if self.task_type == "clm" and self.cp_group is not None:
+            from ..context_parallel import cp_causal_lm_loss, scatter_clm_targets
+
+            tgt = scatter_clm_targets(labels, self.cp_group)
+            loss, true_loss = cp_causal_lm_loss(output, tgt, self.cp_group)
+            with torch.no_grad():
+                metrics["loss"] = metrics.get("loss", 0.0) + float(true_loss)
+                metrics["n_tokens"] = metrics.get("n_tokens", 0) + int(
+                    (tgt != -100).sum()
+                )
+        elif self.task_type == "clm":
             targets = shift_labels(labels, -100)
             loss = causal_lm_loss(output, labels, ignore_index=-100)
             with torch.no_grad():

@@ -30,10 +30,12 @@ class PipelineTrainer:
         max_grad_norm: Optional[float] = 1.0,
         pp_fwd_group=None,
         pp_bwd_group=None,
+        cp_group=None,
     ):
         self.model = model
         self.pp_fwd_group = pp_fwd_group
         self.pp_bwd_group = pp_bwd_group
+        self.cp_group = cp_group
         self.optimizer = optimizer
         self.criterion = criterion
         self.pp_rank = pp_rank

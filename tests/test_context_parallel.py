@@ -350,3 +350,89 @@ def _run_cp_trainer(rank, world):
 
 def test_gpt2_trainer_context_parallel():
     run_distributed(_run_cp_trainer, 2)
+
+
+def _run_cp_pp(rank, world):
+    """[cp=2, pp=2]: pipeline over sequence shards; inter-stage tensors
+    are [B, T/cp, H]; loss trajectory matches single process."""
+    import torch.distributed as dist
+
+    from test_gpt2_pipeline import _shard_full_stage_sd
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        PipelineDataLoader,
+        PipelineParallelWrapper,
+        PipelineTrainer,
+    )
+    from quintnet_amd.utils.data import SyntheticCLM
+    from torch.utils.data import DataLoader
+
+    pg = init_process_groups("cpu", [2, 2], ["dp", "pp"])  # dp axis = cp
+    cp_group = pg.get_group("dp")
+    torch.manual_seed(81)
+    seq = 32
+    cfg = GPT2Config(vocab_size=96, n_positions=seq, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    full = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
+    for p in full.parameters():
+        dist.broadcast(p.data, src=0)
+
+    stage = GPT2Stage(cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+                      tp_group=None, cp_group=cp_group,
+                      tied_group=pg.get_tied_embedding_group())
+    stage.load_state_dict(
+        _shard_full_stage_sd(full.state_dict(), cfg, pg.pp_rank, pg.pp_size,
+                             0, 1, stage),
+        strict=False,
+    )
+    stage.seq_len, stage.hidden_dim = seq, cfg.n_embd
+    wrapper = PipelineParallelWrapper(
+        stage_module=stage, pp_rank=pg.pp_rank, pp_group=pg.get_group("pp"),
+        pp_size=pg.pp_size,
+    )
+    model = DataParallel(
+        wrapper, DistributedConfig(pg.dp_rank, pg.dp_size, cp_group)
+    )
+    opt = torch.optim.Adam(model.parameters(), lr=1e-3)
+    pt = PipelineTrainer(
+        model=model, optimizer=opt, criterion=None,
+        pp_rank=pg.pp_rank, pp_size=pg.pp_size, pp_group=pg.get_group("pp"),
+        pp_group_ranks=pg.get_group_ranks("pp"), schedule="1f1b",
+        task_type="clm", max_grad_norm=None, cp_group=cp_group,
+    )
+    ds = SyntheticCLM(n=8, seq_len=seq, vocab_size=96, seed=12)
+    dl = DataLoader(ds, batch_size=2, shuffle=False)
+    loader = PipelineDataLoader(dl, grad_acc_steps=2, task_type="clm")
+    shapes = (2, seq // pg.dp_size, cfg.n_embd)  # inter-stage seq SHARD
+    losses = []
+    for _ in range(2):
+        m = pt.train_step(loader, shapes, torch.device("cpu"), torch.float32)
+        if pg.pp_rank == pg.pp_size - 1:
+            losses.append(m["loss"])
+
+    if pg.pp_rank == pg.pp_size - 1 and pg.dp_rank == 0:
+        opt_r = torch.optim.Adam(full.parameters(), lr=1e-3)
+        it = iter(PipelineDataLoader(dl, 2, "clm"))
+        ref_losses = []
+        for _ in range(2):
+            opt_r.zero_grad()
+            tot = 0.0
+            for _ in range(2):
+                b = next(it)
+                loss = causal_lm_loss(full(b["input_ids"]), b["labels"], ignore_index=-100)
+                (loss / 2).backward()
+                tot += float(loss.detach())
+            full.sync_tied_weights_grad()
+            opt_r.step()
+            ref_losses.append(tot / 2)
+        for a, b in zip(losses, ref_losses):
+            assert abs(a - b) < 5e-4, (losses, ref_losses)
+
+
+def test_gpt2_cp_pp_composition():
+    run_distributed(_run_cp_pp, 4)
