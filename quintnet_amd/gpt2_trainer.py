@@ -81,19 +81,11 @@ class GPT2Trainer(Trainer):
         seq, hidden = int(seq), int(hidden)
         if mc.get("sequence_parallel") and self.pg is not None and self.pg.tp_size > 1:
             seq //= self.pg.tp_size  # inter-stage activations are seq shards
+        if self._cp_group is not None:
+            seq //= self.pg.axis_size("cp")  # CP: pipeline carries shards
         return seq, hidden
 
     # ------------------------------------------------------------------
-    @property
-    def _cp_group(self):
-        if (
-            self.config.get("context_parallel")
-            and self.pg is not None
-            and "cp" in getattr(self.pg, "mesh_name", ())
-        ):
-            return self.pg.get_group("cp")
-        return None
-
     def _train_epoch_plain(self) -> Dict[str, float]:
         total_loss, total_tokens, steps = 0.0, 0, 0
         accum = 0

@@ -62,6 +62,16 @@ class Trainer:
         if self.is_pipeline:
             self._setup_pipeline()
 
+    @property
+    def _cp_group(self):
+        if (
+            self.config.get("context_parallel")
+            and self.pg is not None
+            and "cp" in getattr(self.pg, "mesh_name", ())
+        ):
+            return self.pg.get_group("cp")
+        return None
+
     # ------------------------------------------------------------------
     def _setup_pipeline(self) -> None:
         micro_b = self._infer_micro_batch()
@@ -82,6 +92,7 @@ class Trainer:
             max_grad_norm=self.max_grad_norm,
             pp_fwd_group=groups.get("pp_fwd"),
             pp_bwd_group=groups.get("pp_bwd"),
+            cp_group=self._cp_group,
         )
 
     def _infer_micro_batch(self) -> int:

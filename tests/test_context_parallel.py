@@ -436,3 +436,49 @@ def _run_cp_pp(rank, world):
 
 def test_gpt2_cp_pp_composition():
     run_distributed(_run_cp_pp, 4)
+
+
+def _run_cp_pp_trainer(rank, world):
+    """Turnkey CP x PP via GPT2Trainer: mesh [cp, pp] + config flag."""
+    import torch.distributed as dist
+
+    from quintnet_amd import GPT2Trainer, init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        PipelineParallelWrapper,
+    )
+    from quintnet_amd.utils.data import SyntheticCLM
+    from torch.utils.data import DataLoader
+
+    pg = init_process_groups("cpu", [2, 2], ["cp", "pp"])
+    torch.manual_seed(91)
+    seq = 32
+    cfg = GPT2Config(vocab_size=96, n_positions=seq, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    stage = GPT2Stage(cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+                      tp_group=None, cp_group=pg.get_group("cp"),
+                      tied_group=pg.get_tied_embedding_group())
+    stage.seq_len, stage.hidden_dim = seq, cfg.n_embd
+    wrapper = PipelineParallelWrapper(
+        stage_module=stage, pp_rank=pg.pp_rank, pp_group=pg.get_group("pp"),
+        pp_size=pg.pp_size,
+    )
+    wrapper.seq_len, wrapper.hidden_dim = seq, cfg.n_embd
+    model = DataParallel(
+        wrapper,
+        DistributedConfig(pg.axis_rank("cp"), pg.axis_size("cp"), pg.get_group("cp")),
+    )
+    ds = SyntheticCLM(n=8, seq_len=seq, vocab_size=96, seed=13)
+    tcfg = {"num_epochs": 1, "grad_acc_steps": 2, "zero1": True,
+            "context_parallel": True, "task_type": "clm",
+            "max_seq_length": seq,
+            "model_config": {"n_embd": cfg.n_embd, "n_positions": seq}}
+    tr = GPT2Trainer(model, DataLoader(ds, batch_size=2), None, tcfg, pg)
+    hist = tr.fit()
+    assert torch.isfinite(torch.tensor(hist["train_loss"]))
+
+
+def test_gpt2_trainer_cp_pp():
+    run_distributed(_run_cp_pp_trainer, 4)
