@@ -28,10 +28,9 @@ class GPT2Block(nn.Module):
         if config.n_experts > 0:
             from ...parallel.expert_parallel import ExpertParallelMLP
 
-            assert tp_group is None, "MoE blocks: TP inside experts is round-2"
             self.mlp = ExpertParallelMLP(
                 config.n_embd, config.n_inner, config.n_experts,
-                top_k=config.moe_top_k, ep_group=ep_group,
+                top_k=config.moe_top_k, ep_group=ep_group, tp_group=tp_group,
                 device=device, dtype=dtype,
             )
         else:

@@ -86,11 +86,26 @@ def all_to_all_var(x, out_splits, in_splits, group):
 
 
 class _Expert(nn.Module):
-    def __init__(self, n_embd: int, n_inner: int, device=None, dtype=None):
+    """One expert MLP; with a tp_group its two GEMMs are Megatron
+    column/row-parallel (TP peers must hold IDENTICAL tokens — routing is
+    deterministic on replicated router weights, so they do)."""
+
+    def __init__(self, n_embd: int, n_inner: int, device=None, dtype=None,
+                 tp_group=None):
         super().__init__()
         kw = {"device": device, "dtype": dtype}
-        self.fc1 = nn.Linear(n_embd, n_inner, **kw)
-        self.fc2 = nn.Linear(n_inner, n_embd, **kw)
+        if tp_group is not None:
+            from .tensor_parallel import ColumnParallelLinear, RowParallelLinear
+
+            self.fc1 = ColumnParallelLinear(
+                n_embd, n_inner, tp_group=tp_group, gather_output=False, **kw
+            )
+            self.fc2 = RowParallelLinear(
+                n_inner, n_embd, tp_group=tp_group, input_is_parallel=True, **kw
+            )
+        else:
+            self.fc1 = nn.Linear(n_embd, n_inner, **kw)
+            self.fc2 = nn.Linear(n_inner, n_embd, **kw)
 
     def forward(self, x):
         return self.fc2(torch.nn.functional.gelu(self.fc1(x), approximate="tanh"))
@@ -104,6 +119,7 @@ class ExpertParallelMLP(nn.Module):
         n_experts: int,
         top_k: int = 2,
         ep_group=None,
+        tp_group=None,
         device=None,
         dtype=None,
     ):
@@ -119,7 +135,8 @@ class ExpertParallelMLP(nn.Module):
         kw = {"device": device, "dtype": dtype}
         self.router = nn.Linear(n_embd, n_experts, bias=False, **kw)
         self.experts = nn.ModuleList(
-            _Expert(n_embd, n_inner, **kw) for _ in range(self.n_local)
+            _Expert(n_embd, n_inner, tp_group=tp_group, **kw)
+            for _ in range(self.n_local)
         )
         self.aux_loss: Optional[torch.Tensor] = None
 

@@ -331,3 +331,49 @@ def _run_moe_pp_ep(rank, world):
 
 def test_moe_pp_ep_composition():
     run_distributed(_run_moe_pp_ep, 4)
+
+
+def _run_moe_tp(rank, world):
+    """TP inside experts: outputs and sliced grads match a dense-TP-free
+    reference MoE exactly (TP peers hold identical tokens)."""
+    import torch.distributed as dist
+
+    from quintnet_amd.parallel import ExpertParallelMLP
+
+    torch.manual_seed(23)
+    ref = ExpertParallelMLP(n_embd=16, n_inner=32, n_experts=2, top_k=1)
+    for p in ref.parameters():
+        dist.broadcast(p.data, src=0)
+
+    tp = ExpertParallelMLP(n_embd=16, n_inner=32, n_experts=2, top_k=1,
+                           tp_group=dist.group.WORLD)
+    inner_loc = 32 // world
+    isl = slice(rank * inner_loc, (rank + 1) * inner_loc)
+    with torch.no_grad():
+        tp.router.weight.copy_(ref.router.weight)
+        for le in range(2):
+            tp.experts[le].fc1.weight.copy_(ref.experts[le].fc1.weight[isl])
+            tp.experts[le].fc1.bias.copy_(ref.experts[le].fc1.bias[isl])
+            tp.experts[le].fc2.weight.copy_(ref.experts[le].fc2.weight[:, isl])
+            tp.experts[le].fc2.bias.copy_(ref.experts[le].fc2.bias)
+
+    torch.manual_seed(77)
+    x = torch.randn(2, 5, 16)
+    dist.broadcast(x, src=0)
+    xr = x.clone().requires_grad_(True)
+    xt = x.clone().requires_grad_(True)
+    yr = ref(xr)
+    yt = tp(xt)
+    assert torch.allclose(yt, yr, atol=1e-5), (yt - yr).abs().max()
+    yr.square().sum().backward()
+    yt.square().sum().backward()
+    assert torch.allclose(xt.grad, xr.grad, atol=1e-5)
+    for le in range(2):
+        assert torch.allclose(tp.experts[le].fc1.weight.grad,
+                              ref.experts[le].fc1.weight.grad[isl], atol=1e-5)
+        assert torch.allclose(tp.experts[le].fc2.weight.grad,
+                              ref.experts[le].fc2.weight.grad[:, isl], atol=1e-5)
+
+
+def test_moe_tp_experts():
+    run_distributed(_run_moe_tp, 2)
