@@ -26,6 +26,20 @@ __all__ = [
 ]
 
 
+def _extra_axes_suffix(pg_manager) -> str:
+    """Filename suffix for model-sharding axes beyond pp/tp (cp, ep, ...)
+    so their shards don't collide; empty for plain [dp, tp, pp] meshes
+    (keeps the reference-compatible naming)."""
+    if pg_manager is None:
+        return ""
+    out = ""
+    for ax in getattr(pg_manager, "mesh_name", ()):
+        if ax in ("dp", "tp", "pp"):
+            continue
+        out += f"_{ax}{pg_manager.axis_rank(ax)}"
+    return out
+
+
 def save_sharded_checkpoint(
     model,
     out_dir: str,
@@ -34,17 +48,23 @@ def save_sharded_checkpoint(
     optimizer=None,
     config: Optional[Dict[str, Any]] = None,
 ) -> Optional[str]:
-    """Write this rank's shard ``{name}_pp{p}_tp{t}.pt`` (dp_rank 0 only)."""
+    """Write this rank's shard ``{name}_pp{p}_tp{t}[{extra}].pt``.
+
+    ``extra`` covers any further model-sharding mesh axes (cp, ep, ...):
+    without it, MoE/CP ranks with equal (pp, tp) would clobber each
+    other's shards.  dp rank 0 of each shard coordinate writes.
+    """
     pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
     tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
     dp_rank = pg_manager.dp_rank if pg_manager is not None else 0
     pp_size = pg_manager.pp_size if pg_manager is not None else 1
     tp_size = pg_manager.tp_size if pg_manager is not None else 1
+    extra = _extra_axes_suffix(pg_manager)
 
     os.makedirs(out_dir, exist_ok=True)
     path = None
     if dp_rank == 0:
-        path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}.pt")
+        path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}{extra}.pt")
         payload = {
             "model_state_dict": {k: v.cpu() for k, v in model.state_dict().items()},
             "parallelism_info": {
@@ -62,7 +82,7 @@ def save_sharded_checkpoint(
     # reference never finished).
     if optimizer is not None and hasattr(optimizer, "state_dict"):
         opath = os.path.join(
-            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}_dp{dp_rank}.pt"
+            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}{extra}_dp{dp_rank}.pt"
         )
         try:
             sd = optimizer.state_dict()
@@ -85,14 +105,15 @@ def load_sharded_checkpoint(
     pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
     tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
     dp_rank = pg_manager.dp_rank if pg_manager is not None else 0
-    path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}.pt")
+    extra = _extra_axes_suffix(pg_manager)
+    path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}{extra}.pt")
     ckpt = torch.load(path, map_location="cpu", weights_only=False)
     model.load_state_dict(ckpt["model_state_dict"], strict=strict)
     if optimizer is not None:
         if hasattr(optimizer, "refresh_master_"):
             optimizer.refresh_master_()
         opath = os.path.join(
-            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}_dp{dp_rank}.pt"
+            out_dir, f"{name}_optim_pp{pp_rank}_tp{tp_rank}{extra}_dp{dp_rank}.pt"
         )
         if os.path.exists(opath):
             osd = torch.load(opath, map_location="cpu", weights_only=False)

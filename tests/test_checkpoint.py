@@ -187,3 +187,37 @@ def test_auto_resume(tmp_path):
     stage2 = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None)
     tr2 = GPT2Trainer(stage2, DataLoader(ds, batch_size=2), None, tcfg, None)
     assert torch.allclose(stage2.state_dict()["embedding.wte.weight"], w0)
+
+
+def _run_moe_ckpt(rank, world, tmpdir):
+    """MoE [ep] mesh: per-ep shard files don't collide; save->resume
+    round-trips the rank-local expert weights exactly."""
+    import os
+
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.checkpoint import (
+        load_sharded_checkpoint,
+        save_sharded_checkpoint,
+    )
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    pg = init_process_groups("cpu", [world], ["ep"])
+    torch.manual_seed(rank)  # deliberately DIFFERENT experts per rank
+    cfg = GPT2Config(vocab_size=64, n_positions=16, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0, n_experts=2, moe_top_k=1)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None,
+                      ep_group=pg.get_group("ep"))
+    save_sharded_checkpoint(stage, tmpdir, name="moe", pg_manager=pg)
+    assert os.path.exists(os.path.join(tmpdir, f"moe_pp0_tp0_ep{rank}.pt"))
+
+    want = stage.blocks[0].mlp.experts[0].fc1.weight.clone()
+    fresh = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None,
+                      ep_group=pg.get_group("ep"))
+    load_sharded_checkpoint(fresh, tmpdir, name="moe", pg_manager=pg)
+    assert torch.allclose(fresh.blocks[0].mlp.experts[0].fc1.weight, want)
+
+
+def test_moe_checkpoint_shards(tmp_path):
+    run_distributed(_run_moe_ckpt, 2, str(tmp_path))
