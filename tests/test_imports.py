@@ -60,3 +60,42 @@ def test_phase_timer_integration():
     )
     hist = tr.fit()
     assert "train_loss" in hist
+
+
+def test_gpt2_config_presets():
+    from quintnet_amd.models import GPT2Config
+
+    base = GPT2Config.from_name("base")
+    assert (base.n_embd, base.n_layer, base.n_head) == (768, 12, 12)
+    med = GPT2Config.from_name("medium")
+    assert (med.n_embd, med.n_layer, med.n_head) == (1024, 24, 16)
+    lg = GPT2Config.from_name("large")
+    assert (lg.n_embd, lg.n_layer, lg.n_head) == (1280, 36, 20)
+    xl = GPT2Config.from_name("xl")
+    assert (xl.n_embd, xl.n_layer, xl.n_head) == (1600, 48, 25)
+    assert base.head_dim == 64 and base.n_inner == 3072
+
+
+def test_example_yaml_configs_parse():
+    import os
+
+    from quintnet_amd import load_config
+
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    for f in ("examples/config.yaml", "examples/gpt2_config.yaml"):
+        cfg = load_config(os.path.join(root, f))
+        assert isinstance(cfg, dict) and cfg
+
+
+def test_memory_stats_cpu():
+    from quintnet_amd.utils.memory import memory_stats
+
+    st = memory_stats()
+    assert isinstance(st, dict)
+
+
+def test_rank_logger(tmp_path, capsys):
+    from quintnet_amd.utils.logger import print_rank_0
+
+    print_rank_0("hello")  # rank 0 in single process
+    assert "hello" in capsys.readouterr().out
