@@ -42,7 +42,6 @@ class _AllToAllVar(torch.autograd.Function):
         ctx.group = group
         ctx.out_splits = out_splits
         ctx.in_splits = in_splits
-        ctx.feat_shape = x.shape[1:]
         return _exchange(x, out_splits, in_splits, group)
 
     @staticmethod

@@ -106,6 +106,8 @@ class ZeRO3Block(nn.Module):
             names.append(name)
             metas.append((p.shape, p.numel()))
             tensors.append(p.detach().reshape(-1))
+        if not tensors:
+            raise ValueError("ZeRO3Block: module has no parameters to shard")
         self._names = names
         self._metas = metas
         total = sum(n for _, n in metas)

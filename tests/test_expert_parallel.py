@@ -377,3 +377,21 @@ def _run_moe_tp(rank, world):
 
 def test_moe_tp_experts():
     run_distributed(_run_moe_tp, 2)
+
+
+def test_moe_empty_expert():
+    """An expert that receives zero tokens must not break fwd/bwd."""
+    from quintnet_amd.parallel import ExpertParallelMLP
+
+    torch.manual_seed(5)
+    moe = ExpertParallelMLP(n_embd=8, n_inner=16, n_experts=2, top_k=1)
+    with torch.no_grad():  # force all tokens to expert 0
+        moe.router.weight[0].fill_(10.0)
+        moe.router.weight[1].fill_(-10.0)
+    x = torch.randn(1, 4, 8, requires_grad=True)
+    y = moe(x)
+    y.sum().backward()
+    assert torch.isfinite(x.grad).all()
+    # expert 1 saw nothing: grads None or zero
+    g = moe.experts[1].fc1.weight.grad
+    assert g is None or torch.count_nonzero(g) == 0
