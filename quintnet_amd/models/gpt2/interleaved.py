@@ -40,13 +40,14 @@ class TiedLMHead(nn.Module):
 
 
 class GPT2ForInterleaving(nn.Module):
-    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None):
+    def __init__(self, config: GPT2Config, tp_group=None, device=None, dtype=None,
+                 ep_group=None):
         super().__init__()
         kw = {"device": device, "dtype": dtype}
         self.config = config
         self.embedding = GPT2Embedding(config, **kw)
         self.blocks = nn.ModuleList(
-            GPT2Block(config, tp_group=tp_group, **kw)
+            GPT2Block(config, tp_group=tp_group, ep_group=ep_group, **kw)
             for _ in range(config.n_layer)
         )
         self.head = nn.Sequential(
