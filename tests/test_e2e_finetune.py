@@ -55,3 +55,15 @@ def test_finetune_merge_roundtrip(tmp_path):
     assert sd["transformer.wte.weight"].shape == (128, 32)
     assert sd["transformer.h.1.attn.c_attn.weight"].shape == (32, 96)
     assert "transformer.ln_f.weight" in sd and "lm_head.weight" in sd
+
+    # numeric round trip: the merged model's val ppl must be in the same
+    # ballpark as the training run's (weights survived shard + merge)
+    r = subprocess.run(
+        [sys.executable, "-m", "examples.verify_model", "--checkpoint",
+         str(merged), "--n", "32"],
+        cwd=root, capture_output=True, text=True, timeout=180,
+    )
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "ppl:" in r.stdout
+    ppl = float(r.stdout.split("ppl:")[1].split()[0])
+    assert ppl < 200, r.stdout  # random-init would be ~vocab_size
