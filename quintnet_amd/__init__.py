@@ -5,7 +5,9 @@ process groups for [dp, tp, pp]; bucketed DP; Megatron TP; 1F1B/AFAB PP;
 7 composable strategies; ViT + GPT-2 model families; trainers; sharded
 checkpoints + merge; staged loading) designed for AMD Instinct MI355X:
 PyTorch-ROCm orchestration, hand-written HIP/CDNA4 (gfx950) kernels for
-the hot ops, and RCCL collectives over xGMI.
+the hot ops, and RCCL collectives over xGMI.  Beyond reference parity:
+interleaved 1F1B, sequence/context/expert parallelism, ZeRO-3,
+KV-cached generation, fp8 forward path, per-phase tracing, watchdog.
 
 Public API parity with the reference package root (__init__.py:17-37).
 """
@@ -36,8 +38,14 @@ from .parallel import (
     RowParallelLinear,
     VocabParallelEmbedding,
     PipelineParallelWrapper,
+    InterleavedPipelineWrapper,
     PipelineTrainer,
     PipelineDataLoader,
+    context_parallel_attention,
+    ring_attention,
+    ExpertParallelMLP,
+    ZeRO3Block,
+    apply_zero3,
 )
 from .optim import ZeroRedundancyAdamW, DistributedAdamW
 
@@ -63,8 +71,14 @@ __all__ = [
     "RowParallelLinear",
     "VocabParallelEmbedding",
     "PipelineParallelWrapper",
+    "InterleavedPipelineWrapper",
     "PipelineTrainer",
     "PipelineDataLoader",
+    "context_parallel_attention",
+    "ring_attention",
+    "ExpertParallelMLP",
+    "ZeRO3Block",
+    "apply_zero3",
     "ZeroRedundancyAdamW",
     "DistributedAdamW",
 ]
