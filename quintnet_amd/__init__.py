@@ -12,7 +12,7 @@ KV-cached generation, fp8 forward path, per-phase tracing, watchdog.
 Public API parity with the reference package root (__init__.py:17-37).
 """
 
-__version__ = "0.1.0"
+__version__ = "0.2.0"
 
 from .core import (
     init_process_groups,
