@@ -1,4 +1,11 @@
-from .data_parallel import DataParallel, BucketConfig, DistributedConfig, GradientBucket
+from .data_parallel import (
+    BucketConfig,
+    DataParallel,
+    DistributedConfig,
+    GradientBucket,
+    create_distributed_ddp,
+    create_local_ddp,
+)
 from .backends import DistributedBackend, TorchDistributedBackend, LocalBackend
 from .tensor_parallel import (
     ColumnParallelLinear,
@@ -31,6 +38,8 @@ from .pipeline import (
 TensorParallel = apply_tensor_parallel
 
 __all__ = [
+    "create_local_ddp",
+    "create_distributed_ddp",
     "ExpertParallelMLP",
     "all_to_all_var",
     "cp_causal_lm_loss",

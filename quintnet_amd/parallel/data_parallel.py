@@ -240,3 +240,19 @@ class DataParallel(nn.Module):
                 return self.flat_params_order.get(dtype)
             return None
         return next(iter(self.flat_params_order.values()))
+
+
+def create_local_ddp(model, rank: int = 0, world_size: int = 1):
+    """Single-process DataParallel (LocalBackend) — reference
+    data_parallel/utils/factory.py:12-22 parity; useful in unit tests."""
+    from .backends import LocalBackend
+
+    ddp = DataParallel(
+        model, DistributedConfig(rank, world_size, None), backend=LocalBackend()
+    )
+    return ddp
+
+
+def create_distributed_ddp(model, rank: int, world_size: int, process_group=None):
+    """torch.distributed-backed DataParallel — reference factory.py:25-38."""
+    return DataParallel(model, DistributedConfig(rank, world_size, process_group))

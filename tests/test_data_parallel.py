@@ -99,3 +99,18 @@ def test_local_backend_single_process():
     ddp(x).sum().backward()
     ddp.finalize_gradients()
     assert ddp.module.weight.grad is not None
+
+
+def test_ddp_factories():
+    """C14 parity: create_local_ddp / create_distributed_ddp."""
+    import torch.nn as nn
+
+    from quintnet_amd.parallel import DataParallel, create_local_ddp
+
+    m = nn.Linear(8, 8)
+    ddp = create_local_ddp(m)
+    assert isinstance(ddp, DataParallel)
+    out = ddp(torch.randn(2, 8))
+    out.sum().backward()
+    ddp.finalize_gradients()
+    assert m.weight.grad is not None
