@@ -19,6 +19,8 @@ from .merge import merge_checkpoints
 from .distributed_loading import load_gpt2_distributed
 
 __all__ = [
+    "save_checkpoint",
+    "load_checkpoint",
     "save_sharded_checkpoint",
     "load_sharded_checkpoint",
     "merge_checkpoints",
@@ -123,4 +125,24 @@ def load_sharded_checkpoint(
                 for k, v in osd.items()
             }
             optimizer.load_state_dict(osd)
+    return ckpt
+
+
+def save_checkpoint(model, path: str, optimizer=None, **extra) -> str:
+    """Single-file save (reference utils/utils.py:54-74 parity): model
+    state dict (+ optional optimizer state and any extra metadata)."""
+    payload: Dict[str, Any] = {"model_state_dict": model.state_dict(), **extra}
+    if optimizer is not None:
+        payload["optimizer_state_dict"] = optimizer.state_dict()
+    os.makedirs(os.path.dirname(os.path.abspath(path)), exist_ok=True)
+    torch.save(payload, path)
+    return path
+
+
+def load_checkpoint(model, path: str, optimizer=None, strict: bool = True):
+    """Single-file load (reference utils/utils.py:77-96 parity)."""
+    ckpt = torch.load(path, map_location="cpu", weights_only=False)
+    model.load_state_dict(ckpt.get("model_state_dict", ckpt), strict=strict)
+    if optimizer is not None and "optimizer_state_dict" in ckpt:
+        optimizer.load_state_dict(ckpt["optimizer_state_dict"])
     return ckpt
