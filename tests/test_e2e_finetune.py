@@ -10,11 +10,15 @@ import torch
 import yaml
 
 
-def test_finetune_merge_roundtrip(tmp_path):
+import pytest
+
+
+@pytest.mark.parametrize("mesh,nproc", [([2, 2, 2], 8), ([1, 2, 2], 4)])
+def test_finetune_merge_roundtrip(tmp_path, mesh, nproc):
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     ckdir = tmp_path / "ck"
     cfg = {
-        "mesh_dim": [2, 2, 2],
+        "mesh_dim": mesh,
         "mesh_name": ["dp", "tp", "pp"],
         "batch_size": 2,
         "max_seq_length": 16,
@@ -35,8 +39,8 @@ def test_finetune_merge_roundtrip(tmp_path):
     env["MASTER_ADDR"] = "127.0.0.1"
     r = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
-         "--nproc-per-node", "8", "--master-addr", "127.0.0.1",
-         "--master-port", "29781", "-m", "examples.gpt2_finetune",
+         "--nproc-per-node", str(nproc), "--master-addr", "127.0.0.1",
+         "--master-port", str(29781 + nproc), "-m", "examples.gpt2_finetune",
          "--config", str(cfg_path)],
         cwd=root, env=env, capture_output=True, text=True, timeout=420,
     )
