@@ -452,7 +452,7 @@ class InterleavedOneFOneBSchedule(PipelineSchedule):
                 # label consumer (last global stage)
                 batch_cache[mu] = next(data_loader)
                 if r != 0 and r != p - 1:
-                    batch = batch_cache.pop(mu)
+                    batch_cache.pop(mu)  # middle ranks only keep lockstep
             if g == 0:
                 x_data = self._stage_input(batch_cache[mu], device)
                 if r != p - 1:
