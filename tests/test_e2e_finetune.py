@@ -13,6 +13,7 @@ import yaml
 import pytest
 
 
+@pytest.mark.slow
 @pytest.mark.parametrize("mesh,nproc", [([2, 2, 2], 8), ([1, 2, 2], 4)])
 def test_finetune_merge_roundtrip(tmp_path, mesh, nproc):
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))

@@ -2,6 +2,7 @@
 PP=2 stages + tied-weight grad sync + GPT2Trainer epoch, vs a
 single-process oracle."""
 
+import pytest
 import torch
 
 from conftest import run_distributed
@@ -241,6 +242,7 @@ def _gpt2_3d(rank, world):
         )
 
 
+@pytest.mark.slow
 def test_gpt2_3d_zero1_matches_single_process():
     run_distributed(_gpt2_3d, 8, timeout=300)
 
@@ -324,5 +326,6 @@ def _gpt2_3d_impl(rank, world, sequence_parallel=False):
         )
 
 
+@pytest.mark.slow
 def test_gpt2_3d_sequence_parallel():
     run_distributed(_gpt2_3d_sp, 8, timeout=300)

@@ -2,6 +2,7 @@
 was an empty stub): strategy composition + a full training epoch whose
 loss matches a single-process oracle."""
 
+import pytest
 import torch
 
 from conftest import run_distributed
@@ -77,6 +78,7 @@ def _full_3d(rank, world):
         )
 
 
+@pytest.mark.slow
 def test_full_3d_matches_single_process():
     run_distributed(_full_3d, 8, timeout=300)
 

@@ -3,6 +3,7 @@ single-process oracle over multiple optimizer steps (grads are implicitly
 verified — a wrong gradient diverges the trajectory).  Beyond reference
 parity: the reference has no interleaved schedule."""
 
+import pytest
 import torch
 
 from conftest import run_distributed
@@ -112,6 +113,7 @@ def test_interleaved_p2_v3():
     run_distributed(_run_p2v3, 2)
 
 
+@pytest.mark.slow
 def test_interleaved_p4_v2():
     run_distributed(_run_p4v2, 4)
 
@@ -183,6 +185,7 @@ def _run_dp_interleaved(rank, world):
         assert torch.allclose(buf, p.data, atol=1e-6), "DP replicas diverged"
 
 
+@pytest.mark.slow
 def test_interleaved_with_data_parallel():
     run_distributed(_run_dp_interleaved, 4)
 

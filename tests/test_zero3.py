@@ -3,6 +3,7 @@ training, shard-only storage, and grad reduce-scatter correctness."""
 
 import copy
 
+import pytest
 import torch
 
 from conftest import run_distributed
@@ -86,6 +87,7 @@ def test_zero3_world2():
     run_distributed(_run_zero3, 2)
 
 
+@pytest.mark.slow
 def test_zero3_world4():
     run_distributed(_run_zero3, 4)
 
