@@ -42,7 +42,11 @@ def load_shards(input_dir: str, prefix: Optional[str] = None) -> Dict[int, Dict[
         pp, tp = int(m.group(1)), int(m.group(2))
         out.setdefault(pp, {})[tp] = torch.load(path, map_location="cpu", weights_only=False)
     if not out:
-        raise FileNotFoundError(f"no *_pp*_tp*.pt shards under {input_dir}")
+        raise FileNotFoundError(
+            f"no *_pp*_tp*.pt shards under {input_dir} (note: shards from "
+            "cp/ep meshes carry an extra _cp*/_ep* suffix and are not "
+            "mergeable by this TP/PP CLI)"
+        )
     return out
 
 
