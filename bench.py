@@ -71,6 +71,13 @@ def build_gpt2(args, pg, device, dtype):
     )
 
     sp = args.sequence_parallel and pg.tp_size > 1
+    interleaved = (
+        os.environ.get("QN_SCHEDULE", "").lower().startswith("interleav")
+        and pg.pp_size > 1
+    )
+    if interleaved:
+        assert not sp, "interleaved bench: sequence parallelism unsupported"
+        return _build_gpt2_interleaved(args, pg, device, dtype)
     if args.tiny:
         cfg = GPT2Config(n_embd=64, n_layer=2, n_head=2, vocab_size=512,
                          n_positions=args.seq_len, dropout=0.0,
@@ -99,6 +106,38 @@ def build_gpt2(args, pg, device, dtype):
                                         pp_group=pg.get_group("pp"), pp_size=pg.pp_size,
                                         device=device)
         model.seq_len, model.hidden_dim = pipe_seq, cfg.n_embd
+    if pg.dp_size > 1:
+        model = DataParallel(model, DistributedConfig(
+            pg.dp_rank, pg.dp_size, pg.get_group("dp")))
+    return model, cfg
+
+
+def _build_gpt2_interleaved(args, pg, device, dtype):
+    """QN_SCHEDULE=interleaved: virtual-pipeline GPT-2 (2 chunks/rank by
+    default, QN_VPP_CHUNKS to change)."""
+    from quintnet_amd.models import GPT2Config, GPT2ForInterleaving
+    from quintnet_amd.parallel import (
+        DataParallel,
+        DistributedConfig,
+        InterleavedPipelineWrapper,
+    )
+
+    if args.tiny:
+        cfg = GPT2Config(n_embd=64, n_layer=4, n_head=2, vocab_size=512,
+                         n_positions=args.seq_len, dropout=0.0)
+    else:
+        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len))
+    tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
+    full = GPT2ForInterleaving(cfg, tp_group=tp_group, device=device, dtype=dtype)
+    chunks = int(os.environ.get("QN_VPP_CHUNKS", "2"))
+    model = InterleavedPipelineWrapper(
+        full, pp_rank=pg.pp_rank, pp_group=pg.get_group("pp"),
+        pp_size=pg.pp_size, num_chunks=chunks, device=device,
+        tied_group=pg.get_tied_embedding_group()
+        if pg.pp_rank in (0, pg.pp_size - 1) else None,
+    )
+    args.pipe_seq = args.seq_len
+    model.seq_len, model.hidden_dim = args.seq_len, cfg.n_embd
     if pg.dp_size > 1:
         model = DataParallel(model, DistributedConfig(
             pg.dp_rank, pg.dp_size, pg.get_group("dp")))
