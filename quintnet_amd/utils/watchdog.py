@@ -67,7 +67,16 @@ class Watchdog:
                     f"{self.timeout_s:.0f}s — dumping thread stacks",
                     file=sys.stderr, flush=True,
                 )
-                faulthandler.dump_traceback(file=sys.stderr)
+                try:
+                    faulthandler.dump_traceback(file=sys.stderr)
+                except Exception:
+                    # captured/wrapped stderr has no fileno (pytest, some
+                    # launcher tees) — fall back to the traceback module
+                    import traceback
+
+                    for tid, frame in sys._current_frames().items():
+                        print(f"--- thread {tid} ---", file=sys.stderr)
+                        traceback.print_stack(frame, file=sys.stderr)
                 if self.kill_on_hang:
                     os._exit(42)  # let the elastic agent restart the job
                 self._last = time.monotonic()  # rearm
