@@ -70,9 +70,15 @@ class GPT2Attention(nn.Module):
 
     def _forward_cp(self, qkv: torch.Tensor) -> torch.Tensor:
         """Context parallelism: x is this rank's sequence shard; attention
-        runs against the all-gathered K/V with the shifted causal diagonal
-        (parallel/context_parallel.py)."""
-        from ...parallel.context_parallel import context_parallel_attention
+        runs against the full sequence — all-gathered K/V by default, or
+        the ring exchange (QN_CP_RING=1 / config.cp_ring) when the
+        gathered KV itself is the memory bound."""
+        import os
+
+        from ...parallel.context_parallel import (
+            context_parallel_attention,
+            ring_attention,
+        )
 
         B, T, _ = qkv.shape
         H, D = self.n_head_local, self.head_dim
@@ -84,7 +90,11 @@ class GPT2Attention(nn.Module):
         q = heads(qkv[:, :, :hl])
         k = heads(qkv[:, :, hl : 2 * hl])
         v = heads(qkv[:, :, 2 * hl :])
-        out = context_parallel_attention(q, k, v, self.cp_group, causal=True)
+        use_ring = getattr(self.config, "cp_ring", False) or os.environ.get(
+            "QN_CP_RING"
+        ) == "1"
+        fn = ring_attention if use_ring else context_parallel_attention
+        out = fn(q, k, v, self.cp_group, causal=True)
         return out.permute(0, 2, 1, 3).reshape(B, T, hl)
 
     def _forward_cached(self, qkv: torch.Tensor, kv_cache: dict) -> torch.Tensor:

@@ -24,6 +24,7 @@ class GPT2Config:
     initializer_range: float = 0.02
     sequence_parallel: bool = False
     fp8: bool = False  # experimental: e4m3 forward GEMMs, bf16 backward
+    cp_ring: bool = False  # context parallelism: ring KV exchange
     n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
     moe_top_k: int = 2  # Megatron-SP over the TP group
 

@@ -92,7 +92,7 @@ def _ring_swap(tensors, rank, world, group):
     for t in tensors:
         ops.append(dist.P2POp(dist.isend, t.contiguous(), peer=nxt, group=group))
     for t in tensors:
-        buf = torch.empty_like(t)
+        buf = torch.empty(t.shape, dtype=t.dtype, device=t.device)
         recvs.append(buf)
         ops.append(dist.P2POp(dist.irecv, buf, peer=prv, group=group))
     for r in dist.batch_isend_irecv(ops):
@@ -122,7 +122,7 @@ class _RingAttention(torch.autograd.Function):
         m = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
         l = torch.zeros(B, H, Tl, 1, device=q.device)
         acc = torch.zeros(B, H, Tl, D, device=q.device)
-        kc, vc = k, v
+        kc, vc = k.contiguous(), v.contiguous()  # ring buffers must be
         for s_hop in range(world):
             j = (rank - s_hop) % world
             if not causal or j <= rank:
@@ -161,7 +161,7 @@ class _RingAttention(torch.autograd.Function):
         lse_ = lse.unsqueeze(-1)
         dq = torch.zeros_like(q32)
         # grad accumulators travel with their chunk for a full cycle
-        kc, vc = k, v
+        kc, vc = k.contiguous(), v.contiguous()
         dk_acc = torch.zeros(B, H, Tl, D, device=q.device)
         dv_acc = torch.zeros(B, H, Tl, D, device=q.device)
         for s_hop in range(world):

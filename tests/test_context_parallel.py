@@ -482,3 +482,36 @@ def _run_cp_pp_trainer(rank, world):
 
 def test_gpt2_trainer_cp_pp():
     run_distributed(_run_cp_pp_trainer, 4)
+
+
+def _run_gpt2_cp_ring(rank, world):
+    """cp_ring flavor produces the same logits as the all-gather flavor."""
+    import torch.distributed as dist
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.parallel import scatter_to_context
+
+    torch.manual_seed(51)
+    cfg_a = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
+                       n_head=2, dropout=0.0)
+    cfg_r = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
+                       n_head=2, dropout=0.0, cp_ring=True)
+    a = GPT2Stage(cfg_a, pp_rank=0, pp_size=1, tp_group=None,
+                  cp_group=dist.group.WORLD)
+    for p in a.parameters():
+        dist.broadcast(p.data, src=0)
+    r = GPT2Stage(cfg_r, pp_rank=0, pp_size=1, tp_group=None,
+                  cp_group=dist.group.WORLD)
+    r.load_state_dict(a.state_dict())
+
+    ids = torch.randint(0, 96, (2, 32))
+    dist.broadcast(ids, src=0)
+    shard = scatter_to_context(ids, dist.group.WORLD, dim=1)
+    with torch.no_grad():
+        la = a(shard)
+        lr = r(shard)
+    assert torch.allclose(la, lr, atol=1e-4), (la - lr).abs().max()
+
+
+def test_gpt2_cp_ring_flavor():
+    run_distributed(_run_gpt2_cp_ring, 2)
