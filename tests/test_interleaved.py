@@ -324,7 +324,7 @@ def _run_gpt2_interleaved_trainer(rank, world):
     for p in model.parameters():
         dist.broadcast(p.data, src=0)
     tcfg = {"schedule": "interleaved", "num_chunks": 2, "num_epochs": 1,
-            "grad_acc_steps": 2, "zero1": False, "task_type": "clm",
+            "grad_acc_steps": 2, "zero1": True, "task_type": "clm",
             "max_seq_length": 16,
             "model_config": {"n_embd": 32, "n_positions": 16}}
     pmodel = get_strategy("pp", pg, tcfg).apply(model)
