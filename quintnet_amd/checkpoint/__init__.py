@@ -92,8 +92,15 @@ def save_sharded_checkpoint(
                 k: (v.cpu() if isinstance(v, torch.Tensor) else v) for k, v in sd.items()
             }
             torch.save(sd, opath)
-        except Exception:  # noqa: BLE001 — optimizer state is best-effort
-            pass
+        except Exception as e:  # noqa: BLE001 — model shards must still land
+            import warnings
+
+            rank = dist.get_rank() if dist.is_initialized() else 0
+            warnings.warn(
+                f"[rank {rank}] optimizer shard save FAILED at {opath}: {e!r} — "
+                "a later resume will restore weights with stale/missing "
+                "optimizer state"
+            )
     if dist.is_initialized():
         dist.barrier()
     return path
@@ -125,6 +132,13 @@ def load_sharded_checkpoint(
                 for k, v in osd.items()
             }
             optimizer.load_state_dict(osd)
+        else:
+            import warnings
+
+            warnings.warn(
+                f"optimizer shard missing at {opath}; resuming with FRESH "
+                "optimizer state (weights restored, Adam moments reset)"
+            )
     return ckpt
 
 

@@ -36,9 +36,18 @@ class GPT2Trainer(Trainer):
             if pg_manager is not None and "dp" in pg_manager.mesh_name and pg_manager.dp_size > 1
             else None
         )
+        def _axis_group(axis, size):
+            if pg_manager is None or axis not in getattr(pg_manager, "mesh_name", ()):
+                return None
+            return pg_manager.get_group(axis) if size > 1 else None
+
+        tp_group = _axis_group("tp", pg_manager.tp_size if pg_manager else 1)
+        pp_group = _axis_group("pp", pg_manager.pp_size if pg_manager else 1)
+        self._tp_group, self._pp_group = tp_group, pp_group
         if use_zero:
             zkw = dict(lr=self.lr, weight_decay=wd, dp_group=dp_group,
-                       max_grad_norm=None)  # clipping handled by the schedule
+                       max_grad_norm=None,  # clipping handled by the schedule
+                       tp_group=tp_group, pp_group=pp_group)
             if isinstance(self.model, DataParallel):
                 self.optimizer = ZeroRedundancyAdamW.from_ddp(self.model, **zkw)
             else:
@@ -93,9 +102,18 @@ class GPT2Trainer(Trainer):
 
         cp_group = self._cp_group
         pt = self._phase_timer()
+        is_ddp = isinstance(self.model, DataParallel)
         for batch in self.train_loader:
             ids = batch["input_ids"].to(self.device, non_blocking=True)
             labels = batch["labels"].to(self.device, non_blocking=True)
+            if is_ddp:
+                # bucket hooks may only fire the async all-reduce on the
+                # LAST micro-batch of the accumulation window; earlier
+                # micro-batches would race the in-flight reduce with the
+                # still-accumulating grads.
+                self.model.require_backward_grad_sync = (
+                    accum + 1 == self.grad_acc_steps
+                )
             if pt:
                 pt.start("forward")
             if cp_group is not None:
@@ -130,33 +148,49 @@ class GPT2Trainer(Trainer):
             steps += 1
             if accum == self.grad_acc_steps:
                 accum = 0
-                if pt:
-                    pt.start("grad_comm")
-                if isinstance(self.model, DataParallel):
-                    self.model.finalize_gradients()
-                inner = _unwrap(self.model)
-                if hasattr(inner, "sync_tied_weights_grad"):
-                    inner.sync_tied_weights_grad()
-                if pt:
-                    pt.stop("grad_comm")
-                    pt.start("optimizer")
-                if self.max_grad_norm:
-                    from .ops import clip_grad_norm_local
-
-                    clip_grad_norm_local(
-                        [p for p in self.model.parameters() if p.requires_grad],
-                        self.max_grad_norm,
-                    )
-                self.optimizer.step()
-                if isinstance(self.model, DataParallel):
-                    self.model.zero_grad()
-                else:
-                    self.optimizer.zero_grad()
-                if pt:
-                    pt.stop("optimizer")
+                self._optim_step(pt)
+        if accum:
+            # tail window (loader length not a multiple of grad_acc_steps):
+            # hooks never fired (sync was off), so enable sync and let
+            # finalize_gradients launch every bucket itself.
+            if is_ddp:
+                self.model.require_backward_grad_sync = True
+            self._optim_step(pt)
         self._report_phases(pt, steps)
         avg = total_loss / max(steps, 1)
         return {"loss": avg, "ppl": math.exp(min(avg, 20.0)), "n_tokens": total_tokens}
+
+    def _optim_step(self, pt) -> None:
+        """finalize grads → tied/SP sync → global-norm clip → step."""
+        if pt:
+            pt.start("grad_comm")
+        if isinstance(self.model, DataParallel):
+            self.model.finalize_gradients()
+        inner = _unwrap(self.model)
+        if hasattr(inner, "sync_tied_weights_grad"):
+            inner.sync_tied_weights_grad()
+        if pt:
+            pt.stop("grad_comm")
+            pt.start("optimizer")
+        if self.max_grad_norm:
+            if hasattr(self.optimizer, "clip_grad_norm_"):
+                self.optimizer.clip_grad_norm_(self.max_grad_norm)
+            else:
+                from .ops import clip_grad_norm_global
+
+                clip_grad_norm_global(
+                    [p for p in self.model.parameters() if p.requires_grad],
+                    self.max_grad_norm,
+                    tp_group=self._tp_group,
+                    pp_group=self._pp_group,
+                )
+        self.optimizer.step()
+        if isinstance(self.model, DataParallel):
+            self.model.zero_grad()
+        else:
+            self.optimizer.zero_grad()
+        if pt:
+            pt.stop("optimizer")
 
     @torch.no_grad()
     def _validate_epoch(self) -> Dict[str, float]:

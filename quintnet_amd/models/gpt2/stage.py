@@ -78,6 +78,10 @@ class GPT2Stage(nn.Module):
                     torch.empty(config.vocab_size, config.n_embd, **kw)
                 )
                 nn.init.normal_(self.lm_head, std=config.initializer_range)
+                # grad-norm bookkeeping: after sync_tied_weights_grad this
+                # grad equals the first stage's wte grad — count it once
+                # (on the first stage) in the PP-reduced global norm
+                self.lm_head._tied_copy = True
 
     # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -18,7 +18,13 @@ from .attention import (
     softmax_bwd,
 )
 from .cross_entropy import cross_entropy, causal_lm_loss, shift_labels, CrossEntropyFunction
-from .adamw import adamw_step_flat, clip_grad_norm_local, l2_norm
+from .adamw import (
+    adamw_step_flat,
+    clip_grad_norm_global,
+    clip_grad_norm_local,
+    grad_sq_norm_contrib,
+    l2_norm,
+)
 from .dropout import fused_dropout, FusedDropout
 from .embedding import embedding_pair
 
@@ -45,6 +51,8 @@ __all__ = [
     "CrossEntropyFunction",
     "adamw_step_flat",
     "clip_grad_norm_local",
+    "clip_grad_norm_global",
+    "grad_sq_norm_contrib",
     "l2_norm",
     "fused_dropout",
     "FusedDropout",

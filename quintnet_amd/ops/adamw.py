@@ -15,7 +15,13 @@ import torch
 
 from . import _backend
 
-__all__ = ["adamw_step_flat", "clip_grad_norm_local", "l2_norm"]
+__all__ = [
+    "adamw_step_flat",
+    "clip_grad_norm_local",
+    "clip_grad_norm_global",
+    "grad_sq_norm_contrib",
+    "l2_norm",
+]
 
 
 def adamw_step_flat(
@@ -70,9 +76,10 @@ def l2_norm(tensors: Iterable[torch.Tensor]) -> torch.Tensor:
 def clip_grad_norm_local(
     params: Iterable[torch.nn.Parameter], max_norm: float
 ) -> torch.Tensor:
-    """Per-rank grad clipping (matches the reference's local-clip
-    semantics — SURVEY.md §8.4; a reduced global-norm variant lives in
-    the ZeRO optimizer)."""
+    """Per-rank grad clipping. Only valid when every rank holds the same
+    (replicated) grads — pure DP after all-reduce.  Under TP/PP use
+    :func:`clip_grad_norm_global` (the reference clips
+    per-rank even under TP, de-syncing replicated params)."""
     grads: List[torch.Tensor] = [p.grad for p in params if p.grad is not None]
     if not grads:
         return torch.zeros(())
@@ -81,4 +88,84 @@ def clip_grad_norm_local(
     if clip < 1.0:
         for g in grads:
             g.mul_(clip.to(g.dtype))
+    return norm
+
+
+def _group_active(group) -> bool:
+    import torch.distributed as dist
+
+    return (
+        group is not None
+        and dist.is_initialized()
+        and dist.get_world_size(group=group) > 1
+    )
+
+
+def grad_sq_norm_contrib(params: Iterable[torch.nn.Parameter], tp_rank: int):
+    """This rank's squared-norm contribution, counting each logical
+    parameter exactly once across the model-parallel axes:
+
+    * ``p._tp_sharded`` (set by the TP layers): every rank's shard is
+      distinct — always included, the TP all-reduce sums them;
+    * replicated params (LayerNorms, row-parallel biases, embeddings
+      without TP): included on tp_rank 0 only;
+    * ``p._tied_copy`` (last-stage LM head sharing the embedding):
+      skipped — the first PP stage counts the embedding.
+
+    Returns (sq_norm_tensor_or_None, all_grads) — all_grads is every
+    grad on this rank (for applying the scale afterwards).
+    """
+    seen = set()
+    contrib: List[torch.Tensor] = []
+    all_grads: List[torch.Tensor] = []
+    for p in params:
+        if p.grad is None or id(p) in seen:
+            continue
+        seen.add(id(p))
+        all_grads.append(p.grad)
+        if getattr(p, "_tied_copy", False):
+            continue
+        if getattr(p, "_tp_sharded", False) or tp_rank == 0:
+            contrib.append(p.grad)
+    if not all_grads:
+        return None, all_grads
+    dev = all_grads[0].device
+    if contrib:
+        sq = l2_norm(contrib).to(dev).float().pow(2)
+    else:
+        sq = torch.zeros((), dtype=torch.float32, device=dev)
+    return sq, all_grads
+
+
+def clip_grad_norm_global(
+    params: Iterable[torch.nn.Parameter],
+    max_norm: float,
+    tp_group=None,
+    pp_group=None,
+) -> torch.Tensor:
+    """TRUE global-norm gradient clipping under TP and PP.
+
+    Squared-norm contributions are summed over the TP group (sharded
+    params) and the PP group (each stage owns distinct params), then the
+    single scale is applied to every local grad — so replicated params
+    get the SAME scale on every model-parallel rank (Megatron-style;
+    fixes the reference's per-rank clip).  Fully
+    device-side: no host sync, hipGraph-capture safe.
+    """
+    import torch.distributed as dist
+
+    tp_on = _group_active(tp_group)
+    pp_on = _group_active(pp_group)
+    tp_rank = dist.get_rank(group=tp_group) if tp_on else 0
+    sq, all_grads = grad_sq_norm_contrib(params, tp_rank)
+    if sq is None:
+        return torch.zeros(())
+    if tp_on:
+        dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=tp_group)
+    if pp_on:
+        dist.all_reduce(sq, op=dist.ReduceOp.SUM, group=pp_group)
+    norm = sq.sqrt()
+    scale = (max_norm / (norm + 1e-6)).clamp_(max=1.0)
+    for g in all_grads:
+        g.mul_(scale.to(g.dtype))
     return norm

@@ -43,6 +43,8 @@ class ZeroRedundancyAdamW:
         dp_group=None,
         max_grad_norm: Optional[float] = None,
         grad_buffer: Optional[torch.Tensor] = None,
+        tp_group=None,
+        pp_group=None,
     ):
         self.params: List[torch.nn.Parameter] = [p for p in params if p.requires_grad]
         if not self.params:
@@ -53,7 +55,10 @@ class ZeroRedundancyAdamW:
         self.weight_decay = weight_decay
         self.dp_group = dp_group
         self.max_grad_norm = max_grad_norm
+        self.tp_group = tp_group
+        self.pp_group = pp_group
         self.step_count = 0
+        self._clipped_this_step = False  # guards against double clipping
 
         # dp_group=None means NO sharding (a PP/TP-only rank must not
         # shard over the world group — ranks hold different params).
@@ -133,13 +138,38 @@ class ZeroRedundancyAdamW:
                 self.flat_grad[off : off + p.numel()].copy_(p.grad.reshape(-1))
 
     def clip_grad_norm_(self, max_norm: float) -> torch.Tensor:
-        """Clip on the flat grad buffer (local norm).  Fully device-side
-        (no host sync — hipGraph-capture safe): always multiplies by
-        min(max/norm, 1)."""
+        """Global-norm clip on the flat grad buffer.  The norm counts
+        each logical parameter once across TP/PP (sharded params summed
+        over tp_group, per-stage params over pp_group, tied LM-head copy
+        skipped — see ops.grad_sq_norm_contrib); the DP axis needs no
+        reduction because grads arrive already averaged.  Fully
+        device-side (no host sync — hipGraph-capture safe): always
+        multiplies by min(max/norm, 1)."""
+        import torch.distributed as _d
+
+        from ..ops.adamw import _group_active
+
         self._gather_grads()
-        norm = l2_norm([self.flat_grad])
+        tp_on = _group_active(self.tp_group)
+        pp_on = _group_active(self.pp_group)
+        if tp_on or pp_on:
+            from ..ops import grad_sq_norm_contrib
+
+            tp_rank = _d.get_rank(group=self.tp_group) if tp_on else 0
+            sq, _ = grad_sq_norm_contrib(self.params, tp_rank)
+            if sq is None:
+                sq = torch.zeros((), dtype=torch.float32,
+                                 device=self.flat_grad.device)
+            if tp_on:
+                _d.all_reduce(sq, op=_d.ReduceOp.SUM, group=self.tp_group)
+            if pp_on:
+                _d.all_reduce(sq, op=_d.ReduceOp.SUM, group=self.pp_group)
+            norm = sq.sqrt()
+        else:
+            norm = l2_norm([self.flat_grad])
         scale = (max_norm / (norm + 1e-6)).clamp_(max=1.0)
         self.flat_grad.mul_(scale.to(self.flat_grad.dtype))
+        self._clipped_this_step = True
         return norm
 
     @torch.no_grad()
@@ -148,8 +178,11 @@ class ZeroRedundancyAdamW:
         if self.step_dev is not None:
             self.step_dev += 1  # device op: correct under graph replay
         self._gather_grads()
-        if self.max_grad_norm:
+        if self.max_grad_norm and not self._clipped_this_step:
+            # skip when the schedule already clipped this step (the
+            # scale would otherwise be applied twice)
             self.clip_grad_norm_(self.max_grad_norm)
+        self._clipped_this_step = False
         shard_param = self.flat_param[self._shard_slice]
         shard_grad = self.flat_grad[self._shard_slice]
         adamw_step_flat(

@@ -23,7 +23,7 @@ from ...core.comm import (
     bidirectional_pipeline_communicate,
     pipeline_communicate,
 )
-from ...ops import causal_lm_loss, clip_grad_norm_local, shift_labels
+from ...ops import causal_lm_loss, shift_labels
 
 __all__ = [
     "PipelineSchedule",
@@ -213,9 +213,13 @@ class PipelineSchedule(abc.ABC):
             if hasattr(t.optimizer, "clip_grad_norm_"):
                 t.optimizer.clip_grad_norm_(t.max_grad_norm)  # flat-buffer clip
             else:
-                clip_grad_norm_local(
+                from ...ops import clip_grad_norm_global
+
+                clip_grad_norm_global(
                     [p for p in self.model.parameters() if p.requires_grad],
                     t.max_grad_norm,
+                    tp_group=getattr(t, "tp_group", None),
+                    pp_group=self.pp_group,
                 )
         t.optimizer.step()
         if hasattr(self.model, "zero_grad"):

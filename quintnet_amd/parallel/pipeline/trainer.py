@@ -31,11 +31,13 @@ class PipelineTrainer:
         pp_fwd_group=None,
         pp_bwd_group=None,
         cp_group=None,
+        tp_group=None,
     ):
         self.model = model
         self.pp_fwd_group = pp_fwd_group
         self.pp_bwd_group = pp_bwd_group
         self.cp_group = cp_group
+        self.tp_group = tp_group  # for the global grad-norm clip
         self.optimizer = optimizer
         self.criterion = criterion
         self.pp_rank = pp_rank

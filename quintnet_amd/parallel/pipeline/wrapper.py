@@ -170,6 +170,11 @@ class InterleavedPipelineWrapper(nn.Module):
             for m in chunk.modules():
                 if hasattr(m, "tied_weight"):
                     self._tied_params.append(m.tied_weight)
+                    if pp_size > 1 and pp_rank != 0:
+                        # head-side copy of the tied weight: after the
+                        # tied-group grad sync it duplicates stage 0's
+                        # embedding grad — count once in the global norm
+                        m.tied_weight._tied_copy = True
                 elif hasattr(m, "wte") and isinstance(getattr(m, "wte"), nn.Embedding):
                     self._tied_params.append(m.wte.weight)
 

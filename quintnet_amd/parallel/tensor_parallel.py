@@ -84,6 +84,13 @@ class ColumnParallelLinear(nn.Module):
         kw = {"device": device, "dtype": dtype}
         self.weight = nn.Parameter(torch.empty(self.out_per_rank, in_features, **kw))
         self.bias = nn.Parameter(torch.empty(self.out_per_rank, **kw)) if bias else None
+        if self.tp_size > 1:
+            # grad-norm bookkeeping: these shards are DISTINCT per TP rank
+            # (clip_grad_norm_global sums their squared norms over the TP
+            # group; unmarked params are replicated and counted once)
+            self.weight._tp_sharded = True
+            if self.bias is not None:
+                self.bias._tp_sharded = True
         self.reset_parameters()
 
     def reset_parameters(self) -> None:
@@ -162,7 +169,11 @@ class RowParallelLinear(nn.Module):
         self.input_is_parallel = input_is_parallel
         kw = {"device": device, "dtype": dtype}
         self.weight = nn.Parameter(torch.empty(out_features, self.in_per_rank, **kw))
+        # NOTE: the row-parallel bias is REPLICATED (added after the
+        # reduction) — only the weight is marked sharded
         self.bias = nn.Parameter(torch.empty(out_features, **kw)) if bias else None
+        if self.tp_size > 1:
+            self.weight._tp_sharded = True
         self.reset_parameters()
 
     def reset_parameters(self) -> None:
@@ -239,6 +250,8 @@ class VocabParallelEmbedding(nn.Module):
         self.vocab_end = self.vocab_start + self.vocab_per_rank
         kw = {"device": device, "dtype": dtype}
         self.weight = nn.Parameter(torch.empty(self.vocab_per_rank, embedding_dim, **kw))
+        if self.tp_size > 1:
+            self.weight._tp_sharded = True
         nn.init.normal_(self.weight, mean=0.0, std=0.02)
 
     def forward(self, ids: torch.Tensor) -> torch.Tensor:

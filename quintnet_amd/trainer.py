@@ -16,7 +16,6 @@ import torch
 import torch.distributed as dist
 import torch.nn as nn
 
-from .ops import clip_grad_norm_local
 from .parallel import DataParallel, PipelineDataLoader, PipelineTrainer
 
 __all__ = ["Trainer"]
@@ -93,6 +92,11 @@ class Trainer:
             pp_fwd_group=groups.get("pp_fwd"),
             pp_bwd_group=groups.get("pp_bwd"),
             cp_group=self._cp_group,
+            tp_group=(
+                self.pg.get_group("tp")
+                if "tp" in self.pg.mesh_name and self.pg.tp_size > 1
+                else None
+            ),
         )
 
     def _infer_micro_batch(self) -> int:
@@ -245,9 +249,19 @@ class Trainer:
             if isinstance(self.model, DataParallel):
                 self.model.finalize_gradients()
             if self.max_grad_norm:
-                clip_grad_norm_local(
+                from .ops import clip_grad_norm_global
+
+                tp_group = (
+                    self.pg.get_group("tp")
+                    if self.pg is not None
+                    and "tp" in getattr(self.pg, "mesh_name", ())
+                    and self.pg.tp_size > 1
+                    else None
+                )
+                clip_grad_norm_global(
                     [p for p in self.model.parameters() if p.requires_grad],
                     self.max_grad_norm,
+                    tp_group=tp_group,
                 )
             self.optimizer.step()
             if hasattr(self.model, "zero_grad") and isinstance(self.model, DataParallel):
