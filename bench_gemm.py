@@ -1,20 +1,33 @@
-"""GEMM kernel microbench: custom NT (glds/reg modes) vs torch.matmul
-on the real GPT-2 bench shapes.  Run on GPU: python bench_gemm.py"""
+"""GEMM microbench: custom NT kernel variants vs hipBLASLt (TunableOp-tuned)
+on the real GPT-2 bench shapes (micro 16 x seq 1024 -> M=16384).
+
+Run on GPU:  python bench_gemm.py [--iters 20] [--quick]
+
+Modes (csrc/gemm.hip gemm_nt_launch):
+  0 auto   1 tile 128x128   2 tile 256x128   3 tile 128x256   4 8-phase 256^2
+"""
+import argparse
 import os
 import time
 
+os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
+_CSV = os.path.join(os.path.dirname(os.path.abspath(__file__)), "profiles",
+                    "tunableop_gfx950_0.csv")
+os.environ.setdefault("PYTORCH_TUNABLEOP_TUNING",
+                      "0" if os.path.exists(_CSV) else "1")
+os.environ.setdefault("PYTORCH_TUNABLEOP_FILENAME", _CSV)
+
 import torch
 
-SHAPES = [  # (M, N, K) of the forward linears + lm_head
-    (8192, 2304, 768),   # c_attn
-    (8192, 768, 768),    # attn c_proj
-    (8192, 3072, 768),   # c_fc
-    (8192, 768, 3072),   # mlp c_proj
-    (8192, 50257, 768),  # lm_head
+SHAPES = [  # (name, M, N, K, act)
+    ("c_attn", 16384, 2304, 768, 0),
+    ("c_proj", 16384, 768, 768, 0),
+    ("c_fc+gelu", 16384, 3072, 768, 1),
+    ("mlp_proj", 16384, 768, 3072, 0),
 ]
 
 
-def bench(fn, iters=20):
+def bench(fn, iters):
     for _ in range(3):
         fn()
     torch.cuda.synchronize()
@@ -26,18 +39,62 @@ def bench(fn, iters=20):
 
 
 def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--iters", type=int, default=20)
+    ap.add_argument("--quick", action="store_true")
+    ap.add_argument("--modes", type=str, default="1,2,3,4")
+    args = ap.parse_args()
+    modes = [int(m) for m in args.modes.split(",")]
+
     from quintnet_amd import _C
 
     dev = torch.device("cuda")
-    print(f"{'shape':>22} {'torch':>9} {'custom':>9}  TF(custom)")
-    for M, N, K in SHAPES:
+    results = {}
+    for name, M, N, K, act in SHAPES:
         a = torch.randn(M, K, device=dev, dtype=torch.bfloat16)
-        b = torch.randn(N, K, device=dev, dtype=torch.bfloat16)
+        b = torch.randn(N, K, device=dev, dtype=torch.bfloat16) * 0.02
         bias = torch.randn(N, device=dev, dtype=torch.bfloat16)
-        t_ref = bench(lambda: torch.nn.functional.linear(a, b, bias))
-        t_cus = bench(lambda: _C.gemm_nt(a, b, bias, 0))
         fl = 2.0 * M * N * K
-        print(f"{(M,N,K)!s:>22} {t_ref*1e6:8.1f}u {t_cus*1e6:8.1f}u  {fl/t_cus/1e12:7.1f} (torch {fl/t_ref/1e12:.1f})")
+
+        def lib():
+            out = torch.nn.functional.linear(a, b, bias)
+            if act == 1:
+                out = torch.nn.functional.gelu(out, approximate="tanh")
+            return out
+
+        ref = lib()
+        t_lib = bench(lib, args.iters)
+        row = {"library": fl / t_lib / 1e12}
+
+        for mode in modes:
+            if mode == 4 and not (M % 256 == 0 and N % 256 == 0 and K % 128 == 0):
+                continue
+            if mode == 2 and M % 256:
+                continue
+            if mode == 3 and N % 256:
+                continue
+            try:
+                res = _C.gemm_nt(a, b, bias, act, mode)
+                out = res[0]
+                err = (out.float() - ref.float()).abs()
+                rel = (err / (ref.float().abs() + 1e-3)).max().item()
+                ok = rel < 0.05
+                t = bench(lambda: _C.gemm_nt(a, b, bias, act, mode), args.iters)
+                row[f"mode{mode}"] = fl / t / 1e12 if ok else float("nan")
+                if not ok:
+                    print(f"  !! mode{mode} WRONG on {name}: relerr {rel:.3f}")
+            except Exception as e:  # noqa: BLE001
+                print(f"  !! mode{mode} failed on {name}: {e}")
+        results[name] = row
+        cols = "  ".join(f"{k}={v:7.1f}" for k, v in row.items())
+        print(f"{name:>10} M{M} N{N} K{K}: {cols} TF", flush=True)
+
+    best = {n: max(((v, k) for k, v in r.items() if k != "library"), default=(0, ""))
+            for n, r in results.items()}
+    for n, (v, k) in best.items():
+        lib_v = results[n]["library"]
+        print(f"{n:>10}: best custom {k} {v:.1f} TF vs library {lib_v:.1f} "
+              f"({'WIN' if v >= lib_v else 'lose'} {v / lib_v:.2f}x)")
 
 
 if __name__ == "__main__":

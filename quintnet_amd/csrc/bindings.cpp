@@ -41,7 +41,8 @@ void adamw_launch(TP*, float*, const TG*, float*, float*, const long long*, long
 template <typename T>
 void sumsq_launch(const T*, long long, float*, hipStream_t);
 void gemm_nt_launch(const unsigned short*, const unsigned short*, const unsigned short*,
-                    unsigned short*, unsigned short*, int, int, int, int, hipStream_t);
+                    unsigned short*, unsigned short*, int, int, int, int, hipStream_t,
+                    int mode = 0);
 struct AttnStrides {
   long long qB, qH, qT, kB, kH, kT, vB, vH, vT, oB, oH, oT;
 };
@@ -78,7 +79,8 @@ unsigned short* bf16p_mut(torch::Tensor& t) {
 
 // ---------------------------------------------------------------------------
 std::vector<torch::Tensor> gemm_nt(torch::Tensor x, torch::Tensor w,
-                                   c10::optional<torch::Tensor> bias, int64_t act) {
+                                   c10::optional<torch::Tensor> bias, int64_t act,
+                                   int64_t mode) {
   CHECK_GPU(x); CHECK_GPU(w);
   TORCH_CHECK(x.dtype() == torch::kBFloat16 && w.dtype() == torch::kBFloat16,
               "gemm_nt: bf16 only");
@@ -103,7 +105,7 @@ std::vector<torch::Tensor> gemm_nt(torch::Tensor x, torch::Tensor w,
     bp = bf16p(bc);
   }
   gemm_nt_launch(bf16p(xc), bf16p(wc), bp, bf16p_mut(out), prep,
-                 (int)M, (int)N, (int)K, (int)act, cur_stream());
+                 (int)M, (int)N, (int)K, (int)act, cur_stream(), (int)mode);
   if (act != 0) return {out, pre};
   return {out};
 }
@@ -511,7 +513,9 @@ void attn_bwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
-  m.def("gemm_nt", &gemm_nt, "bf16 MFMA NT GEMM with fused bias/activation");
+  m.def("gemm_nt", &gemm_nt, "bf16 MFMA NT GEMM with fused bias/activation",
+        py::arg("x"), py::arg("w"), py::arg("bias") = c10::nullopt,
+        py::arg("act") = 0, py::arg("mode") = 0);
   m.def("layernorm_fwd", &layernorm_fwd);
   m.def("layernorm_bwd", &layernorm_bwd);
   m.def("softmax_fwd", &softmax_fwd);

@@ -485,7 +485,7 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_8p_kernel(
 void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
                     const unsigned short* bias, unsigned short* C,
                     unsigned short* Cpre, int M, int N, int K, int act,
-                    hipStream_t stream) {
+                    hipStream_t stream, int mode) {
   static int use_glds = -1;
   if (use_glds < 0) {
     const char* e = getenv("QN_GEMM_GLDS");
@@ -505,9 +505,10 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
   // and amortize the 2-tile prologue: ≥256 workgroups and deep K (measured:
   // +22% vs the 128² kernel at 8192³, +5% at 4096³, −20..−40% on the skinny
   // K=768 / N=768 GPT-2 shapes — see profiles/README.md r04).
-  const bool p8_shape = (M % 256 == 0) && (N % 256 == 0) && (K % 128 == 0) &&
+  const bool p8_ok = (M % 256 == 0) && (N % 256 == 0) && (K % 128 == 0);
+  const bool p8_shape = p8_ok &&
                         ((long long)(M >> 8) * (N >> 8) >= 256) && (K >= 2048);
-  if (use_8p && p8_shape) {
+  if ((mode == 4 && p8_ok) || (mode == 0 && use_8p && p8_shape)) {
     dim3 grid((M >> 8) * (N >> 8)), blk(512);
 #define QN_P8_LAUNCH(A_, S_)                                                   \
   hipLaunchKernelGGL((gemm_nt_8p_kernel<A_, S_>), grid, blk, 0, stream, A, B,  \
@@ -522,8 +523,13 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
 #undef QN_P8_LAUNCH
     return;
   }
-  // 256x128 (8 waves) when the problem fills the chip with it; else 128x128
-  const bool big = big_tile && (M % 256 == 0) && ((long long)(M / 256) * ((N + 127) / 128) >= 256);
+  // tile pick: mode 1/2/3 force 128², 256×128, 128×256; mode 0 keeps the
+  // measured default (128² unless QN_GEMM_BIG)
+  int tile = 1;
+  if (mode == 2 && (M % 256 == 0)) tile = 2;
+  else if (mode == 3 && (N % 256 == 0)) tile = 3;
+  else if (mode == 0 && big_tile && (M % 256 == 0) &&
+           ((long long)(M / 256) * ((N + 127) / 128) >= 256)) tile = 2;
 #define QN_GEMM_LAUNCH(A_, S_, G_, TBM_, TBN_, WR_, WC_)                       \
   hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, G_, TBM_, TBN_, WR_, WC_>),       \
                      dim3(((M + TBM_ - 1) / TBM_) * ((N + TBN_ - 1) / TBN_)),  \
@@ -531,9 +537,12 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
                      N, K)
 #define QN_GEMM_CASE(A_, S_)                                                   \
   do {                                                                         \
-    if (big) {                                                                 \
+    if (tile == 2) {                                                           \
       if (use_glds) QN_GEMM_LAUNCH(A_, S_, true, 256, 128, 4, 2);              \
       else QN_GEMM_LAUNCH(A_, S_, false, 256, 128, 4, 2);                      \
+    } else if (tile == 3) {                                                    \
+      if (use_glds) QN_GEMM_LAUNCH(A_, S_, true, 128, 256, 2, 4);              \
+      else QN_GEMM_LAUNCH(A_, S_, false, 128, 256, 2, 4);                      \
     } else {                                                                   \
       if (use_glds) QN_GEMM_LAUNCH(A_, S_, true, 128, 128, 2, 2);              \
       else QN_GEMM_LAUNCH(A_, S_, false, 128, 128, 2, 2);                      \
