@@ -43,8 +43,18 @@ def main():
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--quick", action="store_true")
     ap.add_argument("--modes", type=str, default="1,2,3,4")
+    ap.add_argument("--shapes", type=str, default=None,
+                    help="semicolon list M,N,K[,act] overriding the default set")
     args = ap.parse_args()
     modes = [int(m) for m in args.modes.split(",")]
+    global SHAPES
+    if args.shapes:
+        SHAPES = []
+        for s in args.shapes.split(";"):
+            parts = [int(x) for x in s.split(",")]
+            m, n, k = parts[:3]
+            act = parts[3] if len(parts) > 3 else 0
+            SHAPES.append((f"M{m}N{n}K{k}", m, n, k, act))
 
     from quintnet_amd import _C
 
