@@ -42,7 +42,7 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--iters", type=int, default=20)
     ap.add_argument("--quick", action="store_true")
-    ap.add_argument("--modes", type=str, default="1,2,3,4")
+    ap.add_argument("--modes", type=str, default="1,4,5")
     ap.add_argument("--shapes", type=str, default=None,
                     help="semicolon list M,N,K[,act] overriding the default set")
     args = ap.parse_args()
@@ -77,7 +77,9 @@ def main():
         row = {"library": fl / t_lib / 1e12}
 
         for mode in modes:
-            if mode == 4 and not (M % 256 == 0 and N % 256 == 0 and K % 128 == 0):
+            if mode in (4, 5) and not (
+                M % 256 == 0 and N % 256 == 0 and K % 128 == 0
+            ):
                 continue
             if mode == 2 and M % 256:
                 continue

@@ -470,6 +470,230 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_8p_kernel(
       }
     }
   }
+#undef P8_STAGE
+}
+
+// ===========================================================================
+// PERSISTENT 8-phase kernel (the library-beating path for the skinny-K
+// GPT-2 forward shapes, M=16384 K=768).  Same per-unit phase schedule as
+// gemm_nt_8p_kernel, but each workgroup owns `cnt` whole 256² tiles and the
+// glds pipeline runs CONTINUOUSLY across tile boundaries — the 2-K-tile
+// prologue (the ~30-50% per-tile overhead at K=768) is paid once per WG
+// instead of once per tile, and there is no inter-launch tail.
+//
+// Scheduling: with P=256 WGs (1/CU), workgroup w runs on XCD w%8 (HW
+// dispatch affinity); tile_of(w,j) = (w%8)*32*cnt + j*32 + w/8 gives each
+// XCD a CONTIGUOUS range of row-major tiles AND keeps its 32 concurrent
+// WGs on 32 ADJACENT tiles in lockstep — at any instant an XCD's L2 only
+// holds the ~400 KB of 64-deep A/B k-slabs those adjacent tiles share,
+// so the nbn-fold A re-reads and nbm-fold B re-reads are L2 hits instead
+// of HBM traffic.  Remainder tiles (T % 256) go to a second launch with
+// P=T_rem, cnt=1 (tile offset t0).
+//
+// Epilogue stores between units mix with the counted vmcnt(6) boundary
+// waits: vmcnt counts stores too on CDNA4, so those waits briefly also
+// drain the C-tile stores (~sub-µs against an ~11 µs tile) — safe
+// (over-waiting only), measured acceptable.
+// ===========================================================================
+template <int ACT, bool SAVE_PRE>
+__global__ __launch_bounds__(512, 1) void gemm_nt_p8p_kernel(
+    const unsigned short* __restrict__ A, const unsigned short* __restrict__ B,
+    const unsigned short* __restrict__ bias, unsigned short* __restrict__ C,
+    unsigned short* __restrict__ Cpre, int M, int N, int K, int nbn, int cnt,
+    int t0) {
+  __shared__ unsigned short smem[P8_LDSEL];
+  const int w = blockIdx.x;
+  const bool full = (gridDim.x == 256);
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 2;
+  const int wc = wave & 3;
+
+  const int skoct = p8_src_koct(lane);
+  const int srow = lane >> 3;
+  int ldst[4];
+  int ra, rb;
+  {
+    int st = 2 * wave;
+    ra = ((st & 8) << 4) + ((st & 7) << 3);
+    rb = ((st >> 2) << 6) + ((st & 3) << 3);
+    // wave-uniform LDS destinations — pin to SGPRs (as VGPRs they can
+    // spill, and a scratch reload inside the loop carries a vmcnt(0)
+    // that drains the whole staging pipeline)
+    ldst[0] = __builtin_amdgcn_readfirstlane(ra * 64);
+    ldst[3] = __builtin_amdgcn_readfirstlane((ra + 64) * 64);
+    ldst[1] = __builtin_amdgcn_readfirstlane((256 + rb) * 64);
+    ldst[2] = __builtin_amdgcn_readfirstlane((256 + rb + 32) * 64);
+  }
+  // staging via SRSRC buffer descriptors (T8): per-lane address work is a
+  // single loop-CONSTANT 32-bit voffset; the tile/half/k-step components
+  // are all scalar soffset math.  (A per-lane 64-bit pointer array here
+  // spills at the 256-VGPR cap, and its scratch reloads inside the loop
+  // each carry a vmcnt(0) that drains the glds pipeline.)
+  const auto rsrcA = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)A, (short)0, (int)((long long)M * K * 2), 0x00020000);
+  const auto rsrcB = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)B, (short)0, (int)((long long)N * K * 2), 0x00020000);
+  const auto rsrcC = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)C, (short)0, (int)((long long)M * N * 2), 0x00020000);
+  const auto rsrcP = __builtin_amdgcn_make_buffer_rsrc(
+      (void*)(SAVE_PRE ? Cpre : C), (short)0, (int)((long long)M * N * 2),
+      0x00020000);
+  const int voffA = ((ra + srow) * K + skoct * 8) * 2;   // bytes, per lane
+  const int voffB = ((rb + srow) * K + skoct * 8) * 2;
+  const int dA1b = 64 * K * 2, dB1b = 32 * K * 2, row8b = 8 * K * 2;
+  int soA = 0, soB = 0;  // per-tile scalar base offsets (bytes)
+#define P8P_TILE(j) (full ? ((w & 7) * 32 * cnt + (j) * 32 + (w >> 3)) : (t0 + w))
+#define P8P_SET_TILE(t)                                                        \
+  do {                                                                         \
+    soA = (((t) / nbn) << 8) * K * 2;                                          \
+    soB = (((t) % nbn) << 8) * K * 2;                                          \
+  } while (0)
+#define P8_LDS(buf, h) ((__attribute__((address_space(3))) void*)(uintptr_t)( \
+    smem + (buf) * (512 * 64) + ldst[h]))
+#define P8_LDS2(buf, h) ((__attribute__((address_space(3))) void*)(uintptr_t)( \
+    smem + (buf) * (512 * 64) + ldst[h] + 512))
+  // half h: 0=AE  1=BE  2=BO(+dB1b)  3=AO(+dA1b); h is a literal → folds
+#define P8_STAGE(h, buf, kel)                                                  \
+  do {                                                                         \
+    const int so_ = (((h) == 0 || (h) == 3) ? soA : soB) +                     \
+                    ((h) == 3 ? dA1b : ((h) == 2 ? dB1b : 0)) + (kel) * 2;     \
+    if ((h) == 0 || (h) == 3) {                                                \
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(rsrcA, P8_LDS(buf, h), 16,      \
+                                               voffA, so_, 0, 0);              \
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(rsrcA, P8_LDS2(buf, h), 16,     \
+                                               voffA, so_ + row8b, 0, 0);      \
+    } else {                                                                   \
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(rsrcB, P8_LDS(buf, h), 16,      \
+                                               voffB, so_, 0, 0);              \
+      __builtin_amdgcn_raw_ptr_buffer_load_lds(rsrcB, P8_LDS2(buf, h), 16,     \
+                                               voffB, so_ + row8b, 0, 0);      \
+    }                                                                          \
+  } while (0)
+
+  // Fragment reads: ONE per-lane byte base per operand; buf/f/extra/ks all
+  // fold into the ds_read offset immediate.  The st_16x32 swizzle term
+  // (kq ^ ((row>>2)&1)<<1) is the same for every f (f*16>>2 is even), and
+  // the ks toggle (^32 elements) lands on a known-zero bit of the base,
+  // so XOR == ADD — this is what keeps the address set at 2 VGPRs
+  // (aoffE[4]/boffE[2] arrays spill in the persistent form).
+  const int frow = lane & 15;
+  const int kq = lane >> 4;
+  const char* smc = (const char*)smem;
+  const int swz2 = ((kq ^ (((frow >> 2) & 1) << 1)) << 3) * 2;
+  const int vA = (wr * 128 + frow) * 128 + swz2;          // bytes
+  const int vB = (256 + wc * 64 + frow) * 128 + swz2;
+#define P8P_READ_A(set, buf, extra)                                           \
+  _Pragma("unroll") for (int f = 0; f < 4; ++f) _Pragma("unroll")             \
+      for (int ks = 0; ks < 2; ++ks)                                          \
+          Ar[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
+              smc + (buf) * 65536 + vA + f * 2048 + (extra) * 2 + ks * 64);
+#define P8P_READ_B(set, buf, extra)                                           \
+  _Pragma("unroll") for (int f = 0; f < 2; ++f) _Pragma("unroll")             \
+      for (int ks = 0; ks < 2; ++ks)                                          \
+          Br[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
+              smc + (buf) * 65536 + vB + f * 2048 + (extra) * 2 + ks * 64);
+
+  bf16x8 Ar[2][8];
+  bf16x8 Br[2][4];
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  const int UT = K >> 7;          // 128-deep units per tile
+  const int total = cnt * UT;     // units this WG
+  P8P_SET_TILE(P8P_TILE(0));
+
+  // prologue: stage unit 0 (K-tiles 0,1) into buf0,buf1
+  P8_STAGE(0, 0, 0);
+  P8_STAGE(1, 0, 0);
+  P8_STAGE(2, 0, 0);
+  P8_STAGE(3, 0, 0);
+  P8_STAGE(0, 1, 64);
+  P8_STAGE(1, 1, 64);
+  P8_STAGE(2, 1, 64);
+  P8_STAGE(3, 1, 64);
+  P8_VM(8);
+  P8_BAR();
+  P8P_READ_A(0, 0, 0);
+  P8P_READ_B(0, 0, 0);
+
+  const int erow = (lane >> 4) * 4;
+  const int ecol = lane & 15;
+
+  // stage cursor = compute unit + 1, clamped at the chunk end; advanced
+  // incrementally (an in-loop integer division spills at 256 VGPRs)
+  int js = 0, kin_s = 0;
+  if (total > 1) {
+    kin_s = 1;
+    if (kin_s == UT) {
+      kin_s = 0;
+      js = 1;
+      P8P_SET_TILE(P8P_TILE(1));
+    }
+  }
+
+  int cu = 0;
+  for (int jc = 0; jc < cnt; ++jc) {
+    for (int kin = 0; kin < UT; ++kin, ++cu) {
+      const int k2 = kin_s << 7;
+      const int k3 = k2 + 64;
+      P8_PHASE(P8P_READ_B(1, 0, 2048), P8_STAGE(0, 0, k2), 0, 0, );
+      P8_PHASE(P8P_READ_A(1, 0, 4096), P8_STAGE(1, 0, k2), 0, 1, );
+      P8_PHASE(, P8_STAGE(2, 0, k2), 1, 0, P8_VM(6));
+      P8_PHASE(P8P_READ_A(0, 1, 0) P8P_READ_B(0, 1, 0), P8_STAGE(3, 0, k2), 1, 1, );
+      P8_PHASE(P8P_READ_B(1, 1, 2048), P8_STAGE(0, 1, k3), 0, 0, );
+      P8_PHASE(P8P_READ_A(1, 1, 4096), P8_STAGE(1, 1, k3), 0, 1, );
+      P8_PHASE(, P8_STAGE(2, 1, k3), 1, 0, P8_VM(6));
+      P8_PHASE(P8P_READ_A(0, 0, 0) P8P_READ_B(0, 0, 0), P8_STAGE(3, 1, k3), 1, 1, );
+      // advance the stage cursor unless already clamped at the end
+      if (cu + 2 < total) {
+        if (++kin_s == UT) {
+          kin_s = 0;
+          ++js;
+          P8P_SET_TILE(P8P_TILE(js));
+        }
+      }
+    }
+    // tile complete: direct epilogue from acc (descriptor stores, 32-bit
+    // byte offsets — 64-bit per-store address math spills), then reset
+    {
+      const int t = P8P_TILE(jc);
+      const int m0 = (t / nbn) << 8;
+      const int n0 = (t % nbn) << 8;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int col = n0 + wc * 64 + j * 16 + ecol;
+        const float bv = (bias != nullptr) ? bf16_to_f32(bias[col]) : 0.f;
+#pragma unroll
+        for (int i = 0; i < 8; ++i) {
+          const int base = (m0 + wr * 128 + i * 16 + erow) * N + col;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float v = acc[i][j][r] + bv;
+            const int off = (base + r * N) * 2;
+            if constexpr (SAVE_PRE)
+              __builtin_amdgcn_raw_buffer_store_b16(f32_to_bf16(v), rsrcP,
+                                                    off, 0, 0);
+            if constexpr (ACT == QN_ACT_GELU) v = gelu_tanh(v);
+            else if constexpr (ACT == QN_ACT_RELU) v = fmaxf(v, 0.f);
+            __builtin_amdgcn_raw_buffer_store_b16(f32_to_bf16(v), rsrcC, off,
+                                                  0, 0);
+          }
+          acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+        }
+      }
+    }
+  }
+#undef P8P_SET_TILE
+#undef P8P_TILE
+#undef P8P_READ_A
+#undef P8P_READ_B
+#undef P8_LDS
+#undef P8_LDS2
 #undef P8_PHASE
 #undef P8_MFMA
 #undef P8_READ_A
@@ -481,6 +705,34 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_8p_kernel(
 }
 
 #include <cstdlib>
+
+// persistent launch: main P=256 x cnt tiles + remainder P=T%256 x 1
+static void gemm_nt_p8p_dispatch(const unsigned short* A, const unsigned short* B,
+                                 const unsigned short* bias, unsigned short* C,
+                                 unsigned short* Cpre, int M, int N, int K,
+                                 int act, hipStream_t stream) {
+  const int nbn = N >> 8;
+  const int T = (M >> 8) * nbn;
+  const int cnt = T / 256;
+  const int trem = T - cnt * 256;
+#define QN_P8P_ONE(A_, S_, P_, CNT_, T0_)                                      \
+  hipLaunchKernelGGL((gemm_nt_p8p_kernel<A_, S_>), dim3(P_), dim3(512), 0,     \
+                     stream, A, B, bias, C, Cpre, M, N, K, nbn, CNT_, T0_)
+#define QN_P8P_CASE(A_, S_)                                                    \
+  do {                                                                         \
+    if (cnt > 0) QN_P8P_ONE(A_, S_, 256, cnt, 0);                              \
+    if (trem > 0) QN_P8P_ONE(A_, S_, trem, 1, cnt * 256);                      \
+  } while (0)
+  if (act == QN_ACT_GELU) {
+    if (Cpre) QN_P8P_CASE(QN_ACT_GELU, true); else QN_P8P_CASE(QN_ACT_GELU, false);
+  } else if (act == QN_ACT_RELU) {
+    if (Cpre) QN_P8P_CASE(QN_ACT_RELU, true); else QN_P8P_CASE(QN_ACT_RELU, false);
+  } else {
+    QN_P8P_CASE(QN_ACT_NONE, false);
+  }
+#undef QN_P8P_CASE
+#undef QN_P8P_ONE
+}
 
 void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
                     const unsigned short* bias, unsigned short* C,
@@ -506,6 +758,10 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
   // +22% vs the 128² kernel at 8192³, +5% at 4096³, −20..−40% on the skinny
   // K=768 / N=768 GPT-2 shapes — see profiles/README.md r04).
   const bool p8_ok = (M % 256 == 0) && (N % 256 == 0) && (K % 128 == 0);
+  if (mode == 5 && p8_ok) {
+    gemm_nt_p8p_dispatch(A, B, bias, C, Cpre, M, N, K, act, stream);
+    return;
+  }
   const bool p8_shape = p8_ok &&
                         ((long long)(M >> 8) * (N >> 8) >= 256) && (K >= 2048);
   if ((mode == 4 && p8_ok) || (mode == 0 && use_8p && p8_shape)) {
