@@ -510,8 +510,11 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_p8p_kernel(
   const int wr = wave >> 2;
   const int wc = wave & 3;
 
-  const int skoct = p8_src_koct(lane);
+  // source granule for the conflict-free image: LDS position (lane&7) of
+  // row srow holds global granule (lane&7) ^ (srow<1:2>·{2,4})
   const int srow = lane >> 3;
+  const int skoct =
+      (lane & 7) ^ ((((srow >> 1) & 1) << 1) | (((srow >> 2) & 1) << 2));
   int ldst[4];
   int ra, rb;
   {
@@ -572,28 +575,34 @@ __global__ __launch_bounds__(512, 1) void gemm_nt_p8p_kernel(
     }                                                                          \
   } while (0)
 
-  // Fragment reads: ONE per-lane byte base per operand; buf/f/extra/ks all
-  // fold into the ds_read offset immediate.  The st_16x32 swizzle term
-  // (kq ^ ((row>>2)&1)<<1) is the same for every f (f*16>>2 is even), and
-  // the ks toggle (^32 elements) lands on a known-zero bit of the base,
-  // so XOR == ADD — this is what keeps the address set at 2 VGPRs
-  // (aoffE[4]/boffE[2] arrays spill in the persistent form).
+  // Fragment reads: one per-lane byte base per (operand, ks); buf/f/extra
+  // fold into ds_read offset immediates.  The granule swizzle here is
+  // CONFLICT-FREE for ds_read_b128's 4×16 lane groups: granule p stored
+  // at position ((ks<<2)|kq) ^ (row<1:2>·{2,4}) — the old row-bit2-only
+  // XOR measured a 50% SQ_LDS_BANK_CONFLICT rate (2-way); this one
+  // simulates and measures clean.  Swizzle bit2 collides with the ks
+  // offset, hence TWO bases per operand instead of a +64 add.
   const int frow = lane & 15;
   const int kq = lane >> 4;
   const char* smc = (const char*)smem;
-  const int swz2 = ((kq ^ (((frow >> 2) & 1) << 1)) << 3) * 2;
-  const int vA = (wr * 128 + frow) * 128 + swz2;          // bytes
-  const int vB = (256 + wc * 64 + frow) * 128 + swz2;
+  const int fr_ = ((((frow >> 1) & 1) << 1) | (((frow >> 2) & 1) << 2));
+  int vA[2], vB[2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks) {
+    const int p = ((ks << 2) | kq) ^ fr_;
+    vA[ks] = (wr * 128 + frow) * 128 + p * 16;
+    vB[ks] = (256 + wc * 64 + frow) * 128 + p * 16;
+  }
 #define P8P_READ_A(set, buf, extra)                                           \
   _Pragma("unroll") for (int f = 0; f < 4; ++f) _Pragma("unroll")             \
       for (int ks = 0; ks < 2; ++ks)                                          \
           Ar[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
-              smc + (buf) * 65536 + vA + f * 2048 + (extra) * 2 + ks * 64);
+              smc + (buf) * 65536 + vA[ks] + f * 2048 + (extra) * 2);
 #define P8P_READ_B(set, buf, extra)                                           \
   _Pragma("unroll") for (int f = 0; f < 2; ++f) _Pragma("unroll")             \
       for (int ks = 0; ks < 2; ++ks)                                          \
           Br[set][f * 2 + ks] = *reinterpret_cast<const bf16x8*>(             \
-              smc + (buf) * 65536 + vB + f * 2048 + (extra) * 2 + ks * 64);
+              smc + (buf) * 65536 + vB[ks] + f * 2048 + (extra) * 2);
 
   bf16x8 Ar[2][8];
   bf16x8 Br[2][4];
