@@ -34,23 +34,33 @@ def _gelu_tanh(x: torch.Tensor) -> torch.Tensor:
 
 import os
 
-# Forward dispatch policy (A/B-able, profiles/README.md r04): hipBLASLt via
-# F.linear wins the GPT-2-sized forwards outright (770-997 TF vs 595-713
-# custom; even GELU-fused, library + a separate act pass measured 58.4 vs
-# 59.6 ms/step) — the MI355X split is "vendor library for plain GEMMs".  The
-# hand-written 8-phase 256² kernel wins once the shape fills the chip at one
-# block/CU with deep K (+22% vs library-class at 8192³), so "auto" routes
-# activation-fused linears of that size to the fused custom epilogue and
-# everything else to the library.  "custom"/"library" force one side.
+# Forward dispatch policy (A/B-measured, profiles/README.md r04 + r2
+# gemm_sweep3): per-shape winner between the hand-written MFMA kernels and
+# hipBLASLt (TunableOp-tuned).  Measured at M=16384 (bench micro 16):
+#   c_attn (N2304 K768):  library 893 vs custom 686  -> library
+#   c_proj (N768  K768):  library 337 vs custom 594  -> CUSTOM (1.76x)
+#   c_fc   (N3072 K768):  library 626 vs custom 543  -> library
+#   mlp    (N768  K3072): library 1158 vs custom 856 -> library
+# "auto" routes the skinny-N square-K class (where hipBLASLt collapses to
+# ~350 TF) to the custom kernel; QN_GEMM_FWD=custom forces the custom path
+# everywhere (per-shape best variant), =library forces hipBLASLt.
 _FWD_MODE = os.environ.get("QN_GEMM_FWD", "auto")
 _WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
 
 
-def _big_fused_shape(m: int, n: int, k: int) -> bool:
-    return (
-        m % 256 == 0 and n % 256 == 0 and k % 128 == 0
-        and (m // 256) * (n // 256) >= 256 and k >= 2048
-    )
+def _custom_wins_shape(m: int, n: int, k: int) -> bool:
+    # the measured win class: narrow-N, shallow square-ish K, tall M
+    return m >= 4096 and n <= 1024 and k <= 1024
+
+
+def _custom_mode(m: int, n: int, k: int) -> int:
+    # per-shape best custom variant: persistent 8-phase for eligible big
+    # shapes with wide N or deep K; 128² two-buffer kernel otherwise
+    if m % 256 == 0 and n % 256 == 0 and k % 128 == 0 and (
+        n >= 3072 or k >= 2048
+    ):
+        return 5
+    return 1
 
 
 def _native_ok(x: torch.Tensor, weight: torch.Tensor) -> bool:
@@ -131,15 +141,16 @@ class LinearFunction(torch.autograd.Function):
             )
             ctx.has_bias = bias is not None
             return out.reshape(*x.shape[:-1], weight.shape[0])
+        m, n, k = x2d.shape[0], weight.shape[0], weight.shape[1]
         want_native = (
             _FWD_MODE == "custom"
-            or (_FWD_MODE == "auto" and act != _ACT_NONE
-                and _big_fused_shape(x2d.shape[0], weight.shape[0], weight.shape[1]))
+            or (_FWD_MODE == "auto" and _custom_wins_shape(m, n, k))
         ) and not prefer_library
         if want_native and _backend.use_native(x) and _native_ok(x, weight):
             res = _backend.ext().gemm_nt(
                 x2d.contiguous(), weight.contiguous(),
                 bias if bias is not None else None, act,
+                _custom_mode(m, n, k),
             )
             out = res[0]
             if act != _ACT_NONE:
