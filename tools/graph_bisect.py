@@ -8,6 +8,8 @@ its replay twice and checks the loss is finite.
 import os
 import sys
 
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
 import torch
 
 
