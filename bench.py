@@ -305,10 +305,11 @@ def main():
         # hipGraph capture of the whole optimizer step (single-GPU path):
         # the launch-bound inner loop replays as one graph; fresh data is
         # copied into the captured static input buffers before each replay.
-        # Enabled for the launch-bound ViT (3.3x);  GPT-2 is compute-bound
-        # (no gain) and its capture trips an HSA aperture fault at some
-        # micro-batch configs — opt in with QN_GRAPHS=1.
-        want_graphs = args.model == "vit" or os.environ.get("QN_GRAPHS") == "1"
+        # ViT (launch-bound): 3.3x.  GPT-2: the round-1 HSA aperture fault
+        # no longer reproduces (tools/graph_bisect.py r2: full-size
+        # fwd+bwd+ZeRO captures clean) and capture is worth ~2 ms/step
+        # (55.98 vs 57.9 ms) — on by default, QN_NO_GRAPHS=1 to disable.
+        want_graphs = os.environ.get("QN_GRAPHS", "1") != "0"
         if (use_cuda and pg.world_size == 1 and want_graphs
                 and os.environ.get("QN_NO_GRAPHS") != "1"):
             try:
@@ -341,6 +342,16 @@ def main():
                 if rank == 0:
                     print(f"# hipGraph capture unavailable ({e!r}); eager path", flush=True)
 
+    # hang watchdog (multi-rank runs): if any collective wedges, kill THIS
+    # process with a stack dump instead of hanging the node until the
+    # driver's own limit (default 600 s per beat; QN_WATCHDOG=0 disables)
+    wd = None
+    if pg.world_size > 1 and os.environ.get("QN_WATCHDOG", "1") != "0":
+        from quintnet_amd.utils.watchdog import Watchdog
+
+        wd = Watchdog(float(os.environ.get("QN_WATCHDOG_S", "600")),
+                      kill_on_hang=True).start()
+
     def sync():
         if dist.is_initialized():
             dist.barrier()
@@ -349,6 +360,8 @@ def main():
 
     for _ in range(args.warmup):
         step()
+        if wd:
+            wd.beat()
     sync()
     if use_cuda and os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
         try:
@@ -358,8 +371,12 @@ def main():
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
+        if wd:
+            wd.beat()
     sync()
     elapsed = time.perf_counter() - t0
+    if wd:
+        wd.stop()
 
     t = torch.tensor([elapsed], dtype=torch.float64)
     if dist.is_initialized():
