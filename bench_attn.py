@@ -43,6 +43,17 @@ def main():
 
     print(f"sdpa fwd {bench(sdpa)*1e6:8.1f}us (torch reference)")
 
+    qr = q.clone().requires_grad_(True)
+    kr = k.clone().requires_grad_(True)
+    vr = v.clone().requires_grad_(True)
+    o = torch.nn.functional.scaled_dot_product_attention(qr, kr, vr, is_causal=True)
+
+    def sdpa_bwd():
+        o.backward(dout, retain_graph=True)
+        qr.grad = kr.grad = vr.grad = None
+
+    print(f"sdpa fwd+bwd-bwd {bench(sdpa_bwd)*1e6:8.1f}us (torch bwd reference)")
+
 
 if __name__ == "__main__":
     main()

@@ -134,7 +134,8 @@ __device__ __forceinline__ void stage_transpose(
 // q/k/v strided [.., T, 64] slices; out written via its own strides
 // (so [B, T, H*64] layout comes out directly); lse2 [BH, T] f32.
 // ===========================================================================
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+template <int MINW>
+__global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, unsigned short* __restrict__ out,
     float* __restrict__ lse2, int Tq, int Tk, int qoff, int H, float scale,
@@ -322,7 +323,8 @@ __global__ void attn_delta_kernel(
 // dQ^T[d][q] += K^T[d][key] · g^T[key][q],
 //   g^T = scale * P^T ⊙ (dP^T - delta[q]),  P^T = exp2(s2 - lse2[q])
 // ===========================================================================
-__global__ __launch_bounds__(256, 2) void attn_bwd_dq_kernel(
+template <int MINW>
+__global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
     const float* __restrict__ lse2, const float* __restrict__ delta,
@@ -696,10 +698,21 @@ void attn_fwd_launch(const unsigned short* q, const unsigned short* k,
                      int B, int H, int Tq, int Tk, int qoff, float scale,
                      int causal, const AttnStrides& st, hipStream_t stream) {
   dim3 grid(Tq / 128, B * H);
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream, q, k, v, out,
-                     lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH, st.qT,
-                     st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB, st.oH,
-                     st.oT);
+  static int occ = -1;
+  if (occ < 0) {
+    const char* e = getenv("QN_ATTN_FWD_OCC");
+    occ = e ? atoi(e) : 3;
+  }
+  if (occ >= 4)
+    hipLaunchKernelGGL(attn_fwd_kernel<4>, grid, dim3(256), 0, stream, q, k, v,
+                       out, lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
+                       st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB,
+                       st.oH, st.oT);
+  else
+    hipLaunchKernelGGL(attn_fwd_kernel<3>, grid, dim3(256), 0, stream, q, k, v,
+                       out, lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
+                       st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB,
+                       st.oH, st.oT);
 }
 
 void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
@@ -722,10 +735,21 @@ void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
                         long long dsB, long long dsH, long long dsT,
                         hipStream_t stream) {
   dim3 grid(Tq / 128, B * H);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream, q, k, v,
-                     dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
-                     st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, dsB, dsH,
-                     dsT, st.oB, st.oH, st.oT);
+  static int occ = -1;
+  if (occ < 0) {
+    const char* e = getenv("QN_ATTN_DQ_OCC");
+    occ = e ? atoi(e) : 2;
+  }
+  if (occ >= 3)
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<3>, grid, dim3(256), 0, stream, q, k,
+                       v, dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal,
+                       st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH,
+                       st.vT, dsB, dsH, dsT, st.oB, st.oH, st.oT);
+  else
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<2>, grid, dim3(256), 0, stream, q, k,
+                       v, dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal,
+                       st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH,
+                       st.vT, dsB, dsH, dsT, st.oB, st.oH, st.oT);
 }
 
 void attn_bwd_dkv_launch(const unsigned short* q, const unsigned short* k,
