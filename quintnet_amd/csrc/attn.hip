@@ -738,7 +738,7 @@ void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
   static int occ = -1;
   if (occ < 0) {
     const char* e = getenv("QN_ATTN_DQ_OCC");
-    occ = e ? atoi(e) : 2;
+    occ = e ? atoi(e) : 3;  // A/B r2: 3 waves/SIMD = bwd 279.8 -> 265.2 us
   }
   if (occ >= 3)
     hipLaunchKernelGGL(attn_bwd_dq_kernel<3>, grid, dim3(256), 0, stream, q, k,
