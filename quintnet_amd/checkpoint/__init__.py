@@ -67,15 +67,25 @@ def save_sharded_checkpoint(
     path = None
     if dp_rank == 0:
         path = os.path.join(out_dir, f"{name}_pp{pp_rank}_tp{tp_rank}{extra}.pt")
+        info = {
+            "pp_rank": pp_rank,
+            "pp_size": pp_size,
+            "tp_rank": tp_rank,
+            "tp_size": tp_size,
+            "dp_rank": dp_rank,
+        }
+        # interleaved (virtual-pipeline) wrapper: record the chunk→global-
+        # stage map so the merge CLI can rename chunk-local parameter
+        # names (chunks.c.i.*) to global layers
+        if hasattr(model, "chunks") and hasattr(model, "layer_distribution"):
+            info["interleaved"] = {
+                "num_chunks": int(model.num_chunks),
+                "pp_size": int(model.pp_size),
+                "layer_distribution": [list(g) for g in model.layer_distribution],
+            }
         payload = {
             "model_state_dict": {k: v.cpu() for k, v in model.state_dict().items()},
-            "parallelism_info": {
-                "pp_rank": pp_rank,
-                "pp_size": pp_size,
-                "tp_rank": tp_rank,
-                "tp_size": tp_size,
-                "dp_rank": dp_rank,
-            },
+            "parallelism_info": info,
             "config": config or {},
         }
         torch.save(payload, path)

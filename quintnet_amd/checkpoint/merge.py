@@ -25,7 +25,7 @@ from typing import Dict, Optional
 
 import torch
 
-__all__ = ["load_shards", "merge_tp_shards", "merge_pp_stages", "convert_to_hf_format", "merge_checkpoints"]
+__all__ = ["load_shards", "merge_tp_shards", "merge_pp_stages", "remap_interleaved_stage", "convert_to_hf_format", "merge_checkpoints"]
 
 _SHARD_RE = re.compile(r"_pp(\d+)_tp(\d+)\.pt$")
 
@@ -125,13 +125,68 @@ def convert_to_hf_format(merged: Dict[str, torch.Tensor]) -> Dict[str, torch.Ten
     return hf
 
 
+def remap_interleaved_stage(
+    state: Dict[str, torch.Tensor], info: Dict
+) -> Dict[str, torch.Tensor]:
+    """Rename an InterleavedPipelineWrapper shard's chunk-local names
+    (``chunks.c.i.*``) to the global merged naming (wte/wpe, h.L.*, ln_f,
+    lm_head) using the recorded chunk→global-stage layer map.
+
+    Chunk module order (parallel/pipeline/wrapper.py): [embedding (global
+    stage 0 only)] + blocks + [head = Sequential(ln_f, TiedLMHead) (last
+    global stage only)].
+    """
+    meta = info["interleaved"]
+    pp_rank = info["pp_rank"]
+    pp_size = meta["pp_size"]
+    num_chunks = meta["num_chunks"]
+    dist_layers = meta["layer_distribution"]
+    n_stages = pp_size * num_chunks
+    out: Dict[str, torch.Tensor] = {}
+    for key, value in state.items():
+        m = re.match(r"chunks\.(\d+)\.(\d+)\.(.*)", key)
+        if not m:
+            out[key] = value
+            continue
+        c, i, rest = int(m.group(1)), int(m.group(2)), m.group(3)
+        g = c * pp_size + pp_rank
+        has_emb = g == 0
+        blocks = dist_layers[g]
+        if has_emb and i == 0:
+            # GPT2Embedding: wte.weight / wpe.weight
+            out[rest] = value
+        elif i - (1 if has_emb else 0) < len(blocks):
+            layer = blocks[i - (1 if has_emb else 0)]
+            out[f"h.{layer}.{rest}"] = value
+        elif g == n_stages - 1:
+            # head Sequential: 0 = ln_f (FusedLayerNorm), 1 = TiedLMHead
+            hm = re.match(r"0\.(.*)", rest)
+            if hm:
+                out[f"ln_f.{hm.group(1)}"] = value
+            else:
+                hm = re.match(r"1\.wte\.(.*)", rest)
+                if hm:  # tied copy of the embedding matrix
+                    out["lm_head.weight"] = value
+        else:
+            raise ValueError(f"unmappable interleaved key {key}")
+    return out
+
+
 def merge_checkpoints(input_dir: str, output_path: str, prefix: str = "final_model") -> str:
     shards = load_shards(input_dir, prefix)
+    first = next(iter(next(iter(shards.values())).values()))
+    interleaved = "interleaved" in first.get("parallelism_info", {})
     pp_stages = {pp: merge_tp_shards({t: s["model_state_dict"] for t, s in tps.items()})
                  for pp, tps in shards.items()}
-    merged = merge_pp_stages(pp_stages)
+    if interleaved:
+        merged: Dict[str, torch.Tensor] = {}
+        for pp, tps in shards.items():
+            info = next(iter(tps.values()))["parallelism_info"]
+            merged.update(remap_interleaved_stage(pp_stages[pp], info))
+    else:
+        merged = merge_pp_stages(pp_stages)
     hf_state = convert_to_hf_format(merged)
-    config = next(iter(next(iter(shards.values())).values())).get("config", {})
+    config = first.get("config", {})
     os.makedirs(os.path.dirname(os.path.abspath(output_path)), exist_ok=True)
     torch.save({"model_state_dict": hf_state, "config": config}, output_path)
     return output_path
