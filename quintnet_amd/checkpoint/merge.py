@@ -25,7 +25,7 @@ from typing import Dict, Optional
 
 import torch
 
-__all__ = ["load_shards", "merge_tp_shards", "merge_pp_stages", "remap_interleaved_stage", "convert_to_hf_format", "merge_checkpoints"]
+__all__ = ["load_shards", "merge_tp_shards", "merge_pp_stages", "remap_interleaved_stage", "merge_ep_shards", "convert_to_hf_format", "merge_checkpoints"]
 
 _SHARD_RE = re.compile(r"_pp(\d+)_tp(\d+)\.pt$")
 
@@ -43,9 +43,9 @@ def load_shards(input_dir: str, prefix: Optional[str] = None) -> Dict[int, Dict[
         out.setdefault(pp, {})[tp] = torch.load(path, map_location="cpu", weights_only=False)
     if not out:
         raise FileNotFoundError(
-            f"no *_pp*_tp*.pt shards under {input_dir} (note: shards from "
-            "cp/ep meshes carry an extra _cp*/_ep* suffix and are not "
-            "mergeable by this TP/PP CLI)"
+            f"no *_pp*_tp*.pt shards under {input_dir} (shards with an _ep* "
+            "suffix must first be folded with merge_ep_shards(); _cp* "
+            "shards are full replicas along cp — rename one set)"
         )
     return out
 
@@ -170,6 +170,59 @@ def remap_interleaved_stage(
         else:
             raise ValueError(f"unmappable interleaved key {key}")
     return out
+
+
+_SHARD_EP_RE = re.compile(r"_pp(\d+)_tp(\d+)_ep(\d+)\.pt$")
+_EXPERT_KEY_RE = re.compile(r"(.*\bexperts)\.(\d+)\.(.*)")
+
+
+def merge_ep_shards(input_dir: str, prefix: str = "final_model",
+                    output_dir: Optional[str] = None) -> str:
+    """Fold expert-parallel shards ``{name}_pp{p}_tp{t}_ep{r}.pt`` into
+    standard ``{name}_pp{p}_tp{t}.pt`` shards with GLOBALLY numbered
+    experts (expert e lives on ep rank e // n_local —
+    parallel/expert_parallel.py), so the regular TP/PP merge CLI and the
+    EP=1 ``ExpertParallelMLP`` can consume them.  Replicated (non-expert)
+    tensors are taken from ep rank 0."""
+    groups: Dict[tuple, Dict[int, dict]] = {}
+    for path in sorted(glob.glob(os.path.join(input_dir, "*.pt"))):
+        base = os.path.basename(path)
+        if prefix and not base.startswith(prefix):
+            continue
+        m = _SHARD_EP_RE.search(base)
+        if not m:
+            continue
+        pp, tp, ep = int(m.group(1)), int(m.group(2)), int(m.group(3))
+        groups.setdefault((pp, tp), {})[ep] = torch.load(
+            path, map_location="cpu", weights_only=False)
+    if not groups:
+        raise FileNotFoundError(f"no *_pp*_tp*_ep*.pt shards under {input_dir}")
+    out_dir = output_dir or os.path.join(input_dir, "ep_merged")
+    os.makedirs(out_dir, exist_ok=True)
+    for (pp, tp), eps in groups.items():
+        ep_size = len(eps)
+        # n_local from the rank-0 shard's expert key set
+        n_local = 0
+        for k in eps[0]["model_state_dict"]:
+            m = _EXPERT_KEY_RE.match(k)
+            if m:
+                n_local = max(n_local, int(m.group(2)) + 1)
+        merged: Dict[str, torch.Tensor] = {}
+        for ep in range(ep_size):
+            for k, v in eps[ep]["model_state_dict"].items():
+                m = _EXPERT_KEY_RE.match(k)
+                if m:
+                    g = ep * n_local + int(m.group(2))
+                    merged[f"{m.group(1)}.{g}.{m.group(3)}"] = v
+                elif ep == 0:
+                    merged[k] = v
+        payload = dict(eps[0])
+        payload["model_state_dict"] = merged
+        info = dict(payload.get("parallelism_info", {}))
+        info["ep_merged"] = {"ep_size": ep_size, "n_local": n_local}
+        payload["parallelism_info"] = info
+        torch.save(payload, os.path.join(out_dir, f"{prefix}_pp{pp}_tp{tp}.pt"))
+    return out_dir
 
 
 def merge_checkpoints(input_dir: str, output_path: str, prefix: str = "final_model") -> str:

@@ -51,3 +51,34 @@ def test_interleaved_save_merge_roundtrip():
     assert torch.equal(hf["transformer.wpe.weight"], full.embedding.wpe.weight)
     assert torch.equal(hf["transformer.ln_f.weight"], full.head[0].weight)
     assert torch.equal(hf["lm_head.weight"], full.embedding.wte.weight)
+
+
+def test_ep_shard_merge_renumbers_experts():
+    """EP shards (rank-local expert indices) fold into standard pp/tp
+    shards with global expert numbering; replicated keys from ep0."""
+    import tempfile
+
+    from quintnet_amd.checkpoint.merge import merge_ep_shards
+
+    with tempfile.TemporaryDirectory() as d:
+        for ep in range(2):
+            sd = {
+                "blocks.0.mlp.router.weight": torch.full((4, 8), float(ep)),
+                "blocks.0.mlp.experts.0.c_fc.weight": torch.full((8, 8), ep * 10.0),
+                "blocks.0.mlp.experts.1.c_fc.weight": torch.full((8, 8), ep * 10.0 + 1),
+                "blocks.0.ln_1.weight": torch.full((8,), float(ep)),
+            }
+            torch.save({"model_state_dict": sd, "parallelism_info": {}},
+                       os.path.join(d, f"final_model_pp0_tp0_ep{ep}.pt"))
+        out_dir = merge_ep_shards(d)
+        merged = torch.load(os.path.join(out_dir, "final_model_pp0_tp0.pt"),
+                            map_location="cpu", weights_only=False)
+        st = merged["model_state_dict"]
+        # global experts 0..3 with the right sources
+        assert float(st["blocks.0.mlp.experts.0.c_fc.weight"][0, 0]) == 0.0
+        assert float(st["blocks.0.mlp.experts.1.c_fc.weight"][0, 0]) == 1.0
+        assert float(st["blocks.0.mlp.experts.2.c_fc.weight"][0, 0]) == 10.0
+        assert float(st["blocks.0.mlp.experts.3.c_fc.weight"][0, 0]) == 11.0
+        # replicated keys come from ep0
+        assert float(st["blocks.0.ln_1.weight"][0]) == 0.0
+        assert merged["parallelism_info"]["ep_merged"]["ep_size"] == 2
