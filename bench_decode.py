@@ -1,0 +1,44 @@
+"""Serving decode benchmark: KV-cached autoregressive generation.
+
+GPU:  python bench_decode.py [--batch 1,8,32] [--new 64] [--prompt 128]
+Reports decode tokens/s per batch size (GPT-2 124M, bf16, int8-cache
+variant included).
+"""
+import argparse
+import time
+
+import torch
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--batch", type=str, default="1,8,32")
+    ap.add_argument("--new", type=int, default=64)
+    ap.add_argument("--prompt", type=int, default=128)
+    args = ap.parse_args()
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    dev = torch.device("cuda")
+    cfg = GPT2Config(dropout=0.0)
+    model = GPT2Stage(cfg, device=dev, dtype=torch.bfloat16).eval()
+
+    for bs in [int(b) for b in args.batch.split(",")]:
+        ids = torch.randint(0, cfg.vocab_size, (bs, args.prompt), device=dev)
+        for cache_dtype, tag in [(None, "bf16kv"), ("int8", "int8kv")]:
+            # warmup
+            model.generate(ids, max_new_tokens=4, cache_dtype=cache_dtype)
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            out = model.generate(ids, max_new_tokens=args.new,
+                                 cache_dtype=cache_dtype)
+            torch.cuda.synchronize()
+            dt = time.perf_counter() - t0
+            ntok = (out.shape[1] - args.prompt) * bs
+            print(f"bs={bs:<3} {tag}: {ntok / dt:8.1f} tok/s "
+                  f"({dt / (out.shape[1] - args.prompt) * 1e3:.2f} ms/token)",
+                  flush=True)
+
+
+if __name__ == "__main__":
+    main()
