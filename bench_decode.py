@@ -23,6 +23,8 @@ def main():
     cfg = GPT2Config(dropout=0.0)
     model = GPT2Stage(cfg, device=dev, dtype=torch.bfloat16).eval()
 
+    from quintnet_amd.models import StaticKVDecoder
+
     for bs in [int(b) for b in args.batch.split(",")]:
         ids = torch.randint(0, cfg.vocab_size, (bs, args.prompt), device=dev)
         for cache_dtype, tag in [(None, "bf16kv"), ("int8", "int8kv")]:
@@ -38,6 +40,22 @@ def main():
             print(f"bs={bs:<3} {tag}: {ntok / dt:8.1f} tok/s "
                   f"({dt / (out.shape[1] - args.prompt) * 1e3:.2f} ms/token)",
                   flush=True)
+        # hipGraph-captured static-cache decode (one replay per token)
+        dec = StaticKVDecoder(model, batch=bs,
+                              max_len=args.prompt + args.new + 8)
+        dec.generate(ids, max_new_tokens=4)  # prefill + capture warmup
+        torch.cuda.synchronize()
+        t0 = time.perf_counter()
+        out = dec.generate(ids, max_new_tokens=args.new)
+        torch.cuda.synchronize()
+        dt = time.perf_counter() - t0
+        ntok = args.new * bs
+        eq = "==" if torch.equal(
+            out[:, : args.prompt + args.new],
+            model.generate(ids, max_new_tokens=args.new)[:, : args.prompt + args.new],
+        ) else "!= eager(!)"
+        print(f"bs={bs:<3} graph : {ntok / dt:8.1f} tok/s "
+              f"({dt / args.new * 1e3:.2f} ms/token) {eq}", flush=True)
 
 
 if __name__ == "__main__":

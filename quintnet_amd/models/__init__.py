@@ -9,6 +9,7 @@ from .vit import (
     ClassificationHead,
 )
 from .gpt2 import (
+    StaticKVDecoder,
     GPT2Config,
     GPT2Embedding,
     GPT2Attention,
@@ -33,4 +34,5 @@ __all__ = [
     "GPT2Block",
     "GPT2ForInterleaving",
     "GPT2Stage",
+    "StaticKVDecoder",
 ]

@@ -5,8 +5,10 @@ from .mlp import GPT2MLP
 from .block import GPT2Block
 from .stage import GPT2Stage
 from .interleaved import GPT2ForInterleaving, TiedLMHead
+from .decode import StaticKVDecoder
 
 __all__ = [
+    "StaticKVDecoder",
     "GPT2Config",
     "GPT2Embedding",
     "GPT2Attention",
