@@ -358,8 +358,23 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
-        step()
+    phase_ms = None
+    for wi in range(args.warmup):
+        if wi == args.warmup - 1 and pg.world_size > 1 and pg.pp_size > 1:
+            # overlap evidence: per-phase HIP-event times for ONE warmup
+            # step (forward/backward/P2P comm/optimizer) — reported in
+            # the JSON, never inside the timed region
+            from quintnet_amd.utils.profiling import PhaseTimer
+
+            pt = PhaseTimer()
+            ptrainer.phase_timer = pt
+            step()
+            if use_cuda:
+                torch.cuda.synchronize()
+            phase_ms = {k: round(v, 2) for k, v in sorted(pt.summary().items())}
+            ptrainer.phase_timer = None
+        else:
+            step()
         if wd:
             wd.beat()
     sync()
@@ -411,6 +426,7 @@ def main():
             "parallelism": parallelism,
             "optimizer": "zero1-adamw",
             "tokens_per_sec": round(value * seq, 1) if task == "clm" else None,
+            "warmup_phase_ms": phase_ms,
         },
     }
     if rank == 0:

@@ -93,3 +93,30 @@ def test_generate_kv_cache_gpu():
     # require the vast majority of tokens to agree
     agree = (out == ref).float().mean().item()
     assert agree > 0.9, (agree, out, ref)
+
+
+@pytest.mark.gpu
+def test_graph_decode_gpu():
+    """StaticKVDecoder graph capture: token-exact vs its eager form."""
+    import torch
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage, StaticKVDecoder
+
+    torch.manual_seed(0)
+    cfg = GPT2Config(n_embd=256, n_layer=3, n_head=4, vocab_size=512,
+                     n_positions=128, dropout=0.0)
+    stage = GPT2Stage(cfg, device=torch.device("cuda"),
+                      dtype=torch.bfloat16).eval()
+    ids = torch.randint(0, cfg.vocab_size, (2, 16), device="cuda")
+
+    eager = StaticKVDecoder(stage, batch=2, max_len=128)
+    cap = StaticKVDecoder.capture
+    StaticKVDecoder.capture = lambda self: self  # keep _graph None
+    try:
+        want = eager.generate(ids, max_new_tokens=12)
+    finally:
+        StaticKVDecoder.capture = cap
+    dec = StaticKVDecoder(stage, batch=2, max_len=128)
+    have = dec.generate(ids, max_new_tokens=12)
+    assert dec._graph is not None
+    assert torch.equal(have, want)
