@@ -33,6 +33,14 @@ class DistributedBackend(abc.ABC):
     @abc.abstractmethod
     def all_reduce_tensor(self, tensor: torch.Tensor, async_op: bool = False): ...
 
+    def reduce_scatter_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                              async_op: bool = False):
+        raise NotImplementedError
+
+    def all_gather_into_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                               async_op: bool = False):
+        raise NotImplementedError
+
 
 class TorchDistributedBackend(DistributedBackend):
     """RCCL (or gloo) through torch.distributed on a given group."""
@@ -68,6 +76,23 @@ class TorchDistributedBackend(DistributedBackend):
             return None
         return dist.all_reduce(tensor, op=dist.ReduceOp.SUM, group=self.group, async_op=async_op)
 
+    def reduce_scatter_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                              async_op: bool = False):
+        if self.get_world_size() <= 1:
+            output.copy_(input[: output.numel()])
+            return None
+        return dist.reduce_scatter_tensor(
+            output, input, op=dist.ReduceOp.SUM, group=self.group,
+            async_op=async_op)
+
+    def all_gather_into_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                               async_op: bool = False):
+        if self.get_world_size() <= 1:
+            output.copy_(input)
+            return None
+        return dist.all_gather_into_tensor(output, input, group=self.group,
+                                           async_op=async_op)
+
 
 class LocalBackend(DistributedBackend):
     """No-op backend for single-process unit tests."""
@@ -85,4 +110,14 @@ class LocalBackend(DistributedBackend):
         return None
 
     def all_reduce_tensor(self, tensor: torch.Tensor, async_op: bool = False):
+        return None
+
+    def reduce_scatter_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                              async_op: bool = False):
+        output.copy_(input[: output.numel()])
+        return None
+
+    def all_gather_into_tensor(self, output: torch.Tensor, input: torch.Tensor,
+                               async_op: bool = False):
+        output.copy_(input)
         return None

@@ -40,6 +40,11 @@ class BucketConfig:
     # kept for API parity with the reference (core/config.py:47-84); flat
     # bucket views are always on here — that was the flag's intent.
     gradient_as_bucket_view: bool = True
+    # "all_reduce" (default, ZeRO-1 pairing) or "reduce_scatter" (ZeRO-2
+    # pairing: each rank receives only its 1/dp chunk of every bucket —
+    # 2/3 of the gradient bytes on the wire vs all-reduce; buckets are
+    # padded to a world_size multiple so chunks divide evenly)
+    grad_reduce_op: str = "all_reduce"
 
 
 @dataclasses.dataclass
@@ -59,14 +64,15 @@ class GradientBucket:
     def __init__(self, params: List[nn.Parameter], flat: torch.Tensor):
         self.params = params
         self.numel = sum(p.numel() for p in params)
-        self.flat = flat
+        self.flat = flat  # may carry a tail pad (reduce_scatter mode)
         self.views: List[torch.Tensor] = []
         off = 0
         for p in params:
             self.views.append(self.flat[off : off + p.numel()].view_as(p))
             off += p.numel()
         self.ready_count = 0
-        self.work = None  # in-flight async all-reduce
+        self.work = None  # in-flight async all-reduce / reduce-scatter
+        self.own_chunk: Optional[torch.Tensor] = None  # RS output (ZeRO-2)
 
     def attach_grads(self) -> None:
         for p, v in zip(self.params, self.views):
@@ -98,6 +104,7 @@ class DataParallel(nn.Module):
             self.backend = LocalBackend()
         self.world_size = self.backend.get_world_size()
         self.require_backward_grad_sync = True
+        self._reduce_scatter = False
         self.flat_grads: Dict[torch.dtype, torch.Tensor] = {}
         self.flat_params_order: Dict[torch.dtype, List[nn.Parameter]] = {}
         self.buckets: List[GradientBucket] = []
@@ -124,8 +131,15 @@ class DataParallel(nn.Module):
         if not params:
             return
         # one flat grad buffer per dtype; buckets are consecutive slices.
-        # reverse order ≈ backward completion order.  Buffer padded to a
-        # world_size multiple so a ZeRO-1 optimizer can shard it evenly.
+        # reverse order ≈ backward completion order.  ZeRO-1 pairing:
+        # buffer padded to a world_size multiple so the optimizer can
+        # shard it evenly.  reduce_scatter (ZeRO-2) pairing: EACH BUCKET
+        # padded to a world_size multiple so its 1/dp chunks divide.
+        rs = (
+            self.bucket_config.grad_reduce_op == "reduce_scatter"
+            and self.world_size > 1
+        )
+        self._reduce_scatter = rs
         params = list(reversed(params))
         by_dtype: Dict[torch.dtype, List[nn.Parameter]] = {}
         for p in params:
@@ -133,37 +147,46 @@ class DataParallel(nn.Module):
         cap = int(self.bucket_config.capacity_mb * 1024 * 1024)
         self.flat_grads: Dict[torch.dtype, torch.Tensor] = {}
         self.flat_params_order: Dict[torch.dtype, List[nn.Parameter]] = {}
+        ws = self.world_size
         for dt, ps in by_dtype.items():
-            total = sum(p.numel() for p in ps)
-            pad = (self.world_size - total % self.world_size) % self.world_size
-            buf = torch.zeros(total + pad, dtype=dt, device=ps[0].device)
-            self.flat_grads[dt] = buf
-            self.flat_params_order[dt] = ps
-            # carve buckets
+            esz = ps[0].element_size()
+            # group params into buckets by capacity
+            groups: List[List[nn.Parameter]] = []
             cur: List[nn.Parameter] = []
             cur_bytes = 0
-            start = 0
-            off = 0
-            esz = ps[0].element_size()
-
-            def flush(cur, start, off):
-                if not cur:
-                    return
-                bucket = GradientBucket(cur, buf[start:off])
-                bucket.attach_grads()
-                self.buckets.append(bucket)
-                for p in cur:
-                    self._param_to_bucket[id(p)] = bucket
-
             for p in ps:
                 sz = p.numel() * esz
                 if cur and cur_bytes + sz > cap:
-                    flush(cur, start, off)
-                    cur, cur_bytes, start = [], 0, off
+                    groups.append(cur)
+                    cur, cur_bytes = [], 0
                 cur.append(p)
                 cur_bytes += sz
-                off += p.numel()
-            flush(cur, start, off)
+            if cur:
+                groups.append(cur)
+            sizes = []
+            for g in groups:
+                n = sum(p.numel() for p in g)
+                if rs:
+                    n = ((n + ws - 1) // ws) * ws
+                sizes.append(n)
+            total = sum(sizes)
+            if not rs:
+                total += (ws - total % ws) % ws
+            buf = torch.zeros(total, dtype=dt, device=ps[0].device)
+            self.flat_grads[dt] = buf
+            self.flat_params_order[dt] = ps
+            start = 0
+            for g, n in zip(groups, sizes):
+                bucket = GradientBucket(g, buf[start : start + n])
+                bucket.attach_grads()
+                if rs:
+                    bucket.own_chunk = torch.zeros(
+                        n // ws, dtype=dt, device=buf.device
+                    )
+                self.buckets.append(bucket)
+                for p in g:
+                    self._param_to_bucket[id(p)] = bucket
+                start += n
 
     def _register_hooks(self) -> None:
         if self.world_size <= 1:
@@ -173,6 +196,13 @@ class DataParallel(nn.Module):
                 h = p.register_post_accumulate_grad_hook(self._make_hook(bucket))
                 self._hooks.append(h)
 
+    def _reduce_bucket(self, bucket: GradientBucket):
+        if self._reduce_scatter:
+            return self.backend.reduce_scatter_tensor(
+                bucket.own_chunk, bucket.flat, async_op=True
+            )
+        return self.backend.all_reduce_tensor(bucket.flat, async_op=True)
+
     def _make_hook(self, bucket: GradientBucket):
         def hook(_param):
             if not self.require_backward_grad_sync:
@@ -180,7 +210,7 @@ class DataParallel(nn.Module):
             bucket.ready_count += 1
             if bucket.ready_count == len(bucket.params) and bucket.work is None:
                 # async: RCCL stream overlaps with the rest of backward
-                bucket.work = self.backend.all_reduce_tensor(bucket.flat, async_op=True)
+                bucket.work = self._reduce_bucket(bucket)
 
         return hook
 
@@ -205,17 +235,24 @@ class DataParallel(nn.Module):
             self.require_backward_grad_sync = prev
 
     def finalize_gradients(self) -> None:
-        """Launch any pending reductions, wait all, apply MEAN scaling."""
+        """Launch any pending reductions, wait all, apply MEAN scaling.
+
+        reduce_scatter mode: only ``bucket.own_chunk`` holds reduced
+        (mean) gradients afterwards — ``param.grad`` views keep the
+        LOCAL sums (the ZeRO-2 optimizer consumes the chunks)."""
         if self.world_size <= 1 or not self.require_backward_grad_sync:
             return
         for bucket in self.buckets:
             if bucket.work is None:
-                bucket.work = self.backend.all_reduce_tensor(bucket.flat, async_op=True)
+                bucket.work = self._reduce_bucket(bucket)
         for bucket in self.buckets:
             if bucket.work is not None:
                 bucket.work.wait()
                 bucket.work = None
-            bucket.flat.div_(self.world_size)
+            if self._reduce_scatter:
+                bucket.own_chunk.div_(self.world_size)
+            else:
+                bucket.flat.div_(self.world_size)
             bucket.ready_count = 0
 
     def zero_grad(self, set_to_none: bool = False):  # noqa: ARG002 — views must persist
@@ -227,7 +264,10 @@ class DataParallel(nn.Module):
 
     def grad_buffer(self, dtype: torch.dtype = None):
         """The single flat grad buffer (for the ZeRO-1 optimizer) or None
-        when params span multiple dtypes."""
+        when params span multiple dtypes or reduce_scatter mode is on
+        (per-bucket padding breaks the contiguous 1/dp sharding)."""
+        if getattr(self, "_reduce_scatter", False):
+            return None
         if len(self.flat_grads) != 1:
             if dtype is not None:
                 return self.flat_grads.get(dtype)
