@@ -93,8 +93,9 @@ __device__ __forceinline__ void write_tile_lds(unsigned short* lds, const s16x8 
   }
 }
 
-template <int ACT, bool SAVE_PRE, bool USE_GLDS, int TBM, int TBN, int WR, int WC>
-__global__ __launch_bounds__(WR * WC * 64) void gemm_nt_kernel(
+template <int ACT, bool SAVE_PRE, bool USE_GLDS, int TBM, int TBN, int WR, int WC,
+          int MINW = 0>
+__global__ __launch_bounds__(WR * WC * 64, MINW) void gemm_nt_kernel(
     const unsigned short* __restrict__ A,  // [M,K]
     const unsigned short* __restrict__ B,  // [N,K]
     const unsigned short* __restrict__ bias,  // [N] or nullptr
@@ -795,11 +796,22 @@ void gemm_nt_launch(const unsigned short* A, const unsigned short* B,
   else if (mode == 3 && (N % 256 == 0)) tile = 3;
   else if (mode == 0 && big_tile && (M % 256 == 0) &&
            ((long long)(M / 256) * ((N + 127) / 128) >= 256)) tile = 2;
+  static int occ128 = -1;
+  if (occ128 < 0) {
+    const char* e = getenv("QN_GEMM_OCC");
+    occ128 = e ? atoi(e) : 0;  // 0 = compiler default (2 blocks/CU at 128²)
+  }
 #define QN_GEMM_LAUNCH(A_, S_, G_, TBM_, TBN_, WR_, WC_)                       \
-  hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, G_, TBM_, TBN_, WR_, WC_>),       \
-                     dim3(((M + TBM_ - 1) / TBM_) * ((N + TBN_ - 1) / TBN_)),  \
-                     dim3(WR_ * WC_ * 64), 0, stream, A, B, bias, C, Cpre, M,  \
-                     N, K)
+  do {                                                                         \
+    dim3 g_(((M + TBM_ - 1) / TBM_) * ((N + TBN_ - 1) / TBN_));                \
+    dim3 b_(WR_ * WC_ * 64);                                                   \
+    if (TBM_ == 128 && TBN_ == 128 && occ128 == 3)                             \
+      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, G_, TBM_, TBN_, WR_, WC_, 3>),\
+                         g_, b_, 0, stream, A, B, bias, C, Cpre, M, N, K);     \
+    else                                                                       \
+      hipLaunchKernelGGL((gemm_nt_kernel<A_, S_, G_, TBM_, TBN_, WR_, WC_>),   \
+                         g_, b_, 0, stream, A, B, bias, C, Cpre, M, N, K);     \
+  } while (0)
 #define QN_GEMM_CASE(A_, S_)                                                   \
   do {                                                                         \
     if (tile == 2) {                                                           \
