@@ -43,6 +43,18 @@ import torch
 import torch.distributed as dist
 
 
+def _write_tunableop_csv(path: str) -> None:
+    res = torch.cuda.tunable.get_results()
+    if not res:
+        return
+    os.makedirs(os.path.dirname(path), exist_ok=True)
+    with open(path, "w") as f:
+        for k, v in torch.cuda.tunable.get_validators():
+            f.write(f"Validator,{k},{v}\n")
+        for r in res:
+            f.write(",".join(str(x) for x in r) + "\n")
+
+
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
@@ -380,7 +392,9 @@ def main():
     sync()
     if use_cuda and os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
         try:
-            torch.cuda.tunable.write_file()  # persist tuned GEMM algorithms
+            # this build has no tunable.write_file() and never writes at
+            # exit — serialize the results ourselves
+            _write_tunableop_csv(_TUNABLE_CSV)
         except Exception:  # noqa: BLE001 — persistence is best-effort
             pass
     t0 = time.perf_counter()
