@@ -34,16 +34,18 @@ def _gelu_tanh(x: torch.Tensor) -> torch.Tensor:
 
 import os
 
-# Forward dispatch policy (A/B-measured, profiles/README.md r04 + r2
-# gemm_sweep3): per-shape winner between the hand-written MFMA kernels and
-# hipBLASLt (TunableOp-tuned).  Measured at M=16384 (bench micro 16):
-#   c_attn (N2304 K768):  library 893 vs custom 686  -> library
-#   c_proj (N768  K768):  library 337 vs custom 594  -> CUSTOM (1.76x)
-#   c_fc   (N3072 K768):  library 626 vs custom 543  -> library
-#   mlp    (N768  K3072): library 1158 vs custom 856 -> library
-# "auto" routes the skinny-N square-K class (where hipBLASLt collapses to
-# ~350 TF) to the custom kernel; QN_GEMM_FWD=custom forces the custom path
-# everywhere (per-shape best variant), =library forces hipBLASLt.
+# Forward dispatch policy (A/B-measured, profiles/gemm_tuned.log r2):
+# per-shape comparison of the hand-written MFMA kernels vs hipBLASLt at
+# its TunableOp-tuned best (M=16384, bench micro 16):
+#   c_attn (N2304 K768):  library 888 vs custom 676
+#   c_proj (N768  K768):  library 645 vs custom 557
+#   c_fc   (N3072 K768):  library 619 vs custom 546 (GELU-fused)
+#   mlp    (N768  K3072): library 1131 vs custom 864
+# The tuned library wins every bench shape, so "auto" keeps it — per the
+# MI355X design split the hand-written kernels own the FUSED hot ops
+# (attention, LN, CE, AdamW, wgrad) and the library serves plain GEMMs.
+# QN_GEMM_FWD=custom forces the MFMA kernels (per-shape best variant);
+# =library forces hipBLASLt everywhere.
 _FWD_MODE = os.environ.get("QN_GEMM_FWD", "auto")
 _WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
 
