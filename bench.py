@@ -29,9 +29,11 @@ import time
 # step).  Tuning runs during the UNTIMED warmup steps; the result CSV is
 # written explicitly after warmup (this build never writes it at exit)
 # and committed under profiles/, so later runs load it and skip tuning.
+# one shared CSV for every rank (local GEMM shapes are identical per
+# rank on these meshes); only rank 0 writes
 _TUNABLE_CSV = os.path.join(
     os.path.dirname(os.path.abspath(__file__)), "profiles",
-    "tunableop_gfx950_%d.csv" % int(os.environ.get("RANK", 0)),
+    "tunableop_gfx950_0.csv",
 )
 os.environ.setdefault("PYTORCH_TUNABLEOP_ENABLED", "1")
 os.environ.setdefault(
@@ -390,7 +392,8 @@ def main():
         if wd:
             wd.beat()
     sync()
-    if use_cuda and os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1":
+    if (use_cuda and os.environ.get("PYTORCH_TUNABLEOP_TUNING") == "1"
+            and int(os.environ.get("RANK", "0")) == 0):
         try:
             # this build has no tunable.write_file() and never writes at
             # exit — serialize the results ourselves
