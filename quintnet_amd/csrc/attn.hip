@@ -134,7 +134,9 @@ __device__ __forceinline__ void stage_transpose(
 // q/k/v strided [.., T, 64] slices; out written via its own strides
 // (so [B, T, H*64] layout comes out directly); lse2 [BH, T] f32.
 // ===========================================================================
-template <int MINW>
+#define KPAD 72  // K-tile LDS row stride (elems): conflict-free b128 frags
+
+template <int MINW, bool KLDS = false>
 __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
     const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
     const unsigned short* __restrict__ v, unsigned short* __restrict__ out,
@@ -144,7 +146,8 @@ __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
     long long ksB, long long ksH, long long ksT,
     long long vsB, long long vsH, long long vsT,
     long long osB, long long osH, long long osT) {
-  __shared__ unsigned short vt[64 * TPAD];
+  __shared__ unsigned short vt[64 * TPAD + (KLDS ? 32 * KPAD : 0)];
+  unsigned short* klds = vt + 64 * TPAD;  // [32][KPAD] row-major K tile
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & 63;
@@ -181,28 +184,53 @@ __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
   // the dominant cost of the unpipelined version.  Source addresses are
   // pointer-bumped (+= 32 rows per tile), never rebuilt.
   const unsigned short* kfp = frag_base(kp, ksT, 0, lane);
+  const unsigned short* ksp = stage_base(kp, ksT, 0);   // KLDS path
   const unsigned short* vsp = stage_base(vp, vsT, 0);
   const long long kstep = 32 * ksT, vstep = 32 * vsT;
   bf16x8 kf_n[4];
   s16x8 v_n = stage_at(vsp);
+  s16x8 k_n = {0, 0, 0, 0, 0, 0, 0, 0};
+  if constexpr (KLDS) {
+    k_n = stage_at(ksp);
+  } else {
 #pragma unroll
-  for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
+    for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
+  }
 
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    // cooperative V^T staging (write the prefetched rows)
+    // cooperative V^T (+ row-major K) staging: write the prefetched rows
     __syncthreads();
     stage_wr(vt, v_n);
+    if constexpr (KLDS) {
+      // [row][KPAD] image, 16B-aligned s16x8 stores
+      int r = threadIdx.x & 31;
+      int d0 = (threadIdx.x >> 5) << 3;
+      *reinterpret_cast<s16x8*>(klds + r * KPAD + d0) = k_n;
+    }
     __syncthreads();
 
     bf16x8 kf_c[4];
+    if constexpr (KLDS) {
+      // shared K fragments from LDS (stride 72: conflict-free)
+      const unsigned short* kl = klds + (lane & 31) * KPAD + ((lane >> 5) << 3);
 #pragma unroll
-    for (int kt = 0; kt < 4; ++kt) kf_c[kt] = kf_n[kt];
+      for (int kt = 0; kt < 4; ++kt)
+        kf_c[kt] = *reinterpret_cast<const bf16x8*>(kl + kt * 16);
+    } else {
+#pragma unroll
+      for (int kt = 0; kt < 4; ++kt) kf_c[kt] = kf_n[kt];
+    }
     if (kv0 + 32 < kv_end) {
       vsp += vstep;
-      kfp += kstep;
       v_n = stage_at(vsp);
+      if constexpr (KLDS) {
+        ksp += kstep;
+        k_n = stage_at(ksp);
+      } else {
+        kfp += kstep;
 #pragma unroll
-      for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
+        for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
+      }
     }
 
     if (!causal || kv0 <= qw + qoff + 31) {  // wave has >= one valid pair
@@ -698,21 +726,24 @@ void attn_fwd_launch(const unsigned short* q, const unsigned short* k,
                      int B, int H, int Tq, int Tk, int qoff, float scale,
                      int causal, const AttnStrides& st, hipStream_t stream) {
   dim3 grid(Tq / 128, B * H);
-  static int occ = -1;
+  static int occ = -1, klds = -1;
   if (occ < 0) {
     const char* e = getenv("QN_ATTN_FWD_OCC");
     occ = e ? atoi(e) : 3;
+    const char* e2 = getenv("QN_ATTN_KLDS");
+    klds = (e2 && e2[0] == '1') ? 1 : 0;
   }
-  if (occ >= 4)
-    hipLaunchKernelGGL(attn_fwd_kernel<4>, grid, dim3(256), 0, stream, q, k, v,
-                       out, lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
-                       st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB,
-                       st.oH, st.oT);
-  else
-    hipLaunchKernelGGL(attn_fwd_kernel<3>, grid, dim3(256), 0, stream, q, k, v,
-                       out, lse2, Tq, Tk, qoff, H, scale, causal, st.qB, st.qH,
-                       st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT, st.oB,
-                       st.oH, st.oT);
+#define QN_AFWD(MINW_, KL_)                                                    \
+  hipLaunchKernelGGL((attn_fwd_kernel<MINW_, KL_>), grid, dim3(256), 0,        \
+                     stream, q, k, v, out, lse2, Tq, Tk, qoff, H, scale,       \
+                     causal, st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB,  \
+                     st.vH, st.vT, st.oB, st.oH, st.oT)
+  if (occ >= 4) {
+    if (klds) QN_AFWD(4, true); else QN_AFWD(4, false);
+  } else {
+    if (klds) QN_AFWD(3, true); else QN_AFWD(3, false);
+  }
+#undef QN_AFWD
 }
 
 void attn_delta_launch(const unsigned short* dout, const unsigned short* out,
