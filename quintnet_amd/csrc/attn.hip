@@ -731,7 +731,7 @@ void attn_fwd_launch(const unsigned short* q, const unsigned short* k,
     const char* e = getenv("QN_ATTN_FWD_OCC");
     occ = e ? atoi(e) : 3;
     const char* e2 = getenv("QN_ATTN_KLDS");
-    klds = (e2 && e2[0] == '1') ? 1 : 0;
+    klds = (e2 && e2[0] == '0') ? 0 : 1;  // default ON: fwd 66.8->47.9 us (r2)
   }
 #define QN_AFWD(MINW_, KL_)                                                    \
   hipLaunchKernelGGL((attn_fwd_kernel<MINW_, KL_>), grid, dim3(256), 0,        \
