@@ -364,6 +364,10 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
     long long dsB, long long dsH, long long dsT,
     long long dqsB, long long dqsH, long long dqsT) {
   __shared__ unsigned short kt_lds[2][64 * TPAD];
+  // row-major K/V tiles: shared fragment source for the S/dP MFMAs
+  // (replaces 4x-redundant per-wave global fragment loads)
+  __shared__ unsigned short krow_l[2][32 * KPAD];
+  __shared__ unsigned short vrow_l[2][32 * KPAD];
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & 63;
@@ -393,32 +397,45 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
   for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
 
   const int kv_end = causal ? min(q0 + qoff + 128, Tk) : Tk;
-  // cross-tile software pipeline (like attn_bwd_dkv): S/dP MFMAs for
-  // tile i+1 interleave with the dQ accumulation of tile i; ONE barrier
-  // per tile; double-buffered Kt.
-  const unsigned short* kfp = frag_base(kp, ksT, 0, lane);
-  const unsigned short* vfp = frag_base(vp, vsT, 0, lane);
+  // cross-tile pipeline: dQ MFMAs of tile i, barrier, then S/dP MFMAs
+  // for tile i+1 from the freshly staged LDS row images; double-buffered.
   const unsigned short* ksp = stage_base(kp, ksT, 0);
+  const unsigned short* vsp = stage_base(vp, vsT, 0);
   const long long kstep = 32 * ksT, vstep = 32 * vsT;
+  const int srow_ = threadIdx.x & 31;
+  const int sd0_ = (threadIdx.x >> 5) << 3;
+  const int fl_ = (lane & 31) * KPAD + ((lane >> 5) << 3);
 
-  // prologue: stage Kt tile 0, prime s/dp for tile 0
-  stage_wr(kt_lds[0], stage_at(ksp));
-  bf16x8 kf_n[4], vf_n[4];
-#pragma unroll
-  for (int t = 0; t < 4; ++t) {
-    kf_n[t] = frag_at(kfp, t);
-    vf_n[t] = frag_at(vfp, t);
+  // prologue: stage tile 0 (Kt transposed + K/V row images)
+  {
+    s16x8 k0 = stage_at(ksp);
+    s16x8 v0 = stage_at(vsp);
+    stage_wr(kt_lds[0], k0);
+    *reinterpret_cast<s16x8*>(&krow_l[0][srow_ * KPAD + sd0_]) = k0;
+    *reinterpret_cast<s16x8*>(&vrow_l[0][srow_ * KPAD + sd0_]) = v0;
   }
-  f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[0], qf[0], zc, 0, 0, 0);
-  f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[0], dof[0], zc, 0, 0, 0);
-#pragma unroll
-  for (int t = 1; t < 4; ++t) {
-    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[t], qf[t], s, 0, 0, 0);
-    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[t], dof[t], dp_, 0, 0, 0);
+  s16x8 kst_n = {0, 0, 0, 0, 0, 0, 0, 0}, vst_n = kst_n;
+  if (32 < kv_end) {
+    kst_n = stage_at(ksp + kstep);
+    vst_n = stage_at(vsp + vstep);
   }
-  s16x8 kst_n = {0, 0, 0, 0, 0, 0, 0, 0};
-  if (32 < kv_end) kst_n = stage_at(ksp + kstep);
-  __syncthreads();  // Kt buf0 visible
+  __syncthreads();  // buf0 visible
+  f32x16 s, dp_;
+  {
+    const unsigned short* kr = &krow_l[0][fl_];
+    const unsigned short* vr = &vrow_l[0][fl_];
+    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        *reinterpret_cast<const bf16x8*>(kr), qf[0], zc, 0, 0, 0);
+    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        *reinterpret_cast<const bf16x8*>(vr), dof[0], zc, 0, 0, 0);
+#pragma unroll
+    for (int t = 1; t < 4; ++t) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(kr + t * 16), qf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(vr + t * 16), dof[t], dp_, 0, 0, 0);
+    }
+  }
 
   int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
@@ -450,14 +467,13 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
 
     if (have_next) {
       stage_wr(kt_lds[cur ^ 1], kst_n);
-      kfp += kstep;
-      vfp += vstep;
+      *reinterpret_cast<s16x8*>(&krow_l[cur ^ 1][srow_ * KPAD + sd0_]) = kst_n;
+      *reinterpret_cast<s16x8*>(&vrow_l[cur ^ 1][srow_ * KPAD + sd0_]) = vst_n;
       ksp += kstep;
-      if (kv0 + 64 < kv_end) kst_n = stage_at(ksp + kstep);
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        kf_n[t] = frag_at(kfp, t);
-        vf_n[t] = frag_at(vfp, t);
+      vsp += vstep;
+      if (kv0 + 64 < kv_end) {
+        kst_n = stage_at(ksp + kstep);
+        vst_n = stage_at(vsp + vstep);
       }
     }
 
@@ -471,16 +487,22 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
         dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, gf1, dqa[mt], 0, 0, 0);
       }
     }
+    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
     if (have_next) {
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[0], qf[0], zc, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[0], dof[0], zc, 0, 0, 0);
+      const unsigned short* kr = &krow_l[cur ^ 1][fl_];
+      const unsigned short* vr = &vrow_l[cur ^ 1][fl_];
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(kr), qf[0], zc, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(vr), dof[0], zc, 0, 0, 0);
 #pragma unroll
       for (int t = 1; t < 4; ++t) {
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf_n[t], qf[t], s, 0, 0, 0);
-        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf_n[t], dof[t], dp_, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(kr + t * 16), qf[t], s, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(vr + t * 16), dof[t], dp_, 0, 0, 0);
       }
     }
-    __syncthreads();
     cur ^= 1;
   }
 
@@ -520,6 +542,11 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   // independent), ONE barrier per tile.
   __shared__ unsigned short dot_lds[2][64 * TPAD];
   __shared__ unsigned short qt_lds[2][64 * TPAD];
+  // row-major q/dO tiles: shared fragment source for the S/dP MFMAs
+  // (replaces 4x-redundant per-wave global fragment loads — the fwd
+  // K-through-LDS result, +40% there)
+  __shared__ unsigned short qrow[2][32 * KPAD];
+  __shared__ unsigned short dorow[2][32 * KPAD];
   __shared__ alignas(16) float lse_t[2][32];
   __shared__ alignas(16) float del_t[2][32];
   const int bh = blockIdx.y;
@@ -561,29 +588,20 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
   const float* delp = delta + (long long)bh * Tq + q_start + threadIdx.x;
   const long long qstep = 32 * qsT, dstep = 32 * dsT;
 
-  // ---- prologue: stage tile 0 into buf0, prime s/dp for tile 0 ------------
+  // ---- prologue: stage tile 0 into buf0 (transposed + row images) ---------
+  const int srow_ = threadIdx.x & 31;
+  const int sd0_ = (threadIdx.x >> 5) << 3;
   {
     s16x8 d0 = stage_at(dosp);
     s16x8 q0 = stage_at(qsp);
     stage_wr(dot_lds[0], d0);
     stage_wr(qt_lds[0], q0);
+    *reinterpret_cast<s16x8*>(&dorow[0][srow_ * KPAD + sd0_]) = d0;
+    *reinterpret_cast<s16x8*>(&qrow[0][srow_ * KPAD + sd0_]) = q0;
     if (threadIdx.x < 32) {
       lse_t[0][threadIdx.x] = *lsep;
       del_t[0][threadIdx.x] = *delp;
     }
-  }
-  bf16x8 qf_n[4], dof_n[4];
-#pragma unroll
-  for (int t = 0; t < 4; ++t) {
-    qf_n[t] = frag_at(qfp, t);
-    dof_n[t] = frag_at(dofp, t);
-  }
-  f32x16 s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[0], kf[0], zc, 0, 0, 0);
-  f32x16 dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[0], vf[0], zc, 0, 0, 0);
-#pragma unroll
-  for (int t = 1; t < 4; ++t) {
-    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[t], kf[t], s, 0, 0, 0);
-    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[t], vf[t], dp_, 0, 0, 0);
   }
   // prefetch tile 1's staging rows + lse/del
   s16x8 dost_n = {0, 0, 0, 0, 0, 0, 0, 0}, qst_n = dost_n;
@@ -597,6 +615,25 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     }
   }
   __syncthreads();  // buf0 visible
+  // prime s/dp for tile 0 from the LDS row images (stride-72 reads are
+  // bank-conflict-free for the b128 fragment pattern)
+  const int fl_ = (lane & 31) * KPAD + ((lane >> 5) << 3);
+  f32x16 s, dp_;
+  {
+    const unsigned short* qr = &qrow[0][fl_];
+    const unsigned short* dr = &dorow[0][fl_];
+    s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        *reinterpret_cast<const bf16x8*>(qr), kf[0], zc, 0, 0, 0);
+    dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+        *reinterpret_cast<const bf16x8*>(dr), vf[0], zc, 0, 0, 0);
+#pragma unroll
+    for (int t = 1; t < 4; ++t) {
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(qr + t * 16), kf[t], s, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(dr + t * 16), vf[t], dp_, 0, 0, 0);
+    }
+  }
 
   int cur = 0;
   for (int qt0 = q_start; qt0 < Tq; qt0 += 32) {
@@ -648,16 +685,18 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
     }
 
     // stage tile i+1 into the other buffer (its last readers finished
-    // before the barrier at the end of the previous iteration)
+    // before the barrier of the previous iteration)
     if (have_next) {
       stage_wr(dot_lds[cur ^ 1], dost_n);
       stage_wr(qt_lds[cur ^ 1], qst_n);
+      *reinterpret_cast<s16x8*>(&dorow[cur ^ 1][srow_ * KPAD + sd0_]) = dost_n;
+      *reinterpret_cast<s16x8*>(&qrow[cur ^ 1][srow_ * KPAD + sd0_]) = qst_n;
       if (threadIdx.x < 32) {
         lse_t[cur ^ 1][threadIdx.x] = lse_n;
         del_t[cur ^ 1][threadIdx.x] = del_n;
       }
-      // issue tile i+2 staging prefetch + tile i+1 fragment loads
-      qfp += qstep; dofp += dstep; qsp += qstep; dosp += dstep;
+      // issue tile i+2 staging prefetch
+      qsp += qstep; dosp += dstep;
       lsep += 32; delp += 32;
       if (qt0 + 64 < Tq) {
         dost_n = stage_at(dosp + dstep);
@@ -666,11 +705,6 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
           lse_n = lsep[32];
           del_n = delp[32];
         }
-      }
-#pragma unroll
-      for (int t = 0; t < 4; ++t) {
-        qf_n[t] = frag_at(qfp, t);
-        dof_n[t] = frag_at(dofp, t);
       }
     }
 
@@ -691,16 +725,25 @@ __global__ __launch_bounds__(256, 2) void attn_bwd_dkv_kernel(
         dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q1f, gf1, dka[mt], 0, 0, 0);
       }
     }
+    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
     if (have_next) {
-      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[0], kf[0], zc, 0, 0, 0);
-      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[0], vf[0], zc, 0, 0, 0);
+      // S/dP for tile i+1 from the freshly staged row images — the
+      // next iteration's writes target buf[cur], whose readers all
+      // finished before the barrier above
+      const unsigned short* qr = &qrow[cur ^ 1][fl_];
+      const unsigned short* dr = &dorow[cur ^ 1][fl_];
+      s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(qr), kf[0], zc, 0, 0, 0);
+      dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+          *reinterpret_cast<const bf16x8*>(dr), vf[0], zc, 0, 0, 0);
 #pragma unroll
       for (int t = 1; t < 4; ++t) {
-        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf_n[t], kf[t], s, 0, 0, 0);
-        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof_n[t], vf[t], dp_, 0, 0, 0);
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(qr + t * 16), kf[t], s, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(dr + t * 16), vf[t], dp_, 0, 0, 0);
       }
     }
-    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
     cur ^= 1;
   }
 
