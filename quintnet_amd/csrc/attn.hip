@@ -146,8 +146,8 @@ __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
     long long ksB, long long ksH, long long ksT,
     long long vsB, long long vsH, long long vsT,
     long long osB, long long osH, long long osT) {
-  // double-buffered V^T + row-major K tiles: one barrier per kv tile
-  __shared__ unsigned short vt_b[2][64 * TPAD + (KLDS ? 32 * KPAD : 0)];
+  __shared__ unsigned short vt[64 * TPAD + (KLDS ? 32 * KPAD : 0)];
+  unsigned short* klds = vt + 64 * TPAD;  // [32][KPAD] row-major K tile
   const int bh = blockIdx.y;
   const int b = bh / H, h = bh % H;
   const int lane = threadIdx.x & 63;
@@ -187,59 +187,46 @@ __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
   const unsigned short* ksp = stage_base(kp, ksT, 0);   // KLDS path
   const unsigned short* vsp = stage_base(vp, vsT, 0);
   const long long kstep = 32 * ksT, vstep = 32 * vsT;
-  const int srow_ = threadIdx.x & 31;
-  const int sd0_ = (threadIdx.x >> 5) << 3;
   bf16x8 kf_n[4];
   s16x8 v_n = stage_at(vsp);
   s16x8 k_n = {0, 0, 0, 0, 0, 0, 0, 0};
   if constexpr (KLDS) {
     k_n = stage_at(ksp);
-    // prologue: stage tile 0 into buf0; prefetch tile 1's rows
-    stage_wr(vt_b[0], v_n);
-    *reinterpret_cast<s16x8*>(vt_b[0] + 64 * TPAD + srow_ * KPAD + sd0_) = k_n;
-    if (32 < kv_end) {
-      vsp += vstep;
-      ksp += kstep;
-      v_n = stage_at(vsp);
-      k_n = stage_at(ksp);
-    }
-    __syncthreads();
   } else {
 #pragma unroll
     for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
   }
 
-  int cur = 0;
   for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+    // cooperative V^T (+ row-major K) staging: write the prefetched rows
+    __syncthreads();
+    stage_wr(vt, v_n);
+    if constexpr (KLDS) {
+      // [row][KPAD] image, 16B-aligned s16x8 stores
+      int r = threadIdx.x & 31;
+      int d0 = (threadIdx.x >> 5) << 3;
+      *reinterpret_cast<s16x8*>(klds + r * KPAD + d0) = k_n;
+    }
+    __syncthreads();
+
     bf16x8 kf_c[4];
     if constexpr (KLDS) {
-      // stage tile i+1 into buf[cur^1] (its readers finished before the
-      // barrier that ended iteration i-1), prefetch tile i+2's rows
-      if (kv0 + 32 < kv_end) {
-        stage_wr(vt_b[cur ^ 1], v_n);
-        *reinterpret_cast<s16x8*>(vt_b[cur ^ 1] + 64 * TPAD + srow_ * KPAD + sd0_) = k_n;
-        if (kv0 + 64 < kv_end) {
-          vsp += vstep;
-          ksp += kstep;
-          v_n = stage_at(vsp);
-          k_n = stage_at(ksp);
-        }
-      }
-      // shared K fragments from LDS buf[cur] (stride 72: conflict-free)
-      const unsigned short* kl =
-          vt_b[cur] + 64 * TPAD + (lane & 31) * KPAD + ((lane >> 5) << 3);
+      // shared K fragments from LDS (stride 72: conflict-free)
+      const unsigned short* kl = klds + (lane & 31) * KPAD + ((lane >> 5) << 3);
 #pragma unroll
       for (int kt = 0; kt < 4; ++kt)
         kf_c[kt] = *reinterpret_cast<const bf16x8*>(kl + kt * 16);
     } else {
-      __syncthreads();
-      stage_wr(vt_b[0], v_n);
-      __syncthreads();
 #pragma unroll
       for (int kt = 0; kt < 4; ++kt) kf_c[kt] = kf_n[kt];
-      if (kv0 + 32 < kv_end) {
-        vsp += vstep;
-        v_n = stage_at(vsp);
+    }
+    if (kv0 + 32 < kv_end) {
+      vsp += vstep;
+      v_n = stage_at(vsp);
+      if constexpr (KLDS) {
+        ksp += kstep;
+        k_n = stage_at(ksp);
+      } else {
         kfp += kstep;
 #pragma unroll
         for (int kt = 0; kt < 4; ++kt) kf_n[kt] = frag_at(kfp, kt);
@@ -302,17 +289,12 @@ __global__ __launch_bounds__(256, MINW) void attn_fwd_kernel(
       // O^T[d][q] += V^T · P : A from vt
 #pragma unroll
       for (int mt = 0; mt < 2; ++mt) {
-        const unsigned short* av =
-            &vt_b[KLDS ? cur : 0][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        const unsigned short* av = &vt[(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
         bf16x8 a0 = *reinterpret_cast<const bf16x8*>(av);
         bf16x8 a1 = *reinterpret_cast<const bf16x8*>(av + 16);
         o[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, pf0, o[mt], 0, 0, 0);
         o[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, pf1, o[mt], 0, 0, 0);
       }
-    }
-    if constexpr (KLDS) {
-      __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
-      cur ^= 1;
     }
   }
 
