@@ -205,7 +205,10 @@ class LinearFunction(torch.autograd.Function):
                 and g.shape[1] % 128 == 0
                 and x2d.shape[1] % 128 == 0
                 and g.shape[0] % 64 == 0
-                and (g.shape[1] // 128) * (x2d.shape[1] // 128) < 128
+                and (
+                    (g.shape[1] // 128) * (x2d.shape[1] // 128) < 128
+                    or _WGRAD_MODE == "custom"  # force: A/B the 144-tile shapes
+                )
                 and _WGRAD_MODE != "library"
             ):
                 grad_w = _backend.ext().wgrad_tn(g.contiguous(), x2d.contiguous())
