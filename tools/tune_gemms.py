@@ -16,11 +16,14 @@ import torch
 
 
 def main():
+    # bound per-solution tuning cost; the full sweep must fit a gpurun call
+    torch.cuda.tunable.set_max_tuning_duration(10)
+    torch.cuda.tunable.set_max_tuning_iterations(30)
     if os.path.exists(CSV):
         torch.cuda.tunable.read_file(CSV)
     dev = torch.device("cuda")
     shapes = []
-    for M in (16384, 8192):          # micro 16 (pp1) and micro 8 (pp2)
+    for M in (32768, 16384, 8192):          # micro 16 (pp1) and micro 8 (pp2)
         for tp in (1, 2):            # full and TP-halved
             E, I, V = 768, 3072, 50257
             shapes += [
