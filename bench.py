@@ -381,9 +381,11 @@ def main():
     # contractual warmup steps and touches no benchmark state.
     if use_cuda:
         _a = torch.randn(8192, 8192, device=device, dtype=torch.bfloat16)
-        for _ in range(40):
-            _a @ _a
-        torch.cuda.synchronize()
+        _t0 = time.perf_counter()
+        while time.perf_counter() - _t0 < 2.0:
+            for _ in range(64):
+                _a @ _a
+            torch.cuda.synchronize()
         del _a
 
     phase_ms = None
