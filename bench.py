@@ -9,12 +9,13 @@ ranks, and rank 0 prints ONE JSON line.
 
 Mesh by GPU count (BASELINE.json configs): 1 → [1,1,1]; 2 → dp2;
 4 → dp4; 8 → the named 3D mesh [2,2,2] (dp×tp×pp, 1F1B).  Per-replica
-work is fixed (weak scaling): micro_batch 8 × grad_acc 4 = 32 sequences
-of 1024 tokens per optimizer step per DP replica, synthetic data,
-random-init weights, bf16 compute, ZeRO-1 AdamW.
+work is fixed (weak scaling): 32 sequences of 1024 tokens per optimizer
+step per DP replica (micro 32 × acc 1 at pp=1; micro 8 × acc 4 on the
+pipeline mesh), synthetic data, random-init weights, bf16 compute,
+ZeRO-1 AdamW.
 
 ``--model vit`` benchmarks the ViT-MNIST config instead
-(micro 8 × grad_acc 8 = 64 images/replica/step).
+(micro 16 × grad_acc 4 = 64 images/replica/step).
 """
 
 from __future__ import annotations
