@@ -11,6 +11,7 @@ from typing import Optional
 import torch.nn as nn
 
 from ..parallel import (
+    BucketConfig,
     DataParallel,
     DistributedConfig,
     PipelineParallelWrapper,
@@ -29,13 +30,17 @@ __all__ = [
 ]
 
 
-def _dp_wrap(model: nn.Module, pg) -> nn.Module:
+def _dp_wrap(model: nn.Module, pg, config=None) -> nn.Module:
     if pg.dp_size <= 1:
         return model
     cfg = DistributedConfig(
         rank=pg.dp_rank, world_size=pg.dp_size, process_group=pg.get_group("dp")
     )
-    return DataParallel(model, config=cfg)
+    # zero_stage 2: reduce-scatter grad buckets (pair with optim.Zero2AdamW)
+    bc = None
+    if int((config or {}).get("zero_stage", 1)) == 2:
+        bc = BucketConfig(grad_reduce_op="reduce_scatter")
+    return DataParallel(model, config=cfg, bucket_config=bc)
 
 
 def _tp_apply(model: nn.Module, pg, device) -> nn.Module:
@@ -82,7 +87,7 @@ def _pp_wrap(model: nn.Module, pg, device, stage_module=None, config=None) -> nn
 class DataParallelCoordinator(BaseCoordinator):
     def parallelize(self) -> nn.Module:
         self.model.to(self.device)
-        return _dp_wrap(self.model, self.pg)
+        return _dp_wrap(self.model, self.pg, self.config)
 
 
 class TensorParallelCoordinator(BaseCoordinator):
@@ -100,13 +105,13 @@ class DPTCoordinator(BaseCoordinator):
 
     def parallelize(self) -> nn.Module:
         m = _tp_apply(self.model, self.pg, self.device)
-        return _dp_wrap(m, self.pg)
+        return _dp_wrap(m, self.pg, self.config)
 
 
 class DPPCoordinator(BaseCoordinator):
     def parallelize(self) -> nn.Module:
         m = _pp_wrap(self.model, self.pg, self.device, config=self.config)
-        return _dp_wrap(m, self.pg)
+        return _dp_wrap(m, self.pg, self.config)
 
 
 class TPPCoordinator(BaseCoordinator):
@@ -134,7 +139,7 @@ class Hybrid3DCoordinator(BaseCoordinator):
     def _parallelize_non_staged(self) -> nn.Module:
         m = _tp_apply(self.model, self.pg, self.device)
         m = _pp_wrap(m, self.pg, self.device, config=self.config)
-        return _dp_wrap(m, self.pg)
+        return _dp_wrap(m, self.pg, self.config)
 
     def _parallelize_staged(self) -> nn.Module:
         from ..checkpoint.distributed_loading import load_gpt2_distributed
@@ -168,4 +173,4 @@ class Hybrid3DCoordinator(BaseCoordinator):
             device=self.device,
         )
         m = _pp_wrap(None, self.pg, self.device, stage_module=stage)
-        return _dp_wrap(m, self.pg)
+        return _dp_wrap(m, self.pg, self.config)

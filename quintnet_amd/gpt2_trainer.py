@@ -48,7 +48,17 @@ class GPT2Trainer(Trainer):
             zkw = dict(lr=self.lr, weight_decay=wd, dp_group=dp_group,
                        max_grad_norm=None,  # clipping handled by the schedule
                        tp_group=tp_group, pp_group=pp_group)
-            if isinstance(self.model, DataParallel):
+            if isinstance(self.model, DataParallel) and getattr(
+                self.model, "_reduce_scatter", False
+            ):
+                # zero_stage 2 (config → coordinator built reduce-scatter
+                # buckets): grads AND optimizer state sharded per bucket
+                from .optim import Zero2AdamW
+
+                self.optimizer = Zero2AdamW(
+                    self.model, lr=self.lr, weight_decay=wd, max_grad_norm=None
+                )
+            elif isinstance(self.model, DataParallel):
                 self.optimizer = ZeroRedundancyAdamW.from_ddp(self.model, **zkw)
             else:
                 self.optimizer = ZeroRedundancyAdamW(self.model.parameters(), **zkw)
