@@ -109,6 +109,53 @@ def _partial(q32, k32, scale, causal_mode, Tl):
     return s
 
 
+# ---------------------------------------------------------------------------
+# Fused-flash per-chunk path (GPU, D=64, Tl%128==0): each ring hop runs the
+# flash kernel on (Q shard, K/V chunk) and the chunk outputs merge with a
+# streaming base-2 log-sum-exp — O(Tl·D) state, no T² score matrix.  The
+# backward reuses the SAME kernels per chunk with the saved GLOBAL lse2
+# (they recompute P = exp2(s·scale·log2e − lse2), which is exactly the
+# globally-normalized probability of that chunk's keys).
+# ---------------------------------------------------------------------------
+def _ring_flash_ok(q: torch.Tensor, k: torch.Tensor) -> bool:
+    from ..ops.attention import _flash_ok
+
+    return (
+        _flash_ok(q)
+        and k.shape[-2] == q.shape[-2]
+        and q.dtype == torch.bfloat16
+    )
+
+
+def _ring_flash_fwd_hop(q, kc, vc, scale, diag, out_run, lse_run):
+    """One chunk through the fused kernel + streaming LSE merge.
+
+    out_run [B,H,Tl,D] fp32 (normalized so far), lse_run [B,H,Tl,1] fp32
+    base-2; returns the updated pair."""
+    from .. import _C
+
+    o_j = torch.empty_like(q)
+    lse_j = _C.attn_fwd(q.contiguous(), kc, vc, o_j, scale, diag, 0)
+    B, H, Tl, _ = q.shape
+    lse_j = lse_j.view(B, H, Tl, 1)
+    m = torch.maximum(lse_run, lse_j)
+    a = torch.exp2(lse_run - m)
+    b = torch.exp2(lse_j - m)
+    denom = a + b
+    out_run = (out_run * a + o_j.float() * b) / denom
+    lse_run = m + torch.log2(denom)
+    return out_run, lse_run
+
+
+def _ring_flash_bwd_hop(q, kc, vc, out, dout, lse2_flat, scale, diag,
+                        dqp, dkp, dvp):
+    """Per-chunk backward via the fused kernels (writes dqp/dkp/dvp)."""
+    from .. import _C
+
+    _C.attn_bwd(q, kc, vc, out, dout, lse2_flat, dqp, dkp, dvp, scale,
+                bool(diag), 0)
+
+
 class _RingAttention(torch.autograd.Function):
     @staticmethod
     def forward(ctx, q, k, v, group, causal):
@@ -118,36 +165,54 @@ class _RingAttention(torch.autograd.Function):
         rank = dist.get_rank(group=group) if world > 1 else 0
         B, H, Tl, D = q.shape
         scale = 1.0 / math.sqrt(D)
-        q32 = q.float()
-        m = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
-        l = torch.zeros(B, H, Tl, 1, device=q.device)
-        acc = torch.zeros(B, H, Tl, D, device=q.device)
+        flash = _ring_flash_ok(q, k)
         kc, vc = k.contiguous(), v.contiguous()  # ring buffers must be
-        for s_hop in range(world):
-            j = (rank - s_hop) % world
-            if not causal or j <= rank:
-                sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
-                # every processed chunk has >=1 unmasked key per row (the
-                # diagonal chunk includes key<=query within-chunk), so mj is
-                # finite and m_new is finite from the first hop; the initial
-                # m=-inf gives alpha=exp(-inf)=0 naturally.
-                mj = sc.amax(dim=-1, keepdim=True)
-                m_new = torch.maximum(m, mj)
-                alpha = torch.exp(m - m_new)
-                p = torch.exp(sc - m_new)
-                acc = acc * alpha + torch.matmul(p, vc.float())
-                l = l * alpha + p.sum(dim=-1, keepdim=True)
-                m = m_new
-            if s_hop < world - 1:
-                kc, vc = _ring_swap([kc, vc], rank, world, group)
-        out = (acc / l.clamp(min=1e-30)).to(q.dtype)
-        lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # [B,H,Tl]
+        if flash:
+            qc = q.contiguous()
+            out_run = torch.zeros(B, H, Tl, D, device=q.device)
+            lse_run = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
+            for s_hop in range(world):
+                j = (rank - s_hop) % world
+                if not causal or j <= rank:
+                    out_run, lse_run = _ring_flash_fwd_hop(
+                        qc, kc, vc, scale, 1 if (causal and j == rank) else 0,
+                        out_run, lse_run,
+                    )
+                if s_hop < world - 1:
+                    kc, vc = _ring_swap([kc, vc], rank, world, group)
+            out = out_run.to(q.dtype)
+            lse = lse_run.reshape(B * H, Tl)  # base-2, the kernels' format
+        else:
+            q32 = q.float()
+            m = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
+            l = torch.zeros(B, H, Tl, 1, device=q.device)
+            acc = torch.zeros(B, H, Tl, D, device=q.device)
+            for s_hop in range(world):
+                j = (rank - s_hop) % world
+                if not causal or j <= rank:
+                    sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
+                    # every processed chunk has >=1 unmasked key per row (the
+                    # diagonal chunk includes key<=query within-chunk), so mj is
+                    # finite and m_new is finite from the first hop; the initial
+                    # m=-inf gives alpha=exp(-inf)=0 naturally.
+                    mj = sc.amax(dim=-1, keepdim=True)
+                    m_new = torch.maximum(m, mj)
+                    alpha = torch.exp(m - m_new)
+                    p = torch.exp(sc - m_new)
+                    acc = acc * alpha + torch.matmul(p, vc.float())
+                    l = l * alpha + p.sum(dim=-1, keepdim=True)
+                    m = m_new
+                if s_hop < world - 1:
+                    kc, vc = _ring_swap([kc, vc], rank, world, group)
+            out = (acc / l.clamp(min=1e-30)).to(q.dtype)
+            lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # natural log
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.group = group
         ctx.causal = causal
         ctx.world = world
         ctx.rank = rank
         ctx.scale = scale
+        ctx.flash = flash
         return out
 
     @staticmethod
@@ -156,14 +221,43 @@ class _RingAttention(torch.autograd.Function):
         group, causal = ctx.group, ctx.causal
         world, rank, scale = ctx.world, ctx.rank, ctx.scale
         B, H, Tl, D = q.shape
+        kc, vc = k.contiguous(), v.contiguous()
+        dk_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        dv_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        if ctx.flash:
+            qc = q.contiguous()
+            doutc = dout.contiguous().to(q.dtype)
+            outc = out.contiguous()
+            lse_flat = lse.contiguous()  # [B*H, Tl] base-2
+            dq32 = torch.zeros(B, H, Tl, D, device=q.device)
+            dqp = torch.empty_like(qc)
+            dkp = torch.empty_like(qc)
+            dvp = torch.empty_like(qc)
+            for s_hop in range(world):
+                j = (rank - s_hop) % world
+                if not causal or j <= rank:
+                    _ring_flash_bwd_hop(
+                        qc, kc, vc, outc, doutc, lse_flat, scale,
+                        1 if (causal and j == rank) else 0, dqp, dkp, dvp,
+                    )
+                    dq32 += dqp.float()
+                    dk_acc += dkp.float()
+                    dv_acc += dvp.float()
+                kc, vc, dk_acc, dv_acc = _ring_swap(
+                    [kc, vc, dk_acc, dv_acc], rank, world, group
+                )
+            return (
+                dq32.to(q.dtype),
+                dk_acc.to(k.dtype),
+                dv_acc.to(v.dtype),
+                None,
+                None,
+            )
         q32, dout32, out32 = q.float(), dout.float(), out.float()
         delta = (dout32 * out32).sum(dim=-1, keepdim=True)  # [B,H,Tl,1]
         lse_ = lse.unsqueeze(-1)
         dq = torch.zeros_like(q32)
         # grad accumulators travel with their chunk for a full cycle
-        kc, vc = k.contiguous(), v.contiguous()
-        dk_acc = torch.zeros(B, H, Tl, D, device=q.device)
-        dv_acc = torch.zeros(B, H, Tl, D, device=q.device)
         for s_hop in range(world):
             j = (rank - s_hop) % world
             if not causal or j <= rank:

@@ -373,3 +373,76 @@ def test_fp8_linear_forward():
     out.sum().backward()
     assert x.grad is not None and w.grad is not None
     assert torch.isfinite(x.grad).all() and torch.isfinite(w.grad).all()
+
+
+def test_ring_flash_chunks_match_direct():
+    """Fused-flash per-chunk ring path (context_parallel): a single-rank
+    simulation of the cp=2 chunk walk (diag chunk + one full chunk with
+    streaming base-2 LSE merge, then per-chunk fused backward with the
+    global lse2) must match the direct flash kernel with q_offset."""
+    import math
+
+    import torch
+
+    from quintnet_amd.ops.attention import attention
+    from quintnet_amd.parallel.context_parallel import (
+        _ring_flash_bwd_hop,
+        _ring_flash_fwd_hop,
+    )
+
+    torch.manual_seed(0)
+    B, H, T, D = 2, 3, 512, 64
+    Tl = T // 2
+    dev = torch.device("cuda")
+    q = torch.randn(B, H, T, D, device=dev, dtype=torch.bfloat16)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    scale = 1.0 / math.sqrt(D)
+
+    # "rank 1" of cp=2: q shard = rows [Tl, 2Tl)
+    qs = q[:, :, Tl:].contiguous()
+    k0, v0 = k[:, :, :Tl].contiguous(), v[:, :, :Tl].contiguous()
+    k1, v1 = k[:, :, Tl:].contiguous(), v[:, :, Tl:].contiguous()
+
+    out_run = torch.zeros(B, H, Tl, D, device=dev)
+    lse_run = torch.full((B, H, Tl, 1), float("-inf"), device=dev)
+    # ring hop order: own (diagonal) chunk first, then the previous rank's
+    out_run, lse_run = _ring_flash_fwd_hop(qs, k1, v1, scale, 1, out_run, lse_run)
+    out_run, lse_run = _ring_flash_fwd_hop(qs, k0, v0, scale, 0, out_run, lse_run)
+    out = out_run.to(qs.dtype)
+
+    # reference: direct fused path over the full KV with q_offset
+    qr = qs.clone().requires_grad_(True)
+    kr = k.clone().requires_grad_(True)
+    vr = v.clone().requires_grad_(True)
+    ref = attention(qr, kr, vr, causal=True, q_offset=Tl)
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2), (
+        (out.float() - ref.float()).abs().max().item()
+    )
+
+    # backward: per-chunk fused kernels with the merged (global) lse2
+    dout = torch.randn_like(out)
+    ref.backward(dout)
+    lse_flat = lse_run.reshape(B * H, Tl).contiguous()
+    dq_acc = torch.zeros(B, H, Tl, D, device=dev)
+    dk_all = torch.zeros(B, H, T, D, device=dev)
+    dv_all = torch.zeros(B, H, T, D, device=dev)
+    dqp = torch.empty_like(qs)
+    dkp = torch.empty_like(qs)
+    dvp = torch.empty_like(qs)
+    for (kc, vc, diag, sl) in [(k1, v1, 1, slice(Tl, T)), (k0, v0, 0, slice(0, Tl))]:
+        _ring_flash_bwd_hop(qs, kc, vc, out, dout, lse_flat, scale, diag,
+                            dqp, dkp, dvp)
+        dq_acc += dqp.float()
+        dk_all[:, :, sl] += dkp.float()
+        dv_all[:, :, sl] += dvp.float()
+
+    assert torch.allclose(dq_acc, qr.grad.float(), atol=5e-2, rtol=5e-2), (
+        (dq_acc - qr.grad.float()).abs().max().item()
+    )
+    assert torch.allclose(dk_all, kr.grad.float(), atol=5e-2, rtol=5e-2), (
+        (dk_all - kr.grad.float()).abs().max().item()
+    )
+    assert torch.allclose(dv_all, vr.grad.float(), atol=5e-2, rtol=5e-2), (
+        (dv_all - vr.grad.float()).abs().max().item()
+    )
