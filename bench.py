@@ -64,6 +64,8 @@ def parse_args():
     p.add_argument("--steps", type=int, default=20)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--model", choices=["gpt2", "vit"], default="gpt2")
+    p.add_argument("--gpt2-size", choices=["base", "medium", "large", "xl"],
+                   default="base", help="GPT-2 preset for --model gpt2")
     p.add_argument("--micro-batch", type=int, default=None)
     p.add_argument("--grad-acc", type=int, default=None)
     p.add_argument("--seq-len", type=int, default=1024)
@@ -100,8 +102,9 @@ def build_gpt2(args, pg, device, dtype):
                          n_positions=args.seq_len, dropout=0.0,
                          sequence_parallel=sp)
     else:
-        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len),
-                         sequence_parallel=sp)
+        cfg = GPT2Config.from_name(args.gpt2_size, dropout=0.0,
+                                   n_positions=max(1024, args.seq_len),
+                                   sequence_parallel=sp)
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     stage = GPT2Stage(
         cfg,
@@ -249,7 +252,9 @@ def main():
         task = "clm"
         seq, hidden = args.seq_len, cfg.n_embd
         pipe_shape_seq = getattr(args, "pipe_seq", seq)
-        model_name = "gpt2-124M" if not args.tiny else "gpt2-tiny"
+        sizes = {"base": "gpt2-124M", "medium": "gpt2-355M",
+                 "large": "gpt2-774M", "xl": "gpt2-1.6B"}
+        model_name = sizes[args.gpt2_size] if not args.tiny else "gpt2-tiny"
     else:
         micro_b = args.micro_batch or 16
         grad_acc = args.grad_acc or 4
