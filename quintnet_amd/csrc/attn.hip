@@ -5,11 +5,15 @@
 //    query row (col = lane&31) and its softmax state (m, l) is
 //    lane-local: the row reduce is a per-reg max/sum + one
 //    __shfl_xor(32), no LDS round trip;
-//  * v_mfma_f32_32x32x16_bf16 tiles; K/Q/V/dO fragments load DIRECTLY
-//    from global memory (A/B fragment layout = 8 contiguous d-elements
-//    per lane, which is exactly a row-major [T, D] slice) — strides are
-//    passed in, so the q/k/v views of the fused QKV projection are
-//    consumed with ZERO transpose/contiguous copies;
+//  * v_mfma_f32_32x32x16_bf16 tiles; the PER-WAVE operands (Q/dO in fwd
+//    and dq; K/V resident in dkv) load directly from global memory
+//    (A/B fragment layout = 8 contiguous d-elements per lane = a
+//    row-major [T, D] slice; strides passed in, so the fused-QKV views
+//    are consumed with zero transpose copies).  The BLOCK-SHARED
+//    operands (K in fwd; K/V in dq; Q/dO in dkv) are staged ONCE per
+//    block into stride-72 LDS row images — bank-conflict-free for the
+//    b128 fragment reads — instead of 4x-redundant per-wave global
+//    loads (r2: fwd +40%, bwd +77%; PMC showed 60-70% WAIT_ANY before);
 //  * P relayout for the PV mfma via v_cvt_pk_bf16_f32 + permlane32_swap
 //    (guide T12): converts the 16 f32 score regs into the bf16 A/B
 //    fragment in-register;

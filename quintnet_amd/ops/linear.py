@@ -1,12 +1,14 @@
 """Linear (GEMM) op with fused bias/activation epilogue.
 
-Forward runs the hand-written CDNA4 MFMA GEMM (csrc/gemm.hip — an NT
-GEMM: C[M,N] = A[M,K]·B[N,K]^T matching nn.Linear's [out,in] weight
-layout) with the bias add and optional GELU/ReLU fused into the
-epilogue.  Backward dgrad/dwgrad are plain (unfused) GEMMs and go
-through rocBLAS/hipBLASLt via torch.matmul — per the MI355X design
-split: hand-written kernels for fused hot ops, the vendor GEMM library
-for plain GEMMs.
+Per-shape measured dispatch between the hand-written CDNA4 MFMA GEMMs
+(csrc/gemm.hip — NT form C[M,N] = A[M,K]·B[N,K]^T matching nn.Linear's
+[out,in] layout, fused bias/GELU/ReLU epilogues, incl. a persistent
+cross-tile-pipelined variant) and TunableOp-tuned hipBLASLt.  The tuned
+library currently wins the plain GPT-2 forward shapes (see the policy
+table below); the custom split-K TN kernel serves the wgrad shapes it
+wins; dgrad is library.  MI355X design split: hand-written kernels own
+the FUSED hot ops (attention, LN, CE, AdamW, wgrad), the vendor library
+serves plain GEMMs.
 
 Replaces the implicit cuBLAS GEMMs at reference
 parallelism/tensor_parallel/layers.py:119,211 and utils/GPT2/*.
