@@ -78,6 +78,7 @@ class GPT2Attention(nn.Module):
         from ...parallel.context_parallel import (
             context_parallel_attention,
             ring_attention,
+            zigzag_ring_attention,
         )
 
         B, T, _ = qkv.shape
@@ -90,10 +91,13 @@ class GPT2Attention(nn.Module):
         q = heads(qkv[:, :, :hl])
         k = heads(qkv[:, :, hl : 2 * hl])
         v = heads(qkv[:, :, 2 * hl :])
-        use_ring = getattr(self.config, "cp_ring", False) or os.environ.get(
-            "QN_CP_RING"
-        ) == "1"
-        fn = ring_attention if use_ring else context_parallel_attention
+        if getattr(self.config, "cp_zigzag", False):
+            fn = zigzag_ring_attention
+        else:
+            use_ring = getattr(self.config, "cp_ring", False) or os.environ.get(
+                "QN_CP_RING"
+            ) == "1"
+            fn = ring_attention if use_ring else context_parallel_attention
         out = fn(q, k, v, self.cp_group, causal=True)
         return out.permute(0, 2, 1, 3).reshape(B, T, hl)
 

@@ -25,6 +25,7 @@ class GPT2Config:
     sequence_parallel: bool = False
     fp8: bool = False  # experimental: e4m3 forward GEMMs, bf16 backward
     cp_ring: bool = False  # context parallelism: ring KV exchange
+    cp_zigzag: bool = False  # CP: zigzag load-balanced ring (implies ring)
     n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
     moe_top_k: int = 2  # Megatron-SP over the TP group
 

@@ -26,12 +26,16 @@ class GPT2Embedding(nn.Module):
         nn.init.normal_(self.wte.weight, std=config.initializer_range)
         nn.init.normal_(self.wpe.weight, std=config.initializer_range)
 
-    def forward(self, input_ids: torch.Tensor, pos_offset: int = 0) -> torch.Tensor:
+    def forward(self, input_ids: torch.Tensor, pos_offset: int = 0,
+                positions: torch.Tensor = None) -> torch.Tensor:
         # pos_offset > 0: incremental decoding — positions start past the
         # KV cache (the fused kernel indexes positions 0..T-1 of whatever
-        # wpe slice it is given, so a slice view is all that's needed)
+        # wpe slice it is given, so a slice view is all that's needed).
+        # positions: explicit NON-CONTIGUOUS position ids (zigzag CP).
         wpe = self.wpe.weight
-        if pos_offset:
+        if positions is not None:
+            wpe = wpe.index_select(0, positions)
+        elif pos_offset:
             wpe = wpe[pos_offset : pos_offset + input_ids.shape[1]]
         x = embedding_pair(input_ids, self.wte.weight, wpe)
         return self.drop(x)

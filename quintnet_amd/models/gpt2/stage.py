@@ -89,11 +89,20 @@ class GPT2Stage(nn.Module):
             if self.cp_group is not None:
                 # context parallelism: x is this rank's ids SHARD (caller
                 # scatters, see parallel/context_parallel.py); positions
-                # start at the shard's global offset
+                # start at the shard's global offset (contiguous shards)
+                # or follow the zigzag chunk map
                 import torch.distributed as dist
 
-                cp_rank = dist.get_rank(group=self.cp_group)
-                x = self.embedding(x, pos_offset=cp_rank * x.shape[1])
+                if getattr(self.config, "cp_zigzag", False):
+                    from ...parallel.context_parallel import zigzag_positions
+
+                    cp = dist.get_world_size(group=self.cp_group)
+                    pos = zigzag_positions(x.shape[1] * cp, self.cp_group,
+                                           x.device)
+                    x = self.embedding(x, positions=pos)
+                else:
+                    cp_rank = dist.get_rank(group=self.cp_group)
+                    x = self.embedding(x, pos_offset=cp_rank * x.shape[1])
             else:
                 x = self.embedding(x)
             if self.sequence_parallel:

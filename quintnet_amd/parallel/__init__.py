@@ -16,6 +16,10 @@ from .tensor_parallel import (
 )
 from .context_parallel import (
     context_parallel_attention,
+    zigzag_ring_attention,
+    zigzag_to_context,
+    zigzag_positions,
+    zigzag_clm_targets,
     cp_causal_lm_loss,
     ring_attention,
     scatter_clm_targets,
@@ -45,6 +49,10 @@ __all__ = [
     "cp_causal_lm_loss",
     "scatter_clm_targets",
     "ring_attention",
+    "zigzag_ring_attention",
+    "zigzag_to_context",
+    "zigzag_positions",
+    "zigzag_clm_targets",
     "ZeRO3Block",
     "apply_zero3",
     "context_parallel_attention",

@@ -24,7 +24,11 @@ from ..ops.attention import attention
 __all__ = [
     "context_parallel_attention",
     "ring_attention",
+    "zigzag_ring_attention",
     "scatter_to_context",
+    "zigzag_to_context",
+    "zigzag_positions",
+    "zigzag_clm_targets",
     "scatter_clm_targets",
     "cp_causal_lm_loss",
 ]
@@ -85,6 +89,8 @@ def context_parallel_attention(
 # ---------------------------------------------------------------------------
 def _ring_swap(tensors, rank, world, group):
     """Send tensors to rank+1, receive the same shapes from rank-1."""
+    if world == 1:
+        return tensors
     nxt = (rank + 1) % world
     prv = (rank - 1 + world) % world
     ops = []
@@ -280,6 +286,213 @@ class _RingAttention(torch.autograd.Function):
             None,
             None,
         )
+
+
+# ---------------------------------------------------------------------------
+# Zigzag (load-balanced) ring attention: the sequence is split into 2·cp
+# chunks and rank r holds chunks (r, 2cp−1−r) — under the causal mask every
+# rank then processes the SAME number of (q-half, kv-half) blocks per ring
+# cycle (plain ring gives rank r work ∝ r+1).  Per block the fused flash
+# kernel runs at half-shard size (fp32 composed fallback on CPU); merges
+# are streaming per q-half.
+# ---------------------------------------------------------------------------
+def _zz_ids(rank: int, world: int):
+    return (rank, 2 * world - 1 - rank)
+
+
+def zigzag_to_context(x: torch.Tensor, cp_group, dim: int = 1) -> torch.Tensor:
+    """Keep this rank's ZIGZAG sequence shard: chunks (r, 2cp−1−r) of the
+    2·cp-way split, concatenated."""
+    world = _ws(cp_group)
+    if world == 1:
+        return x
+    if x.shape[dim] % (2 * world) != 0:
+        raise ValueError(
+            f"zigzag CP: dim {dim} ({x.shape[dim]}) must divide by 2*cp={2*world}"
+        )
+    rank = dist.get_rank(group=cp_group)
+    lo, hi = _zz_ids(rank, world)
+    ch = x.chunk(2 * world, dim=dim)
+    return torch.cat([ch[lo], ch[hi]], dim=dim).contiguous()
+
+
+def zigzag_positions(T: int, cp_group, device) -> torch.Tensor:
+    """Global position ids of this rank's zigzag shard (for wpe)."""
+    world = _ws(cp_group)
+    rank = dist.get_rank(group=cp_group) if world > 1 else 0
+    h = T // (2 * world)
+    lo, hi = _zz_ids(rank, world)
+    return torch.cat([
+        torch.arange(lo * h, (lo + 1) * h, device=device),
+        torch.arange(hi * h, (hi + 1) * h, device=device),
+    ])
+
+
+def zigzag_clm_targets(labels: torch.Tensor, cp_group, ignore_index: int = -100):
+    """Shift the FULL label sequence, then keep the zigzag shard."""
+    from ..ops import shift_labels
+
+    return zigzag_to_context(shift_labels(labels, ignore_index), cp_group, dim=1)
+
+
+class _ZigzagRingAttention(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q, k, v, group, causal):
+        import math
+
+        world = _ws(group)
+        rank = dist.get_rank(group=group) if world > 1 else 0
+        B, H, Tl, D = q.shape
+        assert Tl % 2 == 0, "zigzag shard must hold two equal half-chunks"
+        h = Tl // 2
+        scale = 1.0 / math.sqrt(D)
+        flash = _ring_flash_ok(q, k) and (h % 128 == 0)
+        cq = _zz_ids(rank, world)
+        kc, vc = k.contiguous(), v.contiguous()
+        if flash:
+            qc = q.contiguous()
+            qh = [qc[:, :, :h].contiguous(), qc[:, :, h:].contiguous()]
+            out_run = [torch.zeros(B, H, h, D, device=q.device) for _ in range(2)]
+            lse_run = [
+                torch.full((B, H, h, 1), float("-inf"), device=q.device)
+                for _ in range(2)
+            ]
+        else:
+            q32 = q.float()
+            m = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
+            l = torch.zeros(B, H, Tl, 1, device=q.device)
+            acc = torch.zeros(B, H, Tl, D, device=q.device)
+        for s_hop in range(world):
+            j = (rank - s_hop) % world
+            ck = _zz_ids(j, world)
+            for qi in range(2):
+                for ki in range(2):
+                    if causal and ck[ki] > cq[qi]:
+                        continue
+                    diag = 1 if (causal and ck[ki] == cq[qi]) else 0
+                    ksl = kc[:, :, ki * h : (ki + 1) * h]
+                    vsl = vc[:, :, ki * h : (ki + 1) * h]
+                    if flash:
+                        out_run[qi], lse_run[qi] = _ring_flash_fwd_hop(
+                            qh[qi], ksl.contiguous(), vsl.contiguous(), scale,
+                            diag, out_run[qi], lse_run[qi],
+                        )
+                    else:
+                        sl = slice(qi * h, (qi + 1) * h)
+                        sc = _partial(q32[:, :, sl], ksl.float(), scale, diag, h)
+                        mj = sc.amax(dim=-1, keepdim=True)
+                        m_new = torch.maximum(m[:, :, sl], mj)
+                        alpha = torch.exp(m[:, :, sl] - m_new)
+                        p = torch.exp(sc - m_new)
+                        acc[:, :, sl] = acc[:, :, sl] * alpha + torch.matmul(
+                            p, vsl.float())
+                        l[:, :, sl] = l[:, :, sl] * alpha + p.sum(
+                            dim=-1, keepdim=True)
+                        m[:, :, sl] = m_new
+            if s_hop < world - 1:
+                kc, vc = _ring_swap([kc, vc], rank, world, group)
+        if flash:
+            out = torch.cat(out_run, dim=2).to(q.dtype)
+            lse = torch.cat(lse_run, dim=2).reshape(B * H, Tl)  # base-2
+        else:
+            out = (acc / l.clamp(min=1e-30)).to(q.dtype)
+            lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # natural
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.group, ctx.causal = group, causal
+        ctx.world, ctx.rank, ctx.scale = world, rank, scale
+        ctx.flash = flash
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        q, k, v, out, lse = ctx.saved_tensors
+        group, causal = ctx.group, ctx.causal
+        world, rank, scale = ctx.world, ctx.rank, ctx.scale
+        B, H, Tl, D = q.shape
+        h = Tl // 2
+        cq = _zz_ids(rank, world)
+        kc, vc = k.contiguous(), v.contiguous()
+        dk_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        dv_acc = torch.zeros(B, H, Tl, D, device=q.device)
+        if ctx.flash:
+            qc = q.contiguous()
+            doutc = dout.contiguous().to(q.dtype)
+            outc = out.contiguous()
+            lse4 = lse.view(B, H, Tl)
+            dq32 = torch.zeros(B, H, Tl, D, device=q.device)
+            dqp = torch.empty(B, H, h, D, device=q.device, dtype=q.dtype)
+            dkp = torch.empty_like(dqp)
+            dvp = torch.empty_like(dqp)
+            for s_hop in range(world):
+                j = (rank - s_hop) % world
+                ck = _zz_ids(j, world)
+                for qi in range(2):
+                    for ki in range(2):
+                        if causal and ck[ki] > cq[qi]:
+                            continue
+                        diag = 1 if (causal and ck[ki] == cq[qi]) else 0
+                        qsl = slice(qi * h, (qi + 1) * h)
+                        ksl = slice(ki * h, (ki + 1) * h)
+                        _ring_flash_bwd_hop(
+                            qc[:, :, qsl].contiguous(),
+                            kc[:, :, ksl].contiguous(),
+                            vc[:, :, ksl].contiguous(),
+                            outc[:, :, qsl].contiguous(),
+                            doutc[:, :, qsl].contiguous(),
+                            lse4[:, :, qsl].reshape(B * H, h).contiguous(),
+                            scale, diag, dqp, dkp, dvp,
+                        )
+                        dq32[:, :, qsl] += dqp.float()
+                        dk_acc[:, :, ksl] += dkp.float()
+                        dv_acc[:, :, ksl] += dvp.float()
+                kc, vc, dk_acc, dv_acc = _ring_swap(
+                    [kc, vc, dk_acc, dv_acc], rank, world, group)
+            return (dq32.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype),
+                    None, None)
+        q32, dout32, out32 = q.float(), dout.float(), out.float()
+        delta = (dout32 * out32).sum(dim=-1, keepdim=True)
+        lse_ = lse.unsqueeze(-1)
+        dq = torch.zeros_like(q32)
+        for s_hop in range(world):
+            j = (rank - s_hop) % world
+            ck = _zz_ids(j, world)
+            for qi in range(2):
+                for ki in range(2):
+                    if causal and ck[ki] > cq[qi]:
+                        continue
+                    diag = 1 if (causal and ck[ki] == cq[qi]) else 0
+                    qsl = slice(qi * h, (qi + 1) * h)
+                    ksl = slice(ki * h, (ki + 1) * h)
+                    sc = _partial(q32[:, :, qsl], kc[:, :, ksl].float(), scale,
+                                  diag, h)
+                    p = torch.exp(sc - lse_[:, :, qsl])
+                    dp = torch.matmul(dout32[:, :, qsl],
+                                      vc[:, :, ksl].float().transpose(-2, -1))
+                    ds = p * (dp - delta[:, :, qsl]) * scale
+                    dq[:, :, qsl] += torch.matmul(ds, kc[:, :, ksl].float())
+                    dk_acc[:, :, ksl] += torch.matmul(
+                        ds.transpose(-2, -1), q32[:, :, qsl])
+                    dv_acc[:, :, ksl] += torch.matmul(
+                        p.transpose(-2, -1), dout32[:, :, qsl])
+            kc, vc, dk_acc, dv_acc = _ring_swap(
+                [kc, vc, dk_acc, dv_acc], rank, world, group)
+        return (dq.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype),
+                None, None)
+
+
+def zigzag_ring_attention(
+    q: torch.Tensor,
+    k: torch.Tensor,
+    v: torch.Tensor,
+    cp_group,
+    causal: bool = True,
+) -> torch.Tensor:
+    """Load-balanced ring CP attention over ZIGZAG shards (chunks
+    (r, 2cp−1−r) of a 2·cp split — use ``zigzag_to_context`` /
+    ``zigzag_positions`` for the inputs).  Every rank processes the same
+    block count under the causal mask; per-block compute is the fused
+    flash kernel on GPU."""
+    return _ZigzagRingAttention.apply(q, k, v, cp_group, causal)
 
 
 def ring_attention(
