@@ -87,10 +87,9 @@ def context_parallel_attention(
 # path (streaming log-sum-exp merge); the fused flash kernel can be slotted
 # per chunk later.
 # ---------------------------------------------------------------------------
-def _ring_swap(tensors, rank, world, group):
-    """Send tensors to rank+1, receive the same shapes from rank-1."""
-    if world == 1:
-        return tensors
+def _ring_swap_begin(tensors, rank, world, group):
+    """Issue send-to-rank+1 / recv-from-rank-1 without waiting (overlap
+    with compute: sends only READ the live buffers)."""
     nxt = (rank + 1) % world
     prv = (rank - 1 + world) % world
     ops = []
@@ -101,7 +100,15 @@ def _ring_swap(tensors, rank, world, group):
         buf = torch.empty(t.shape, dtype=t.dtype, device=t.device)
         recvs.append(buf)
         ops.append(dist.P2POp(dist.irecv, buf, peer=prv, group=group))
-    for r in dist.batch_isend_irecv(ops):
+    return dist.batch_isend_irecv(ops), recvs
+
+
+def _ring_swap(tensors, rank, world, group):
+    """Send tensors to rank+1, receive the same shapes from rank-1."""
+    if world == 1:
+        return tensors
+    reqs, recvs = _ring_swap_begin(tensors, rank, world, group)
+    for r in reqs:
         r.wait()
     return recvs
 
@@ -179,13 +186,18 @@ class _RingAttention(torch.autograd.Function):
             lse_run = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
             for s_hop in range(world):
                 j = (rank - s_hop) % world
+                reqs = None
+                if s_hop < world - 1:  # overlap the hop exchange w/ compute
+                    reqs, bufs = _ring_swap_begin([kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     out_run, lse_run = _ring_flash_fwd_hop(
                         qc, kc, vc, scale, 1 if (causal and j == rank) else 0,
                         out_run, lse_run,
                     )
-                if s_hop < world - 1:
-                    kc, vc = _ring_swap([kc, vc], rank, world, group)
+                if reqs is not None:
+                    for r_ in reqs:
+                        r_.wait()
+                    kc, vc = bufs
             out = out_run.to(q.dtype)
             lse = lse_run.reshape(B * H, Tl)  # base-2, the kernels' format
         else:
@@ -195,6 +207,9 @@ class _RingAttention(torch.autograd.Function):
             acc = torch.zeros(B, H, Tl, D, device=q.device)
             for s_hop in range(world):
                 j = (rank - s_hop) % world
+                reqs = None
+                if s_hop < world - 1:  # overlap the hop exchange w/ compute
+                    reqs, bufs = _ring_swap_begin([kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
                     # every processed chunk has >=1 unmasked key per row (the
@@ -208,8 +223,10 @@ class _RingAttention(torch.autograd.Function):
                     acc = acc * alpha + torch.matmul(p, vc.float())
                     l = l * alpha + p.sum(dim=-1, keepdim=True)
                     m = m_new
-                if s_hop < world - 1:
-                    kc, vc = _ring_swap([kc, vc], rank, world, group)
+                if reqs is not None:
+                    for r_ in reqs:
+                        r_.wait()
+                    kc, vc = bufs
             out = (acc / l.clamp(min=1e-30)).to(q.dtype)
             lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # natural log
         ctx.save_for_backward(q, k, v, out, lse)
