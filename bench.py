@@ -102,9 +102,14 @@ def build_gpt2(args, pg, device, dtype):
                          n_positions=args.seq_len, dropout=0.0,
                          sequence_parallel=sp)
     else:
+        # vocab_pad_to=128: 50257 -> 50304 logits width (128-aligned
+        # rows), IDENTICAL math — pad columns masked to -inf, zero grad
+        # (models/gpt2/config.py; measured +1.16 ms/step at 50257,
+        # tools/probe_vocab_pad.py)
         cfg = GPT2Config.from_name(args.gpt2_size, dropout=0.0,
                                    n_positions=max(1024, args.seq_len),
-                                   sequence_parallel=sp)
+                                   sequence_parallel=sp,
+                                   vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", "128")))
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     stage = GPT2Stage(
         cfg,
@@ -146,7 +151,8 @@ def _build_gpt2_interleaved(args, pg, device, dtype):
         cfg = GPT2Config(n_embd=64, n_layer=4, n_head=2, vocab_size=512,
                          n_positions=args.seq_len, dropout=0.0)
     else:
-        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len))
+        cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len),
+                         vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", "128")))
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     full = GPT2ForInterleaving(cfg, tp_group=tp_group, device=device, dtype=dtype)
     chunks = int(os.environ.get("QN_VPP_CHUNKS", "2"))

@@ -87,14 +87,23 @@ def load_gpt2_distributed(
     is_first = pp_rank == 0
     is_last = pp_rank == pp_size - 1
 
+    def _pad_vocab(w: torch.Tensor) -> torch.Tensor:
+        # padded-vocab layout (config.vocab_pad_to): HF files carry the
+        # logical vocab rows; zero-fill the pad rows (they are masked out
+        # of the logits and never receive gradient — models/gpt2/stage.py)
+        pv = getattr(config, "padded_vocab_size", w.shape[0])
+        if pv == w.shape[0]:
+            return w
+        return torch.cat([w, w.new_zeros(pv - w.shape[0], w.shape[1])], dim=0)
+
     if is_first:
-        out["embedding.wte.weight"] = get("wte.weight")
+        out["embedding.wte.weight"] = _pad_vocab(get("wte.weight"))
         out["embedding.wpe.weight"] = get("wpe.weight")
     if is_last:
         out["ln_f.weight"] = get("ln_f.weight")
         out["ln_f.bias"] = get("ln_f.bias")
         if pp_size > 1:
-            out["lm_head"] = get("wte.weight").clone()
+            out["lm_head"] = _pad_vocab(get("wte.weight").clone())
 
     for local_idx, gl in enumerate(layers):
         src = f"h.{gl}"

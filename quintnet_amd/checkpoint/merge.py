@@ -109,7 +109,9 @@ def merge_pp_stages(
     return merged
 
 
-def convert_to_hf_format(merged: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+def convert_to_hf_format(
+    merged: Dict[str, torch.Tensor], vocab_size: Optional[int] = None
+) -> Dict[str, torch.Tensor]:
     hf: Dict[str, torch.Tensor] = {}
     for key, value in merged.items():
         new_key = key if key.startswith("lm_head") else "transformer." + key
@@ -119,6 +121,14 @@ def convert_to_hf_format(merged: Dict[str, torch.Tensor]) -> Dict[str, torch.Ten
             or "c_fc.weight" in key
         ):
             value = value.t().contiguous()  # Linear [out,in] -> HF Conv1D [in,out]
+        if (
+            vocab_size is not None
+            and (key.startswith("lm_head") or key.startswith("wte."))
+            and value.shape[0] > vocab_size
+        ):
+            # drop the padded-vocab layout rows (config vocab_pad_to):
+            # they are zero / never trained — HF files carry logical vocab
+            value = value[:vocab_size].contiguous()
         hf[new_key] = value
     if "lm_head.weight" not in hf and "transformer.wte.weight" in hf:
         hf["lm_head.weight"] = hf["transformer.wte.weight"]
@@ -238,8 +248,9 @@ def merge_checkpoints(input_dir: str, output_path: str, prefix: str = "final_mod
             merged.update(remap_interleaved_stage(pp_stages[pp], info))
     else:
         merged = merge_pp_stages(pp_stages)
-    hf_state = convert_to_hf_format(merged)
     config = first.get("config", {})
+    mc = config.get("model_config", {}) if isinstance(config, dict) else {}
+    hf_state = convert_to_hf_format(merged, vocab_size=mc.get("vocab_size"))
     os.makedirs(os.path.dirname(os.path.abspath(output_path)), exist_ok=True)
     torch.save({"model_state_dict": hf_state, "config": config}, output_path)
     return output_path

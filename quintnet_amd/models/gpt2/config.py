@@ -28,6 +28,16 @@ class GPT2Config:
     cp_zigzag: bool = False  # CP: zigzag load-balanced ring (implies ring)
     n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
     moe_top_k: int = 2  # Megatron-SP over the TP group
+    # Pad the EMBEDDING TABLE (and hence the logits width) up to a
+    # multiple of this, keeping vocab_size as the logical width.  50257
+    # gives every logits-sized tensor odd-element rows — unaligned
+    # vectorized access in the LM-head GEMMs and CE kernels.  128-pad
+    # (50304) measures 1.16 ms/step faster on the whole LM-head block
+    # (tools/probe_vocab_pad.py).  Logits columns >= vocab_size are
+    # masked to -inf before any loss/argmax, so the math is EXACTLY the
+    # unpadded model's (pad rows get zero grad).  0 = off (checkpoint-
+    # compatible with real HF shapes).
+    vocab_pad_to: int = 0
 
     def __post_init__(self):
         if self.n_inner is None:
@@ -36,6 +46,13 @@ class GPT2Config:
     @property
     def head_dim(self) -> int:
         return self.n_embd // self.n_head
+
+    @property
+    def padded_vocab_size(self) -> int:
+        p = self.vocab_pad_to
+        if not p:
+            return self.vocab_size
+        return (self.vocab_size + p - 1) // p * p
 
     @classmethod
     def gpt2_base(cls, **kw) -> "GPT2Config":

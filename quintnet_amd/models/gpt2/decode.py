@@ -93,7 +93,10 @@ class StaticKVDecoder:
             x = self._block_step(i, x, t0, T)
         x = st.ln_f(x[:, -1:])
         w = st.embedding.wte.weight if st.lm_head is None else st.lm_head
-        return torch.nn.functional.linear(x, w)[:, -1]
+        logits = torch.nn.functional.linear(x, w)[:, -1]
+        from .stage import mask_pad_logits
+
+        return mask_pad_logits(logits, st.config)
 
     # ------------------------------------------------------------------
     def _token_step(self) -> None:

@@ -20,10 +20,16 @@ class GPT2Embedding(nn.Module):
         super().__init__()
         kw = {"device": device, "dtype": dtype}
         self.config = config
-        self.wte = nn.Embedding(config.vocab_size, config.n_embd, **kw)
+        # padded_vocab_size == vocab_size unless vocab_pad_to is set; the
+        # pad rows are zero-init, masked out of the logits, and get zero
+        # gradient — pure layout, identical math (config.py).
+        self.wte = nn.Embedding(config.padded_vocab_size, config.n_embd, **kw)
         self.wpe = nn.Embedding(config.n_positions, config.n_embd, **kw)
         self.drop = FusedDropout(config.dropout)
         nn.init.normal_(self.wte.weight, std=config.initializer_range)
+        if config.padded_vocab_size != config.vocab_size:
+            with torch.no_grad():
+                self.wte.weight[config.vocab_size:].zero_()
         nn.init.normal_(self.wpe.weight, std=config.initializer_range)
 
     def forward(self, input_ids: torch.Tensor, pos_offset: int = 0,

@@ -27,16 +27,19 @@ __all__ = ["TiedLMHead", "GPT2ForInterleaving"]
 class TiedLMHead(nn.Module):
     """LM head projecting with the embedding matrix (weight tying)."""
 
-    def __init__(self, wte: nn.Embedding):
+    def __init__(self, wte: nn.Embedding, config: GPT2Config = None):
         super().__init__()
         self.wte = wte  # shared module -> shared Parameter
+        self._config = config
 
     @property
     def tied_weight(self) -> nn.Parameter:
         return self.wte.weight
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return fused_linear(x, self.wte.weight, None, None, prefer_library=True)
+        lim = self._config.vocab_size if self._config is not None else None
+        return fused_linear(x, self.wte.weight, None, None,
+                            prefer_library=True, logical_out=lim)
 
 
 class GPT2ForInterleaving(nn.Module):
@@ -52,7 +55,7 @@ class GPT2ForInterleaving(nn.Module):
         )
         self.head = nn.Sequential(
             FusedLayerNorm(config.n_embd, eps=config.layer_norm_epsilon, **kw),
-            TiedLMHead(self.embedding.wte),
+            TiedLMHead(self.embedding.wte, config),
         )
         self.seq_len = config.n_positions
         self.hidden_dim = config.n_embd

@@ -28,7 +28,25 @@ from .block import GPT2Block
 from .config import GPT2Config
 from .embeddings import GPT2Embedding
 
-__all__ = ["GPT2Stage"]
+__all__ = ["GPT2Stage", "mask_pad_logits"]
+
+
+def mask_pad_logits(logits: torch.Tensor, config: GPT2Config) -> torch.Tensor:
+    """-inf the pad columns of a padded-vocab logits tensor (no-op when
+    vocab_pad_to is off).  exp(-inf)=0 makes the CE/logsumexp/argmax math
+    EXACTLY the unpadded model's, and the pad columns get zero gradient
+    (so the zero-init pad rows of wte/lm_head never train).  Writes 47
+    columns of 50304 — ~3 MB at bench shape, noise next to the 1.16 ms
+    the aligned GEMM/CE rows save (tools/probe_vocab_pad.py)."""
+    real = config.vocab_size
+    if logits.shape[-1] == real:
+        return logits
+    assert not logits.requires_grad, (
+        "grad-mode logits must be masked via linear(..., logical_out=) "
+        "(in-place on a reshape view costs a full-logits CopySlices clone)"
+    )
+    logits[..., real:] = float("-inf")
+    return logits
 
 
 class GPT2Stage(nn.Module):
@@ -75,9 +93,12 @@ class GPT2Stage(nn.Module):
                 self.lm_head = None
             else:
                 self.lm_head = nn.Parameter(
-                    torch.empty(config.vocab_size, config.n_embd, **kw)
+                    torch.empty(config.padded_vocab_size, config.n_embd, **kw)
                 )
                 nn.init.normal_(self.lm_head, std=config.initializer_range)
+                if config.padded_vocab_size != config.vocab_size:
+                    with torch.no_grad():
+                        self.lm_head[config.vocab_size:].zero_()
                 # grad-norm bookkeeping: after sync_tied_weights_grad this
                 # grad equals the first stage's wte grad — count it once
                 # (on the first stage) in the PP-reduced global norm
@@ -125,7 +146,9 @@ class GPT2Stage(nn.Module):
                 # the complete grad — reduce-scatter would double-count.
                 x = All_Gather.apply(x, self.tp_group, 1, "slice")
             w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
-            x = fused_linear(x, w, None, None, prefer_library=True)  # plain GEMM: hipBLASLt
+            # plain GEMM: hipBLASLt; logical_out masks padded-vocab columns
+            x = fused_linear(x, w, None, None, prefer_library=True,
+                             logical_out=self.config.vocab_size)
         elif pending is not None:
             x = x + pending  # fold before the PP send
         return x
@@ -172,6 +195,7 @@ class GPT2Stage(nn.Module):
                 x = self.ln_f(x[:, -1:])
                 w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
                 logits = fused_linear(x, w, None, None, prefer_library=True)[:, -1]
+                logits = mask_pad_logits(logits, self.config)
                 if temperature and temperature > 0:
                     logits = logits.float() / temperature
                     if top_k and top_k > 0:

@@ -129,6 +129,28 @@ def _fp8_linear(x2d: torch.Tensor, weight: torch.Tensor, bias):
     )
 
 
+class _MaskColsInplace(torch.autograd.Function):
+    """-inf fill of out[:, limit:] IN-PLACE with mark_dirty.
+
+    Must run on the NON-VIEW 2-D output of LinearFunction, before any
+    reshape: an in-place op on a view makes autograd rebase it as
+    CopySlices, whose backward clones the full logits tensor (measured
+    +6.8 ms/step at GPT-2 bench shape — profiles/ vocab-pad diff).  On
+    the direct Function output it is a plain in-place node: no copies,
+    pass-through gradient (downstream of -inf columns every consumer
+    computes exp(-inf)=0, so their grad is exactly zero already)."""
+
+    @staticmethod
+    def forward(ctx, out, limit):
+        out[:, limit:] = float("-inf")
+        ctx.mark_dirty(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, None
+
+
 class LinearFunction(torch.autograd.Function):
     @staticmethod
     def forward(ctx, x, weight, bias, activation, prefer_library=False, fp8=False):
@@ -149,7 +171,7 @@ class LinearFunction(torch.autograd.Function):
                 x2d, weight, pre_act if pre_act is not None else torch.empty(0)
             )
             ctx.has_bias = bias is not None
-            return out.reshape(*x.shape[:-1], weight.shape[0])
+            return out
         m, n, k = x2d.shape[0], weight.shape[0], weight.shape[1]
         want_native = (
             _FWD_MODE == "custom"
@@ -175,7 +197,7 @@ class LinearFunction(torch.autograd.Function):
                 out = torch.relu(out)
         ctx.save_for_backward(x2d, weight, pre_act if pre_act is not None else torch.empty(0))
         ctx.has_bias = bias is not None
-        return out.reshape(*x.shape[:-1], weight.shape[0])
+        return out
 
     @staticmethod
     def backward(ctx, grad_out):
@@ -231,16 +253,29 @@ def linear(
     activation: Optional[str] = None,
     prefer_library: bool = False,
     fp8: bool = False,
+    logical_out: Optional[int] = None,
 ) -> torch.Tensor:
     """y = activation(x @ weight.T + bias), fused on gfx950.
 
     ``prefer_library=True`` sends a PLAIN (no bias, no activation) GEMM
     through hipBLASLt instead — per the MI355X design split, the vendor
     library serves plain GEMMs (e.g. the tied LM head) while fused ones
-    run the hand-written MFMA kernel.  ``fp8=True`` (experimental) runs
+    run the hand-written MFMA kernel.  ``logical_out`` (padded-vocab
+    models): -inf-fill output columns >= logical_out in place before
+    the caller sees the tensor.  ``fp8=True`` (experimental) runs
     the FORWARD in OCP e4m3 with cached weight casts + delayed activation
     scaling (~1.5-1.8x raw GEMM throughput; pays off once K is large
     enough that the activation cast pass is small next to the GEMM —
     GPT-2-small shapes measure net-negative, see NOTES_ROUND2.md);
     backward stays bf16."""
-    return LinearFunction.apply(x, weight, bias, activation, prefer_library, fp8)
+    out = LinearFunction.apply(x, weight, bias, activation, prefer_library, fp8)
+    if logical_out is not None and logical_out < weight.shape[0]:
+        # padded-vocab logits: mask the pad columns BEFORE the reshape
+        # (on the non-view 2-D output) — see _MaskColsInplace
+        if out.requires_grad:
+            out = _MaskColsInplace.apply(out, logical_out)
+        else:
+            out[:, logical_out:] = float("-inf")
+    # reshape OUTSIDE the custom Function: keeps the output a plain
+    # autograd view so callers may do in-place ops on it
+    return out.reshape(*x.shape[:-1], weight.shape[0])
