@@ -40,6 +40,8 @@ template <typename TP, typename TG>
 void adamw_launch(TP*, float*, const TG*, float*, float*, const long long*, long long, float, float, float, float, float, int, hipStream_t);
 template <typename T>
 void sumsq_launch(const T*, long long, float*, hipStream_t);
+std::vector<at::Tensor> gemm_bias_gelu_aux(at::Tensor x, at::Tensor w,
+                                           at::Tensor bias);  // blaslt.cpp
 void gemm_nt_launch(const unsigned short*, const unsigned short*, const unsigned short*,
                     unsigned short*, unsigned short*, int, int, int, int, hipStream_t,
                     int mode = 0);
@@ -538,6 +540,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dv"), py::arg("scale"), py::arg("causal"),
         py::arg("q_offset") = 0);
   m.def("adamw_step", &adamw_step);
+  m.def("gemm_bias_gelu_aux", &gemm_bias_gelu_aux,
+        "hipBLASLt GEMM + fused bias/GELU epilogue (aux = pre-activation)");
   m.def("multi_tensor_sumsq", &multi_tensor_sumsq);
   m.attr("gfx") = "gfx950";
 }

@@ -50,6 +50,11 @@ import os
 # =library forces hipBLASLt everywhere.
 _FWD_MODE = os.environ.get("QN_GEMM_FWD", "auto")
 _WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
+# GELU-epilogue GEMM (csrc/blaslt.cpp): one hipBLASLt kernel computes
+# gemm+bias+gelu and stores the pre-activation aux for the fused
+# act_bwd — replaces the separate eager GELU pass (~90 us/layer).
+_GELU_EPI = os.environ.get("QN_GELU_EPI", "1") != "0"
+_gelu_epi_broken = False
 
 
 def _custom_wins_shape(m: int, n: int, k: int) -> bool:
@@ -172,6 +177,27 @@ class LinearFunction(torch.autograd.Function):
             )
             ctx.has_bias = bias is not None
             return out
+        global _gelu_epi_broken
+        if (
+            _GELU_EPI
+            and not _gelu_epi_broken
+            and act == _ACT_GELU
+            and bias is not None
+            and _backend.use_native(x)
+            and _backend.has_ext()
+            and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and hasattr(_backend.ext(), "gemm_bias_gelu_aux")
+        ):
+            try:
+                out, pre_act = _backend.ext().gemm_bias_gelu_aux(
+                    x2d.contiguous(), weight.contiguous(), bias.contiguous()
+                )
+                ctx.save_for_backward(x2d, weight, pre_act)
+                ctx.has_bias = True
+                return out
+            except RuntimeError:
+                _gelu_epi_broken = True  # no algo for this config: eager path
         m, n, k = x2d.shape[0], weight.shape[0], weight.shape[1]
         want_native = (
             _FWD_MODE == "custom"

@@ -28,6 +28,7 @@ sources = [
     os.path.join(CSRC, "embedding.hip"),
     os.path.join(CSRC, "gemm.hip"),
     os.path.join(CSRC, "wgrad.hip"),
+    os.path.join(CSRC, "blaslt.cpp"),
 ]
 
 setup(
@@ -38,6 +39,7 @@ setup(
         CUDAExtension(
             name="quintnet_amd._C",
             sources=sources,
+            libraries=["hipblaslt"],
             extra_compile_args={
                 "cxx": ["-O3", "-std=c++17"],
                 "nvcc": ["-O3", "-std=c++17", "--offload-arch=gfx950",

@@ -475,3 +475,42 @@ def test_zigzag_flash_world1_matches_direct():
     for a, b in ((q, qr), (k, kr), (v, vr)):
         assert torch.allclose(a.grad.float(), b.grad.float(), atol=5e-2,
                               rtol=5e-2), (a.grad - b.grad).abs().max()
+
+
+def test_gelu_epilogue_gemm_numerics():
+    """hipBLASLt GELU_AUX_BIAS epilogue (csrc/blaslt.cpp): out equals
+    gelu_tanh(x@w.T+b), aux equals the pre-activation, and the full
+    autograd path (fused act_bwd on the aux) matches an fp32 oracle."""
+    from quintnet_amd.ops import ext, has_ext, linear
+
+    if not has_ext() or not hasattr(ext(), "gemm_bias_gelu_aux"):
+        pytest.skip("extension without gemm_bias_gelu_aux")
+    torch.manual_seed(3)
+    M, K, N = 512, 768, 3072
+    x = torch.randn(M, K, device=DEV, dtype=torch.bfloat16)
+    w = torch.randn(N, K, device=DEV, dtype=torch.bfloat16) * 0.02
+    b = torch.randn(N, device=DEV, dtype=torch.bfloat16)
+    out, aux = ext().gemm_bias_gelu_aux(x, w, b)
+    pre_ref = torch.nn.functional.linear(x.float(), w.float(), b.float())
+    out_ref = torch.nn.functional.gelu(pre_ref, approximate="tanh")
+    scale = max(float(pre_ref.abs().max()), 1.0)
+    assert (aux.float() - pre_ref).abs().max() / scale < 3e-2
+    assert (out.float() - out_ref).abs().max() / scale < 3e-2
+
+    # end-to-end through LinearFunction (epilogue branch) vs fp32
+    xg = x.clone().requires_grad_(True)
+    wg = w.clone().requires_grad_(True)
+    bg = b.clone().requires_grad_(True)
+    y = linear(xg, wg, bg, activation="gelu")
+    g = torch.randn_like(y)
+    y.backward(g)
+    xr = x.float().requires_grad_(True)
+    wr = w.float().requires_grad_(True)
+    br = b.float().requires_grad_(True)
+    yr = torch.nn.functional.gelu(
+        torch.nn.functional.linear(xr, wr, br), approximate="tanh"
+    )
+    yr.backward(g.float())
+    for got, ref in ((xg.grad, xr.grad), (wg.grad, wr.grad), (bg.grad, br.grad)):
+        s = max(float(ref.abs().max()), 1.0)
+        assert (got.float() - ref).abs().max() / s < 5e-2
