@@ -480,9 +480,15 @@ def test_zigzag_flash_world1_matches_direct():
 def test_gelu_epilogue_gemm_numerics():
     """hipBLASLt GELU_AUX_BIAS epilogue (csrc/blaslt.cpp): out equals
     gelu_tanh(x@w.T+b), aux equals the pre-activation, and the full
-    autograd path (fused act_bwd on the aux) matches an fp32 oracle."""
+    autograd path (fused act_bwd on the aux) matches an fp32 oracle.
+    EXPERIMENTAL (opt-in like the feature itself): run with
+    QN_GELU_EPI=1."""
+    import os
+
     from quintnet_amd.ops import ext, has_ext, linear
 
+    if os.environ.get("QN_GELU_EPI") != "1":
+        pytest.skip("experimental epilogue path (enable with QN_GELU_EPI=1)")
     if not has_ext() or not hasattr(ext(), "gemm_bias_gelu_aux"):
         pytest.skip("extension without gemm_bias_gelu_aux")
     torch.manual_seed(3)
