@@ -53,10 +53,10 @@ _WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
 # GELU-epilogue GEMM (csrc/blaslt.cpp): one hipBLASLt kernel computes
 # gemm+bias+gelu and stores the pre-activation aux for the fused
 # act_bwd — replaces the separate eager GELU pass (~90 us/layer).
-# EXPERIMENTAL, default off: the r2 A/B measured no step-time win
-# (48.9 vs 48.7 ms — the heuristic algo without TunableOp tuning gives
-# back what the fused epilogue saves) and the numerics test is gated
-# behind QN_GELU_EPI=1 pending validation on hardware.
+# EXPERIMENTAL, default off: this ROCm 7.2 hipBLASLt build returns NO
+# algo for the GELU_AUX_BIAS bf16 epilogue on gfx950 (heuristic count
+# 0 — profiles/gelu_epi_diag note), so the branch raises and falls back
+# to the eager GELU; revisit when the library gains the epilogue.
 _GELU_EPI = os.environ.get("QN_GELU_EPI", "0") == "1"
 _gelu_epi_broken = False
 
