@@ -32,7 +32,9 @@ def main():
                 (M, I // tp, E),             # c_fc fwd
                 (M, E, I // tp),             # mlp c_proj fwd
             ]
-        shapes += [(M, 50257, 768)]          # lm_head (tp1 benches)
+        # lm_head shapes: logical 50257 (vocab_pad_to=0) and padded
+        # 50304 (bench default) — dp meshes run M=32768, pp meshes 8192
+        shapes += [(M, 50257, 768), (M, 50304, 768)]
     done = set()
     for (M, N, K) in shapes:
         if (M, N, K) in done:
