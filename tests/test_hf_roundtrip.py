@@ -205,3 +205,41 @@ def test_merge_cli_emits_hf_keyset():
                     have = have.t()
                 assert have.shape == want.shape, k
                 assert torch.allclose(have.float(), want.float(), atol=1e-5), k
+
+
+def test_hf_load_into_padded_config_and_export_slices_back():
+    """HF file (logical vocab) -> padded-vocab stage: logits over the
+    real vocab match HF exactly, pad columns are -inf; and the merge
+    CLI's HF export slices wte/lm_head back to the logical vocab."""
+    from quintnet_amd.checkpoint.distributed_loading import load_gpt2_distributed
+    from quintnet_amd.checkpoint.merge import convert_to_hf_format
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    with tempfile.TemporaryDirectory() as d:
+        hf = _make_hf_checkpoint(d, vocab=100)
+        cfg = GPT2Config(
+            vocab_size=100, n_positions=64, n_embd=64, n_layer=3, n_head=2,
+            dropout=0.0, vocab_pad_to=64,
+        )
+        assert cfg.padded_vocab_size == 128
+        sd = load_gpt2_distributed(d, cfg, pp_rank=0, pp_size=1, tp_rank=0, tp_size=1)
+        assert sd["embedding.wte.weight"].shape[0] == 128
+        assert bool((sd["embedding.wte.weight"][100:] == 0).all())
+        stage = GPT2Stage.from_sharded_state_dict(cfg, sd, pp_rank=0, pp_size=1)
+        stage.eval()
+        ids = torch.randint(0, 100, (2, 16))
+        with torch.no_grad():
+            ours = stage(ids)
+            theirs = hf(ids).logits
+        assert ours.shape[-1] == 128
+        assert torch.allclose(ours[..., :100], theirs, atol=2e-4)
+        assert bool(torch.isneginf(ours[..., 100:]).all())
+
+        # HF export: padded rows must be sliced off
+        merged = {
+            "wte.weight": stage.embedding.wte.weight.detach(),
+            "wpe.weight": stage.embedding.wpe.weight.detach(),
+        }
+        out = convert_to_hf_format(merged, vocab_size=100)
+        assert out["transformer.wte.weight"].shape[0] == 100
+        assert out["lm_head.weight"].shape[0] == 100
