@@ -164,11 +164,13 @@ def _shard_full_stage_sd(sd, cfg, pp_rank, pp_size, tp_rank, tp_size, stage):
     return out
 
 
-def _gpt2_3d(rank, world):
+def _gpt2_3d(rank, world, vocab=128, vocab_pad_to=0):
     """The bench/finetune assembly at mesh [2,2,2] (dp x tp x pp) with
     ZeRO-1: tiny GPT-2, one epoch, loss must match a single-process run
     (TP/PP are exact decompositions; both DP replicas see the same
-    stream here)."""
+    stream here).  vocab_pad_to>0 runs the padded-vocab layout the GPU
+    bench defaults to (bench.py QN_VOCAB_PAD) through the full 3D
+    assembly."""
     import copy
 
     import torch.distributed as dist
@@ -185,8 +187,8 @@ def _gpt2_3d(rank, world):
 
     pg = init_process_groups("cpu", [2, 2, 2], ["dp", "tp", "pp"])
     torch.manual_seed(77)
-    cfg = GPT2Config(vocab_size=128, n_positions=16, n_embd=32, n_layer=2,
-                     n_head=2, dropout=0.0)
+    cfg = GPT2Config(vocab_size=vocab, n_positions=16, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0, vocab_pad_to=vocab_pad_to)
     stage = GPT2Stage(
         cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
         tp_group=pg.get_group("tp"), tied_group=pg.get_tied_embedding_group(),
@@ -214,7 +216,7 @@ def _gpt2_3d(rank, world):
             "grad_acc_steps": 2, "max_grad_norm": None, "zero1": True,
             "task_type": "clm", "max_seq_length": 16,
             "model_config": {"n_embd": cfg.n_embd, "n_positions": 16}}
-    ds = SyntheticCLM(n=8, seq_len=16, vocab_size=128, seed=9)  # same per dp!
+    ds = SyntheticCLM(n=8, seq_len=16, vocab_size=vocab, seed=9)  # same per dp!
     dl = torch.utils.data.DataLoader(ds, batch_size=2, shuffle=False)
     trainer = GPT2Trainer(pmodel, dl, None, tcfg, pg)
     metrics = trainer.fit()
@@ -245,6 +247,13 @@ def _gpt2_3d(rank, world):
 @pytest.mark.slow
 def test_gpt2_3d_zero1_matches_single_process():
     run_distributed(_gpt2_3d, 8, timeout=300)
+
+
+@pytest.mark.slow
+def test_gpt2_3d_padded_vocab():
+    """mesh [2,2,2] with the padded-vocab layout (the GPU bench's
+    default config) — exact vs the single-process oracle."""
+    run_distributed(_gpt2_3d, 8, 100, 64, timeout=300)
 
 
 def _gpt2_3d_sp(rank, world):
