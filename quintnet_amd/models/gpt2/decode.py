@@ -106,7 +106,10 @@ class StaticKVDecoder:
         self.tok.copy_(logits.argmax(dim=-1, keepdim=True))
         self.past.add_(1)
 
+    @torch.no_grad()
     def capture(self) -> "StaticKVDecoder":
+        # no_grad: decode never needs autograd, and the padded-vocab
+        # logits mask requires grad-free in-place (models/gpt2/stage.py)
         # warmup + capture advance past and clobber tok; restore both.
         # (the cache rows the warmup writes sit beyond `past` and are
         # never read until a real step rewrites them)

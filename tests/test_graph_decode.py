@@ -32,3 +32,30 @@ def test_static_decoder_two_calls_reset():
     a = dec.generate(ids, max_new_tokens=6)
     b = dec.generate(ids, max_new_tokens=6)
     assert torch.equal(a, b)
+
+
+def test_static_decoder_padded_vocab_matches_unpadded():
+    """Padded-vocab config decodes the same tokens as the unpadded one
+    (pad logits masked to -inf before every argmax)."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage, StaticKVDecoder
+
+    base = dict(n_embd=64, n_layer=2, n_head=2, vocab_size=100,
+                n_positions=64, dropout=0.0)
+    torch.manual_seed(5)
+    s0 = GPT2Stage(GPT2Config(**base)).eval()
+    torch.manual_seed(5)
+    s1 = GPT2Stage(GPT2Config(**base, vocab_pad_to=64)).eval()
+    with torch.no_grad():
+        sd0 = s0.state_dict()
+        sd1 = s1.state_dict()
+        for k, v in sd0.items():
+            if sd1[k].shape != v.shape:
+                sd1[k].zero_()
+                sd1[k][: v.shape[0]] = v
+            else:
+                sd1[k] = v
+        s1.load_state_dict(sd1)
+    ids = torch.randint(0, 100, (2, 8))
+    want = StaticKVDecoder(s0, batch=2, max_len=64).generate(ids, max_new_tokens=8)
+    have = StaticKVDecoder(s1, batch=2, max_len=64).generate(ids, max_new_tokens=8)
+    assert torch.equal(have, want)
