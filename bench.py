@@ -104,12 +104,15 @@ def build_gpt2(args, pg, device, dtype):
     else:
         # vocab_pad_to=128: 50257 -> 50304 logits width (128-aligned
         # rows), IDENTICAL math — pad columns masked to -inf, zero grad
-        # (models/gpt2/config.py; measured +1.16 ms/step at 50257,
-        # tools/probe_vocab_pad.py)
+        # (models/gpt2/config.py; ~1 ms/step, profiles/vocab_pad_ab.md).
+        # pp meshes default to 0: their M=8192 vocab GEMMs are tuned in
+        # the CSV at 50257 but not at 50304 (tuning budget), and an
+        # untuned library algo can cost more than alignment saves.
+        pad_default = "128" if pg.pp_size == 1 else "0"
         cfg = GPT2Config.from_name(args.gpt2_size, dropout=0.0,
                                    n_positions=max(1024, args.seq_len),
                                    sequence_parallel=sp,
-                                   vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", "128")))
+                                   vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", pad_default)))
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     stage = GPT2Stage(
         cfg,
@@ -151,8 +154,9 @@ def _build_gpt2_interleaved(args, pg, device, dtype):
         cfg = GPT2Config(n_embd=64, n_layer=4, n_head=2, vocab_size=512,
                          n_positions=args.seq_len, dropout=0.0)
     else:
+        # interleaved runs only at pp>1 — same tuned-shape rationale
         cfg = GPT2Config(dropout=0.0, n_positions=max(1024, args.seq_len),
-                         vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", "128")))
+                         vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", "0")))
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     full = GPT2ForInterleaving(cfg, tp_group=tp_group, device=device, dtype=dtype)
     chunks = int(os.environ.get("QN_VPP_CHUNKS", "2"))
