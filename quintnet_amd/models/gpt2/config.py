@@ -28,6 +28,12 @@ class GPT2Config:
     cp_zigzag: bool = False  # CP: zigzag load-balanced ring (implies ring)
     n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
     moe_top_k: int = 2  # Megatron-SP over the TP group
+    # Recompute each block's activations in backward instead of storing
+    # them (torch.utils.checkpoint, non-reentrant): activation memory
+    # drops from O(n_layer) to O(1) blocks at ~1.33x forward FLOPs.
+    # Incompatible with MoE blocks (the aux load-balancing loss is read
+    # outside the checkpointed region).
+    activation_checkpointing: bool = False
     # Pad the EMBEDDING TABLE (and hence the logits width) up to a
     # multiple of this, keeping vocab_size as the logical width.  50257
     # gives every logits-sized tensor odd-element rows — unaligned

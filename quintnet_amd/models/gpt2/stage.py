@@ -71,6 +71,11 @@ class GPT2Stage(nn.Module):
         self.cp_group = cp_group
         if cp_group is not None:
             assert not config.sequence_parallel, "CP and Megatron-SP are exclusive"
+        if getattr(config, "activation_checkpointing", False):
+            assert not config.n_experts, (
+                "activation_checkpointing is incompatible with MoE blocks "
+                "(aux_loss is read outside the recomputed region)"
+            )
         self.sequence_parallel = config.sequence_parallel
         self.is_first_stage = pp_rank == 0
         self.is_last_stage = pp_rank == pp_size - 1
@@ -131,9 +136,21 @@ class GPT2Stage(nn.Module):
                 x = scatter_to_sequence(x, self.tp_group, 1)
         # residual-fused block chain: each block's trailing residual add
         # rides the NEXT LayerNorm kernel (ops/layernorm.py)
+        ckpt = (
+            getattr(self.config, "activation_checkpointing", False)
+            and self.training
+            and torch.is_grad_enabled()
+        )
         pending = None
         for blk in self.blocks:
-            pending, x = blk.forward_fused(x, pending)
+            if ckpt:
+                from torch.utils.checkpoint import checkpoint
+
+                pending, x = checkpoint(
+                    blk.forward_fused, x, pending, use_reentrant=False
+                )
+            else:
+                pending, x = blk.forward_fused(x, pending)
         if self.is_last_stage:
             if pending is None:
                 x = self.ln_f(x)
