@@ -194,6 +194,7 @@ class GPT2Trainer(Trainer):
                     tp_group=self._tp_group,
                     pp_group=self._pp_group,
                 )
+        self._lr_step()
         self.optimizer.step()
         if isinstance(self.model, DataParallel):
             self.model.zero_grad()

@@ -221,6 +221,8 @@ class PipelineSchedule(abc.ABC):
                     tp_group=getattr(t, "tp_group", None),
                     pp_group=self.pp_group,
                 )
+        if hasattr(t, "_lr_step"):
+            t._lr_step()
         t.optimizer.step()
         if hasattr(self.model, "zero_grad"):
             self.model.zero_grad()

@@ -46,7 +46,12 @@ class PipelineTrainer:
         self.pp_group_ranks = pp_group_ranks
         self.max_grad_norm = max_grad_norm
         self.task_type = task_type
+        self.lr_scheduler = None  # set by the owning Trainer at fit()
         self.schedule = get_schedule(schedule, self, task_type)
+
+    def _lr_step(self) -> None:
+        if self.lr_scheduler is not None:
+            self.lr_scheduler.step()
 
     @property
     def is_last_stage(self) -> bool:

@@ -123,7 +123,42 @@ class Trainer:
         return next(self.model.parameters()).dtype
 
     # ------------------------------------------------------------------
+    def _build_lr_schedule(self):
+        """Optional step-based LR schedule (config: lr_schedule in
+        constant|linear|cosine, warmup_steps, min_lr).  Built lazily at
+        fit() time — subclasses replace self.optimizer in __init__."""
+        kind = self.config.get("lr_schedule") or "constant"
+        if kind == "constant" and not int(self.config.get("warmup_steps", 0)):
+            return None
+        from .optim import LRSchedule
+
+        try:
+            steps_per_epoch = max(
+                1, len(self.train_loader) // max(self.grad_acc_steps, 1)
+            )
+        except TypeError:  # sized loaders only; fall back to config
+            steps_per_epoch = int(self.config.get("steps_per_epoch", 1000))
+        total = int(self.config.get(
+            "total_steps", self.num_epochs * steps_per_epoch
+        ))
+        return LRSchedule(
+            self.optimizer,
+            base_lr=self.lr,
+            total_steps=total,
+            warmup_steps=int(self.config.get("warmup_steps", 0)),
+            kind=kind,
+            min_lr=float(self.config.get("min_lr", 0.0)),
+        )
+
+    def _lr_step(self) -> None:
+        sched = getattr(self, "lr_scheduler", None)
+        if sched is not None:
+            sched.step()
+
     def fit(self) -> Dict[str, float]:
+        self.lr_scheduler = self._build_lr_schedule()
+        if self.pipeline_trainer is not None:
+            self.pipeline_trainer.lr_scheduler = self.lr_scheduler
         wd = None
         if self.config.get("watchdog_timeout_s"):
             from .utils.watchdog import Watchdog
@@ -263,6 +298,7 @@ class Trainer:
                     self.max_grad_norm,
                     tp_group=tp_group,
                 )
+            self._lr_step()
             self.optimizer.step()
             if hasattr(self.model, "zero_grad") and isinstance(self.model, DataParallel):
                 self.model.zero_grad()
