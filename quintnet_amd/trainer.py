@@ -141,7 +141,7 @@ class Trainer:
         total = int(self.config.get(
             "total_steps", self.num_epochs * steps_per_epoch
         ))
-        return LRSchedule(
+        sched = LRSchedule(
             self.optimizer,
             base_lr=self.lr,
             total_steps=total,
@@ -149,6 +149,10 @@ class Trainer:
             kind=kind,
             min_lr=float(self.config.get("min_lr", 0.0)),
         )
+        # mid-training resume: the ZeRO optimizers checkpoint their own
+        # step counter — restart the schedule from there
+        sched._step = int(getattr(self.optimizer, "step_count", 0))
+        return sched
 
     def _lr_step(self) -> None:
         sched = getattr(self, "lr_scheduler", None)
