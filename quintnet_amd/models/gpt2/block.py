@@ -35,9 +35,21 @@ class GPT2Block(nn.Module):
             )
         else:
             self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)
+        # recompute-in-backward for the PLAIN forward path (interleaved
+        # chunks call blocks directly); GPT2Stage handles its own fused
+        # chain (models/gpt2/stage.py)
+        self._ckpt = bool(
+            getattr(config, "activation_checkpointing", False)
+            and not config.n_experts
+        )
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        m, s2 = self.forward_fused(x, None)
+        if self._ckpt and self.training and torch.is_grad_enabled():
+            from torch.utils.checkpoint import checkpoint
+
+            m, s2 = checkpoint(self.forward_fused, x, None, use_reentrant=False)
+        else:
+            m, s2 = self.forward_fused(x, None)
         return s2 + m
 
     def forward_cached(self, x: torch.Tensor, kv_cache: dict) -> torch.Tensor:

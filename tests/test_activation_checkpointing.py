@@ -71,3 +71,26 @@ def test_checkpointing_with_dropout_matches():
     loss.backward()  # raises inside checkpoint if recompute diverges
     for _, p in m.named_parameters():
         assert torch.isfinite(p.grad).all()
+
+
+def test_checkpointing_interleaved_model_matches():
+    """GPT2ForInterleaving (plain block path) under checkpointing."""
+    from quintnet_amd.models import GPT2ForInterleaving
+
+    base = dict(n_embd=64, n_layer=2, n_head=2, vocab_size=64,
+                n_positions=32, dropout=0.0)
+    torch.manual_seed(3)
+    m0 = GPT2ForInterleaving(GPT2Config(**base))
+    torch.manual_seed(3)
+    m1 = GPT2ForInterleaving(GPT2Config(**base, activation_checkpointing=True))
+    m1.load_state_dict(m0.state_dict())
+    m0.train(), m1.train()
+    ids = torch.randint(0, 64, (2, 16))
+    labels = torch.randint(0, 64, (2, 16))
+    l0 = causal_lm_loss(m0(ids), labels)
+    l1 = causal_lm_loss(m1(ids), labels)
+    assert torch.equal(l0, l1)
+    l0.backward()
+    l1.backward()
+    for (k, p0), (_, p1) in zip(m0.named_parameters(), m1.named_parameters()):
+        assert torch.allclose(p0.grad, p1.grad, rtol=1e-5, atol=1e-7), k
