@@ -72,6 +72,8 @@ def main():
             n_layer=mc.get("n_layer", 12),
             n_head=mc.get("n_head", 12),
             dropout=mc.get("dropout", 0.1),
+            vocab_pad_to=mc.get("vocab_pad_to", 0),
+            activation_checkpointing=mc.get("activation_checkpointing", False),
         )
         dtype = torch.bfloat16 if dev_type == "cuda" else torch.float32
         stage = GPT2Stage(
