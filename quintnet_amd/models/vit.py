@@ -155,6 +155,7 @@ class Model(nn.Module):
         n_heads=4,
         depth=8,
         n_classes=10,
+        activation_checkpointing=False,
     ):
         super().__init__()
         self.embedding = ViTEmbedding(img_size, patch_size, in_channels, hidden_dim)
@@ -163,6 +164,9 @@ class Model(nn.Module):
         )
         self.classification_head = ClassificationHead(hidden_dim, n_classes)
         self.hidden_dim = hidden_dim
+        # recompute blocks in backward (same semantics as the GPT-2
+        # path, models/gpt2/stage.py)
+        self.activation_checkpointing = activation_checkpointing
 
     @property
     def seq_len(self) -> int:
@@ -170,8 +174,18 @@ class Model(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = self.embedding(x)
+        ckpt = (
+            self.activation_checkpointing
+            and self.training
+            and torch.is_grad_enabled()
+        )
         for blk in self.blocks:
-            x = blk(x)
+            if ckpt:
+                from torch.utils.checkpoint import checkpoint
+
+                x = checkpoint(blk, x, use_reentrant=False)
+            else:
+                x = blk(x)
         return self.classification_head(x)
 
 

@@ -126,3 +126,24 @@ def test_checkpointing_reduces_saved_activations():
     # counts equally, so compare the difference, not a ratio.
     assert b1 < b0, (b0, b1)
     assert b0 - b1 > 100_000, (b0, b1)
+
+
+def test_vit_checkpointing_matches():
+    from quintnet_amd.models import Model as ViT
+
+    torch.manual_seed(4)
+    m0 = ViT(depth=3)
+    torch.manual_seed(4)
+    m1 = ViT(depth=3, activation_checkpointing=True)
+    m1.load_state_dict(m0.state_dict())
+    m0.train(), m1.train()
+    x = torch.randn(2, 1, 28, 28)
+    y = torch.randint(0, 10, (2,))
+    crit = torch.nn.CrossEntropyLoss()
+    l0 = crit(m0(x), y)
+    l1 = crit(m1(x), y)
+    assert torch.equal(l0, l1)
+    l0.backward()
+    l1.backward()
+    for (k, p0), (_, p1) in zip(m0.named_parameters(), m1.named_parameters()):
+        assert torch.allclose(p0.grad, p1.grad, rtol=1e-5, atol=1e-7), k
