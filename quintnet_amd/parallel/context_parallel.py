@@ -258,6 +258,14 @@ class _RingAttention(torch.autograd.Function):
             dvp = torch.empty_like(qc)
             for s_hop in range(world):
                 j = (rank - s_hop) % world
+                # kv only gets READ by the block compute: start its hop
+                # exchange now and overlap; the grad accumulators are
+                # WRITTEN during compute, so they travel afterwards
+                # (split swap; P2P pair order kv-then-grads everywhere)
+                reqs_kv = None
+                if world > 1:
+                    reqs_kv, bufs_kv = _ring_swap_begin(
+                        [kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     _ring_flash_bwd_hop(
                         qc, kc, vc, outc, doutc, lse_flat, scale,
@@ -266,9 +274,15 @@ class _RingAttention(torch.autograd.Function):
                     dq32 += dqp.float()
                     dk_acc += dkp.float()
                     dv_acc += dvp.float()
-                kc, vc, dk_acc, dv_acc = _ring_swap(
-                    [kc, vc, dk_acc, dv_acc], rank, world, group
-                )
+                if reqs_kv is not None:
+                    reqs_g, bufs_g = _ring_swap_begin(
+                        [dk_acc, dv_acc], rank, world, group)
+                    for r_ in reqs_kv:
+                        r_.wait()
+                    kc, vc = bufs_kv
+                    for r_ in reqs_g:
+                        r_.wait()
+                    dk_acc, dv_acc = bufs_g
             return (
                 dq32.to(q.dtype),
                 dk_acc.to(k.dtype),
@@ -292,7 +306,9 @@ class _RingAttention(torch.autograd.Function):
                 dk_acc = dk_acc + torch.matmul(ds.transpose(-2, -1), q32)
                 dv_acc = dv_acc + torch.matmul(p.transpose(-2, -1), dout32)
             # rotate kv + their grads one more hop; after `world` hops the
-            # accumulators are back at the chunk owner
+            # accumulators are back at the chunk owner (kv pre-rotated
+            # before compute in the flash branch above; here the hop is
+            # synchronous — the fp32 branch is the small-shape fallback)
             kc, vc, dk_acc, dv_acc = _ring_swap(
                 [kc, vc, dk_acc, dv_acc], rank, world, group
             )
@@ -443,6 +459,9 @@ class _ZigzagRingAttention(torch.autograd.Function):
             for s_hop in range(world):
                 j = (rank - s_hop) % world
                 ck = _zz_ids(j, world)
+                if world > 1:  # pre-rotate kv under the block compute
+                    reqs_kv, bufs_kv = _ring_swap_begin(
+                        [kc, vc], rank, world, group)
                 for qi in range(2):
                     for ki in range(2):
                         if causal and ck[ki] > cq[qi]:
@@ -462,8 +481,15 @@ class _ZigzagRingAttention(torch.autograd.Function):
                         dq32[:, :, qsl] += dqp.float()
                         dk_acc[:, :, ksl] += dkp.float()
                         dv_acc[:, :, ksl] += dvp.float()
-                kc, vc, dk_acc, dv_acc = _ring_swap(
-                    [kc, vc, dk_acc, dv_acc], rank, world, group)
+                if world > 1:
+                    reqs_g, bufs_g = _ring_swap_begin(
+                        [dk_acc, dv_acc], rank, world, group)
+                    for r_ in reqs_kv:
+                        r_.wait()
+                    kc, vc = bufs_kv
+                    for r_ in reqs_g:
+                        r_.wait()
+                    dk_acc, dv_acc = bufs_g
             return (dq32.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype),
                     None, None)
         q32, dout32, out32 = q.float(), dout.float(), out.float()
