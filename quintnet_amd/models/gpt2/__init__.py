@@ -6,9 +6,11 @@ from .block import GPT2Block
 from .stage import GPT2Stage
 from .interleaved import GPT2ForInterleaving, TiedLMHead
 from .decode import StaticKVDecoder
+from .speculative import speculative_generate
 
 __all__ = [
     "StaticKVDecoder",
+    "speculative_generate",
     "GPT2Config",
     "GPT2Embedding",
     "GPT2Attention",
