@@ -178,13 +178,14 @@ class GPT2Stage(nn.Module):
         max_new_tokens: int = 32,
         temperature: float = 0.0,
         top_k: int = 0,
+        top_p: float = 0.0,
         eos_token_id: int = None,
         cache_dtype: str = None,
     ) -> torch.Tensor:
         """KV-cached autoregressive generation (serving path; pp==1).
 
         temperature 0 = greedy; otherwise softmax sampling with optional
-        top-k truncation.  Works under TP (all TP ranks call this and run
+        top-k and/or top-p (nucleus) truncation.  Works under TP (all TP ranks call this and run
         the same collectives; sampling is made rank-consistent by seeding
         from the tokens).  The cache attends via the q_offset attention
         path (models/gpt2/attention.py), positions via wpe slicing.
@@ -218,6 +219,17 @@ class GPT2Stage(nn.Module):
                     if top_k and top_k > 0:
                         kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
                         logits = logits.masked_fill(logits < kth, float("-inf"))
+                    if top_p and 0.0 < top_p < 1.0:
+                        # nucleus: keep the smallest prefix of the sorted
+                        # distribution whose mass reaches top_p (always
+                        # keeps the argmax)
+                        sl, si = torch.sort(logits, dim=-1, descending=True)
+                        cum = torch.softmax(sl, dim=-1).cumsum(dim=-1)
+                        drop_sorted = cum - torch.softmax(sl, -1) >= top_p
+                        drop = torch.zeros_like(drop_sorted).scatter(
+                            -1, si, drop_sorted
+                        )
+                        logits = logits.masked_fill(drop, float("-inf"))
                     # rank-consistent sampling under TP: derive the RNG from
                     # the current sequence so every TP rank draws the same
                     gen = torch.Generator(device="cpu")

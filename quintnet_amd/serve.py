@@ -35,6 +35,7 @@ def build_app(stage):
             max_new_tokens=int(req.get("max_new_tokens", 32)),
             temperature=float(req.get("temperature", 0.0)),
             top_k=int(req.get("top_k", 0)),
+            top_p=float(req.get("top_p", 0.0)),
             eos_token_id=req.get("eos_token_id"),
             cache_dtype=req.get("cache_dtype"),
         )
