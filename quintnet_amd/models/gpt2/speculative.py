@@ -29,6 +29,41 @@ from .stage import GPT2Stage, mask_pad_logits
 __all__ = ["speculative_generate"]
 
 
+def _dist(logits_row: torch.Tensor, temperature: float, top_k: int,
+          top_p: float) -> torch.Tensor:
+    """logits [V] -> probability vector under the sampling transform."""
+    x = logits_row.float() / max(temperature, 1e-6)
+    if top_k and top_k > 0:
+        kth = torch.topk(x, top_k).values[-1]
+        x = x.masked_fill(x < kth, float("-inf"))
+    if top_p and 0.0 < top_p < 1.0:
+        sl, si = torch.sort(x, descending=True)
+        probs = torch.softmax(sl, dim=-1)
+        drop_sorted = probs.cumsum(-1) - probs >= top_p
+        drop = torch.zeros_like(drop_sorted).scatter(-1, si, drop_sorted)
+        x = x.masked_fill(drop, float("-inf"))
+    return torch.softmax(x, dim=-1)
+
+
+def _spec_accept(p: torch.Tensor, q: torch.Tensor, tok: int,
+                 gen: torch.Generator):
+    """Speculative-sampling acceptance (Leviathan et al.): given target
+    dist p, draft dist q and a token sampled from q, accept with prob
+    min(1, p/q); on rejection return a token from norm(max(0, p - q)).
+    The emitted token is then distributed EXACTLY as p
+    (tests/test_speculative.py::test_accept_resample_lemma)."""
+    pq = float(p[tok]) / max(float(q[tok]), 1e-20)
+    u = float(torch.rand((), generator=gen))
+    if u < pq:
+        return True, tok
+    resid = (p - q).clamp_min(0)
+    tot = float(resid.sum())
+    if tot <= 0:  # p <= q everywhere can only happen with p == q
+        return True, tok
+    new = int(torch.multinomial(resid / tot, 1, generator=gen))
+    return False, new
+
+
 def _new_caches(stage: GPT2Stage) -> List[dict]:
     return [{"k": None, "v": None, "k8": None, "quant": None}
             for _ in stage.blocks]
@@ -66,12 +101,19 @@ def speculative_generate(
     max_new_tokens: int = 32,
     draft_k: int = 4,
     eos_token_id: int = None,
+    temperature: float = 0.0,
+    top_k: int = 0,
+    top_p: float = 0.0,
+    seed: int = None,
 ) -> torch.Tensor:
-    """Greedy speculative decode; returns [1, T0 + <=max_new_tokens].
+    """Speculative decode; returns [1, T0 + <=max_new_tokens].
 
     ``target`` and ``draft`` must share the tokenizer (vocab); the draft
-    is typically a much smaller model.  Output is token-identical to
-    ``target.generate(..., temperature=0)``.
+    is typically a much smaller model.  temperature==0: greedy
+    acceptance, token-identical to ``target.generate(temperature=0)``.
+    temperature>0: stochastic speculative sampling — each emitted token
+    is distributed exactly as a sample from the TARGET's (temperature/
+    top-k/top-p transformed) distribution (_spec_accept lemma).
     """
     assert input_ids.shape[0] == 1, "speculative decoding is per-sequence"
     assert target.is_first_stage and target.is_last_stage, "pp==1 only"
@@ -81,8 +123,15 @@ def speculative_generate(
         out = input_ids
         tc, dc = _new_caches(target), _new_caches(draft)
         limit = min(target.config.n_positions, draft.config.n_positions)
-        # prefill: target's greedy next token seeds the loop
-        nxt = _cached_logits(target, tc, out)[:, -1].argmax(dim=-1, keepdim=True)
+        gen = torch.Generator()
+        gen.manual_seed(seed if seed is not None else 0x5eed)
+        # prefill: the target's next token seeds the loop
+        pre = _cached_logits(target, tc, out)[:, -1]
+        if temperature and temperature > 0:
+            p0 = _dist(pre[0], temperature, top_k, top_p)
+            nxt = out.new_tensor([[int(torch.multinomial(p0, 1, generator=gen))]])
+        else:
+            nxt = pre.argmax(dim=-1, keepdim=True)
         out = torch.cat([out, nxt], dim=1)
         while (
             out.shape[1] - input_ids.shape[1] < max_new_tokens
@@ -94,11 +143,18 @@ def speculative_generate(
             # draft proposes k tokens greedily (first call also catches
             # its cache up on every token it has not forwarded yet)
             proposal = []
+            q_dists = []
             d_in = out[:, _clen(dc):]
             for _ in range(k):
-                d_in = _cached_logits(draft, dc, d_in)[:, -1].argmax(
-                    dim=-1, keepdim=True
-                )
+                dl = _cached_logits(draft, dc, d_in)[:, -1]
+                if temperature and temperature > 0:
+                    q = _dist(dl[0], temperature, top_k, top_p)
+                    q_dists.append(q)
+                    d_in = out.new_tensor(
+                        [[int(torch.multinomial(q, 1, generator=gen))]]
+                    )
+                else:
+                    d_in = dl.argmax(dim=-1, keepdim=True)
                 proposal.append(d_in)
             prop = torch.cat(proposal, dim=1)  # [1, k]
             # ONE target forward over its own catch-up + the proposal;
@@ -106,11 +162,29 @@ def speculative_generate(
             # last accepted token and after each proposed token
             t_in = torch.cat([out[:, _clen(tc):], prop], dim=1)
             tl = _cached_logits(target, tc, t_in)
-            choice = tl[:, -(k + 1):].argmax(dim=-1)  # [1, k+1]
-            n_acc = 0
-            while n_acc < k and int(choice[0, n_acc]) == int(prop[0, n_acc]):
-                n_acc += 1
-            nxt = choice[:, n_acc : n_acc + 1]  # target's own next token
+            if temperature and temperature > 0:
+                # stochastic acceptance; q dists were stashed by the
+                # proposal loop above
+                n_acc, forced = 0, None
+                for i in range(k):
+                    p_i = _dist(tl[0, -(k + 1) + i], temperature, top_k, top_p)
+                    ok, tok2 = _spec_accept(p_i, q_dists[i],
+                                            int(prop[0, i]), gen)
+                    if ok:
+                        n_acc += 1
+                    else:
+                        forced = tok2
+                        break
+                if forced is None:
+                    p_last = _dist(tl[0, -1], temperature, top_k, top_p)
+                    forced = int(torch.multinomial(p_last, 1, generator=gen))
+                nxt = out.new_tensor([[forced]])
+            else:
+                choice = tl[:, -(k + 1):].argmax(dim=-1)  # [1, k+1]
+                n_acc = 0
+                while n_acc < k and int(choice[0, n_acc]) == int(prop[0, n_acc]):
+                    n_acc += 1
+                nxt = choice[:, n_acc : n_acc + 1]  # target's next token
             out = torch.cat([out, prop[:, :n_acc], nxt], dim=1)
             # drop rejected proposal entries; the caches then cover at
             # most the emitted sequence minus the trailing token

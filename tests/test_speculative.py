@@ -62,3 +62,33 @@ def test_speculative_eos_stop():
     have = speculative_generate(target, draft, ids, max_new_tokens=20,
                                 draft_k=3, eos_token_id=eos)
     assert torch.equal(have, want), (have, want)
+
+
+def test_accept_resample_lemma():
+    """The speculative acceptance rule must emit tokens distributed
+    EXACTLY as the target dist p, regardless of the draft dist q
+    (Leviathan et al. correctness lemma) — checked empirically."""
+    from quintnet_amd.models.gpt2.speculative import _spec_accept
+
+    p = torch.tensor([0.5, 0.3, 0.15, 0.05])
+    q = torch.tensor([0.1, 0.2, 0.3, 0.4])  # deliberately mismatched
+    gen = torch.Generator().manual_seed(123)
+    n = 40000
+    toks = torch.multinomial(q.expand(n, -1), 1, generator=gen).squeeze(1)
+    counts = torch.zeros(4)
+    for i in range(n):
+        _, tok = _spec_accept(p, q, int(toks[i]), gen)
+        counts[tok] += 1
+    emp = counts / n
+    assert (emp - p).abs().max() < 0.015, emp
+
+
+def test_stochastic_speculative_runs_and_is_deterministic():
+    target, draft = _models()
+    ids = torch.randint(0, 96, (1, 8))
+    a = speculative_generate(target, draft, ids, max_new_tokens=12,
+                             draft_k=3, temperature=0.8, top_p=0.9, seed=7)
+    b = speculative_generate(target, draft, ids, max_new_tokens=12,
+                             draft_k=3, temperature=0.8, top_p=0.9, seed=7)
+    assert torch.equal(a, b)
+    assert a.shape[1] == 20 and int(a.max()) < 96
