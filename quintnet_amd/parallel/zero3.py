@@ -78,17 +78,10 @@ class ZeRO3Block(nn.Module):
 
     def __init__(self, module: nn.Module, dp_group=None):
         super().__init__()
-        # checkpoint's RNG preservation covers torch RNG only; the fused
-        # counter-based dropout kernel would draw a DIFFERENT mask in the
-        # backward recompute, corrupting gradients silently.
-        for m in module.modules():
-            if type(m).__name__ == "FusedDropout" and getattr(m, "p", 0) > 0:
-                raise ValueError(
-                    "ZeRO3Block: module uses FusedDropout with p>0 — its "
-                    "counter-based mask is not recompute-stable under "
-                    "activation checkpointing; set dropout=0 or use "
-                    "torch.nn.Dropout"
-                )
+        # dropout IS recompute-stable here: FusedDropout draws its
+        # per-call seed from the torch CPU RNG (ops/dropout.py), which
+        # non-reentrant checkpoint stashes and restores — proven exact
+        # in tests/test_activation_checkpointing.py dropout test.
         self.module = module
         self.dp_group = dp_group
         self.world = _group_size(dp_group)
@@ -145,6 +138,29 @@ class ZeRO3Block(nn.Module):
                 self._run, self.shard, *args, use_reentrant=False, **kwargs
             )
         return self._run(self.shard, *args, **kwargs)
+
+    # residual-fused entry (GPT2Stage's block chain calls this instead
+    # of forward — models/gpt2/stage.py); same gather/checkpoint shape
+    def _run_fused(self, shard: torch.Tensor, x, pending):
+        full = _GatherFlat.apply(shard, self.dp_group)
+        return self._call_method(
+            self._param_views(full), "forward_fused", x, pending
+        )
+
+    def _call_method(self, views, name, *args):
+        """functional_call only invokes module.forward — reparametrize
+        manually for other entry points (forward_fused)."""
+        from torch.nn.utils.stateless import _reparametrize_module
+
+        with _reparametrize_module(self.module, views):
+            return getattr(self.module, name)(*args)
+
+    def forward_fused(self, x, pending):
+        if torch.is_grad_enabled() and self.shard.requires_grad:
+            return checkpoint(
+                self._run_fused, self.shard, x, pending, use_reentrant=False
+            )
+        return self._run_fused(self.shard, x, pending)
 
     # ------------------------------------------------------------------
     def full_state_dict_tensors(self) -> Dict[str, torch.Tensor]:
