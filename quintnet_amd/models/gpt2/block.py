@@ -32,6 +32,7 @@ class GPT2Block(nn.Module):
                 config.n_embd, config.n_inner, config.n_experts,
                 top_k=config.moe_top_k, ep_group=ep_group, tp_group=tp_group,
                 device=device, dtype=dtype,
+                capacity_factor=getattr(config, "moe_capacity_factor", 0.0),
             )
         else:
             self.mlp = GPT2MLP(config, tp_group=tp_group, device=device, dtype=dtype)

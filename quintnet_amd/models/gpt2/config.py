@@ -28,6 +28,11 @@ class GPT2Config:
     cp_zigzag: bool = False  # CP: zigzag load-balanced ring (implies ring)
     n_experts: int = 0  # >0: MoE MLP blocks (expert parallelism)
     moe_top_k: int = 2  # Megatron-SP over the TP group
+    # >0: Switch-style expert capacity — each expert serves at most
+    # ceil(cf * tokens * top_k / n_experts) assignments per rank;
+    # overflow contributes zero (residual carries the token).  0 = the
+    # default exact variable-size routing (nothing dropped).
+    moe_capacity_factor: float = 0.0
     # Recompute each block's activations in backward instead of storing
     # them (torch.utils.checkpoint, non-reentrant): activation memory
     # drops from O(n_layer) to O(1) blocks at ~1.33x forward FLOPs.

@@ -121,6 +121,7 @@ class ExpertParallelMLP(nn.Module):
         tp_group=None,
         device=None,
         dtype=None,
+        capacity_factor: float = 0.0,
     ):
         super().__init__()
         self.ep_group = ep_group
@@ -138,6 +139,13 @@ class ExpertParallelMLP(nn.Module):
             for _ in range(self.n_local)
         )
         self.aux_loss: Optional[torch.Tensor] = None
+        # capacity_factor > 0: Switch/GShard-style token dropping — each
+        # expert processes at most ceil(cf * n_tok * k / n_experts) of
+        # THIS RANK's assignments (position-priority within the stable
+        # expert sort); dropped assignments contribute ZERO (the block's
+        # residual stream carries the token through).  0 = exact
+        # variable-size routing, nothing dropped.
+        self.capacity_factor = float(capacity_factor)
 
     # ------------------------------------------------------------------
     def forward(self, x: torch.Tensor) -> torch.Tensor:
@@ -162,10 +170,22 @@ class ExpertParallelMLP(nn.Module):
         # dispatch: one row per (token, k) assignment, sorted by expert
         flat_idx = top_idx.reshape(-1)  # [n_tok*k]
         order = torch.argsort(flat_idx, stable=True)
-        src_token = torch.div(order, self.top_k, rounding_mode="floor")
-        routed = flat[src_token]  # [n_tok*k, H], grouped by expert
         sorted_exp = flat_idx[order]
         per_expert = torch.bincount(sorted_exp, minlength=self.n_experts)
+        n_assign = n_tok * self.top_k
+        dropped = None
+        if self.capacity_factor > 0:
+            cap = int(-(-self.capacity_factor * n_assign // self.n_experts))
+            if bool((per_expert > cap).any()):
+                start = per_expert.cumsum(0) - per_expert  # exclusive cumsum
+                within = torch.arange(n_assign, device=flat.device) - start[sorted_exp]
+                keep = within < cap
+                dropped = order[~keep]  # original assignment slots dropped
+                order = order[keep]
+                sorted_exp = sorted_exp[keep]
+                per_expert = per_expert.clamp(max=cap)
+        src_token = torch.div(order, self.top_k, rounding_mode="floor")
+        routed = flat[src_token]  # [kept, H], grouped by expert
 
         # exchange: expert e lives on rank e // n_local
         out_splits = [
@@ -190,7 +210,13 @@ class ExpertParallelMLP(nn.Module):
         # via inverse-permutation gathers (clean autograd)
         back = processed[torch.argsort(regroup_idx)]
         returned = all_to_all_var(back, in_splits, out_splits, self.ep_group)
-        unsorted = returned[torch.argsort(order)]
+        if dropped is None:
+            unsorted = returned[torch.argsort(order)]
+        else:
+            # dropped assignment slots contribute zero
+            unsorted = returned.new_zeros(n_assign, H).index_copy(
+                0, order, returned
+            )
         weighted = unsorted.view(n_tok, self.top_k, H) * gates.unsqueeze(-1).to(
             unsorted.dtype
         )

@@ -396,3 +396,35 @@ def test_moe_empty_expert():
     # expert 1 saw nothing: grads None or zero
     g = moe.experts[1].fc1.weight.grad
     assert g is None or torch.count_nonzero(g) == 0
+
+
+def test_capacity_factor_large_equals_exact():
+    """A capacity that never binds must reproduce the exact router."""
+    from quintnet_amd.parallel.expert_parallel import ExpertParallelMLP
+
+    torch.manual_seed(3)
+    m = ExpertParallelMLP(32, 64, 4, top_k=2)
+    m2 = ExpertParallelMLP(32, 64, 4, top_k=2, capacity_factor=100.0)
+    m2.load_state_dict(m.state_dict())
+    x = torch.randn(2, 8, 32)
+    y, y2 = m(x), m2(x)
+    assert torch.allclose(y, y2, atol=1e-6)
+
+
+def test_capacity_factor_binds_and_backprops():
+    """cf=1.0 on a skewed router must drop overflow (zero contribution)
+    while gradients still flow through the kept assignments."""
+    from quintnet_amd.parallel.expert_parallel import ExpertParallelMLP
+
+    torch.manual_seed(4)
+    m = ExpertParallelMLP(16, 32, 4, top_k=1, capacity_factor=1.0)
+    with torch.no_grad():  # tie all logits -> topk picks expert 0
+        m.router.weight.zero_()
+    x = torch.randn(1, 8, 16, requires_grad=True)
+    y = m(x)
+    # capacity = ceil(1.0 * 8 * 1 / 4) = 2 -> 6 of 8 assignments dropped
+    n_zero_rows = int((y.reshape(-1, 16).abs().sum(-1) == 0).sum())
+    assert n_zero_rows == 6, n_zero_rows
+    y.sum().backward()
+    assert torch.isfinite(x.grad).all()
+    assert float(x.grad.abs().sum()) > 0
