@@ -56,6 +56,15 @@ def save_sharded_checkpoint(
     without it, MoE/CP ranks with equal (pp, tp) would clobber each
     other's shards.  dp rank 0 of each shard coordinate writes.
     """
+    for m in model.modules():
+        if type(m).__name__ == "ZeRO3Block":
+            raise ValueError(
+                "save_sharded_checkpoint: model is ZeRO-3 wrapped — its "
+                "state dict holds flat shards, not the named-parameter "
+                "contract the merge CLI expects.  Re-gather with "
+                "ZeRO3Block.full_state_dict_tensors() per block (see "
+                "parallel/zero3.py) before saving."
+            )
     pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
     tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
     dp_rank = pg_manager.dp_rank if pg_manager is not None else 0
@@ -121,6 +130,15 @@ def load_sharded_checkpoint(
 ):
     """Reload this rank's model shard (and, for mid-training resume, its
     ZeRO optimizer shard when ``optimizer`` is given)."""
+    for m in model.modules():
+        if type(m).__name__ == "ZeRO3Block":
+            raise ValueError(
+                "save_sharded_checkpoint: model is ZeRO-3 wrapped — its "
+                "state dict holds flat shards, not the named-parameter "
+                "contract the merge CLI expects.  Re-gather with "
+                "ZeRO3Block.full_state_dict_tensors() per block (see "
+                "parallel/zero3.py) before saving."
+            )
     pp_rank = pg_manager.pp_rank if pg_manager is not None else 0
     tp_rank = pg_manager.tp_rank if pg_manager is not None else 0
     dp_rank = pg_manager.dp_rank if pg_manager is not None else 0

@@ -196,3 +196,17 @@ def _run_zero3_gpt2(rank, world):
 
 def test_zero3_gpt2_fused_chain_world2():
     run_distributed(_run_zero3_gpt2, 2)
+
+
+def test_zero3_checkpoint_save_refuses_sharded_model(tmp_path):
+    """A ZeRO-3-wrapped model must not silently save flat shards under
+    the named-parameter checkpoint contract."""
+    from quintnet_amd.checkpoint import save_sharded_checkpoint
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.parallel.zero3 import apply_zero3
+
+    m = GPT2Stage(GPT2Config(n_embd=32, n_layer=2, n_head=2, vocab_size=64,
+                             n_positions=32, dropout=0.0))
+    apply_zero3(m, dp_group=None)
+    with pytest.raises(ValueError, match="ZeRO-3"):
+        save_sharded_checkpoint(m, str(tmp_path))
