@@ -15,7 +15,7 @@ parity): ``embedding.wte.weight``, ``embedding.wpe.weight``,
 
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import torch
 import torch.distributed as dist
