@@ -17,7 +17,10 @@ from typing import Optional
 import torch
 
 
-def build_app(stage):
+def build_app(stage, draft=None):
+    """FastAPI app over ``stage``; pass ``draft`` (a smaller GPT2Stage
+    sharing the vocab) to serve speculative decoding on requests that
+    set ``"speculative": true``."""
     from fastapi import Body, FastAPI
 
     app = FastAPI(title="quintnet_amd GPT-2 server")
@@ -30,6 +33,23 @@ def build_app(stage):
     def generate(req: dict = Body(...)):
         dev = next(stage.parameters()).device
         ids = torch.tensor([req["input_ids"]], dtype=torch.long, device=dev)
+        if draft is not None and req.get("speculative"):
+            from .models import speculative_generate
+
+            out = speculative_generate(
+                stage, draft, ids,
+                max_new_tokens=int(req.get("max_new_tokens", 32)),
+                draft_k=int(req.get("draft_k", 4)),
+                temperature=float(req.get("temperature", 0.0)),
+                top_k=int(req.get("top_k", 0)),
+                top_p=float(req.get("top_p", 0.0)),
+                eos_token_id=req.get("eos_token_id"),
+                seed=req.get("seed"),
+            )
+            return {
+                "output_ids": out[0].tolist(),
+                "new_ids": out[0, ids.shape[1]:].tolist(),
+            }
         out = stage.generate(
             ids,
             max_new_tokens=int(req.get("max_new_tokens", 32)),

@@ -32,3 +32,27 @@ def test_serve_generate_endpoint():
                                         "temperature": 0.8, "top_k": 10,
                                         "cache_dtype": "int8"})
     assert r2.status_code == 200 and len(r2.json()["new_ids"]) == 4
+
+
+def test_serve_speculative_endpoint():
+    import torch
+    from fastapi.testclient import TestClient
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.serve import build_app
+
+    torch.manual_seed(0)
+    base = dict(vocab_size=96, n_positions=64, dropout=0.0)
+    target = GPT2Stage(GPT2Config(n_embd=64, n_layer=2, n_head=2, **base)).eval()
+    draft = GPT2Stage(GPT2Config(n_embd=32, n_layer=1, n_head=2, **base)).eval()
+    client = TestClient(build_app(target, draft=draft))
+    ids = [3, 5, 7, 9]
+    r = client.post("/generate", json={
+        "input_ids": ids, "max_new_tokens": 8, "speculative": True,
+    })
+    assert r.status_code == 200
+    spec = r.json()["new_ids"]
+    plain = client.post("/generate", json={
+        "input_ids": ids, "max_new_tokens": 8,
+    }).json()["new_ids"]
+    assert spec == plain  # greedy speculative is exact
