@@ -131,10 +131,20 @@ class GPT2Trainer(Trainer):
                     cp_causal_lm_loss,
                     scatter_clm_targets,
                     scatter_to_context,
+                    zigzag_clm_targets,
+                    zigzag_to_context,
                 )
 
-                ids_shard = scatter_to_context(ids, cp_group, dim=1)
-                tgt_shard = scatter_clm_targets(labels, cp_group)
+                # shard layout must match the stage's position math:
+                # zigzag configs expect chunks (r, 2cp-1-r), contiguous
+                # configs expect rank-ordered slices
+                inner = _unwrap(self.model)
+                if getattr(getattr(inner, "config", None), "cp_zigzag", False):
+                    ids_shard = zigzag_to_context(ids, cp_group, dim=1)
+                    tgt_shard = zigzag_clm_targets(labels, cp_group)
+                else:
+                    ids_shard = scatter_to_context(ids, cp_group, dim=1)
+                    tgt_shard = scatter_clm_targets(labels, cp_group)
                 logits = self.model(ids_shard)
                 loss, true_loss = cp_causal_lm_loss(logits, tgt_shard, cp_group)
             else:

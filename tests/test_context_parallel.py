@@ -643,3 +643,56 @@ def test_gpt2_cp_zigzag_cp2():
 
 def test_gpt2_cp_zigzag_cp4():
     run_distributed(_run_gpt2_cp_zigzag, 4)
+
+
+def _run_cp_trainer_zigzag(rank, world):
+    """GPT2Trainer with a cp_zigzag config: the trainer must scatter
+    ZIGZAG shards (chunks (r, 2cp-1-r)) to match the stage's position
+    math; loss matches the plain single-process epoch."""
+    import torch.distributed as dist
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd import GPT2Trainer, init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel import DataParallel, DistributedConfig
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    pg = init_process_groups("cpu", [world], ["cp"])
+    torch.manual_seed(62)
+    cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0, cp_zigzag=True)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None,
+                      cp_group=pg.get_group("cp"))
+    for p in stage.parameters():
+        dist.broadcast(p.data, src=0)
+    ref = GPT2Stage(GPT2Config(vocab_size=96, n_positions=32, n_embd=32,
+                               n_layer=2, n_head=2, dropout=0.0))
+    ref.load_state_dict(stage.state_dict())
+
+    model = DataParallel(
+        stage, DistributedConfig(rank, world, pg.get_group("cp"))
+    )
+    ds = SyntheticCLM(n=4, seq_len=32, vocab_size=96, seed=8)
+    tcfg = {"num_epochs": 1, "grad_acc_steps": 2, "zero1": False,
+            "context_parallel": True, "learning_rate": 1e-3,
+            "max_grad_norm": None, "task_type": "clm"}
+    tr = GPT2Trainer(model, DataLoader(ds, batch_size=2), None, tcfg, pg)
+    hist = tr.fit()
+
+    opt = torch.optim.AdamW(ref.parameters(), lr=1e-3, weight_decay=0.01)
+    dl = DataLoader(ds, batch_size=2)
+    tot, steps, accum = 0.0, 0, 0
+    for b in dl:
+        loss = causal_lm_loss(ref(b["input_ids"]), b["labels"], ignore_index=-100)
+        (loss / 2).backward()
+        tot += float(loss.detach()); steps += 1
+        accum += 1
+        if accum == 2:
+            accum = 0
+            opt.step(); opt.zero_grad()
+    assert abs(hist["train_loss"] - tot / steps) < 5e-4, (hist, tot / steps)
+
+
+def test_gpt2_trainer_context_parallel_zigzag():
+    run_distributed(_run_cp_trainer_zigzag, 2)
