@@ -278,14 +278,31 @@ class Trainer:
 
     def _train_epoch_plain(self) -> Dict[str, float]:
         total_loss, correct, total, steps = 0.0, 0, 0, 0
+        accum = 0
+        is_ddp = isinstance(self.model, DataParallel)
         for batch in self.train_loader:
             batch = PipelineDataLoader._normalize(batch)
             x = batch["images"].to(self.device, non_blocking=True)
             y = batch["labels"].to(self.device, non_blocking=True)
+            if is_ddp:
+                # async bucket reduce only on the LAST micro-batch of the
+                # window (same rule as gpt2_trainer.py — earlier micro-
+                # batches would race the reduce with accumulation)
+                self.model.require_backward_grad_sync = (
+                    accum + 1 == self.grad_acc_steps
+                )
             out = self.model(x)
             loss = self.criterion(out, y)
-            loss.backward()
-            if isinstance(self.model, DataParallel):
+            (loss / self.grad_acc_steps).backward()
+            accum += 1
+            total_loss += float(loss.detach())
+            correct += int((out.argmax(-1) == y).sum())
+            total += int(y.numel())
+            steps += 1
+            if accum < self.grad_acc_steps:
+                continue
+            accum = 0
+            if is_ddp:
                 self.model.finalize_gradients()
             if self.max_grad_norm:
                 from .ops import clip_grad_norm_global
@@ -308,10 +325,17 @@ class Trainer:
                 self.model.zero_grad()
             else:
                 self.optimizer.zero_grad()
-            total_loss += float(loss.detach())
-            correct += int((out.argmax(-1) == y).sum())
-            total += y.numel()
-            steps += 1
+        if accum:
+            # tail window (loader length not a multiple of grad_acc_steps)
+            if is_ddp:
+                self.model.require_backward_grad_sync = True
+                self.model.finalize_gradients()
+            self._lr_step()
+            self.optimizer.step()
+            if is_ddp:
+                self.model.zero_grad()
+            else:
+                self.optimizer.zero_grad()
         return {
             "loss": total_loss / max(steps, 1),
             "accuracy": 100.0 * correct / max(total, 1),

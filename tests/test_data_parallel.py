@@ -114,3 +114,39 @@ def test_ddp_factories():
     out.sum().backward()
     ddp.finalize_gradients()
     assert m.weight.grad is not None
+
+
+def test_plain_trainer_grad_accumulation_oracle():
+    """Base Trainer classification path honors grad_acc_steps: one
+    optimizer step per window, mean-of-window gradient (was silently
+    ignored before — stepped per micro-batch)."""
+    import copy
+
+    import torch
+
+    from quintnet_amd.models import Model as ViT
+    from quintnet_amd.trainer import Trainer
+    from quintnet_amd.utils.data import SyntheticMNIST
+
+    torch.manual_seed(5)
+    m = ViT(depth=2, hidden_dim=32, n_heads=2)
+    ref = copy.deepcopy(m)
+    ds = SyntheticMNIST(n=8, seed=3)
+    dl = torch.utils.data.DataLoader(ds, batch_size=2, shuffle=False)
+    tr = Trainer(m, dl, None,
+                 {"num_epochs": 1, "grad_acc_steps": 2, "max_grad_norm": None,
+                  "learning_rate": 1e-3},
+                 None)
+    tr.fit()
+
+    opt = torch.optim.Adam(ref.parameters(), lr=1e-3)
+    crit = torch.nn.CrossEntropyLoss()
+    it = iter(torch.utils.data.DataLoader(ds, batch_size=2, shuffle=False))
+    for _ in range(2):  # 4 batches / window 2 = 2 steps
+        opt.zero_grad()
+        for _ in range(2):
+            b = next(it)
+            (crit(ref(b["images"]), b["labels"]) / 2).backward()
+        opt.step()
+    for (k, p), (_, r) in zip(m.named_parameters(), ref.named_parameters()):
+        assert torch.allclose(p, r, atol=1e-6), k
