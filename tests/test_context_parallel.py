@@ -595,7 +595,8 @@ def _run_gpt2_cp_zigzag(rank, world):
 
     torch.manual_seed(33)
     cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
-                     n_head=2, dropout=0.0, cp_zigzag=True)
+                     n_head=2, dropout=0.0, cp_zigzag=True,
+                     vocab_pad_to=vocab_pad_to)
     full_cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
                           n_head=2, dropout=0.0)
     full = GPT2Stage(full_cfg, pp_rank=0, pp_size=1, tp_group=None)
@@ -645,10 +646,11 @@ def test_gpt2_cp_zigzag_cp4():
     run_distributed(_run_gpt2_cp_zigzag, 4)
 
 
-def _run_cp_trainer_zigzag(rank, world):
+def _run_cp_trainer_zigzag(rank, world, vocab_pad_to=0):
     """GPT2Trainer with a cp_zigzag config: the trainer must scatter
     ZIGZAG shards (chunks (r, 2cp-1-r)) to match the stage's position
-    math; loss matches the plain single-process epoch."""
+    math; loss matches the plain single-process epoch.  vocab_pad_to
+    composes the padded-vocab layout with CP."""
     import torch.distributed as dist
     from torch.utils.data import DataLoader
 
@@ -661,13 +663,15 @@ def _run_cp_trainer_zigzag(rank, world):
     pg = init_process_groups("cpu", [world], ["cp"])
     torch.manual_seed(62)
     cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
-                     n_head=2, dropout=0.0, cp_zigzag=True)
+                     n_head=2, dropout=0.0, cp_zigzag=True,
+                     vocab_pad_to=vocab_pad_to)
     stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None,
                       cp_group=pg.get_group("cp"))
     for p in stage.parameters():
         dist.broadcast(p.data, src=0)
     ref = GPT2Stage(GPT2Config(vocab_size=96, n_positions=32, n_embd=32,
-                               n_layer=2, n_head=2, dropout=0.0))
+                               n_layer=2, n_head=2, dropout=0.0,
+                               vocab_pad_to=vocab_pad_to))
     ref.load_state_dict(stage.state_dict())
 
     model = DataParallel(
@@ -696,3 +700,7 @@ def _run_cp_trainer_zigzag(rank, world):
 
 def test_gpt2_trainer_context_parallel_zigzag():
     run_distributed(_run_cp_trainer_zigzag, 2)
+
+
+def test_gpt2_trainer_zigzag_padded_vocab():
+    run_distributed(_run_cp_trainer_zigzag, 2, 64)
