@@ -112,7 +112,11 @@ def build_gpt2(args, pg, device, dtype):
         cfg = GPT2Config.from_name(args.gpt2_size, dropout=0.0,
                                    n_positions=max(1024, args.seq_len),
                                    sequence_parallel=sp,
-                                   vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", pad_default)))
+                                   vocab_pad_to=int(os.environ.get("QN_VOCAB_PAD", pad_default)),
+                                   # QN_ACT_CKPT=1: recompute blocks in backward
+                                   # (memory for ~1.33x fwd FLOPs — lets the
+                                   # larger presets raise micro-batch)
+                                   activation_checkpointing=os.environ.get("QN_ACT_CKPT") == "1")
     tp_group = pg.get_group("tp") if pg.tp_size > 1 else None
     stage = GPT2Stage(
         cfg,
