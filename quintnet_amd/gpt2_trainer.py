@@ -181,7 +181,7 @@ class GPT2Trainer(Trainer):
         return {"loss": avg, "ppl": math.exp(min(avg, 20.0)), "n_tokens": total_tokens}
 
     def _optim_step(self, pt) -> None:
-        """finalize grads → tied/SP sync → global-norm clip → step."""
+        """finalize grads → tied/SP sync → [NaN guard] → clip → step."""
         if pt:
             pt.start("grad_comm")
         if isinstance(self.model, DataParallel):
@@ -192,6 +192,10 @@ class GPT2Trainer(Trainer):
         if pt:
             pt.stop("grad_comm")
             pt.start("optimizer")
+        if self.config.get("detect_nan_grads"):
+            from .utils.watchdog import assert_finite_grads
+
+            assert_finite_grads(self.model)
         if self.max_grad_norm:
             if hasattr(self.optimizer, "clip_grad_norm_"):
                 self.optimizer.clip_grad_norm_(self.max_grad_norm)
