@@ -55,6 +55,11 @@ def main():
     ap.add_argument("--staged", action="store_true")
     args = ap.parse_args()
     cfg = load_config(args.config)
+    from quintnet_amd.core.config import validate_config
+
+    problems = validate_config(cfg, world_size=int(os.environ.get("WORLD_SIZE", "1")))
+    if problems:
+        raise SystemExit("config problems:\n  " + "\n  ".join(problems))
 
     dev_type = "cuda" if torch.cuda.is_available() else "cpu"
     pg = init_process_groups(

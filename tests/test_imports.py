@@ -99,3 +99,24 @@ def test_rank_logger(tmp_path, capsys):
 
     print_rank_0("hello")  # rank 0 in single process
     assert "hello" in capsys.readouterr().out
+
+
+def test_validate_config_catches_mistakes():
+    from quintnet_amd.core.config import load_config, validate_config
+
+    # both shipped example configs must validate clean
+    for f in ("examples/config.yaml", "examples/gpt2_config.yaml"):
+        cfg = load_config(f)
+        assert validate_config(cfg, world_size=8) == [], f
+
+    bad = {
+        "mesh_dim": [2, 3], "mesh_name": ["dp", "tp"],
+        "strategy_name": "4d", "schedule": "zigzag",
+        "lr_schedule": "exponential",
+        "model_config": {"n_embd": 30, "n_head": 4, "n_experts": 3},
+    }
+    errs = validate_config(bad, world_size=8)
+    joined = "\n".join(errs)
+    for frag in ("needs 6 ranks", "unknown strategy", "unknown pipeline schedule",
+                 "unknown lr_schedule", "not divisible by n_head"):
+        assert frag in joined, (frag, errs)
