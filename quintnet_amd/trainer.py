@@ -191,6 +191,21 @@ class Trainer:
                 for k, v in metrics.items():
                     msg += f" {k}={v:.4f}"
                 print(msg, flush=True)
+                mf = self.config.get("metrics_file")
+                if mf:
+                    # machine-readable training record: one JSON line per
+                    # epoch, append-mode (survives resume)
+                    import json
+
+                    lr_now = None
+                    sched = getattr(self, "lr_scheduler", None)
+                    if sched is not None:
+                        lr_now = sched.lr_at(max(sched._step - 1, 0))
+                    with open(mf, "a") as f:
+                        f.write(json.dumps({
+                            "epoch": epoch + 1, "seconds": round(dt, 2),
+                            "lr": lr_now, **{k: float(v) for k, v in metrics.items()},
+                        }) + "\n")
             history = metrics
             if wd:
                 wd.beat()

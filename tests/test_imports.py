@@ -120,3 +120,30 @@ def test_validate_config_catches_mistakes():
     for frag in ("needs 6 ranks", "unknown strategy", "unknown pipeline schedule",
                  "unknown lr_schedule", "not divisible by n_head"):
         assert frag in joined, (frag, errs)
+
+
+def test_metrics_file_jsonl(tmp_path):
+    import json
+
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1, n_head=2)
+    mf = tmp_path / "metrics.jsonl"
+    tr = GPT2Trainer(
+        GPT2Stage(cfg),
+        DataLoader(SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0), batch_size=2),
+        None,
+        {"num_epochs": 2, "grad_acc_steps": 1, "zero1": False,
+         "metrics_file": str(mf), "lr_schedule": "cosine"},
+        None,
+    )
+    tr.fit()
+    lines = [json.loads(l) for l in mf.read_text().splitlines()]
+    assert len(lines) == 2
+    assert lines[0]["epoch"] == 1 and "train_loss" in lines[0]
+    assert lines[1]["lr"] is not None and lines[1]["lr"] > 0
