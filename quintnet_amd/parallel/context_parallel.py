@@ -398,6 +398,10 @@ class _ZigzagRingAttention(torch.autograd.Function):
         for s_hop in range(world):
             j = (rank - s_hop) % world
             ck = _zz_ids(j, world)
+            reqs_kv = None
+            if s_hop < world - 1:  # overlap the hop exchange w/ compute
+                reqs_kv, bufs_kv = _ring_swap_begin(
+                    [kc, vc], rank, world, group)
             for qi in range(2):
                 for ki in range(2):
                     if causal and ck[ki] > cq[qi]:
@@ -422,8 +426,10 @@ class _ZigzagRingAttention(torch.autograd.Function):
                         l[:, :, sl] = l[:, :, sl] * alpha + p.sum(
                             dim=-1, keepdim=True)
                         m[:, :, sl] = m_new
-            if s_hop < world - 1:
-                kc, vc = _ring_swap([kc, vc], rank, world, group)
+            if reqs_kv is not None:
+                for r_ in reqs_kv:
+                    r_.wait()
+                kc, vc = bufs_kv
         if flash:
             out = torch.cat(out_run, dim=2).to(q.dtype)
             lse = torch.cat(lse_run, dim=2).reshape(B * H, Tl)  # base-2

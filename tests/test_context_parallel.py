@@ -595,8 +595,7 @@ def _run_gpt2_cp_zigzag(rank, world):
 
     torch.manual_seed(33)
     cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
-                     n_head=2, dropout=0.0, cp_zigzag=True,
-                     vocab_pad_to=vocab_pad_to)
+                     n_head=2, dropout=0.0, cp_zigzag=True)
     full_cfg = GPT2Config(vocab_size=96, n_positions=32, n_embd=32, n_layer=2,
                           n_head=2, dropout=0.0)
     full = GPT2Stage(full_cfg, pp_rank=0, pp_size=1, tp_group=None)
