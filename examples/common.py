@@ -32,7 +32,11 @@ def build_model(cfg) -> Model:
     )
 
 
-def build_loaders(cfg, args):
+def build_loaders(cfg, args, pg=None):
+    """Train/val loaders; with a process-group manager whose dp axis is
+    >1, each DP replica gets its OWN data shard via DistributedSampler
+    (reference examples/full_3d.py:129 parity — without it every
+    replica trains on identical batches and DP adds no data throughput)."""
     if cfg.get("dataset_path"):
         train = CustomDataset(cfg["dataset_path"], "train")
         val = CustomDataset(cfg["dataset_path"], "test")
@@ -41,9 +45,22 @@ def build_loaders(cfg, args):
         val = SyntheticMNIST(n=args.n_val, seed=1)
     bs = cfg.get("batch_size", 8)
     nw = cfg.get("num_workers", 0)
+    tr_sampler = va_sampler = None
+    if pg is not None and getattr(pg, "dp_size", 1) > 1:
+        from torch.utils.data.distributed import DistributedSampler
+
+        tr_sampler = DistributedSampler(
+            train, num_replicas=pg.dp_size, rank=pg.dp_rank,
+            shuffle=True, seed=42,
+        )
+        va_sampler = DistributedSampler(
+            val, num_replicas=pg.dp_size, rank=pg.dp_rank, shuffle=False,
+        )
     return (
-        DataLoader(train, batch_size=bs, shuffle=False, num_workers=nw),
-        DataLoader(val, batch_size=bs, shuffle=False, num_workers=nw),
+        DataLoader(train, batch_size=bs, num_workers=nw,
+                   shuffle=False, sampler=tr_sampler),
+        DataLoader(val, batch_size=bs, num_workers=nw,
+                   shuffle=False, sampler=va_sampler),
     )
 
 

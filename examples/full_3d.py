@@ -27,7 +27,7 @@ def main():
         dist.broadcast(p.data, src=0)
     strategy = get_strategy(cfg.get("strategy_name", "3d"), pg, cfg)
     pmodel = strategy.apply(model)
-    train, val = build_loaders(cfg, args)
+    train, val = build_loaders(cfg, args, pg)
     t0 = time.time()
     Trainer(pmodel, train, val, cfg, pg).fit()
     if pg.rank == 0:

@@ -21,7 +21,7 @@ def main():
     for p in model.parameters():
         dist.broadcast(p.data, src=0)
     pmodel = get_strategy("pp", pg, cfg).apply(model)
-    train, val = build_loaders(cfg, args)
+    train, val = build_loaders(cfg, args, pg)
     Trainer(pmodel, train, val, cfg, pg).fit()
 
 
