@@ -237,13 +237,13 @@ class GPT2Trainer(Trainer):
         return {"loss": avg, "ppl": math.exp(min(avg, 20.0))}
 
     # ------------------------------------------------------------------
-    def _save_checkpoint(self) -> None:
+    def _save_checkpoint(self, name: str = None) -> None:
         out_dir = self.config.get("checkpoint_dir") or self.config.get("output_dir")
         if not out_dir:
             return
         from .checkpoint import save_sharded_checkpoint
 
-        name = self.config.get("checkpoint_name", "final_model")
+        name = name or self.config.get("checkpoint_name", "final_model")
         save_sharded_checkpoint(
             _unwrap(self.model),
             out_dir,

@@ -58,6 +58,7 @@ class Trainer:
         self.is_pipeline = self.pp_size > 1
         self.pipeline_trainer: Optional[PipelineTrainer] = None
         self.tensor_shapes = None
+        self._best_metric: Optional[float] = None
         if self.is_pipeline:
             self._setup_pipeline()
 
@@ -212,6 +213,17 @@ class Trainer:
             every = int(self.config.get("save_every", 0))
             if every and (epoch + 1) % every == 0 and epoch + 1 < self.num_epochs:
                 self._save_checkpoint()
+            if self.config.get("save_best"):
+                # keep a separate best-validation shard set alongside the
+                # rolling checkpoint (lower is better: loss-like metrics)
+                key = self.config.get("best_metric", "val_loss")
+                cur = metrics.get(key)
+                if cur is not None and (
+                    self._best_metric is None or cur < self._best_metric
+                ):
+                    self._best_metric = float(cur)
+                    name = self.config.get("checkpoint_name", "final_model")
+                    self._save_checkpoint(name=f"{name}_best")
         self._save_checkpoint()
         return history
 
@@ -388,7 +400,7 @@ class Trainer:
         return metrics
 
     # ------------------------------------------------------------------
-    def _save_checkpoint(self) -> None:
+    def _save_checkpoint(self, name: str = None) -> None:
         path = self.config.get("checkpoint_dir")
         if not path:
             return
@@ -396,5 +408,5 @@ class Trainer:
         if self._is_rank0():
             torch.save(
                 {"model_state_dict": _unwrap(self.model).state_dict(), "config": self.config},
-                os.path.join(path, "checkpoint.pt"),
+                os.path.join(path, f"{name}.pt" if name else "checkpoint.pt"),
             )

@@ -221,3 +221,33 @@ def _run_moe_ckpt(rank, world, tmpdir):
 
 def test_moe_checkpoint_shards(tmp_path):
     run_distributed(_run_moe_ckpt, 2, str(tmp_path))
+
+
+def test_save_best_keeps_best_val_shards(tmp_path):
+    """save_best: a separate *_best shard set tracks the lowest val
+    loss across epochs (base trainer semantics through GPT2Trainer)."""
+    import os
+
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    vs = SyntheticCLM(n=2, seq_len=16, vocab_size=64, seed=7)
+    tr = GPT2Trainer(
+        GPT2Stage(cfg), DataLoader(ds, batch_size=2),
+        DataLoader(vs, batch_size=2),
+        {"num_epochs": 2, "grad_acc_steps": 1, "zero1": False,
+         "checkpoint_dir": str(tmp_path), "save_best": True,
+         "learning_rate": 1e-3},
+        None,
+    )
+    tr.fit()
+    assert os.path.exists(tmp_path / "final_model_pp0_tp0.pt")
+    assert os.path.exists(tmp_path / "final_model_best_pp0_tp0.pt")
+    assert tr._best_metric is not None
