@@ -59,6 +59,8 @@ class Trainer:
         self.pipeline_trainer: Optional[PipelineTrainer] = None
         self.tensor_shapes = None
         self._best_metric: Optional[float] = None
+        self._es_best: Optional[float] = None
+        self._es_bad = 0
         if self.is_pipeline:
             self._setup_pipeline()
 
@@ -224,6 +226,22 @@ class Trainer:
                     self._best_metric = float(cur)
                     name = self.config.get("checkpoint_name", "final_model")
                     self._save_checkpoint(name=f"{name}_best")
+            patience = int(self.config.get("early_stop_patience", 0))
+            if patience:
+                key = self.config.get("best_metric", "val_loss")
+                cur = metrics.get(key)
+                if cur is not None:
+                    if self._es_best is None or cur < self._es_best - float(
+                        self.config.get("early_stop_min_delta", 0.0)
+                    ):
+                        self._es_best, self._es_bad = float(cur), 0
+                    else:
+                        self._es_bad += 1
+                        if self._es_bad >= patience:
+                            if self._is_rank0():
+                                print(f"[early stop] {key} flat for "
+                                      f"{patience} epochs", flush=True)
+                            break
         self._save_checkpoint()
         return history
 

@@ -251,3 +251,29 @@ def test_save_best_keeps_best_val_shards(tmp_path):
     assert os.path.exists(tmp_path / "final_model_pp0_tp0.pt")
     assert os.path.exists(tmp_path / "final_model_best_pp0_tp0.pt")
     assert tr._best_metric is not None
+
+
+def test_early_stopping_breaks_epoch_loop(tmp_path):
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    vs = SyntheticCLM(n=2, seq_len=16, vocab_size=64, seed=7)
+    tr = GPT2Trainer(
+        GPT2Stage(cfg), DataLoader(ds, batch_size=2),
+        DataLoader(vs, batch_size=2),
+        {"num_epochs": 50, "grad_acc_steps": 1, "zero1": False,
+         # lr=0: val loss can never improve -> stop after patience
+         "learning_rate": 0.0, "early_stop_patience": 2,
+         "metrics_file": str(tmp_path / "m.jsonl")},
+        None,
+    )
+    tr.fit()
+    n_epochs_run = len((tmp_path / "m.jsonl").read_text().splitlines())
+    assert n_epochs_run <= 4, n_epochs_run  # 1 best + 2 flat, not 50
