@@ -95,16 +95,21 @@ class TorchDistributedBackend(DistributedBackend):
 
 
 class LocalBackend(DistributedBackend):
-    """No-op backend for single-process unit tests."""
+    """No-op backend for single-process unit tests.  ``world_size`` can
+    be faked so bucket-LAYOUT logic (padding, chunking) is testable
+    without spawning processes — the comm ops remain no-ops."""
+
+    def __init__(self, world_size: int = 1, rank: int = 0):
+        self._world, self._rank = world_size, rank
 
     def is_initialized(self) -> bool:
         return True
 
     def get_world_size(self) -> int:
-        return 1
+        return self._world
 
     def get_rank(self) -> int:
-        return 0
+        return self._rank
 
     def broadcast_tensor(self, tensor: torch.Tensor, src: int = 0) -> None:
         return None

@@ -116,3 +116,57 @@ def test_merge_qkv_roundtrip(heads, dh, tp, seed):
         sl = slice(r * H // tp, (r + 1) * H // tp)
         shards.append(torch.cat([q[sl], k[sl], v[sl]], dim=0))
     assert torch.equal(_merge_qkv_rows(shards), w)
+
+
+@_FAST
+@given(
+    sizes=st.lists(st.integers(1, 5000), min_size=1, max_size=20),
+    cap_kb=st.sampled_from([1, 4, 16]),
+    rs=st.booleans(),
+    ws=st.sampled_from([1, 2, 4, 8]),
+)
+def test_ddp_bucket_partition_invariants(sizes, cap_kb, rs, ws):
+    """Bucket layout: every param lands in exactly one bucket, flats
+    cover all params, reduce-scatter buckets are world-divisible."""
+    import torch.nn as nn
+
+    from quintnet_amd.parallel.backends import LocalBackend
+    from quintnet_amd.parallel.data_parallel import (
+        BucketConfig,
+        DataParallel,
+        DistributedConfig,
+    )
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.ps = nn.ParameterList(
+                nn.Parameter(torch.zeros(n)) for n in sizes
+            )
+
+    m = M()
+    ddp = DataParallel(
+        m,
+        DistributedConfig(0, ws, None),
+        bucket_config=BucketConfig(
+            capacity_mb=cap_kb / 1024.0,
+            grad_reduce_op="reduce_scatter" if rs else "all_reduce",
+        ),
+        backend=LocalBackend(world_size=ws),
+    )
+    covered = set()
+    total_params = 0
+    for b in ddp.buckets:
+        for p in b.params:
+            assert id(p) not in covered, "param in two buckets"
+            covered.add(id(p))
+        total_params += len(b.params)
+        if rs and ws > 1:
+            assert b.flat.numel() % ws == 0, "RS bucket not world-divisible"
+            assert b.own_chunk is not None
+            assert b.own_chunk.numel() == b.flat.numel() // ws
+    assert total_params == len(sizes)
+    # every param's grad view must alias the flat buffer
+    for b in ddp.buckets:
+        n = sum(p.numel() for p in b.params)
+        assert n <= b.flat.numel() <= n + ws
