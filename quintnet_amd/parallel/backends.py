@@ -119,7 +119,7 @@ class LocalBackend(DistributedBackend):
 
     def reduce_scatter_tensor(self, output: torch.Tensor, input: torch.Tensor,
                               async_op: bool = False):
-        output.copy_(input[: output.numel()])
+        output.copy_(input.view(self._world, -1)[self._rank])
         return None
 
     def all_gather_into_tensor(self, output: torch.Tensor, input: torch.Tensor,
