@@ -170,3 +170,21 @@ def test_ddp_bucket_partition_invariants(sizes, cap_kb, rs, ws):
     for b in ddp.buckets:
         n = sum(p.numel() for p in b.params)
         assert n <= b.flat.numel() <= n + ws
+
+
+@_FAST
+@given(
+    n=st.integers(2, 30),
+    v=st.integers(3, 60),
+    eps=st.floats(0.01, 0.5),
+    seed=st.integers(0, 10**6),
+)
+def test_label_smoothing_matches_torch(n, v, eps, seed):
+    g = torch.Generator().manual_seed(seed)
+    logits = (torch.randn(n, v, generator=g) * 2).requires_grad_(True)
+    tgt = torch.randint(0, v, (n,), generator=g)
+    ours = cross_entropy(logits, tgt, label_smoothing=eps)
+    ref = torch.nn.functional.cross_entropy(logits, tgt, label_smoothing=eps)
+    assert torch.allclose(ours, ref, rtol=1e-5, atol=1e-6)
+    ours.backward()
+    assert torch.isfinite(logits.grad).all()
