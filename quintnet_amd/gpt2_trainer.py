@@ -66,7 +66,10 @@ class GPT2Trainer(Trainer):
             self.optimizer = torch.optim.AdamW(
                 self.model.parameters(), lr=self.lr, weight_decay=wd
             )
-        self.criterion = nn.CrossEntropyLoss(ignore_index=-100)
+        self._label_smoothing = float(config.get("label_smoothing", 0.0))
+        self.criterion = nn.CrossEntropyLoss(
+            ignore_index=-100, label_smoothing=self._label_smoothing
+        )
         if self.is_pipeline:
             # re-bind the pipeline trainer to the new optimizer/criterion
             self.pipeline_trainer.optimizer = self.optimizer
@@ -149,7 +152,8 @@ class GPT2Trainer(Trainer):
                 loss, true_loss = cp_causal_lm_loss(logits, tgt_shard, cp_group)
             else:
                 logits = self.model(ids)
-                loss = causal_lm_loss(logits, labels, ignore_index=-100)
+                loss = causal_lm_loss(logits, labels, ignore_index=-100,
+                                      label_smoothing=self._label_smoothing)
             aux_w = float(self.config.get("moe_aux_weight", 0.0))
             total = loss
             if aux_w:

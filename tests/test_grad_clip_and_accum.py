@@ -171,3 +171,27 @@ def test_tied_copy_excluded_from_norm():
     # both scaled by 1/6
     assert torch.allclose(a.grad, torch.full((4,), 0.5), atol=1e-4)
     assert torch.allclose(b.grad, torch.full((4,), 0.5), atol=1e-4)
+
+
+def test_label_smoothing_config_flows_to_loss():
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0)
+    ds = SyntheticCLM(n=2, seq_len=16, vocab_size=64, seed=0)
+
+    def run(ls):
+        torch.manual_seed(3)
+        tr = GPT2Trainer(GPT2Stage(cfg), DataLoader(ds, batch_size=2), None,
+                         {"num_epochs": 1, "grad_acc_steps": 1, "zero1": False,
+                          "learning_rate": 0.0, "label_smoothing": ls}, None)
+        return tr.fit()["train_loss"]
+
+    plain, smoothed = run(0.0), run(0.2)
+    assert abs(plain - smoothed) > 1e-4  # smoothing changes the loss
+    assert smoothed > 0
