@@ -202,16 +202,19 @@ class GPT2Trainer(Trainer):
             assert_finite_grads(self.model)
         if self.max_grad_norm:
             if hasattr(self.optimizer, "clip_grad_norm_"):
-                self.optimizer.clip_grad_norm_(self.max_grad_norm)
+                gn = self.optimizer.clip_grad_norm_(self.max_grad_norm)
             else:
                 from .ops import clip_grad_norm_global
 
-                clip_grad_norm_global(
+                gn = clip_grad_norm_global(
                     [p for p in self.model.parameters() if p.requires_grad],
                     self.max_grad_norm,
                     tp_group=self._tp_group,
                     pp_group=self._pp_group,
                 )
+            # observability: last pre-clip global grad norm (float lazily
+            # — avoid a sync inside the hot loop; .item() on read)
+            self._last_grad_norm = gn
         self._lr_step()
         self.optimizer.step()
         if isinstance(self.model, DataParallel):

@@ -204,10 +204,14 @@ class Trainer:
                     sched = getattr(self, "lr_scheduler", None)
                     if sched is not None:
                         lr_now = sched.lr_at(max(sched._step - 1, 0))
+                    gnorm = getattr(self, "_last_grad_norm", None)
+                    if gnorm is not None:
+                        gnorm = round(float(gnorm), 6)
                     with open(mf, "a") as f:
                         f.write(json.dumps({
                             "epoch": epoch + 1, "seconds": round(dt, 2),
-                            "lr": lr_now, **{k: float(v) for k, v in metrics.items()},
+                            "lr": lr_now, "grad_norm": gnorm,
+                            **{k: float(v) for k, v in metrics.items()},
                         }) + "\n")
             history = metrics
             if wd:
