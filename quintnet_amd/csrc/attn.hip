@@ -764,6 +764,194 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dkv_kernel(
   }
 }
 
+
+// Non-pipelined dkv variant for the 3-waves/SIMD occupancy point
+// (QN_ATTN_DKV_OCC=3).  The shipped kernel above software-pipelines
+// S/dP of tile i+1 against dV/dK of tile i, which needs ~205 VGPRs —
+// 2 waves/SIMD; instantiating IT at 3 waves spills 39 regs (offline
+// dump, profiles/kernel_resources.md).  Here the whole tile-i chain
+// (S/dP -> softmax -> dV/dK) runs inside one iteration so s/dp are
+// TRANSIENT, trading in-wave MFMA overlap for one extra resident wave
+// to hide the latency instead.  Same math, same LDS layout, single
+// barrier per tile.  Numerics gate: tests/test_ops_gpu.py attention
+// bwd oracles with QN_ATTN_DKV_OCC=3 (r3).
+__global__ __launch_bounds__(256, 3) void attn_bwd_dkv_np_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse2, const float* __restrict__ delta,
+    unsigned short* __restrict__ dk, unsigned short* __restrict__ dv,
+    int Tq, int Tk, int qoff, int H, float scale, int causal,
+    long long qsB, long long qsH, long long qsT,
+    long long ksB, long long ksH, long long ksT,
+    long long vsB, long long vsH, long long vsT,
+    long long dsB, long long dsH, long long dsT,
+    long long dksB, long long dksH, long long dksT,
+    long long dvsB, long long dvsH, long long dvsT) {
+  __shared__ unsigned short dot_lds[2][64 * TPAD];
+  __shared__ unsigned short qt_lds[2][64 * TPAD];
+  __shared__ unsigned short qrow[2][32 * KPAD];
+  __shared__ unsigned short dorow[2][32 * KPAD];
+  __shared__ alignas(16) float lse_t[2][32];
+  __shared__ alignas(16) float del_t[2][32];
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int kv0b = blockIdx.x * 128;
+  const int kw = kv0b + wave * 32;
+  const int mykey = kw + (lane & 31);
+
+  const unsigned short* qp = q + b * qsB + h * qsH;
+  const unsigned short* kp = k + b * ksB + h * ksH;
+  const unsigned short* vp = v + b * vsB + h * vsH;
+  const unsigned short* dop = dout + b * dsB + h * dsH;
+  unsigned short* dkp = dk + b * dksB + h * dksH;
+  unsigned short* dvp = dv + b * dvsB + h * dvsH;
+
+  bf16x8 kf[4], vf[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    kf[t] = frag_ld(kp, ksT, kw, t * 16, lane);
+    vf[t] = frag_ld(vp, vsT, kw, t * 16, lane);
+  }
+  const float s2scale = scale * LOG2E;
+
+  f32x16 dka[2], dva[2], zc;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { dka[0][i] = dka[1][i] = dva[0][i] = dva[1][i] = 0.f; zc[i] = 0.f; }
+
+  const int q_start = causal ? max(kv0b - qoff, 0) : 0;
+  const unsigned short* qsp = stage_base(qp, qsT, q_start);
+  const unsigned short* dosp = stage_base(dop, dsT, q_start);
+  const float* lsep = lse2 + (long long)bh * Tq + q_start + threadIdx.x;
+  const float* delp = delta + (long long)bh * Tq + q_start + threadIdx.x;
+  const long long qstep = 32 * qsT, dstep = 32 * dsT;
+
+  const int srow_ = threadIdx.x & 31;
+  const int sd0_ = (threadIdx.x >> 5) << 3;
+  {
+    s16x8 d0 = stage_at(dosp);
+    s16x8 q0 = stage_at(qsp);
+    stage_wr(dot_lds[0], d0);
+    stage_wr(qt_lds[0], q0);
+    *reinterpret_cast<s16x8*>(&dorow[0][srow_ * KPAD + sd0_]) = d0;
+    *reinterpret_cast<s16x8*>(&qrow[0][srow_ * KPAD + sd0_]) = q0;
+    if (threadIdx.x < 32) {
+      lse_t[0][threadIdx.x] = *lsep;
+      del_t[0][threadIdx.x] = *delp;
+    }
+  }
+  __syncthreads();
+
+  const int fl_ = (lane & 31) * KPAD + ((lane >> 5) << 3);
+  int cur = 0;
+  for (int qt0 = q_start; qt0 < Tq; qt0 += 32) {
+    const bool have_next = qt0 + 32 < Tq;
+    const bool active = !(causal && qt0 + qoff + 31 < kw);
+
+    if (active) {
+      // S/dP for THIS tile from the row images (transient chains)
+      f32x16 s, dp_;
+      {
+        const unsigned short* qr = &qrow[cur][fl_];
+        const unsigned short* dr = &dorow[cur][fl_];
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(qr), kf[0], zc, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(dr), vf[0], zc, 0, 0, 0);
+#pragma unroll
+        for (int t = 1; t < 4; ++t) {
+          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8*>(qr + t * 16), kf[t], s, 0, 0, 0);
+          dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8*>(dr + t * 16), vf[t], dp_, 0, 0, 0);
+        }
+      }
+      // softmax + relayout, HALF at a time: pv/gv live 8 floats each
+      // instead of 16 (register pressure — this variant trades in-wave
+      // ILP for occupancy everywhere)
+      bf16x8 pf0, pf1, gf0, gf1;
+      {
+        const bool diag = causal && (qt0 + qoff < kw + 31);
+        const int lb = (lane >> 5) << 2;
+#pragma unroll
+        for (int half = 0; half < 2; ++half) {
+          float pv[8], gv[8];
+#pragma unroll
+          for (int g = 0; g < 2; ++g) {
+            const int gg = half * 2 + g;
+            f32x4 lse4 = *reinterpret_cast<const f32x4*>(&lse_t[cur][lb + gg * 8]);
+            f32x4 del4 = *reinterpret_cast<const f32x4*>(&del_t[cur][lb + gg * 8]);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              int r = gg * 4 + j;
+              float p;
+              if (diag) {
+                int qrow_ = drow(r, lane);
+                p = (mykey > qt0 + qoff + qrow_)
+                        ? 0.f
+                        : __builtin_amdgcn_exp2f(s[r] * s2scale - lse4[j]);
+              } else {
+                p = __builtin_amdgcn_exp2f(s[r] * s2scale - lse4[j]);
+              }
+              pv[g * 4 + j] = p;
+              gv[g * 4 + j] = scale * p * (dp_[r] - del4[j]);
+            }
+          }
+          if (half == 0) { pf0 = relayout8(pv); gf0 = relayout8(gv); }
+          else           { pf1 = relayout8(pv); gf1 = relayout8(gv); }
+        }
+      }
+      // dV/dK from the transpose images
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt) {
+        const unsigned short* adot = &dot_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        const unsigned short* aqt = &qt_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        bf16x8 d0 = *reinterpret_cast<const bf16x8*>(adot);
+        bf16x8 d1 = *reinterpret_cast<const bf16x8*>(adot + 16);
+        bf16x8 q0f = *reinterpret_cast<const bf16x8*>(aqt);
+        bf16x8 q1f = *reinterpret_cast<const bf16x8*>(aqt + 16);
+        dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d0, pf0, dva[mt], 0, 0, 0);
+        dva[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(d1, pf1, dva[mt], 0, 0, 0);
+        dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q0f, gf0, dka[mt], 0, 0, 0);
+        dka[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(q1f, gf1, dka[mt], 0, 0, 0);
+      }
+    }
+
+    if (have_next) {
+      // stage tile i+1 into the other buffer.  Loads are issued HERE,
+      // after the tile's MFMAs, so their registers are not live across
+      // the compute — at 3 waves/SIMD the other waves' compute covers
+      // the load latency that in-tile overlap covered at 2 waves.
+      s16x8 dost_n = stage_at(dosp + dstep);
+      s16x8 qst_n = stage_at(qsp + qstep);
+      stage_wr(dot_lds[cur ^ 1], dost_n);
+      stage_wr(qt_lds[cur ^ 1], qst_n);
+      *reinterpret_cast<s16x8*>(&dorow[cur ^ 1][srow_ * KPAD + sd0_]) = dost_n;
+      *reinterpret_cast<s16x8*>(&qrow[cur ^ 1][srow_ * KPAD + sd0_]) = qst_n;
+      if (threadIdx.x < 32) {
+        lse_t[cur ^ 1][threadIdx.x] = lsep[32];
+        del_t[cur ^ 1][threadIdx.x] = delp[32];
+      }
+      qsp += qstep; dosp += dstep;
+      lsep += 32; delp += 32;
+    }
+    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
+    cur ^= 1;
+  }
+
+  if (mykey < Tk) {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = mt * 32 + drow(r, lane);
+        dkp[(long long)mykey * dksT + d] = f32_to_bf16(dka[mt][r]);
+        dvp[(long long)mykey * dvsT + d] = f32_to_bf16(dva[mt][r]);
+      }
+  }
+}
+
 // ---- launchers -------------------------------------------------------------
 struct AttnStrides {
   long long qB, qH, qT, kB, kH, kT, vB, vH, vT, oB, oH, oT;
@@ -850,7 +1038,7 @@ void attn_bwd_dkv_launch(const unsigned short* q, const unsigned short* k,
     return e ? atoi(e) : 2;
   }();
   if (occ >= 3)
-    hipLaunchKernelGGL(attn_bwd_dkv_kernel<3>, grid, dim3(256), 0, stream, q, k, v,
+    hipLaunchKernelGGL(attn_bwd_dkv_np_kernel, grid, dim3(256), 0, stream, q, k, v,
                        dout, lse2, delta, dk, dv, Tq, Tk, qoff, H, scale, causal, st.qB,
                        st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH, st.vT,
                        dsB, dsH, dsT, dkB, dkH, dkT, dvB, dvH, dvT);
