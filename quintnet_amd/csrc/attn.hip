@@ -1007,7 +1007,12 @@ void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
     const char* e = getenv("QN_ATTN_DQ_OCC");
     occ = e ? atoi(e) : 3;  // A/B r2: 3 waves/SIMD = bwd 279.8 -> 265.2 us
   }
-  if (occ >= 3)
+  if (occ >= 4)  // offline: <4> = 128 VGPR, see dump tool output
+    hipLaunchKernelGGL(attn_bwd_dq_kernel<4>, grid, dim3(256), 0, stream, q, k,
+                       v, dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal,
+                       st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH,
+                       st.vT, dsB, dsH, dsT, st.oB, st.oH, st.oT);
+  else if (occ >= 3)
     hipLaunchKernelGGL(attn_bwd_dq_kernel<3>, grid, dim3(256), 0, stream, q, k,
                        v, dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal,
                        st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH,
