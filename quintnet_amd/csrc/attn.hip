@@ -521,6 +521,152 @@ __global__ __launch_bounds__(256, MINW) void attn_bwd_dq_kernel(
   }
 }
 
+// Non-pipelined dq variant for the 4-waves/SIMD point
+// (QN_ATTN_DQ_OCC=4): same trade as attn_bwd_dkv_np_kernel — the S/dP
+// chains become transient inside the tile (no cross-tile pipelining),
+// staging loads issue after the tile's MFMAs, and the extra resident
+// wave hides the latency the in-wave overlap used to.  The pipelined
+// <3> form needs 162 VGPRs; instantiating it at 4 waves spills 22.
+__global__ __launch_bounds__(256, 4) void attn_bwd_dq_np_kernel(
+    const unsigned short* __restrict__ q, const unsigned short* __restrict__ k,
+    const unsigned short* __restrict__ v, const unsigned short* __restrict__ dout,
+    const float* __restrict__ lse2, const float* __restrict__ delta,
+    unsigned short* __restrict__ dq, int Tq, int Tk, int qoff, int H,
+    float scale, int causal,
+    long long qsB, long long qsH, long long qsT,
+    long long ksB, long long ksH, long long ksT,
+    long long vsB, long long vsH, long long vsT,
+    long long dsB, long long dsH, long long dsT,
+    long long dqsB, long long dqsH, long long dqsT) {
+  __shared__ unsigned short kt_lds[2][64 * TPAD];
+  __shared__ unsigned short krow_l[2][32 * KPAD];
+  __shared__ unsigned short vrow_l[2][32 * KPAD];
+  const int bh = blockIdx.y;
+  const int b = bh / H, h = bh % H;
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int q0 = blockIdx.x * 128;
+  const int qw = q0 + wave * 32;
+  const int myq = qw + (lane & 31);
+
+  const unsigned short* qp = q + b * qsB + h * qsH;
+  const unsigned short* kp = k + b * ksB + h * ksH;
+  const unsigned short* vp = v + b * vsB + h * vsH;
+  const unsigned short* dop = dout + b * dsB + h * dsH;
+  unsigned short* dqp = dq + b * dqsB + h * dqsH;
+
+  bf16x8 qf[4], dof[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) {
+    qf[t] = frag_ld(qp, qsT, qw, t * 16, lane);
+    dof[t] = frag_ld(dop, dsT, qw, t * 16, lane);
+  }
+  const float my_lse = lse2[(long long)bh * Tq + min(myq, Tq - 1)];
+  const float my_delta = delta[(long long)bh * Tq + min(myq, Tq - 1)];
+  const float s2scale = scale * LOG2E;
+
+  f32x16 dqa[2], zc;
+#pragma unroll
+  for (int i = 0; i < 16; ++i) { dqa[0][i] = 0.f; dqa[1][i] = 0.f; zc[i] = 0.f; }
+
+  const int kv_end = causal ? min(q0 + qoff + 128, Tk) : Tk;
+  const unsigned short* ksp = stage_base(kp, ksT, 0);
+  const unsigned short* vsp = stage_base(vp, vsT, 0);
+  const long long kstep = 32 * ksT, vstep = 32 * vsT;
+  const int srow_ = threadIdx.x & 31;
+  const int sd0_ = (threadIdx.x >> 5) << 3;
+  const int fl_ = (lane & 31) * KPAD + ((lane >> 5) << 3);
+
+  {
+    s16x8 k0 = stage_at(ksp);
+    s16x8 v0 = stage_at(vsp);
+    stage_wr(kt_lds[0], k0);
+    *reinterpret_cast<s16x8*>(&krow_l[0][srow_ * KPAD + sd0_]) = k0;
+    *reinterpret_cast<s16x8*>(&vrow_l[0][srow_ * KPAD + sd0_]) = v0;
+  }
+  __syncthreads();  // buf0 visible
+
+  int cur = 0;
+  for (int kv0 = 0; kv0 < kv_end; kv0 += 32) {
+    const bool have_next = kv0 + 32 < kv_end;
+    const bool active = !(causal && kv0 > qw + qoff + 31);
+
+    if (active) {
+      // transient S/dP chains for THIS tile
+      f32x16 s, dp_;
+      {
+        const unsigned short* kr = &krow_l[cur][fl_];
+        const unsigned short* vr = &vrow_l[cur][fl_];
+        s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(kr), qf[0], zc, 0, 0, 0);
+        dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+            *reinterpret_cast<const bf16x8*>(vr), dof[0], zc, 0, 0, 0);
+#pragma unroll
+        for (int t = 1; t < 4; ++t) {
+          s = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8*>(kr + t * 16), qf[t], s, 0, 0, 0);
+          dp_ = __builtin_amdgcn_mfma_f32_32x32x16_bf16(
+              *reinterpret_cast<const bf16x8*>(vr + t * 16), dof[t], dp_, 0, 0, 0);
+        }
+      }
+      // softmax half at a time (8-float g liveness) + dq MFMAs
+      const bool diag = causal && (kv0 + 31 > qw + qoff);
+      bf16x8 gf0, gf1;
+#pragma unroll
+      for (int half = 0; half < 2; ++half) {
+        float g[8];
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          int r = half * 8 + j;
+          float p;
+          if (diag) {
+            int key = kv0 + drow(r, lane);
+            p = (key > myq + qoff)
+                    ? 0.f
+                    : __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+          } else {
+            p = __builtin_amdgcn_exp2f(s[r] * s2scale - my_lse);
+          }
+          g[j] = scale * p * (dp_[r] - my_delta);
+        }
+        if (half == 0) gf0 = relayout8(g); else gf1 = relayout8(g);
+      }
+#pragma unroll
+      for (int mt = 0; mt < 2; ++mt) {
+        const unsigned short* ak = &kt_lds[cur][(mt * 32 + (lane & 31)) * TPAD + ((lane >> 5) << 3)];
+        bf16x8 a0 = *reinterpret_cast<const bf16x8*>(ak);
+        bf16x8 a1 = *reinterpret_cast<const bf16x8*>(ak + 16);
+        dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a0, gf0, dqa[mt], 0, 0, 0);
+        dqa[mt] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(a1, gf1, dqa[mt], 0, 0, 0);
+      }
+    }
+
+    if (have_next) {
+      // stage tile i+1 (loads issued after the MFMAs — registers are
+      // not live across compute; the 4th wave hides the latency)
+      s16x8 kst_n = stage_at(ksp + kstep);
+      s16x8 vst_n = stage_at(vsp + vstep);
+      stage_wr(kt_lds[cur ^ 1], kst_n);
+      *reinterpret_cast<s16x8*>(&krow_l[cur ^ 1][srow_ * KPAD + sd0_]) = kst_n;
+      *reinterpret_cast<s16x8*>(&vrow_l[cur ^ 1][srow_ * KPAD + sd0_]) = vst_n;
+      ksp += kstep;
+      vsp += vstep;
+    }
+    __syncthreads();  // buf[cur^1] writes visible; buf[cur] reads done
+    cur ^= 1;
+  }
+
+  if (myq < Tq) {
+#pragma unroll
+    for (int mt = 0; mt < 2; ++mt)
+#pragma unroll
+      for (int r = 0; r < 16; ++r) {
+        int d = mt * 32 + drow(r, lane);
+        dqp[(long long)myq * dqsT + d] = f32_to_bf16(dqa[mt][r]);
+      }
+  }
+}
+
 // ===========================================================================
 // backward dK/dV: block owns a 128-key kv tile (wave per 32 keys),
 // loops q tiles.  Mirrored orientation: lane owns a KEY column.
@@ -1007,8 +1153,8 @@ void attn_bwd_dq_launch(const unsigned short* q, const unsigned short* k,
     const char* e = getenv("QN_ATTN_DQ_OCC");
     occ = e ? atoi(e) : 3;  // A/B r2: 3 waves/SIMD = bwd 279.8 -> 265.2 us
   }
-  if (occ >= 4)  // offline: <4> = 128 VGPR, see dump tool output
-    hipLaunchKernelGGL(attn_bwd_dq_kernel<4>, grid, dim3(256), 0, stream, q, k,
+  if (occ >= 4)  // non-pipelined 4-wave variant (see kernel comment)
+    hipLaunchKernelGGL(attn_bwd_dq_np_kernel, grid, dim3(256), 0, stream, q, k,
                        v, dout, lse2, delta, dq, Tq, Tk, qoff, H, scale, causal,
                        st.qB, st.qH, st.qT, st.kB, st.kH, st.kT, st.vB, st.vH,
                        st.vT, dsB, dsH, dsT, st.oB, st.oH, st.oT);
