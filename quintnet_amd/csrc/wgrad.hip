@@ -160,6 +160,103 @@ __global__ __launch_bounds__(256) void wgrad_tn_kernel(
 #undef WG_STAGE
 }
 
+// 32-m-row stage variant: halves the LDS footprint (2x2x8 KiB = 32 KiB)
+// so the CU fits 4 workgroups instead of 2 — 3 waves/SIMD after the
+// VGPR cap instead of the 2 the 64 KiB default allows (the default
+// kernel is LDS-limited: profiles/kernel_resources.md).  Twice the
+// steps, half the MFMA work per step, same counted-vmcnt two-buffer
+// pipeline.  Opt-in via QN_WGRAD_LDS=32 until A/B'd on hardware (r3).
+#define WG32_IMG_EL 4096  // elements per operand image (8 KiB)
+
+__global__ __launch_bounds__(256, 3) void wgrad_tn32_kernel(
+    const unsigned short* __restrict__ dY,  // [M,N]
+    const unsigned short* __restrict__ X,   // [M,K]
+    float* __restrict__ slab,               // [splits, N, K]
+    int M, int N, int K, int chunk) {
+  __shared__ unsigned short smem[2 * 2 * WG32_IMG_EL];
+
+  const int nbk = K >> 7;
+  const int bn = blockIdx.x / nbk, bk = blockIdx.x % nbk;
+  const int n0 = bn << 7, k0 = bk << 7;
+  const long long m0 = (long long)blockIdx.y * chunk;
+
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int wr = wave >> 1, wc = wave & 1;  // 2x2 waves; per-wave 64n x 64k
+
+  // image per operand: 128 out x 32 m = 64 blocks (B = o16*8 + m4,
+  // m4 0..7); 8 glds instrs fill it -> 2 per wave per operand
+  const unsigned short* gpa[2];
+  const unsigned short* gpb[2];
+#pragma unroll
+  for (int j = 0; j < 2; ++j) {
+    int I = wave * 2 + j;
+    int Bl = I * 8 + (lane >> 3);
+    int m4 = Bl & 7, o16 = Bl >> 3;
+    int gb = lane & 7;
+    long long mrow = m0 + m4 * 4 + (gb >> 1);
+    gpa[j] = dY + mrow * N + n0 + o16 * 16 + (gb & 1) * 8;
+    gpb[j] = X + mrow * K + k0 + o16 * 16 + (gb & 1) * 8;
+  }
+
+#define WG32_STAGE(buf, sa, sb)                                                  do {                                                                             _Pragma("unroll") for (int j = 0; j < 2; ++j) {                                  int I = wave * 2 + j;                                                          wglds16(gpa[j] + (sa), smem + (buf) * 2 * WG32_IMG_EL + I * 512);              wglds16(gpb[j] + (sb), smem + (buf) * 2 * WG32_IMG_EL + WG32_IMG_EL + I * 512);     }                                                                            } while (0)
+
+  // fragment offsets: o16 group stride is now 8 blocks x 64 = 512 el;
+  // single 32-m depth per step (the ks loop of the 64-m kernel is gone)
+  const int frow = lane & 15;
+  const int kq = lane >> 4;
+  const int abase = (wr * 4) * 8 * 64 + kq * 128 + frow * 4;
+  const int bbase = WG32_IMG_EL + (wc * 4) * 8 * 64 + kq * 128 + frow * 4;
+
+  f32x4 acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  union frag8 { bf16x8w v; bf16x4w h[2]; };
+
+#define WG32_MFMA(buf)                                                           do {                                                                             const unsigned short* base = smem + (buf) * 2 * WG32_IMG_EL;                   frag8 af[4], bf[4];                                                            _Pragma("unroll") for (int f = 0; f < 4; ++f) {                                  af[f].h[0] = tr16(base + abase + f * 512);                                     af[f].h[1] = tr16(base + abase + f * 512 + 64);                                bf[f].h[0] = tr16(base + bbase + f * 512);                                     bf[f].h[1] = tr16(base + bbase + f * 512 + 64);                              }                                                                              _Pragma("unroll") for (int i = 0; i < 4; ++i)                                      _Pragma("unroll") for (int j = 0; j < 4; ++j)                                      acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(                               af[i].v, bf[j].v, acc[i][j], 0, 0, 0);                           } while (0)
+
+  const int nsteps = chunk >> 5;
+  const long long da = (long long)32 * N, db = (long long)32 * K;
+  WG32_STAGE(0, 0, 0);
+  if (nsteps > 1) WG32_STAGE(1, da, db);
+  long long sa = da, sb = db;
+  for (int s = 0; s < nsteps - 1; ++s) {
+    // 4 glds/wave per tile: tile s landed once the s+1 tile's 4 are the
+    // only ones outstanding
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    WG32_MFMA(s & 1);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    if (s + 2 < nsteps) {
+      sa += da; sb += db;
+      WG32_STAGE(s & 1, sa, sb);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  WG32_MFMA((nsteps - 1) & 1);
+
+  const int erow = (lane >> 4) * 4;
+  const int ecol = lane & 15;
+  float* out = slab + (long long)blockIdx.y * N * K;
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int n = n0 + wr * 64 + i * 16 + erow;
+      const int k = k0 + wc * 64 + j * 16 + ecol;
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        out[(long long)(n + r) * K + k] = acc[i][j][r];
+    }
+#undef WG32_MFMA
+#undef WG32_STAGE
+}
+
 // fold: dW_bf16[i] = sum_s slab[s][i]
 __global__ void wgrad_fold_kernel(const float* __restrict__ slab,
                                   unsigned short* __restrict__ out,
@@ -201,8 +298,17 @@ void wgrad_tn_launch(const unsigned short* dY, const unsigned short* X,
                      int splits, hipStream_t stream) {
   const int chunk = M / splits;
   dim3 grid((N >> 7) * (K >> 7), splits), blk(256);
-  hipLaunchKernelGGL(wgrad_tn_kernel, grid, blk, 0, stream, dY, X, slab, M, N,
-                     K, chunk);
+  static int lds32 = -1;
+  if (lds32 < 0) {
+    const char* e = getenv("QN_WGRAD_LDS");
+    lds32 = (e && atoi(e) == 32) ? 1 : 0;
+  }
+  if (lds32 && (chunk & 31) == 0)
+    hipLaunchKernelGGL(wgrad_tn32_kernel, grid, blk, 0, stream, dY, X, slab,
+                       M, N, K, chunk);
+  else
+    hipLaunchKernelGGL(wgrad_tn_kernel, grid, blk, 0, stream, dY, X, slab, M,
+                       N, K, chunk);
   const long long nk = (long long)N * K;
   hipLaunchKernelGGL(wgrad_fold_kernel, dim3((nk / 4 + 255) / 256), dim3(256),
                      0, stream, slab, out, nk, splits);
