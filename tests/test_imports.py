@@ -147,3 +147,22 @@ def test_metrics_file_jsonl(tmp_path):
     assert len(lines) == 2
     assert lines[0]["epoch"] == 1 and "train_loss" in lines[0]
     assert lines[1]["lr"] is not None and lines[1]["lr"] > 0
+
+
+def test_info_report_runs():
+    from quintnet_amd.info import report
+
+    out = report()
+    assert "native extension" in out and "QN_ATTN_DKV_OCC" in out
+
+
+def test_all_examples_parse():
+    import ast
+    import glob
+    import os
+
+    root = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))), "examples")
+    files = glob.glob(os.path.join(root, "*.py"))
+    assert len(files) >= 13
+    for f in files:
+        ast.parse(open(f).read(), filename=f)
