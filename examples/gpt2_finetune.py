@@ -35,7 +35,9 @@ def build_loaders(cfg, pg):
         tok = GPT2Tokenizer.from_pretrained(cfg.get("tokenizer_path", "gpt2"))
         train = SummarizationDataset(path)
         return (
-            SummarizationDataLoader(train, tok, batch_size=bs, max_length=seq, shuffle=False),
+            SummarizationDataLoader(train, tok, batch_size=bs, max_length=seq,
+                                    shuffle=False, dp_rank=pg.dp_rank,
+                                    dp_size=pg.dp_size),
             None,
         )
     vocab = cfg.get("model_config", {}).get("vocab_size", 50257)

@@ -140,11 +140,24 @@ class SummarizationCollator:
         }
 
 
-def SummarizationDataLoader(dataset, tokenizer, batch_size=8, max_length=512, shuffle=True, **kw):
+def SummarizationDataLoader(dataset, tokenizer, batch_size=8, max_length=512,
+                            shuffle=True, dp_rank: int = 0, dp_size: int = 1,
+                            **kw):
+    """Collated loader for the summarization set; with dp_size > 1 each
+    DP replica reads its own shard (DistributedSampler — matches the
+    reference's full_3d.py:129 DP data sharding)."""
+    sampler = None
+    if dp_size > 1:
+        from torch.utils.data.distributed import DistributedSampler
+
+        sampler = DistributedSampler(dataset, num_replicas=dp_size,
+                                     rank=dp_rank, shuffle=shuffle, seed=42)
+        shuffle = False
     return DataLoader(
         dataset,
         batch_size=batch_size,
         shuffle=shuffle,
+        sampler=sampler,
         collate_fn=SummarizationCollator(tokenizer, max_length),
         **kw,
     )
