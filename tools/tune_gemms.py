@@ -35,6 +35,16 @@ def main():
         # lm_head shapes: logical 50257 (vocab_pad_to=0) and padded
         # 50304 (bench default) — dp meshes run M=32768, pp meshes 8192
         shapes += [(M, 50257, 768), (M, 50304, 768)]
+    def flush():
+        # write INCREMENTALLY so a timeout keeps everything tuned so far
+        res = torch.cuda.tunable.get_results()
+        with open(CSV, "w") as f:
+            for k, v in torch.cuda.tunable.get_validators():
+                f.write(f"Validator,{k},{v}\n")
+            for r in res:
+                f.write(",".join(str(x) for x in r) + "\n")
+        return len(res)
+
     done = set()
     for (M, N, K) in shapes:
         if (M, N, K) in done:
@@ -49,14 +59,9 @@ def main():
         g @ b                                    # dgrad
         g.t() @ a                                # wgrad
         torch.cuda.synchronize()
-        print(f"tuned {M}x{N}x{K}", flush=True)
-    res = torch.cuda.tunable.get_results()
-    with open(CSV, "w") as f:
-        for k, v in torch.cuda.tunable.get_validators():
-            f.write(f"Validator,{k},{v}\n")
-        for r in res:
-            f.write(",".join(str(x) for x in r) + "\n")
-    print(f"wrote {len(res)} entries -> {CSV}")
+        n = flush()
+        print(f"tuned {M}x{N}x{K} ({n} entries)", flush=True)
+    print(f"wrote {flush()} entries -> {CSV}")
 
 
 if __name__ == "__main__":
