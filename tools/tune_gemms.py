@@ -23,6 +23,10 @@ def main():
         torch.cuda.tunable.read_file(CSV)
     dev = torch.device("cuda")
     shapes = []
+    # NEW (likely-untuned) shapes first so a short timeout still banks
+    # them: the padded-vocab lm-head GEMMs at every mesh M
+    for M in (8192, 16384, 32768):
+        shapes += [(M, 50304, 768)]
     for M in (32768, 16384, 8192):          # micro 16 (pp1) and micro 8 (pp2)
         for tp in (1, 2):            # full and TP-halved
             E, I, V = 768, 3072, 50257
