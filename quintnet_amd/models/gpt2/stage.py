@@ -140,6 +140,10 @@ class GPT2Stage(nn.Module):
             getattr(self.config, "activation_checkpointing", False)
             and self.training
             and torch.is_grad_enabled()
+            # ZeRO-3 blocks already recompute inside their own gather
+            # region — a second wrapper would recompute the recompute
+            and not (len(self.blocks)
+                     and type(self.blocks[0]).__name__ == "ZeRO3Block")
         )
         pending = None
         for blk in self.blocks:

@@ -210,3 +210,30 @@ def test_zero3_checkpoint_save_refuses_sharded_model(tmp_path):
     apply_zero3(m, dp_group=None)
     with pytest.raises(ValueError, match="ZeRO-3"):
         save_sharded_checkpoint(m, str(tmp_path))
+
+
+def test_zero3_skips_double_activation_checkpointing():
+    """ZeRO-3 already recomputes inside the gather region; the stage's
+    own activation_checkpointing flag must not re-wrap it (grads still
+    correct, single recompute)."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.ops import causal_lm_loss
+    from quintnet_amd.parallel.zero3 import apply_zero3
+
+    torch.manual_seed(2)
+    cfg = GPT2Config(n_embd=32, n_layer=2, n_head=2, vocab_size=64,
+                     n_positions=32, dropout=0.0,
+                     activation_checkpointing=True)
+    m = GPT2Stage(cfg)
+    ref = copy.deepcopy(m)
+    apply_zero3(m, dp_group=None)
+    m.train(), ref.train()
+    ids = torch.randint(0, 64, (2, 16))
+    labels = torch.randint(0, 64, (2, 16))
+    l1 = causal_lm_loss(m(ids), labels)
+    l0 = causal_lm_loss(ref(ids), labels)
+    assert torch.allclose(l0, l1, atol=1e-6)
+    l1.backward()
+    for blk in m.blocks:
+        assert blk.shard.grad is not None
+        assert torch.isfinite(blk.shard.grad).all()
