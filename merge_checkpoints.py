@@ -17,7 +17,7 @@ from quintnet_amd.checkpoint import merge_checkpoints
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--input-dir", required=True, help="directory with *_pp{p}_tp{t}.pt shards")
-    ap.add_argument("--output", required=True, help="output .pt path (HF-format state dict)")
+    ap.add_argument("--output", required=True, help="output path: .pt (state dict + config) or .safetensors (transformers-loadable)")
     ap.add_argument("--prefix", default="final_model", help="shard filename prefix")
     args = ap.parse_args()
     out = merge_checkpoints(args.input_dir, args.output, prefix=args.prefix)

@@ -252,5 +252,15 @@ def merge_checkpoints(input_dir: str, output_path: str, prefix: str = "final_mod
     mc = config.get("model_config", {}) if isinstance(config, dict) else {}
     hf_state = convert_to_hf_format(merged, vocab_size=mc.get("vocab_size"))
     os.makedirs(os.path.dirname(os.path.abspath(output_path)), exist_ok=True)
-    torch.save({"model_state_dict": hf_state, "config": config}, output_path)
+    if output_path.endswith(".safetensors"):
+        # the on-disk format transformers loads directly: tensors only,
+        # HF key names minus the tied lm_head (HF re-ties from wte), and
+        # Linear weights already transposed back to Conv1D layout above
+        from safetensors.torch import save_file
+
+        st = {k: v.contiguous() for k, v in hf_state.items()
+              if k != "lm_head.weight"}
+        save_file(st, output_path, metadata={"format": "pt"})
+    else:
+        torch.save({"model_state_dict": hf_state, "config": config}, output_path)
     return output_path

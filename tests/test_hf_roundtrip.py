@@ -243,3 +243,41 @@ def test_hf_load_into_padded_config_and_export_slices_back():
         out = convert_to_hf_format(merged, vocab_size=100)
         assert out["transformer.wte.weight"].shape[0] == 100
         assert out["lm_head.weight"].shape[0] == 100
+
+
+def test_merge_to_safetensors_loads_in_transformers():
+    """merge_checkpoints --output x.safetensors produces a file
+    transformers loads directly (from_pretrained on the directory)."""
+    from transformers import GPT2Config as HFConfig
+    from transformers import GPT2LMHeadModel
+
+    from quintnet_amd.checkpoint.distributed_loading import load_gpt2_distributed
+    from quintnet_amd.checkpoint.merge import merge_checkpoints
+    from quintnet_amd.models import GPT2Stage
+
+    with tempfile.TemporaryDirectory() as d:
+        hf = _make_hf_checkpoint(d)
+        cfg = _our_config()
+        shard_dir = os.path.join(d, "shards")
+        os.makedirs(shard_dir)
+        sd = load_gpt2_distributed(d, cfg, pp_rank=0, pp_size=1, tp_rank=0, tp_size=1)
+        st = GPT2Stage.from_sharded_state_dict(cfg, sd, pp_rank=0, pp_size=1)
+        torch.save(
+            {"model_state_dict": st.state_dict(),
+             "parallelism_info": {"pp_rank": 0, "pp_size": 1, "tp_rank": 0,
+                                  "tp_size": 1, "dp_rank": 0}},
+            os.path.join(shard_dir, "final_model_pp0_tp0.pt"),
+        )
+        outdir = os.path.join(d, "export")
+        os.makedirs(outdir)
+        merge_checkpoints(shard_dir, os.path.join(outdir, "model.safetensors"),
+                          prefix="final_model")
+        # config.json so from_pretrained can instantiate
+        HFConfig(n_embd=64, n_layer=3, n_head=2, vocab_size=128,
+                 n_positions=64).save_pretrained(outdir)
+        reloaded = GPT2LMHeadModel.from_pretrained(outdir).eval()
+        ids = torch.randint(0, 128, (1, 16))
+        with torch.no_grad():
+            a = hf(ids).logits
+            b = reloaded(ids).logits
+        assert torch.allclose(a, b, atol=1e-5)
