@@ -7,10 +7,12 @@ from .stage import GPT2Stage
 from .interleaved import GPT2ForInterleaving, TiedLMHead
 from .decode import StaticKVDecoder
 from .speculative import speculative_generate
+from .beam import beam_search
 
 __all__ = [
     "StaticKVDecoder",
     "speculative_generate",
+    "beam_search",
     "GPT2Config",
     "GPT2Embedding",
     "GPT2Attention",

@@ -33,6 +33,20 @@ def build_app(stage, draft=None):
     def generate(req: dict = Body(...)):
         dev = next(stage.parameters()).device
         ids = torch.tensor([req["input_ids"]], dtype=torch.long, device=dev)
+        if int(req.get("num_beams", 0)) > 1:
+            from .models import beam_search
+
+            out = beam_search(
+                stage, ids,
+                max_new_tokens=int(req.get("max_new_tokens", 32)),
+                num_beams=int(req.get("num_beams")),
+                length_penalty=float(req.get("length_penalty", 1.0)),
+                eos_token_id=req.get("eos_token_id"),
+            )
+            return {
+                "output_ids": out[0].tolist(),
+                "new_ids": out[0, ids.shape[1]:].tolist(),
+            }
         if draft is not None and req.get("speculative"):
             from .models import speculative_generate
 

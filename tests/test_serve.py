@@ -56,3 +56,22 @@ def test_serve_speculative_endpoint():
         "input_ids": ids, "max_new_tokens": 8,
     }).json()["new_ids"]
     assert spec == plain  # greedy speculative is exact
+
+
+def test_serve_beam_endpoint():
+    import torch
+    from fastapi.testclient import TestClient
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.serve import build_app
+
+    torch.manual_seed(1)
+    stage = GPT2Stage(GPT2Config(n_embd=64, n_layer=2, n_head=2,
+                                 vocab_size=96, n_positions=64,
+                                 dropout=0.0)).eval()
+    client = TestClient(build_app(stage))
+    r = client.post("/generate", json={
+        "input_ids": [1, 2, 3], "max_new_tokens": 6, "num_beams": 3,
+    })
+    assert r.status_code == 200
+    assert len(r.json()["new_ids"]) == 6
