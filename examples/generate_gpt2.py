@@ -24,6 +24,9 @@ def main():
     ap.add_argument("--max-new", type=int, default=32)
     ap.add_argument("--temperature", type=float, default=0.8)
     ap.add_argument("--top-k", type=int, default=40)
+    ap.add_argument("--top-p", type=float, default=0.0)
+    ap.add_argument("--num-beams", type=int, default=0,
+                    help=">1: beam search (overrides sampling)")
     ap.add_argument("--tiny", action="store_true", help="tiny random model (CPU demo)")
     args = ap.parse_args()
 
@@ -42,8 +45,15 @@ def main():
     stage.eval()
 
     ids = torch.randint(0, cfg.vocab_size, (1, args.prompt_len), device=dev)
-    out = stage.generate(ids, max_new_tokens=args.max_new,
-                         temperature=args.temperature, top_k=args.top_k)
+    if args.num_beams > 1:
+        from quintnet_amd.models import beam_search
+
+        out = beam_search(stage, ids, max_new_tokens=args.max_new,
+                          num_beams=args.num_beams)
+    else:
+        out = stage.generate(ids, max_new_tokens=args.max_new,
+                             temperature=args.temperature, top_k=args.top_k,
+                             top_p=args.top_p)
     print("prompt :", ids[0].tolist())
     print("output :", out[0, args.prompt_len:].tolist())
 
