@@ -183,6 +183,7 @@ class GPT2Stage(nn.Module):
         temperature: float = 0.0,
         top_k: int = 0,
         top_p: float = 0.0,
+        repetition_penalty: float = 1.0,
         eos_token_id: int = None,
         cache_dtype: str = None,
     ) -> torch.Tensor:
@@ -218,6 +219,13 @@ class GPT2Stage(nn.Module):
                 w = self.embedding.wte.weight if self.lm_head is None else self.lm_head
                 logits = fused_linear(x, w, None, None, prefer_library=True)[:, -1]
                 logits = mask_pad_logits(logits, self.config)
+                if repetition_penalty and repetition_penalty != 1.0:
+                    # CTRL-style: push down tokens already in the output
+                    logits = logits.float()
+                    seen = logits.gather(1, out)
+                    seen = torch.where(seen > 0, seen / repetition_penalty,
+                                       seen * repetition_penalty)
+                    logits = logits.scatter(1, out, seen)
                 if temperature and temperature > 0:
                     logits = logits.float() / temperature
                     if top_k and top_k > 0:

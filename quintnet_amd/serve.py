@@ -70,6 +70,7 @@ def build_app(stage, draft=None):
             temperature=float(req.get("temperature", 0.0)),
             top_k=int(req.get("top_k", 0)),
             top_p=float(req.get("top_p", 0.0)),
+            repetition_penalty=float(req.get("repetition_penalty", 1.0)),
             eos_token_id=req.get("eos_token_id"),
             cache_dtype=req.get("cache_dtype"),
         )

@@ -81,3 +81,25 @@ def test_generate_top_p_masks_tail():
     # p=0.05 typically keeps only the top token; allow equality check on
     # the first generated token at least
     assert int(out[0, 8]) == int(want[0, 8])
+
+
+def test_repetition_penalty_reduces_repeats():
+    """A strong penalty must yield fewer immediate repeats than greedy
+    (random-init GPT-2 collapses to one token greedily)."""
+    import torch
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    torch.manual_seed(4)
+    m = GPT2Stage(GPT2Config(n_embd=32, n_layer=1, n_head=2, vocab_size=64,
+                             n_positions=64, dropout=0.0)).eval()
+    ids = torch.randint(0, 64, (1, 6))
+
+    def repeats(seq):
+        new = seq[0, 6:]
+        return int((new[1:] == new[:-1]).sum())
+
+    greedy = m.generate(ids, max_new_tokens=12, temperature=0.0)
+    pen = m.generate(ids, max_new_tokens=12, temperature=1e-4,
+                     repetition_penalty=5.0)
+    assert repeats(pen) < max(repeats(greedy), 1)
