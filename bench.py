@@ -436,6 +436,18 @@ def main():
             _write_tunableop_csv(_TUNABLE_CSV)
         except Exception:  # noqa: BLE001 — persistence is best-effort
             pass
+    prof = None
+    if os.environ.get("QN_TORCH_PROFILE") == "1" and use_cuda:
+        # overlap-evidence timeline (chrome trace): NOT for timing runs —
+        # profiling overhead lands inside the measured region.  Open the
+        # trace at ui.perfetto.dev; comm kernels on their own stream
+        # overlapping compute = the bucket/1F1B overlap proof.
+        from torch.profiler import ProfilerActivity, profile
+
+        prof = profile(
+            activities=[ProfilerActivity.CPU, ProfilerActivity.CUDA],
+        )
+        prof.__enter__()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         step()
@@ -443,6 +455,12 @@ def main():
             wd.beat()
     sync()
     elapsed = time.perf_counter() - t0
+    if prof is not None:
+        prof.__exit__(None, None, None)
+        os.makedirs("gpurun_out", exist_ok=True)
+        tr = f"gpurun_out/trace_rank{os.environ.get('RANK', '0')}.json"
+        prof.export_chrome_trace(tr)
+        print(f"# torch-profiler trace -> {tr}", flush=True)
     if wd:
         wd.stop()
 
