@@ -120,3 +120,31 @@ def test_graph_decode_gpu():
     have = dec.generate(ids, max_new_tokens=12)
     assert dec._graph is not None
     assert torch.equal(have, want)
+
+
+def test_beam_and_speculative_on_gpu():
+    """Serving extensions on real bf16 kernels: beam=1 == greedy, and
+    greedy speculative == greedy (exactness survives the fused
+    attention + library GEMM path)."""
+    import torch
+
+    from quintnet_amd.models import GPT2Config, GPT2Stage, beam_search
+    from quintnet_amd.models.gpt2.speculative import speculative_generate
+
+    torch.manual_seed(11)
+    base = dict(vocab_size=128, n_positions=96, dropout=0.0)
+    target = GPT2Stage(GPT2Config(n_embd=128, n_layer=2, n_head=2, **base),
+                       device="cuda", dtype=torch.bfloat16).eval()
+    draft = GPT2Stage(GPT2Config(n_embd=64, n_layer=1, n_head=2, **base),
+                      device="cuda", dtype=torch.bfloat16).eval()
+    ids = torch.randint(0, 128, (1, 8), device="cuda")
+    want = target.generate(ids, max_new_tokens=8, temperature=0.0)
+    assert torch.equal(
+        beam_search(target, ids, max_new_tokens=8, num_beams=1), want
+    )
+    assert torch.equal(
+        speculative_generate(target, draft, ids, max_new_tokens=8, draft_k=3),
+        want,
+    )
+    wide = beam_search(target, ids, max_new_tokens=8, num_beams=4)
+    assert wide.shape == want.shape and int(wide.max()) < 128
