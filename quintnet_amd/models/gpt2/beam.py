@@ -62,13 +62,18 @@ def beam_search(
                   for _ in stage.blocks]
         logits = _step_logits(stage, caches, input_ids, 0)  # [1, V]
         logprobs = torch.log_softmax(logits.float(), dim=-1)[0]
-        scores, toks = logprobs.topk(num_beams)  # [beams]
-        # expand the prompt cache to the beam batch
+        # the active set starts at min(beams, |finite vocab|) — only V
+        # distinct one-token prefixes exist — and grows back toward
+        # num_beams as candidates multiply (found by the exhaustive-
+        # optimum property test: topk(num_beams) with num_beams > V
+        # both crashed and silently capped the width)
+        k0 = min(num_beams, int(torch.isfinite(logprobs).sum()))
+        scores, toks = logprobs.topk(k0)  # [k0]
         for c in caches:
-            c["k"] = c["k"].expand(num_beams, -1, -1, -1).contiguous()
-            c["v"] = c["v"].expand(num_beams, -1, -1, -1).contiguous()
+            c["k"] = c["k"].expand(k0, -1, -1, -1).contiguous()
+            c["v"] = c["v"].expand(k0, -1, -1, -1).contiguous()
         seqs = torch.cat(
-            [input_ids.expand(num_beams, -1), toks.unsqueeze(1)], dim=1
+            [input_ids.expand(k0, -1), toks.unsqueeze(1)], dim=1
         )
         finished: List[tuple] = []  # (score/penalty, tensor)
 
@@ -76,7 +81,7 @@ def beam_search(
             return float(s) / max(length, 1) ** length_penalty
 
         if eos_token_id is not None:
-            for b in range(num_beams):
+            for b in range(k0):
                 if int(toks[b]) == eos_token_id:
                     finished.append((fin_score(scores[b], 1), seqs[b]))
                     scores[b] = float("-inf")
@@ -89,7 +94,8 @@ def beam_search(
             V = lp.shape[-1]
             total = scores.unsqueeze(1) + lp  # [-inf rows drop out]
             # 2*beams candidates so eos hits don't starve the active set
-            cand_scores, flat = total.reshape(-1).topk(2 * num_beams)
+            k_cand = min(2 * num_beams, total.numel())
+            cand_scores, flat = total.reshape(-1).topk(k_cand)
             beam_idx = flat // V
             tok_idx = flat % V
             new_scores, new_beams, new_toks = [], [], []
@@ -128,7 +134,7 @@ def beam_search(
                 if len(finished) >= num_beams and best_fin >= best_active:
                     break
 
-        for b in range(num_beams):
+        for b in range(scores.shape[0]):
             if not bool(torch.isinf(scores[b])):
                 finished.append(
                     (fin_score(scores[b], seqs.shape[1] - T0), seqs[b])

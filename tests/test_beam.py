@@ -68,3 +68,30 @@ def test_beam_padded_vocab_never_emits_pad():
     ids = torch.randint(0, 100, (1, 6))
     out = beam_search(m, ids, max_new_tokens=8, num_beams=4)
     assert int(out.max()) < 100
+
+
+def test_wide_beam_equals_exhaustive_optimum():
+    """With num_beams >= V^(L-1) nothing can be pruned, so beam search
+    must return the EXACT argmax over all V^L continuations."""
+    import itertools
+
+    torch.manual_seed(31)
+    V, L = 5, 3
+    m = GPT2Stage(GPT2Config(
+        n_embd=32, n_layer=1, n_head=2, vocab_size=V, n_positions=32,
+        dropout=0.0,
+    )).eval()
+    ids = torch.randint(0, V, (1, 4))
+
+    def seq_score(cont):
+        seq = torch.cat([ids, torch.tensor([list(cont)])], dim=1)
+        with torch.no_grad():
+            lp = torch.log_softmax(m(seq)[0, :-1].float(), dim=-1)
+        s = 0.0
+        for i, t in enumerate(cont):
+            s += float(lp[ids.shape[1] - 1 + i, t])
+        return s
+
+    best = max(itertools.product(range(V), repeat=L), key=seq_score)
+    out = beam_search(m, ids, max_new_tokens=L, num_beams=V * V)
+    assert tuple(out[0, 4:].tolist()) == best, (out, best)
