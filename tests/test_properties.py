@@ -47,7 +47,7 @@ def test_shift_labels_alignment(b, t, seed):
 
 
 @_FAST
-@given(world=st.sampled_from([1, 2, 4, 8]), tl=st.integers(1, 16))
+@given(world=st.sampled_from([1, 2, 3, 4, 5, 6, 8]), tl=st.integers(1, 16))
 def test_zigzag_chunk_map_is_partition(world, tl):
     """Every rank gets chunks (r, 2w-1-r); together they cover 0..2w-1
     exactly once, and zigzag positions are a permutation of 0..T-1."""

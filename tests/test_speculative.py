@@ -22,7 +22,7 @@ def _models(vocab=96, pad=0):
     return target, draft
 
 
-@pytest.mark.parametrize("draft_k", [1, 3, 4])
+@pytest.mark.parametrize("draft_k", [1, 3, 4, 8])
 def test_speculative_matches_greedy(draft_k):
     target, draft = _models()
     ids = torch.randint(0, 96, (1, 10))
