@@ -277,3 +277,34 @@ def test_early_stopping_breaks_epoch_loop(tmp_path):
     tr.fit()
     n_epochs_run = len((tmp_path / "m.jsonl").read_text().splitlines())
     assert n_epochs_run <= 4, n_epochs_run  # 1 best + 2 flat, not 50
+
+
+def test_resume_continues_lr_schedule(tmp_path):
+    """After auto-resume the cosine schedule restarts from the ZeRO
+    optimizer's checkpointed step counter, not from zero."""
+    import torch
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    cfg = GPT2Config(vocab_size=64, n_positions=32, n_embd=16, n_layer=1,
+                     n_head=2, dropout=0.0)
+    ds = SyntheticCLM(n=4, seq_len=16, vocab_size=64, seed=0)
+    base = {"num_epochs": 1, "grad_acc_steps": 1, "zero1": True,
+            "learning_rate": 1e-3, "lr_schedule": "cosine",
+            "total_steps": 10, "checkpoint_dir": str(tmp_path)}
+    torch.manual_seed(0)
+    tr1 = GPT2Trainer(GPT2Stage(cfg), DataLoader(ds, batch_size=2), None,
+                      dict(base), None)
+    tr1.fit()  # 2 optimizer steps, then shards saved
+    assert tr1.optimizer.step_count == 2
+
+    torch.manual_seed(0)
+    tr2 = GPT2Trainer(GPT2Stage(cfg), DataLoader(ds, batch_size=2), None,
+                      dict(base, resume_from="auto"), None)
+    assert tr2.optimizer.step_count == 2  # state restored
+    sched = tr2._build_lr_schedule()
+    assert sched._step == 2  # curve continues, not restarted
+    assert sched.lr_at(2) < sched.lr_at(0)
