@@ -103,3 +103,36 @@ def test_repetition_penalty_reduces_repeats():
     pen = m.generate(ids, max_new_tokens=12, temperature=1e-4,
                      repetition_penalty=5.0)
     assert repeats(pen) < max(repeats(greedy), 1)
+
+
+def _tp2_generate_rank_consistent(rank, world):
+    """TP-sharded generate: every TP rank must emit the SAME tokens,
+    including under temperature sampling (rank-consistent RNG) and the
+    new top-p / repetition-penalty transforms."""
+    import torch
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    pg = init_process_groups("cpu", [world], ["tp"])
+    torch.manual_seed(14)
+    cfg = GPT2Config(vocab_size=96, n_positions=64, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=pg.get_group("tp"))
+    for p in stage.parameters():
+        dist.broadcast(p.data, src=0)
+    stage.eval()
+    ids = torch.randint(0, 96, (1, 6))
+    dist.broadcast(ids, src=0)
+    out = stage.generate(ids, max_new_tokens=10, temperature=0.8,
+                         top_k=20, top_p=0.9, repetition_penalty=1.3)
+    ref = out.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.equal(out, ref), "TP ranks diverged during sampling"
+
+
+def test_tp2_generate_rank_consistent():
+    from conftest import run_distributed
+
+    run_distributed(_tp2_generate_rank_consistent, 2)
