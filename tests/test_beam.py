@@ -95,3 +95,30 @@ def test_wide_beam_equals_exhaustive_optimum():
     best = max(itertools.product(range(V), repeat=L), key=seq_score)
     out = beam_search(m, ids, max_new_tokens=L, num_beams=V * V)
     assert tuple(out[0, 4:].tolist()) == best, (out, best)
+
+
+def _tp2_beam(rank, world):
+    import torch.distributed as dist
+
+    from quintnet_amd import init_process_groups
+
+    pg = init_process_groups("cpu", [world], ["tp"])
+    torch.manual_seed(15)
+    cfg = GPT2Config(vocab_size=96, n_positions=64, n_embd=32, n_layer=2,
+                     n_head=2, dropout=0.0)
+    stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=pg.get_group("tp"))
+    for p in stage.parameters():
+        dist.broadcast(p.data, src=0)
+    stage.eval()
+    ids = torch.randint(0, 96, (1, 6))
+    dist.broadcast(ids, src=0)
+    out = beam_search(stage, ids, max_new_tokens=8, num_beams=3)
+    ref = out.clone()
+    dist.broadcast(ref, src=0)
+    assert torch.equal(out, ref), "TP ranks diverged in beam search"
+
+
+def test_tp2_beam_rank_consistent():
+    from conftest import run_distributed
+
+    run_distributed(_tp2_beam, 2)
