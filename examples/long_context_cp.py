@@ -3,8 +3,10 @@ across ranks end to end (attention sees the full context through the
 gathered-KV kernels), gradients are exact.
 
     torchrun --nproc_per_node=2 --master-addr 127.0.0.1 -m examples.long_context_cp
+    # flavors: --ring (O(T/cp) KV memory) or --zigzag (balanced causal work)
 """
 
+import argparse
 import sys, os
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
@@ -19,6 +21,12 @@ from quintnet_amd.utils.data import SyntheticCLM
 
 
 def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--ring", action="store_true",
+                    help="ring KV exchange (peak KV memory O(T/cp))")
+    ap.add_argument("--zigzag", action="store_true",
+                    help="zigzag load-balanced ring (equal causal work/rank)")
+    args = ap.parse_args()
     dev_type = "cuda" if torch.cuda.is_available() else "cpu"
     world = int(os.environ.get("WORLD_SIZE", 1))
     pg = init_process_groups(dev_type, [world], ["cp"])
@@ -26,7 +34,9 @@ def main():
     seq = 4096 if dev_type == "cuda" else 256
     torch.manual_seed(0)
     cfg = GPT2Config(vocab_size=512, n_positions=seq, n_embd=64, n_layer=2,
-                     n_head=2, dropout=0.0)
+                     n_head=2, dropout=0.0,
+                     cp_ring=args.ring or args.zigzag,
+                     cp_zigzag=args.zigzag)
     dtype = torch.bfloat16 if dev_type == "cuda" else torch.float32
     stage = GPT2Stage(cfg, pp_rank=0, pp_size=1, tp_group=None,
                       cp_group=pg.get_group("cp") if world > 1 else None,
