@@ -87,30 +87,44 @@ def context_parallel_attention(
 # path (streaming log-sum-exp merge); the fused flash kernel can be slotted
 # per chunk later.
 # ---------------------------------------------------------------------------
+class _SwapHandle:
+    """In-flight ring exchange: keeps the SEND buffers referenced until
+    wait() (``.contiguous()`` may have created temporaries; some
+    backends do not pin send storage for async P2P)."""
+
+    __slots__ = ("reqs", "recvs", "_sends")
+
+    def __init__(self, reqs, recvs, sends):
+        self.reqs, self.recvs, self._sends = reqs, recvs, sends
+
+    def wait(self):
+        for r in self.reqs:
+            r.wait()
+        return self.recvs
+
+
 def _ring_swap_begin(tensors, rank, world, group):
     """Issue send-to-rank+1 / recv-from-rank-1 without waiting (overlap
     with compute: sends only READ the live buffers)."""
     nxt = (rank + 1) % world
     prv = (rank - 1 + world) % world
+    sends = [t.contiguous() for t in tensors]
     ops = []
     recvs = []
-    for t in tensors:
-        ops.append(dist.P2POp(dist.isend, t.contiguous(), peer=nxt, group=group))
+    for t in sends:
+        ops.append(dist.P2POp(dist.isend, t, peer=nxt, group=group))
     for t in tensors:
         buf = torch.empty(t.shape, dtype=t.dtype, device=t.device)
         recvs.append(buf)
         ops.append(dist.P2POp(dist.irecv, buf, peer=prv, group=group))
-    return dist.batch_isend_irecv(ops), recvs
+    return _SwapHandle(dist.batch_isend_irecv(ops), recvs, sends)
 
 
 def _ring_swap(tensors, rank, world, group):
     """Send tensors to rank+1, receive the same shapes from rank-1."""
     if world == 1:
         return tensors
-    reqs, recvs = _ring_swap_begin(tensors, rank, world, group)
-    for r in reqs:
-        r.wait()
-    return recvs
+    return _ring_swap_begin(tensors, rank, world, group).wait()
 
 
 def _partial(q32, k32, scale, causal_mode, Tl):
@@ -186,18 +200,16 @@ class _RingAttention(torch.autograd.Function):
             lse_run = torch.full((B, H, Tl, 1), float("-inf"), device=q.device)
             for s_hop in range(world):
                 j = (rank - s_hop) % world
-                reqs = None
+                hand = None
                 if s_hop < world - 1:  # overlap the hop exchange w/ compute
-                    reqs, bufs = _ring_swap_begin([kc, vc], rank, world, group)
+                    hand = _ring_swap_begin([kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     out_run, lse_run = _ring_flash_fwd_hop(
                         qc, kc, vc, scale, 1 if (causal and j == rank) else 0,
                         out_run, lse_run,
                     )
-                if reqs is not None:
-                    for r_ in reqs:
-                        r_.wait()
-                    kc, vc = bufs
+                if hand is not None:
+                    kc, vc = hand.wait()
             out = out_run.to(q.dtype)
             lse = lse_run.reshape(B * H, Tl)  # base-2, the kernels' format
         else:
@@ -207,9 +219,9 @@ class _RingAttention(torch.autograd.Function):
             acc = torch.zeros(B, H, Tl, D, device=q.device)
             for s_hop in range(world):
                 j = (rank - s_hop) % world
-                reqs = None
+                hand = None
                 if s_hop < world - 1:  # overlap the hop exchange w/ compute
-                    reqs, bufs = _ring_swap_begin([kc, vc], rank, world, group)
+                    hand = _ring_swap_begin([kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     sc = _partial(q32, kc.float(), scale, 1 if (causal and j == rank) else 0, Tl)
                     # every processed chunk has >=1 unmasked key per row (the
@@ -223,10 +235,8 @@ class _RingAttention(torch.autograd.Function):
                     acc = acc * alpha + torch.matmul(p, vc.float())
                     l = l * alpha + p.sum(dim=-1, keepdim=True)
                     m = m_new
-                if reqs is not None:
-                    for r_ in reqs:
-                        r_.wait()
-                    kc, vc = bufs
+                if hand is not None:
+                    kc, vc = hand.wait()
             out = (acc / l.clamp(min=1e-30)).to(q.dtype)
             lse = (m + torch.log(l.clamp(min=1e-30))).squeeze(-1)  # natural log
         ctx.save_for_backward(q, k, v, out, lse)
@@ -262,10 +272,9 @@ class _RingAttention(torch.autograd.Function):
                 # exchange now and overlap; the grad accumulators are
                 # WRITTEN during compute, so they travel afterwards
                 # (split swap; P2P pair order kv-then-grads everywhere)
-                reqs_kv = None
+                hand_kv = None
                 if world > 1:
-                    reqs_kv, bufs_kv = _ring_swap_begin(
-                        [kc, vc], rank, world, group)
+                    hand_kv = _ring_swap_begin([kc, vc], rank, world, group)
                 if not causal or j <= rank:
                     _ring_flash_bwd_hop(
                         qc, kc, vc, outc, doutc, lse_flat, scale,
@@ -274,15 +283,11 @@ class _RingAttention(torch.autograd.Function):
                     dq32 += dqp.float()
                     dk_acc += dkp.float()
                     dv_acc += dvp.float()
-                if reqs_kv is not None:
-                    reqs_g, bufs_g = _ring_swap_begin(
+                if hand_kv is not None:
+                    hand_g = _ring_swap_begin(
                         [dk_acc, dv_acc], rank, world, group)
-                    for r_ in reqs_kv:
-                        r_.wait()
-                    kc, vc = bufs_kv
-                    for r_ in reqs_g:
-                        r_.wait()
-                    dk_acc, dv_acc = bufs_g
+                    kc, vc = hand_kv.wait()
+                    dk_acc, dv_acc = hand_g.wait()
             return (
                 dq32.to(q.dtype),
                 dk_acc.to(k.dtype),
@@ -398,10 +403,9 @@ class _ZigzagRingAttention(torch.autograd.Function):
         for s_hop in range(world):
             j = (rank - s_hop) % world
             ck = _zz_ids(j, world)
-            reqs_kv = None
+            hand_kv = None
             if s_hop < world - 1:  # overlap the hop exchange w/ compute
-                reqs_kv, bufs_kv = _ring_swap_begin(
-                    [kc, vc], rank, world, group)
+                hand_kv = _ring_swap_begin([kc, vc], rank, world, group)
             for qi in range(2):
                 for ki in range(2):
                     if causal and ck[ki] > cq[qi]:
@@ -426,10 +430,8 @@ class _ZigzagRingAttention(torch.autograd.Function):
                         l[:, :, sl] = l[:, :, sl] * alpha + p.sum(
                             dim=-1, keepdim=True)
                         m[:, :, sl] = m_new
-            if reqs_kv is not None:
-                for r_ in reqs_kv:
-                    r_.wait()
-                kc, vc = bufs_kv
+            if hand_kv is not None:
+                kc, vc = hand_kv.wait()
         if flash:
             out = torch.cat(out_run, dim=2).to(q.dtype)
             lse = torch.cat(lse_run, dim=2).reshape(B * H, Tl)  # base-2
@@ -466,8 +468,7 @@ class _ZigzagRingAttention(torch.autograd.Function):
                 j = (rank - s_hop) % world
                 ck = _zz_ids(j, world)
                 if world > 1:  # pre-rotate kv under the block compute
-                    reqs_kv, bufs_kv = _ring_swap_begin(
-                        [kc, vc], rank, world, group)
+                    hand_kv = _ring_swap_begin([kc, vc], rank, world, group)
                 for qi in range(2):
                     for ki in range(2):
                         if causal and ck[ki] > cq[qi]:
@@ -488,14 +489,10 @@ class _ZigzagRingAttention(torch.autograd.Function):
                         dk_acc[:, :, ksl] += dkp.float()
                         dv_acc[:, :, ksl] += dvp.float()
                 if world > 1:
-                    reqs_g, bufs_g = _ring_swap_begin(
+                    hand_g = _ring_swap_begin(
                         [dk_acc, dv_acc], rank, world, group)
-                    for r_ in reqs_kv:
-                        r_.wait()
-                    kc, vc = bufs_kv
-                    for r_ in reqs_g:
-                        r_.wait()
-                    dk_acc, dv_acc = bufs_g
+                    kc, vc = hand_kv.wait()
+                    dk_acc, dv_acc = hand_g.wait()
             return (dq32.to(q.dtype), dk_acc.to(k.dtype), dv_acc.to(v.dtype),
                     None, None)
         q32, dout32, out32 = q.float(), dout.float(), out.float()
