@@ -7,7 +7,7 @@ kernels against.
 """
 
 from ._backend import ext, has_ext, use_native, force_eager
-from .linear import linear, LinearFunction
+from .linear import linear, LinearFunction, defer_wgrads, flush_deferred_wgrads
 from .layernorm import layer_norm, layer_norm_residual, FusedLayerNorm, LayerNormFunction
 from .attention import (
     attention,
@@ -35,6 +35,8 @@ __all__ = [
     "force_eager",
     "linear",
     "LinearFunction",
+    "defer_wgrads",
+    "flush_deferred_wgrads",
     "layer_norm",
     "layer_norm_residual",
     "FusedLayerNorm",

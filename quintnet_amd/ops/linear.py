@@ -22,7 +22,7 @@ import torch
 
 from . import _backend
 
-__all__ = ["linear", "LinearFunction"]
+__all__ = ["linear", "LinearFunction", "defer_wgrads", "flush_deferred_wgrads"]
 
 _ACT_NONE = 0
 _ACT_GELU = 1
@@ -59,6 +59,78 @@ _WGRAD_MODE = os.environ.get("QN_WGRAD", "auto")
 # to the eager GELU; revisit when the library gains the epilogue.
 _GELU_EPI = os.environ.get("QN_GELU_EPI", "0") == "1"
 _gelu_epi_broken = False
+
+# ---------------------------------------------------------------------------
+# Deferred weight gradients (zero-bubble building block): with the mode
+# on, LinearFunction.backward computes grad_x (the critical path that
+# feeds the upstream pipeline stage) immediately but QUEUES the weight
+# GEMM; flush_deferred_wgrads() runs the queued dW GEMMs later — e.g.
+# while a pipeline stage would otherwise idle in a bubble or block on a
+# P2P wait — and ACCUMULATES into param.grad exactly as autograd would
+# have (sum over micro-batches; DDP bucket views receive the adds, so
+# finalize_gradients() reduces complete gradients).  Requires hook-fired
+# bucket reduction to be OFF during backward (the hooks key on autograd
+# accumulation, which deferral bypasses); DataParallel.finalize_gradients
+# launches every bucket itself, which is the ZB reduction path.
+_DEFER_WGRADS = False
+_DEFERRED: list = []
+
+
+class _DeferScope:
+    def __enter__(self):
+        defer_wgrads(True)
+        return self
+
+    def __exit__(self, *a):
+        defer_wgrads(False)
+        return False
+
+
+def defer_wgrads(on: bool = True):
+    """Toggle deferred-dW mode.  Usable as a context manager:
+    ``with defer_wgrads.scope(): loss.backward()`` then
+    ``flush_deferred_wgrads()`` before any grad consumer."""
+    global _DEFER_WGRADS
+    _DEFER_WGRADS = bool(on)
+
+
+defer_wgrads.scope = _DeferScope
+
+
+def flush_deferred_wgrads() -> int:
+    """Run every queued weight-gradient GEMM, accumulating into
+    ``param.grad`` (creating it when absent).  Returns the number of
+    flushed entries.  MUST run before grad reduction / clipping /
+    optimizer step."""
+    global _DEFERRED
+    queue, _DEFERRED = _DEFERRED, []
+    with torch.no_grad():  # the queued activations carry autograd history
+        for g, x2d, weight in queue:
+            dw = _wgrad(g, x2d)
+            if weight.grad is None:
+                weight.grad = dw
+            else:
+                weight.grad += dw
+    return len(queue)
+
+
+def _wgrad(g: torch.Tensor, x2d: torch.Tensor) -> torch.Tensor:
+    """The same per-shape dW dispatch backward() uses."""
+    if (
+        _backend.use_native(g)
+        and _backend.has_ext()
+        and g.dtype == torch.bfloat16
+        and g.shape[1] % 128 == 0
+        and x2d.shape[1] % 128 == 0
+        and g.shape[0] % 64 == 0
+        and (
+            (g.shape[1] // 128) * (x2d.shape[1] // 128) < 128
+            or _WGRAD_MODE == "custom"
+        )
+        and _WGRAD_MODE != "library"
+    ):
+        return _backend.ext().wgrad_tn(g.contiguous(), x2d.contiguous())
+    return g.t() @ x2d
 
 
 def _custom_wins_shape(m: int, n: int, k: int) -> bool:
@@ -250,24 +322,16 @@ class LinearFunction(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             grad_x = (g @ weight).reshape(ctx.x_shape)
         if ctx.needs_input_grad[1]:
-            # hand-written split-K TN kernel: hipBLASLt runs these deep-
-            # contraction small-output shapes at 208-513 TF (profiles r04)
-            if (
-                _backend.use_native(g)
-                and _backend.has_ext()
-                and g.dtype == torch.bfloat16
-                and g.shape[1] % 128 == 0
-                and x2d.shape[1] % 128 == 0
-                and g.shape[0] % 64 == 0
-                and (
-                    (g.shape[1] // 128) * (x2d.shape[1] // 128) < 128
-                    or _WGRAD_MODE == "custom"  # force: A/B the 144-tile shapes
-                )
-                and _WGRAD_MODE != "library"
-            ):
-                grad_w = _backend.ext().wgrad_tn(g.contiguous(), x2d.contiguous())
+            if _DEFER_WGRADS:
+                # zero-bubble mode: queue the dW GEMM; autograd gets None
+                # (no accumulation) and flush_deferred_wgrads() adds the
+                # contribution into weight.grad later
+                _DEFERRED.append((g, x2d, weight))
             else:
-                grad_w = g.t() @ x2d
+                # hand-written split-K TN kernel for the shapes it wins
+                # (hipBLASLt runs these deep-contraction small-output
+                # shapes at 208-513 TF, profiles r04)
+                grad_w = _wgrad(g, x2d)
         if ctx.has_bias and ctx.needs_input_grad[2]:
             if _backend.use_native(g) and _backend.has_ext():
                 grad_b = _backend.ext().colsum(g)

@@ -42,6 +42,9 @@ class PipelineSchedule(abc.ABC):
     ):
         self.trainer = trainer
         self.task_type = task_type
+        # zero-bubble dW deferral (opt-in via trainer config; the
+        # trainer object carries the flag from GPT2Trainer/Trainer)
+        self._defer_wgrads = bool(getattr(trainer, "defer_wgrads", False))
 
     # -- conveniences over the owning PipelineTrainer -------------------
     @property
@@ -198,6 +201,18 @@ class PipelineSchedule(abc.ABC):
         if pt:
             pt.start("backward")
         try:
+            if self._defer_wgrads:
+                # zero-bubble mode: the backward computes only dX (the
+                # critical path feeding the upstream stage); the weight
+                # GEMMs queue and run in flush_deferred_wgrads() at the
+                # optimizer step — on GPU their kernels overlap whatever
+                # the stage would otherwise idle on (ops/linear.py)
+                from ...ops import defer_wgrads
+
+                with defer_wgrads.scope():
+                    return self.model.backward(
+                        input_tensor, output_tensor, output_grad
+                    )
             return self.model.backward(input_tensor, output_tensor, output_grad)
         finally:
             if pt:
@@ -208,6 +223,10 @@ class PipelineSchedule(abc.ABC):
         pt = self._pt
         if pt:
             pt.start("optimizer")
+        if self._defer_wgrads:
+            from ...ops import flush_deferred_wgrads
+
+            flush_deferred_wgrads()  # BEFORE reduction/tied-sync/clip
         self._finalize_grads()
         if t.max_grad_norm is not None and t.max_grad_norm > 0:
             if hasattr(t.optimizer, "clip_grad_norm_"):

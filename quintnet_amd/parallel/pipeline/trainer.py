@@ -32,6 +32,7 @@ class PipelineTrainer:
         pp_bwd_group=None,
         cp_group=None,
         tp_group=None,
+        defer_wgrads: bool = False,
     ):
         self.model = model
         self.pp_fwd_group = pp_fwd_group
@@ -47,6 +48,7 @@ class PipelineTrainer:
         self.max_grad_norm = max_grad_norm
         self.task_type = task_type
         self.lr_scheduler = None  # set by the owning Trainer at fit()
+        self.defer_wgrads = bool(defer_wgrads)
         self.schedule = get_schedule(schedule, self, task_type)
 
     def _lr_step(self) -> None:

@@ -85,6 +85,7 @@ class Trainer:
             model=self.model,
             optimizer=self.optimizer,
             criterion=self.criterion,
+            defer_wgrads=bool(self.config.get("defer_wgrads", False)),
             pp_rank=self.pg.pp_rank,
             pp_size=self.pg.pp_size,
             pp_group=self.pg.get_group("pp"),
