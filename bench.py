@@ -307,6 +307,8 @@ def main():
             schedule=os.environ.get("QN_SCHEDULE", "1f1b"),
             task_type=task, max_grad_norm=1.0,
             pp_fwd_group=groups.get("pp_fwd"), pp_bwd_group=groups.get("pp_bwd"),
+            # QN_DEFER_WGRADS=1: zero-bubble dW deferral (A/B on pp meshes)
+            defer_wgrads=os.environ.get("QN_DEFER_WGRADS") == "1",
         )
         shapes = (micro_b, pipe_shape_seq if task == "clm" else seq, hidden)
 
