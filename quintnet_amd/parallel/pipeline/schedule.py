@@ -357,6 +357,12 @@ class OneFOneBSchedule(PipelineSchedule):
             if self.is_last:
                 out_grad = None
             else:
+                if self._defer_wgrads:
+                    from ...ops import flush_deferred_wgrads
+
+                    # ZB: launch the queued dW GEMMs before blocking on
+                    # the bwd-grad recv — their kernels fill the wait
+                    flush_deferred_wgrads()
                 out_grad = bidirectional_pipeline_communicate(
                     "send_fwd_recv_bwd", self.pp_rank, self.pp_size, self.group_ranks,
                     send_tensor=out, recv_shapes=tensor_shapes, dtype=dtype,
@@ -394,6 +400,10 @@ class OneFOneBSchedule(PipelineSchedule):
 
         # -- cooldown backwards -----------------------------------------
         for _ in range(warmup):
+            if self._defer_wgrads:
+                from ...ops import flush_deferred_wgrads
+
+                flush_deferred_wgrads()  # ZB: dW kernels under the recv wait
             grad = pipeline_communicate(
                 "recv_backward", self.pp_rank, self.pp_size, self.group_ranks,
                 shapes=tensor_shapes, dtype=dtype, device=device, group=self.pp_group,
