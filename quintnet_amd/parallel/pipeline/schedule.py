@@ -294,6 +294,10 @@ class AllFwdAllBwdSchedule(PipelineSchedule):
         for i in range(num_micro):
             if i == num_micro - 1:
                 self._set_grad_sync(True)
+            if self._defer_wgrads and i:
+                from ...ops import flush_deferred_wgrads
+
+                flush_deferred_wgrads()  # ZB: dW kernels under the recv wait
             grad = pipeline_communicate(
                 "recv_backward", self.pp_rank, self.pp_size, self.group_ranks,
                 shapes=tensor_shapes, dtype=dtype, device=device, group=self.pp_group,
@@ -521,11 +525,21 @@ class InterleavedOneFOneBSchedule(PipelineSchedule):
             if g == last_g:
                 grad = None
             else:
+                if self._defer_wgrads:
+                    from ...ops import flush_deferred_wgrads
+
+                    flush_deferred_wgrads()  # ZB: dW under the ring recv
                 grad = ring_recv(
                     r, p, self.group_ranks, +1, tensor_shapes, dtype, device,
                     group=bwd_group,
                 )
-            in_grad = self.model.backward(b_in, b_out, grad)
+            if self._defer_wgrads:
+                from ...ops import defer_wgrads
+
+                with defer_wgrads.scope():
+                    in_grad = self.model.backward(b_in, b_out, grad)
+            else:
+                in_grad = self.model.backward(b_in, b_out, grad)
             if g != 0 and in_grad is not None:
                 send_keep.append(in_grad)
                 send_reqs.extend(

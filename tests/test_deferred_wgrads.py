@@ -151,3 +151,37 @@ def test_pp2_deferred_wgrads_matches():
     from conftest import run_distributed
 
     run_distributed(_pp2_defer, 2)
+
+
+def _interleaved_defer(rank, world):
+    """Interleaved 1F1B with defer_wgrads: loss identical to the same
+    run without deferral."""
+    from torch.utils.data import DataLoader
+
+    from quintnet_amd import GPT2Trainer, get_strategy, init_process_groups
+    from quintnet_amd.models import GPT2ForInterleaving
+    from quintnet_amd.utils.data import SyntheticCLM
+
+    pg = init_process_groups("cpu", [world], ["pp"])
+    cfg = GPT2Config(vocab_size=96, n_positions=16, n_embd=32, n_layer=4,
+                     n_head=2, dropout=0.0)
+
+    def run(flag):
+        torch.manual_seed(29 + rank)
+        model = GPT2ForInterleaving(cfg)
+        tcfg = {"schedule": "interleaved", "num_chunks": 2, "num_epochs": 1,
+                "grad_acc_steps": 2, "zero1": True, "task_type": "clm",
+                "max_seq_length": 16, "defer_wgrads": flag,
+                "model_config": {"n_embd": 32, "n_positions": 16}}
+        pmodel = get_strategy("pp", pg, tcfg).apply(model)
+        ds = SyntheticCLM(n=8, seq_len=16, vocab_size=96, seed=2)
+        tr = GPT2Trainer(pmodel, DataLoader(ds, batch_size=2), None, tcfg, pg)
+        return tr.fit()["train_loss"]
+
+    assert abs(run(False) - run(True)) < 1e-6
+
+
+def test_interleaved_deferred_wgrads_matches():
+    from conftest import run_distributed
+
+    run_distributed(_interleaved_defer, 2)
