@@ -227,6 +227,10 @@ class PipelineSchedule(abc.ABC):
             from ...ops import flush_deferred_wgrads
 
             flush_deferred_wgrads()  # BEFORE reduction/tied-sync/clip
+            # hooks stayed off for the whole window (see the gated
+            # _set_grad_sync sites); enable now so finalize launches
+            # every bucket itself with the flushed, complete gradients
+            self._set_grad_sync(True)
         self._finalize_grads()
         if t.max_grad_norm is not None and t.max_grad_norm > 0:
             if hasattr(t.optimizer, "clip_grad_norm_"):
@@ -292,7 +296,7 @@ class AllFwdAllBwdSchedule(PipelineSchedule):
             outputs.append(out)
 
         for i in range(num_micro):
-            if i == num_micro - 1:
+            if i == num_micro - 1 and not self._defer_wgrads:
                 self._set_grad_sync(True)
             if self._defer_wgrads and i:
                 from ...ops import flush_deferred_wgrads
@@ -375,7 +379,10 @@ class OneFOneBSchedule(PipelineSchedule):
             inputs.append(inp_used)
             outputs.append(out)
             b_in, b_out = inputs.pop(0), outputs.pop(0)
-            if backwards_done + 1 == num_micro:
+            if backwards_done + 1 == num_micro and not self._defer_wgrads:
+                # (ZB: hooks stay off — post-accumulate hooks fire even
+                # for deferred params, which would launch the bucket
+                # reduction BEFORE the final flush; finalize launches)
                 self._set_grad_sync(True)
             in_grad = self._backward_step(b_in, b_out, out_grad)
             maybe_enable_sync()
@@ -413,7 +420,10 @@ class OneFOneBSchedule(PipelineSchedule):
                 shapes=tensor_shapes, dtype=dtype, device=device, group=self.pp_group,
             )
             b_in, b_out = inputs.pop(0), outputs.pop(0)
-            if backwards_done + 1 == num_micro:
+            if backwards_done + 1 == num_micro and not self._defer_wgrads:
+                # (ZB: hooks stay off — post-accumulate hooks fire even
+                # for deferred params, which would launch the bucket
+                # reduction BEFORE the final flush; finalize launches)
                 self._set_grad_sync(True)
             in_grad = self._backward_step(b_in, b_out, grad)
             maybe_enable_sync()

@@ -185,3 +185,59 @@ def test_interleaved_deferred_wgrads_matches():
     from conftest import run_distributed
 
     run_distributed(_interleaved_defer, 2)
+
+
+def _zb_zero2_pp_dp(rank, world):
+    """defer_wgrads x zero_stage=2 x pp2 x dp2: the flush must land the
+    dW adds in the reduce-scatter bucket views BEFORE finalize launches
+    them; params stay dp-identical and match the non-deferred run."""
+    from quintnet_amd import init_process_groups
+    from quintnet_amd.gpt2_trainer import GPT2Trainer
+    from quintnet_amd.optim import Zero2AdamW
+    from quintnet_amd.parallel import (
+        BucketConfig,
+        DataParallel,
+        DistributedConfig,
+        PipelineParallelWrapper,
+    )
+
+    pg = init_process_groups("cpu", [2, 2], ["dp", "pp"])
+    cfg = GPT2Config(n_embd=64, n_layer=4, n_head=2, vocab_size=128,
+                     n_positions=32, dropout=0.0)
+
+    def run(flag):
+        torch.manual_seed(5 + pg.pp_rank)  # dp pair identical init
+        stage = GPT2Stage(cfg, pp_rank=pg.pp_rank, pp_size=pg.pp_size,
+                          tied_group=pg.get_tied_embedding_group())
+        stage.seq_len, stage.hidden_dim = 32, 64
+        config = {"zero_stage": 2, "learning_rate": 1e-3, "num_epochs": 1,
+                  "grad_acc_steps": 2, "max_grad_norm": 1.0,
+                  "defer_wgrads": flag,
+                  "model_config": {"n_positions": 32, "n_embd": 64}}
+        pmodel = PipelineParallelWrapper(
+            stage_module=stage, pp_rank=pg.pp_rank,
+            pp_group=pg.get_group("pp"), pp_size=pg.pp_size)
+        pmodel.seq_len, pmodel.hidden_dim = 32, 64
+        pmodel = DataParallel(
+            pmodel,
+            DistributedConfig(pg.dp_rank, pg.dp_size, pg.get_group("dp")),
+            bucket_config=BucketConfig(grad_reduce_op="reduce_scatter"))
+        g = torch.Generator().manual_seed(7)
+        ids = torch.randint(0, 128, (8, 32), generator=g)
+        data = [{"input_ids": ids[i : i + 2], "labels": ids[i : i + 2].clone()}
+                for i in range(0, 8, 2)]
+        tr = GPT2Trainer(pmodel, data, None, config, pg)
+        assert isinstance(tr.optimizer, Zero2AdamW)
+        tr._train_epoch(0)
+        return [p.detach().clone() for p in stage.parameters()]
+
+    plain = run(False)
+    zb = run(True)
+    for a, b in zip(plain, zb):
+        assert torch.allclose(a, b, atol=1e-6)
+
+
+def test_zb_zero2_pp_dp_world4():
+    from conftest import run_distributed
+
+    run_distributed(_zb_zero2_pp_dp, 4)
