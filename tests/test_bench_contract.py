@@ -34,3 +34,16 @@ def test_bench_emits_contract_json():
     for ck in ("model", "global_batch", "seq_len", "parallelism"):
         assert ck in j["config"], ck
     assert j["value"] > 0 and j["ms_per_step"] > 0
+
+
+def test_bench_vit_contract():
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"), "--model", "vit",
+         "--tiny", "--gpus", "1", "--steps", "1", "--warmup", "0"],
+        capture_output=True, text=True, timeout=600, cwd=REPO,
+    )
+    assert out.returncode == 0, out.stderr[-800:]
+    payloads = [json.loads(l) for l in out.stdout.splitlines()
+                if l.strip().startswith("{")]
+    assert len(payloads) == 1
+    assert payloads[0]["config"]["model"].startswith("vit")
