@@ -188,3 +188,28 @@ def test_label_smoothing_matches_torch(n, v, eps, seed):
     assert torch.allclose(ours, ref, rtol=1e-5, atol=1e-6)
     ours.backward()
     assert torch.isfinite(logits.grad).all()
+
+
+@settings(max_examples=15, deadline=None)
+@given(
+    seed=st.integers(0, 10**6),
+    draft_k=st.integers(1, 6),
+    horizon=st.integers(1, 20),
+)
+def test_speculative_greedy_exactness_randomized(seed, draft_k, horizon):
+    """Greedy speculative == plain greedy for arbitrary tiny models and
+    draft quality (fp32; the verify forward batches tokens while
+    generate steps singly, so this also probes argmax-tie stability)."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.models.gpt2.speculative import speculative_generate
+
+    g = torch.Generator().manual_seed(seed)
+    torch.manual_seed(int(torch.randint(0, 2**31 - 1, (1,), generator=g)))
+    base = dict(vocab_size=64, n_positions=64, dropout=0.0)
+    target = GPT2Stage(GPT2Config(n_embd=32, n_layer=2, n_head=2, **base)).eval()
+    draft = GPT2Stage(GPT2Config(n_embd=16, n_layer=1, n_head=2, **base)).eval()
+    ids = torch.randint(0, 64, (1, 5), generator=g)
+    want = target.generate(ids, max_new_tokens=horizon, temperature=0.0)
+    have = speculative_generate(target, draft, ids, max_new_tokens=horizon,
+                                draft_k=draft_k)
+    assert torch.equal(have, want), (seed, draft_k, horizon)
