@@ -213,3 +213,30 @@ def test_speculative_greedy_exactness_randomized(seed, draft_k, horizon):
     have = speculative_generate(target, draft, ids, max_new_tokens=horizon,
                                 draft_k=draft_k)
     assert torch.equal(have, want), (seed, draft_k, horizon)
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    seed=st.integers(0, 10**6),
+    beams=st.integers(1, 6),
+    eos=st.integers(0, 47),
+    lp=st.floats(0.5, 2.0),
+)
+def test_beam_search_fuzz_never_crashes(seed, beams, eos, lp):
+    """Beam search across random models/eos/length-penalties: always a
+    valid [1, <=T0+new] sequence, eos only terminal, tokens in vocab."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+    from quintnet_amd.models.gpt2.beam import beam_search
+
+    torch.manual_seed(seed)
+    m = GPT2Stage(GPT2Config(n_embd=16, n_layer=1, n_head=2, vocab_size=48,
+                             n_positions=48, dropout=0.0)).eval()
+    ids = torch.randint(0, 48, (1, 4))
+    out = beam_search(m, ids, max_new_tokens=10, num_beams=beams,
+                      eos_token_id=eos, length_penalty=lp)
+    assert out.shape[0] == 1 and 4 <= out.shape[1] <= 14
+    assert int(out.max()) < 48 and int(out.min()) >= 0
+    new = out[0, 4:]
+    hits = (new == eos).nonzero()
+    if hits.numel():
+        assert int(hits[0]) == new.shape[0] - 1  # eos is terminal
