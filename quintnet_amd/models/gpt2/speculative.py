@@ -32,14 +32,15 @@ __all__ = ["speculative_generate"]
 def _dist(logits_row: torch.Tensor, temperature: float, top_k: int,
           top_p: float) -> torch.Tensor:
     """logits [V] -> probability vector under the sampling transform."""
-    x = logits_row.float() / max(temperature, 1e-6)
+    x = logits_row.float() / max(temperature, 1e-5)
     if top_k and top_k > 0:
-        kth = torch.topk(x, top_k).values[-1]
+        kth = torch.topk(x, min(int(top_k), x.shape[-1])).values[-1]
         x = x.masked_fill(x < kth, float("-inf"))
     if top_p and 0.0 < top_p < 1.0:
         sl, si = torch.sort(x, descending=True)
         probs = torch.softmax(sl, dim=-1)
         drop_sorted = probs.cumsum(-1) - probs >= top_p
+        drop_sorted[..., 0] = False  # fp32-underflow guard: argmax kept
         drop = torch.zeros_like(drop_sorted).scatter(-1, si, drop_sorted)
         x = x.masked_fill(drop, float("-inf"))
     return torch.softmax(x, dim=-1)

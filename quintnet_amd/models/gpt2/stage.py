@@ -226,10 +226,13 @@ class GPT2Stage(nn.Module):
                     seen = torch.where(seen > 0, seen / repetition_penalty,
                                        seen * repetition_penalty)
                     logits = logits.scatter(1, out, seen)
-                if temperature and temperature > 0:
+                # temperatures below 1e-5 behave as greedy (dividing by
+                # them overflows the logits to inf -> NaN softmax)
+                if temperature and temperature > 1e-5:
                     logits = logits.float() / temperature
                     if top_k and top_k > 0:
-                        kth = torch.topk(logits, top_k, dim=-1).values[..., -1:]
+                        k_ = min(int(top_k), logits.shape[-1])
+                        kth = torch.topk(logits, k_, dim=-1).values[..., -1:]
                         logits = logits.masked_fill(logits < kth, float("-inf"))
                     if top_p and 0.0 < top_p < 1.0:
                         # nucleus: keep the smallest prefix of the sorted
@@ -238,6 +241,9 @@ class GPT2Stage(nn.Module):
                         sl, si = torch.sort(logits, dim=-1, descending=True)
                         cum = torch.softmax(sl, dim=-1).cumsum(dim=-1)
                         drop_sorted = cum - torch.softmax(sl, -1) >= top_p
+                        # tiny top_p underflows to 0 in fp32 (0 >= 0 would
+                        # drop EVERYTHING): the argmax is always kept
+                        drop_sorted[..., 0] = False
                         drop = torch.zeros_like(drop_sorted).scatter(
                             -1, si, drop_sorted
                         )

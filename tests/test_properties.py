@@ -240,3 +240,27 @@ def test_beam_search_fuzz_never_crashes(seed, beams, eos, lp):
     hits = (new == eos).nonzero()
     if hits.numel():
         assert int(hits[0]) == new.shape[0] - 1  # eos is terminal
+
+
+@settings(max_examples=20, deadline=None)
+@given(
+    seed=st.integers(0, 10**6),
+    temp=st.floats(0.0, 2.0),
+    top_k=st.integers(0, 50),
+    top_p=st.floats(0.0, 1.0),
+    pen=st.floats(0.8, 3.0),
+)
+def test_generate_sampling_fuzz(seed, temp, top_k, top_p, pen):
+    """generate() across the whole sampling-knob space: valid tokens,
+    right length, int8 cache path included every other example."""
+    from quintnet_amd.models import GPT2Config, GPT2Stage
+
+    torch.manual_seed(seed)
+    m = GPT2Stage(GPT2Config(n_embd=16, n_layer=1, n_head=2, vocab_size=48,
+                             n_positions=48, dropout=0.0)).eval()
+    ids = torch.randint(0, 48, (1, 4))
+    out = m.generate(ids, max_new_tokens=6, temperature=temp, top_k=top_k,
+                     top_p=top_p, repetition_penalty=pen,
+                     cache_dtype="int8" if seed % 2 else None)
+    assert out.shape == (1, 10)
+    assert int(out.max()) < 48 and int(out.min()) >= 0
