@@ -264,3 +264,30 @@ def test_generate_sampling_fuzz(seed, temp, top_k, top_p, pen):
                      cache_dtype="int8" if seed % 2 else None)
     assert out.shape == (1, 10)
     assert int(out.max()) < 48 and int(out.min()) >= 0
+
+
+@_FAST
+@given(
+    n=st.integers(2, 20),
+    v=st.integers(4, 60),
+    n_masked=st.integers(1, 3),
+    seed=st.integers(0, 10**6),
+)
+def test_cross_entropy_with_minus_inf_columns(n, v, n_masked, seed):
+    """CE over logits with -inf columns (the padded-vocab layout):
+    matches torch, grads finite and zero on the masked columns."""
+    g = torch.Generator().manual_seed(seed)
+    n_masked = min(n_masked, v - 1)
+    logits = torch.randn(n, v, generator=g) * 3
+    logits[:, v - n_masked:] = float("-inf")
+    tgt = torch.randint(0, v - n_masked, (n,), generator=g)
+    l1 = logits.clone().requires_grad_(True)
+    ours = cross_entropy(l1, tgt)
+    l2 = logits.clone().requires_grad_(True)
+    ref = torch.nn.functional.cross_entropy(l2, tgt)
+    assert torch.allclose(ours, ref, rtol=1e-5, atol=1e-6)
+    ours.backward()
+    ref.backward()
+    assert torch.isfinite(l1.grad).all()
+    assert bool((l1.grad[:, v - n_masked:] == 0).all())
+    assert torch.allclose(l1.grad, l2.grad, rtol=1e-5, atol=1e-6)
